@@ -1,0 +1,130 @@
+"""Layer base class + registry.
+
+Equivalent surface to `caffe::Layer<Dtype>` + `REGISTER_LAYER_CLASS`
+(consumed in the reference by cos_data_layer.cpp:81 and the layer census of
+SURVEY.md §3.6), re-designed for torch tensors and explicit fp32-master /
+bf16-compute dtype handling.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Type
+
+import torch
+
+from ...proto import caffe_pb
+from ..blob import Blob
+
+LAYER_REGISTRY: Dict[str, Type["Layer"]] = {}
+
+
+def register_layer(name: str):
+    def deco(cls):
+        LAYER_REGISTRY[name] = cls
+        cls.TYPE = name
+        return cls
+    return deco
+
+
+def create_layer(param: caffe_pb.LayerParameter, net) -> "Layer":
+    cls = LAYER_REGISTRY.get(param.type)
+    if cls is None:
+        raise ValueError(f"unknown layer type {param.type!r} (layer {param.name!r})")
+    return cls(param, net)
+
+
+class Layer:
+    TYPE = ""
+
+    def __init__(self, param: caffe_pb.LayerParameter, net):
+        self.param = param
+        self.name = param.name
+        self.net = net
+        self.phase = net.phase
+        self.blobs: List[Blob] = []      # learnable params, fp32 master copies
+        self._inplace: List[bool] = []
+
+    # dtype the layer computes in (activations)
+    @property
+    def dtype(self) -> torch.dtype:
+        return self.net.dtype
+
+    @property
+    def device(self) -> torch.device:
+        return self.net.device
+
+    def param_spec(self, i: int) -> caffe_pb.ParamSpec:
+        specs = self.param.param
+        return specs[i] if i < len(specs) else caffe_pb.ParamSpec()
+
+    def add_param(self, shape, filler: caffe_pb.FillerParameter = None,
+                  name: str = "") -> Blob:
+        from .. import fillers
+        b = Blob(shape, name=name or f"{self.name}_p{len(self.blobs)}",
+                 dtype=torch.float32, device=self.device, alloc_diff=True)
+        if filler is not None:
+            fillers.fill(b.data, filler, generator=self.net.generator)
+        spec = self.param_spec(len(self.blobs))
+        b._lr_mult = spec.lr_mult
+        b._decay_mult = spec.decay_mult
+        self.blobs.append(b)
+        return b
+
+    # -- lifecycle -----------------------------------------------------------
+    def setup(self, bottom: List[Blob], top: List[Blob]) -> None:
+        pass
+
+    def reshape(self, bottom: List[Blob], top: List[Blob]) -> None:
+        pass
+
+    def forward(self, bottom: List[Blob], top: List[Blob]) -> float:
+        """Returns this layer's loss contribution (0 for non-loss layers)."""
+        raise NotImplementedError
+
+    def backward(self, top: List[Blob], propagate_down: List[bool],
+                 bottom: List[Blob]) -> None:
+        if any(propagate_down):
+            raise NotImplementedError(
+                f"{type(self).__name__} has no backward")
+
+    # -- helpers -------------------------------------------------------------
+    def cast(self, t: torch.Tensor) -> torch.Tensor:
+        return t.to(self.dtype) if t.dtype != self.dtype else t
+
+    def weight(self, i: int = 0) -> torch.Tensor:
+        """Param i's data in compute dtype."""
+        return self.cast(self.blobs[i].data)
+
+    def acc_param_diff(self, i: int, grad: torch.Tensor) -> None:
+        d = self.blobs[i].ensure_diff()
+        d.add_(grad.to(d.dtype))
+
+    @staticmethod
+    def acc_blob_diff(blob: Blob, dx: torch.Tensor, inplace: bool) -> None:
+        """Accumulate dx into blob.diff (replace when the layer ran in-place:
+        top and bottom share the tensor, so += would double-count)."""
+        if inplace:
+            blob.diff = dx
+        else:
+            d = blob.ensure_diff()
+            if d.dtype != dx.dtype:
+                blob.diff = d.to(dx.dtype)
+                d = blob.diff
+            d.add_(dx)
+
+    def loss_weight(self, i: int) -> float:
+        lw = self.param.loss_weight
+        if i < len(lw):
+            return lw[i]
+        if "Loss" in self.param.type and i == 0:
+            return 1.0
+        return 0.0
+
+    @property
+    def has_loss(self) -> bool:
+        return any(self.loss_weight(i) != 0.0
+                   for i in range(max(1, len(self.param.top))))
+
+    def auto_top_blobs(self) -> int:
+        """Min number of top blobs (used by Net to allocate)."""
+        return len(self.param.top)

@@ -1,0 +1,199 @@
+"""Structural layers: Concat, Slice, Eltwise, Flatten, Reshape, Split,
+Silence, Power, Exp, Log — plumbing needed by GoogLeNet (Concat/Inception)
+and LRCN (Silence) per SURVEY.md §2.5/§3.6.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from ...proto import caffe_pb
+from .base import Layer, register_layer
+
+
+@register_layer("Concat")
+class ConcatLayer(Layer):
+    def setup(self, bottom, top):
+        p = self.param.concat_param
+        self.axis = p.axis if p.has_field("axis") or not p.has_field("concat_dim") \
+            else int(p.concat_dim)
+
+    def forward(self, bottom, top):
+        top[0].data = torch.cat([b.data for b in bottom], dim=self.axis)
+        return 0.0
+
+    def backward(self, top, propagate_down, bottom):
+        sizes = [b.data.shape[self.axis] for b in bottom]
+        pieces = torch.split(top[0].diff, sizes, dim=self.axis)
+        for b, pd, piece in zip(bottom, propagate_down, pieces):
+            if pd:
+                self.acc_blob_diff(b, piece.contiguous(), False)
+
+
+@register_layer("Slice")
+class SliceLayer(Layer):
+    def setup(self, bottom, top):
+        p = self.param.slice_param
+        self.axis = p.axis
+        self.points = [int(x) for x in p.slice_point]
+
+    def _sizes(self, x):
+        total = x.shape[self.axis]
+        if self.points:
+            pts = [0] + self.points + [total]
+            return [pts[i + 1] - pts[i] for i in range(len(pts) - 1)]
+        n = len(self.param.top)
+        assert total % n == 0
+        return [total // n] * n
+
+    def forward(self, bottom, top):
+        pieces = torch.split(bottom[0].data, self._sizes(bottom[0].data),
+                             dim=self.axis)
+        for t, piece in zip(top, pieces):
+            t.data = piece.contiguous()
+        return 0.0
+
+    def backward(self, top, propagate_down, bottom):
+        if propagate_down[0]:
+            dx = torch.cat([t.diff if t.diff is not None
+                            else torch.zeros_like(t.data) for t in top],
+                           dim=self.axis)
+            self.acc_blob_diff(bottom[0], dx, False)
+
+
+@register_layer("Eltwise")
+class EltwiseLayer(Layer):
+    def setup(self, bottom, top):
+        p = self.param.eltwise_param
+        self.op = p.operation
+        self.coeff = list(p.coeff) or [1.0] * len(self.param.bottom)
+        self._argmax = None
+
+    def forward(self, bottom, top):
+        Op = caffe_pb.EltwiseParameter.EltwiseOp
+        if self.op == Op.SUM:
+            y = bottom[0].data * self.coeff[0]
+            for b, c in zip(bottom[1:], self.coeff[1:]):
+                y = y + b.data * c
+        elif self.op == Op.PROD:
+            y = bottom[0].data
+            for b in bottom[1:]:
+                y = y * b.data
+        else:  # MAX
+            stacked = torch.stack([b.data for b in bottom])
+            y, self._argmax = stacked.max(dim=0)
+        top[0].data = y
+        return 0.0
+
+    def backward(self, top, propagate_down, bottom):
+        Op = caffe_pb.EltwiseParameter.EltwiseOp
+        dy = top[0].diff
+        for i, (b, pd) in enumerate(zip(bottom, propagate_down)):
+            if not pd:
+                continue
+            if self.op == Op.SUM:
+                dx = dy * self.coeff[i]
+            elif self.op == Op.PROD:
+                dx = dy.clone()
+                for j, ob in enumerate(bottom):
+                    if j != i:
+                        dx = dx * ob.data
+            else:
+                dx = dy * (self._argmax == i).to(dy.dtype)
+            self.acc_blob_diff(b, dx, False)
+
+
+@register_layer("Flatten")
+class FlattenLayer(Layer):
+    def setup(self, bottom, top):
+        p = self.param.flatten_param
+        self.axis = p.axis
+        self.end_axis = p.end_axis
+
+    def forward(self, bottom, top):
+        x = bottom[0].data
+        end = self.end_axis if self.end_axis >= 0 else x.dim() + self.end_axis
+        shape = list(x.shape[:self.axis]) + [-1] + list(x.shape[end + 1:])
+        top[0].data = x.reshape(shape)
+        return 0.0
+
+    def backward(self, top, propagate_down, bottom):
+        if propagate_down[0]:
+            self.acc_blob_diff(bottom[0],
+                               top[0].diff.reshape(bottom[0].data.shape), False)
+
+
+@register_layer("Reshape")
+class ReshapeLayer(Layer):
+    def setup(self, bottom, top):
+        self.dims = [int(d) for d in self.param.reshape_param.shape.dim]
+
+    def forward(self, bottom, top):
+        x = bottom[0].data
+        shape = []
+        for i, d in enumerate(self.dims):
+            if d == 0:
+                shape.append(x.shape[i])
+            else:
+                shape.append(d)
+        top[0].data = x.reshape(shape)
+        return 0.0
+
+    def backward(self, top, propagate_down, bottom):
+        if propagate_down[0]:
+            self.acc_blob_diff(bottom[0],
+                               top[0].diff.reshape(bottom[0].data.shape), False)
+
+
+@register_layer("Split")
+class SplitLayer(Layer):
+    """Explicit fan-out (Caffe inserts these automatically; our Net
+    accumulates diffs instead, but the layer type is still supported)."""
+
+    def forward(self, bottom, top):
+        for t in top:
+            t.data = bottom[0].data
+        return 0.0
+
+    def backward(self, top, propagate_down, bottom):
+        if propagate_down[0]:
+            dx = None
+            for t in top:
+                if t.diff is not None:
+                    dx = t.diff.clone() if dx is None else dx + t.diff
+            if dx is not None:
+                self.acc_blob_diff(bottom[0], dx, False)
+
+
+@register_layer("Silence")
+class SilenceLayer(Layer):
+    def forward(self, bottom, top):
+        return 0.0
+
+    def backward(self, top, propagate_down, bottom):
+        for b, pd in zip(bottom, propagate_down):
+            if pd:
+                b.ensure_diff()  # zeros
+
+
+@register_layer("Power")
+class PowerLayer(Layer):
+    def setup(self, bottom, top):
+        p = self.param.power_param
+        self.power, self.scale, self.shift = p.power, p.scale, p.shift
+
+    def forward(self, bottom, top):
+        x = bottom[0].data
+        inner = self.scale * x + self.shift
+        top[0].data = inner if self.power == 1.0 else inner.pow(self.power)
+        self._inner = inner
+        return 0.0
+
+    def backward(self, top, propagate_down, bottom):
+        if propagate_down[0]:
+            dy = top[0].diff
+            if self.power == 1.0:
+                dx = dy * self.scale
+            else:
+                dx = dy * self.power * self.scale * self._inner.pow(self.power - 1)
+            self.acc_blob_diff(bottom[0], dx, top[0] is bottom[0])

@@ -1,0 +1,292 @@
+"""Solver family: SGD / Nesterov / AdaGrad / RMSProp / AdaDelta / Adam.
+
+Re-implements the `caffe::Solver` surface the reference drives through JNI
+(CaffeNet.cpp:196-205, 592-738 — SURVEY.md §2.5): Step(n) with
+on_start / on_gradients_ready callbacks around each iteration (the hook the
+distributed engine attaches to), lr policies, momentum, L1/L2 weight decay,
+gradient clipping, iter_size accumulation, and snapshot/restore to
+`.caffemodel` / `.solverstate` (binary proto or HDF5).
+"""
+
+from __future__ import annotations
+
+import math
+import os
+from typing import Callable, List, Optional
+
+import torch
+
+from .. import ops
+from ..proto import caffe_pb, read_binary_proto, text_format, write_binary_proto
+from .blob import Blob
+from .net import Net
+
+
+class Callback:
+    def on_start(self) -> None: ...
+    def on_gradients_ready(self) -> None: ...
+
+
+class Solver:
+    def __init__(self, param: caffe_pb.SolverParameter, *,
+                 device: Optional[torch.device] = None,
+                 dtype: torch.dtype = torch.float32,
+                 proto_dir: str = "."):
+        self.param = param
+        self.device = device or torch.device("cpu")
+        self.dtype = dtype
+        self.iter = 0
+        self.current_step = 0
+        self.callbacks: List[Callback] = []
+        self.solver_count = 1
+        self.rank = 0
+        self.type = (param.type or "SGD")
+        if param.has_field("solver_type") and not param.has_field("type"):
+            self.type = caffe_pb.SolverType.by_number.get(
+                param.solver_type, "SGD").capitalize()
+        seed = param.random_seed if param.random_seed >= 0 else None
+
+        net_param = self._resolve_net_param(proto_dir)
+        train_state = caffe_pb.NetState(phase=caffe_pb.Phase.TRAIN)
+        if param.has_field("train_state"):
+            train_state._merge(param.train_state)
+            train_state.phase = caffe_pb.Phase.TRAIN
+        self.net = Net(net_param, train_state, device=self.device,
+                       dtype=self.dtype, seed=seed)
+        self.test_nets: List[Net] = []
+        n_test = len(param.test_iter)
+        for i in range(n_test):
+            ts = caffe_pb.NetState(phase=caffe_pb.Phase.TEST)
+            if i < len(param.test_state):
+                ts._merge(param.test_state[i])
+                ts.phase = caffe_pb.Phase.TEST
+            if param.test_iter[i] <= 0:
+                continue
+            tn = Net(net_param, ts, device=self.device, dtype=self.dtype,
+                     seed=seed)
+            tn.share_trained_layers_with(self.net)
+            self.test_nets.append(tn)
+
+        self.params: List[Blob] = self.net.learnable_params()
+        self.history = [torch.zeros_like(b.data) for b in self.params]
+        self.history2 = [torch.zeros_like(b.data) for b in self.params] \
+            if self.type in ("Adam", "AdaDelta") else []
+        self._losses: List[float] = []
+        self.smoothed_loss = 0.0
+
+    def _resolve_net_param(self, proto_dir: str) -> caffe_pb.NetParameter:
+        p = self.param
+        if p.has_field("net_param"):
+            return p.net_param
+        if p.has_field("train_net_param"):
+            return p.train_net_param
+        path = p.net or p.train_net
+        if not path:
+            raise ValueError("solver has no net")
+        if not os.path.exists(path):
+            alt = os.path.join(proto_dir, os.path.basename(path))
+            if os.path.exists(alt):
+                path = alt
+        return text_format.parse_file(path, caffe_pb.NetParameter)
+
+    # -------------------------------------------------------------- lr policy
+    def get_lr(self) -> float:
+        p = self.param
+        policy = p.lr_policy or "fixed"
+        base = p.base_lr
+        it = self.iter
+        if policy == "fixed":
+            return base
+        if policy == "step":
+            self.current_step = it // max(1, p.stepsize)
+            return base * (p.gamma ** self.current_step)
+        if policy == "exp":
+            return base * (p.gamma ** it)
+        if policy == "inv":
+            return base * (1.0 + p.gamma * it) ** (-p.power)
+        if policy == "multistep":
+            if (self.current_step < len(p.stepvalue)
+                    and it >= p.stepvalue[self.current_step]):
+                self.current_step += 1
+            return base * (p.gamma ** self.current_step)
+        if policy == "poly":
+            return base * (1.0 - it / max(1, p.max_iter)) ** p.power
+        if policy == "sigmoid":
+            return base / (1.0 + math.exp(-p.gamma * (it - p.stepsize)))
+        raise ValueError(f"unknown lr_policy {policy!r}")
+
+    # ------------------------------------------------------------------- step
+    def step(self, iters: int) -> float:
+        loss = 0.0
+        for _ in range(iters):
+            loss = self._step_one()
+        return loss
+
+    def _step_one(self) -> float:
+        p = self.param
+        for cb in self.callbacks:
+            cb.on_start()
+        self.net.zero_param_diffs()
+        loss = 0.0
+        for _ in range(max(1, p.iter_size)):
+            loss += self.net.forward_backward()
+        loss /= max(1, p.iter_size)
+        for cb in self.callbacks:
+            cb.on_gradients_ready()
+        self.apply_update()
+        self.iter += 1
+        self._update_smoothed_loss(loss)
+        if p.display and self.iter % p.display == 0 and self.rank == 0:
+            print(f"[cos-amd] iter {self.iter} loss {self.smoothed_loss:.6f} "
+                  f"lr {self.get_lr():.6g}", flush=True)
+        return loss
+
+    def _update_smoothed_loss(self, loss: float) -> None:
+        avg = max(1, self.param.average_loss)
+        self._losses.append(loss)
+        if len(self._losses) > avg:
+            self._losses.pop(0)
+        self.smoothed_loss = sum(self._losses) / len(self._losses)
+
+    # ----------------------------------------------------------------- update
+    def apply_update(self) -> None:
+        p = self.param
+        rate = self.get_lr()
+        # normalize for iter_size
+        scale = 1.0 / (max(1, p.iter_size) * self.solver_count)
+        if scale != 1.0:
+            for b in self.params:
+                if b.diff is not None:
+                    b.diff.mul_(scale)
+        # gradient clipping (global L2 norm)
+        clip = p.clip_gradients
+        if clip > 0:
+            sumsq = 0.0
+            for b in self.params:
+                if b.diff is not None:
+                    sumsq += float(b.diff.float().pow(2).sum())
+            norm = math.sqrt(sumsq)
+            if norm > clip:
+                cs = clip / norm
+                for b in self.params:
+                    if b.diff is not None:
+                        b.diff.mul_(cs)
+        wd = p.weight_decay
+        reg = p.regularization_type
+        for i, b in enumerate(self.params):
+            if b._lr_mult == 0 or b.diff is None:
+                continue
+            local_rate = rate * b._lr_mult
+            local_decay = wd * b._decay_mult
+            grad = b.diff
+            if reg == "L1" and local_decay:
+                grad = grad + local_decay * torch.sign(b.data)
+                local_decay = 0.0
+            self._update_one(i, b, grad, local_rate, local_decay)
+
+    def _update_one(self, i, b, grad, lr, wd) -> None:
+        t = self.type
+        if t == "SGD":
+            ops.sgd_update(b.data, grad, self.history[i], lr,
+                           self.param.momentum, wd)
+        elif t == "Nesterov":
+            ops.nesterov_update(b.data, grad, self.history[i], lr,
+                                self.param.momentum, wd)
+        elif t == "Adam":
+            ops.adam_update(b.data, grad, self.history[i], self.history2[i],
+                            lr, self.param.momentum, self.param.momentum2,
+                            self.param.delta, wd, self.iter + 1)
+        elif t == "AdaGrad":
+            g = grad + wd * b.data if wd else grad
+            self.history[i].addcmul_(g, g, value=1.0)
+            b.data.sub_(lr * g / (self.history[i].sqrt() + self.param.delta))
+        elif t == "RMSProp":
+            g = grad + wd * b.data if wd else grad
+            rd = self.param.rms_decay
+            self.history[i].mul_(rd).addcmul_(g, g, value=1 - rd)
+            b.data.sub_(lr * g / (self.history[i].sqrt() + self.param.delta))
+        elif t == "AdaDelta":
+            g = grad + wd * b.data if wd else grad
+            mom, delta = self.param.momentum, self.param.delta
+            self.history[i].mul_(mom).addcmul_(g, g, value=1 - mom)
+            update = g * (self.history2[i] + delta).sqrt() \
+                / (self.history[i] + delta).sqrt()
+            self.history2[i].mul_(mom).addcmul_(update, update, value=1 - mom)
+            b.data.sub_(lr * update)
+        else:
+            raise ValueError(f"unknown solver type {self.type!r}")
+
+    # ------------------------------------------------------------- validation
+    def test(self, test_net_id: int = 0) -> dict:
+        net = self.test_nets[test_net_id]
+        iters = self.param.test_iter[test_net_id]
+        sums: dict = {}
+        for _ in range(iters):
+            net.forward()
+            for name in net.output_blob_names():
+                v = net.blob_by_name(name).data.float().mean().item()
+                sums[name] = sums.get(name, 0.0) + v
+        return {k: v / iters for k, v in sums.items()}
+
+    # --------------------------------------------------------------- snapshot
+    def snapshot_filename(self, kind: str) -> str:
+        prefix = self.param.snapshot_prefix or "snapshot"
+        ext = {"model": ".caffemodel", "state": ".solverstate"}[kind]
+        h5 = self.param.snapshot_format == caffe_pb.SnapshotFormat.HDF5
+        return f"{prefix}_iter_{self.iter}{ext}" + (".h5" if h5 else "")
+
+    def snapshot(self) -> str:
+        model_file = self.snapshot_filename("model")
+        state_file = self.snapshot_filename("state")
+        h5 = self.param.snapshot_format == caffe_pb.SnapshotFormat.HDF5
+        if h5:
+            from ..utils import hdf5 as h5util
+            h5util.save_net(model_file, self.net)
+            h5util.save_solver_state(state_file, self, model_file)
+        else:
+            write_binary_proto(model_file, self.net.to_proto())
+            state = caffe_pb.SolverState(
+                iter=self.iter, learned_net=model_file,
+                current_step=self.current_step)
+            state.history = [Blob._tensor_to_proto(h) for h in
+                             self.history + self.history2]
+            write_binary_proto(state_file, state)
+        return model_file
+
+    def restore(self, state_file: str) -> None:
+        if state_file.endswith(".h5"):
+            from ..utils import hdf5 as h5util
+            h5util.load_solver_state(state_file, self)
+            return
+        state = read_binary_proto(state_file, caffe_pb.SolverState)
+        self.iter = state.iter
+        self.current_step = state.current_step
+        hist = self.history + self.history2
+        for h, bp in zip(hist, state.history):
+            tmp = Blob([0], device=h.device)
+            tmp.from_proto(bp)
+            h.copy_(tmp.data.reshape(h.shape))
+        if state.learned_net and os.path.exists(state.learned_net):
+            self.load_weights(state.learned_net)
+
+    def load_weights(self, model_file: str) -> None:
+        """Finetune path (reference: CopyTrainedLayersFrom,
+        CaffeNet.cpp:320-331).  Accepts comma-separated file list."""
+        for f in str(model_file).split(","):
+            f = f.strip()
+            if not f:
+                continue
+            if f.endswith(".h5"):
+                from ..utils import hdf5 as h5util
+                h5util.load_net(f, self.net)
+            else:
+                net_param = read_binary_proto(f, caffe_pb.NetParameter)
+                self.net.copy_trained_layers_from(net_param)
+            for tn in self.test_nets:
+                tn.share_trained_layers_with(self.net)
+
+
+def solver_from_prototxt(path: str, **kwargs) -> Solver:
+    param = text_format.parse_file(path, caffe_pb.SolverParameter)
+    kwargs.setdefault("proto_dir", os.path.dirname(os.path.abspath(path)))
+    return Solver(param, **kwargs)
