@@ -1,0 +1,114 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: AlexNet (bvlc_reference / CaffeNet class) training
+step, synthetic 227x227 data, bs=256/GPU, bf16 compute — the headline
+metric named by BASELINE.json (whole-node images/sec at 1/2/4/8 MI355X).
+
+Single GPU:   python bench.py --gpus 1 --steps 20 --warmup 5
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N ...
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch  # noqa: E402
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=256)
+    ap.add_argument("--model", type=str, default="alexnet")
+    args = ap.parse_args()
+
+    from caffeonspark_amd.core import solver_from_prototxt
+    from caffeonspark_amd.parallel import DistributedSync, init_distributed
+
+    ws = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = init_distributed()
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_gpu = torch.cuda.is_available()
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+        dev = torch.device("cuda", local_rank)
+        dtype = torch.bfloat16
+    else:
+        dev = torch.device("cpu")
+        dtype = torch.float32
+
+    root = os.path.dirname(os.path.abspath(__file__))
+    solver = solver_from_prototxt(
+        os.path.join(root, "caffeonspark_amd", "models",
+                     f"{args.model}_solver.prototxt"),
+        device=dev, dtype=dtype)
+    solver.param.display = 0
+
+    sync = None
+    if ws > 1:
+        sync = DistributedSync(solver)
+        sync.broadcast_params()
+
+    # synthetic batch of the named config's shape, resident on device
+    n = args.batch
+    g = torch.Generator().manual_seed(1234 + rank)
+    dl = solver.net.data_layers()[0]
+    c, h, w = dl.channels, dl.height, dl.width
+    x = torch.randn(n, c, h, w, generator=g).to(dev, dtype)
+    y = torch.randint(0, 1000, (n,), generator=g).float().to(dev)
+    dl.batch_size = n
+    dl.reset(x, y)
+
+    def barrier_sync():
+        if ws > 1:
+            torch.distributed.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        solver._step_one()
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        solver._step_one()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max step time over ranks
+    if ws > 1:
+        t = torch.tensor([elapsed], device=dev if use_gpu else None)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    value = (n * ws * args.steps) / elapsed
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images/sec",
+            "value": round(value, 2),
+            "unit": "images/sec",
+            "n_gpus": ws,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {"model": "alexnet(bvlc_reference/CaffeNet)",
+                       "global_batch": n * ws, "seq_len": None,
+                       "parallelism": f"dp{ws}"},
+        }), flush=True)
+    if ws > 1:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

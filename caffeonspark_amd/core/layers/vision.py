@@ -65,9 +65,10 @@ class ConvolutionLayer(Layer):
         x = bottom[0].data
         w = self.weight(0)
         b = self.cast(self.blobs[1].data) if self.bias_term else None
+        self._ctx = {}
         top[0].data = ops.conv2d_forward(
             x, w, b, (self.sh, self.sw), (self.ph, self.pw),
-            (self.dil, self.dil), self.groups)
+            (self.dil, self.dil), self.groups, ctx=self._ctx)
         return 0.0
 
     def backward(self, top, propagate_down, bottom):
@@ -79,7 +80,8 @@ class ConvolutionLayer(Layer):
         dx, dw, db = ops.conv2d_backward(
             x, w, dy, (self.sh, self.sw), (self.ph, self.pw),
             (self.dil, self.dil), self.groups,
-            need_dx=propagate_down[0], need_dw=need_dw, bias=need_db)
+            need_dx=propagate_down[0], need_dw=need_dw, bias=need_db,
+            ctx=getattr(self, "_ctx", None))
         if dw is not None:
             self.acc_param_diff(0, dw)
         if db is not None:

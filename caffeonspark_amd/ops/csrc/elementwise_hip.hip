@@ -1,0 +1,387 @@
+#include "hip/hip_runtime.h"
+// Elementwise kernel family: ReLU fwd/bwd, dropout (counter-based RNG),
+// fused SGD-momentum update, LSTM unit fwd/bwd, embed gather/scatter,
+// column-sum reduction (bias grads), softmax+NLL loss.
+// All bf16 traffic is vectorized 8-wide (G13: scalar bf16 ~2x slower).
+
+#include "common.h"
+
+namespace cosamd {
+
+typedef unsigned short u16;
+typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+__device__ __forceinline__ float ldbf(const u16* p) {
+  return bf2f(*reinterpret_cast<const bf16*>(p));
+}
+__device__ __forceinline__ void stbf(u16* p, float v) {
+  bf16 b = f2bf(v);
+  *p = *reinterpret_cast<u16*>(&b);
+}
+
+// ---------------------------------------------------------------- ReLU
+
+__global__ void relu_fwd_kernel(const u16* __restrict__ x, u16* __restrict__ y,
+                                float slope, int64_t n8, int64_t n) {
+  if (blockIdx.x == 0 && threadIdx.x == 0)
+    for (int64_t k = n8 * 8; k < n; ++k) {
+      float f = ldbf(x + k);
+      stbf(y + k, f > 0.f ? f : f * slope);
+    }
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n8; i += (int64_t)gridDim.x * blockDim.x) {
+    u16x8 v = *reinterpret_cast<const u16x8*>(x + i * 8);
+    u16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      u16 raw = v[j];
+      float f = bf2f(*reinterpret_cast<const bf16*>(&raw));
+      f = f > 0.f ? f : f * slope;
+      bf16 b = f2bf(f);
+      o[j] = *reinterpret_cast<u16*>(&b);
+    }
+    *reinterpret_cast<u16x8*>(y + i * 8) = o;
+  }
+}
+
+__global__ void relu_bwd_kernel(const u16* __restrict__ y,
+                                const u16* __restrict__ dy,
+                                u16* __restrict__ dx, float slope,
+                                int64_t n8, int64_t n) {
+  if (blockIdx.x == 0 && threadIdx.x == 0)
+    for (int64_t k = n8 * 8; k < n; ++k) {
+      float fy = ldbf(y + k), fd = ldbf(dy + k);
+      stbf(dx + k, fy > 0.f ? fd : fd * slope);
+    }
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n8; i += (int64_t)gridDim.x * blockDim.x) {
+    u16x8 vy = *reinterpret_cast<const u16x8*>(y + i * 8);
+    u16x8 vd = *reinterpret_cast<const u16x8*>(dy + i * 8);
+    u16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      u16 ry = vy[j], rd = vd[j];
+      float fy = bf2f(*reinterpret_cast<const bf16*>(&ry));
+      float fd = bf2f(*reinterpret_cast<const bf16*>(&rd));
+      float g = fy > 0.f ? fd : fd * slope;
+      bf16 b = f2bf(g);
+      o[j] = *reinterpret_cast<u16*>(&b);
+    }
+    *reinterpret_cast<u16x8*>(dx + i * 8) = o;
+  }
+}
+
+// -------------------------------------------------------------- dropout
+// counter-based RNG (splitmix-style hash of seed+index): reproducible,
+// stateless, good enough for dropout masks.
+
+__device__ __forceinline__ unsigned hash_rng(uint64_t seed, uint64_t idx) {
+  uint64_t z = seed + idx * 0x9E3779B97F4A7C15ull;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  return (unsigned)(z >> 32);
+}
+
+__global__ void dropout_fwd_kernel(const u16* __restrict__ x,
+                                   u16* __restrict__ y, u16* __restrict__ mask,
+                                   float keep, float inv_keep, uint64_t seed,
+                                   int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += (int64_t)gridDim.x * blockDim.x) {
+    float u = (hash_rng(seed, i) >> 8) * (1.f / 16777216.f);
+    float m = (u < keep) ? inv_keep : 0.f;
+    stbf(mask + i, m);
+    stbf(y + i, ldbf(x + i) * m);
+  }
+}
+
+__global__ void mul_bf16_kernel(const u16* __restrict__ a,
+                                const u16* __restrict__ b,
+                                u16* __restrict__ y, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += (int64_t)gridDim.x * blockDim.x)
+    stbf(y + i, ldbf(a + i) * ldbf(b + i));
+}
+
+// ------------------------------------------------------- SGD fused update
+// V = mu*V + lr*(g + wd*p); p -= V   (fp32 master weights)
+
+__global__ void sgd_update_kernel(float* __restrict__ p, const float* __restrict__ g,
+                                  float* __restrict__ v, float lr, float mu,
+                                  float wd, int64_t n4, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n4; i += (int64_t)gridDim.x * blockDim.x) {
+    if (i * 4 + 4 <= n) {
+      f32x4 pv = *reinterpret_cast<f32x4*>(p + i * 4);
+      f32x4 gv = *reinterpret_cast<const f32x4*>(g + i * 4);
+      f32x4 vv = *reinterpret_cast<f32x4*>(v + i * 4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float gg = gv[j] + wd * pv[j];
+        vv[j] = mu * vv[j] + lr * gg;
+        pv[j] -= vv[j];
+      }
+      *reinterpret_cast<f32x4*>(v + i * 4) = vv;
+      *reinterpret_cast<f32x4*>(p + i * 4) = pv;
+    } else {
+      for (int64_t k = i * 4; k < n; ++k) {
+        float gg = g[k] + wd * p[k];
+        v[k] = mu * v[k] + lr * gg;
+        p[k] -= v[k];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------- column reduce
+// out[c] += sum_r in[r*ld + c]  (bias gradients; in bf16, out fp32)
+
+__global__ void colsum_kernel(const u16* __restrict__ in, float* __restrict__ out,
+                              int64_t rows, int cols, int ld) {
+  // block handles 256 columns chunk; grid.y strides rows
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= cols) return;
+  float acc = 0.f;
+  for (int64_t r = blockIdx.y; r < rows; r += gridDim.y)
+    acc += ldbf(in + r * ld + c);
+  atomicAdd(out + c, acc);
+}
+
+// --------------------------------------------------------------- LSTM unit
+// gates [N,4H] pre-activation (i,f,o,g); fp32 cell state.
+
+__global__ void lstm_unit_fwd_kernel(
+    const float* __restrict__ c_prev, const u16* __restrict__ gates,
+    const u16* __restrict__ cont, float* __restrict__ c_out,
+    u16* __restrict__ h_out, float* __restrict__ act,  // [N,4,H] activated
+    int64_t n, int H) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t row = i / H;
+    int hh = i % H;
+    const u16* g = gates + row * 4 * H;
+    float gi = 1.f / (1.f + __expf(-ldbf(g + hh)));
+    float gf = 1.f / (1.f + __expf(-ldbf(g + H + hh)));
+    float go = 1.f / (1.f + __expf(-ldbf(g + 2 * H + hh)));
+    float gg = tanhf(ldbf(g + 3 * H + hh));
+    float ct = ldbf(cont + row);
+    float c = gf * c_prev[i] * ct + gi * gg;
+    float tc = tanhf(c);
+    c_out[i] = c;
+    stbf(h_out + i, go * tc);
+    float* a = act + row * 4 * H;
+    a[hh] = gi; a[H + hh] = gf; a[2 * H + hh] = go; a[3 * H + hh] = gg;
+    // tc recomputed in bwd from c_out
+  }
+}
+
+__global__ void lstm_unit_bwd_kernel(
+    const float* __restrict__ c_prev, const float* __restrict__ c_out,
+    const float* __restrict__ act, const u16* __restrict__ cont,
+    const float* __restrict__ dc_next, const u16* __restrict__ dh,
+    float* __restrict__ dc_prev, u16* __restrict__ dgates,
+    int64_t n, int H) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t row = i / H;
+    int hh = i % H;
+    const float* a = act + row * 4 * H;
+    float gi = a[hh], gf = a[H + hh], go = a[2 * H + hh], gg = a[3 * H + hh];
+    float tc = tanhf(c_out[i]);
+    float ct = ldbf(cont + row);
+    float dhv = ldbf(dh + i);
+    float dc = dc_next[i] + dhv * go * (1.f - tc * tc);
+    float dov = dhv * tc;
+    float div = dc * gg;
+    float dgv = dc * gi;
+    float dfv = dc * c_prev[i] * ct;
+    dc_prev[i] = dc * gf * ct;
+    u16* dg = dgates + row * 4 * H;
+    stbf(dg + hh, div * gi * (1.f - gi));
+    stbf(dg + H + hh, dfv * gf * (1.f - gf));
+    stbf(dg + 2 * H + hh, dov * go * (1.f - go));
+    stbf(dg + 3 * H + hh, dgv * (1.f - gg * gg));
+  }
+}
+
+// ------------------------------------------------------------------ embed
+
+__global__ void embed_fwd_kernel(const float* __restrict__ idx,
+                                 const u16* __restrict__ w,
+                                 u16* __restrict__ y, int64_t n, int E,
+                                 int V) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n * E; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t row = i / E;
+    int e = i % E;
+    int v = (int)idx[row];
+    v = v < 0 ? 0 : (v >= V ? V - 1 : v);
+    y[i] = w[(int64_t)v * E + e];
+  }
+}
+
+__global__ void embed_bwd_kernel(const float* __restrict__ idx,
+                                 const u16* __restrict__ dy,
+                                 float* __restrict__ dw, int64_t n, int E,
+                                 int V) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n * E; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t row = i / E;
+    int e = i % E;
+    int v = (int)idx[row];
+    if (v < 0 || v >= V) continue;
+    atomicAdd(dw + (int64_t)v * E + e, ldbf(dy + i));
+  }
+}
+
+// ----------------------------------------------------- softmax + NLL loss
+// x [N, C] bf16 -> prob fp32, per-row loss accumulated into loss[0],
+// valid count into count[0].  One wave per row.
+
+__global__ void softmax_loss_fwd_kernel(
+    const u16* __restrict__ x, const float* __restrict__ label,
+    float* __restrict__ prob, float* __restrict__ loss,
+    int* __restrict__ count, int64_t nrows, int C, int ignore,
+    int has_ignore) {
+  int64_t row = (int64_t)blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+  int lane = threadIdx.x & 63;
+  if (row >= nrows) return;
+  const u16* xp = x + row * C;
+  float mx = -3.4e38f;
+  for (int c = lane; c < C; c += 64) mx = fmaxf(mx, ldbf(xp + c));
+#pragma unroll
+  for (int off = 32; off; off >>= 1)
+    mx = fmaxf(mx, __shfl_down(mx, off));
+  mx = __shfl(mx, 0);
+  float sum = 0.f;
+  for (int c = lane; c < C; c += 64) sum += __expf(ldbf(xp + c) - mx);
+#pragma unroll
+  for (int off = 32; off; off >>= 1) sum += __shfl_down(sum, off);
+  sum = __shfl(sum, 0);
+  float inv = 1.f / sum;
+  float* pp = prob + row * C;
+  for (int c = lane; c < C; c += 64)
+    pp[c] = __expf(ldbf(xp + c) - mx) * inv;
+  if (lane == 0) {
+    int lab = (int)label[row];
+    if (!(has_ignore && lab == ignore)) {
+      float lp = ldbf(xp + lab) - mx - __logf(sum);
+      atomicAdd(loss, -lp);
+      atomicAdd(count, 1);
+    }
+  }
+}
+
+__global__ void softmax_loss_bwd_kernel(
+    const float* __restrict__ prob, const float* __restrict__ label,
+    u16* __restrict__ dx, float scale, int64_t nrows, int C, int ignore,
+    int has_ignore) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < nrows * C; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t row = i / C;
+    int c = i % C;
+    int lab = (int)label[row];
+    float v;
+    if (has_ignore && lab == ignore) {
+      v = 0.f;
+    } else {
+      v = prob[i] - (c == lab ? 1.f : 0.f);
+    }
+    stbf(dx + i, v * scale);
+  }
+}
+
+// ------------------------------------------------------------------- hosts
+
+static int nb(int64_t total) {
+  return (int)hmin<int64_t>(4096, (total + 255) / 256);
+}
+
+void relu_fwd(const void* x, void* y, float slope, int64_t n,
+              hipStream_t stream) {
+  int64_t n8 = n / 8;
+ hipLaunchKernelGGL(( relu_fwd_kernel), dim3(nb(hmax<int64_t>(n8, 1))), dim3(256), 0, stream, 
+      (const u16*)x, (u16*)y, slope, n8, n);
+}
+
+void relu_bwd(const void* y, const void* dy, void* dx, float slope, int64_t n,
+              hipStream_t stream) {
+  int64_t n8 = n / 8;
+ hipLaunchKernelGGL(( relu_bwd_kernel), dim3(nb(hmax<int64_t>(n8, 1))), dim3(256), 0, stream, 
+      (const u16*)y, (const u16*)dy, (u16*)dx, slope, n8, n);
+}
+
+void dropout_fwd(const void* x, void* y, void* mask, float ratio,
+                 uint64_t seed, int64_t n, hipStream_t stream) {
+  float keep = 1.f - ratio;
+ hipLaunchKernelGGL(( dropout_fwd_kernel), dim3(nb(n)), dim3(256), 0, stream, 
+      (const u16*)x, (u16*)y, (u16*)mask, keep, 1.f / keep, seed, n);
+}
+
+void mul_bf16(const void* a, const void* b, void* y, int64_t n,
+              hipStream_t stream) {
+ hipLaunchKernelGGL(( mul_bf16_kernel), dim3(nb(n)), dim3(256), 0, stream, 
+      (const u16*)a, (const u16*)b, (u16*)y, n);
+}
+
+void sgd_update(float* p, const float* g, float* v, float lr, float mu,
+                float wd, int64_t n, hipStream_t stream) {
+  int64_t n4 = (n + 3) / 4;
+ hipLaunchKernelGGL(( sgd_update_kernel), dim3(nb(n4)), dim3(256), 0, stream, p, g, v, lr, mu, wd, n4, n);
+}
+
+void colsum(const void* in, float* out, int64_t rows, int cols, int ld,
+            hipStream_t stream) {
+  dim3 grid((cols + 255) / 256, (unsigned)hmin<int64_t>(rows, 64));
+ hipLaunchKernelGGL(( colsum_kernel), dim3(grid), dim3(256), 0, stream, (const u16*)in, out, rows, cols, ld);
+}
+
+void lstm_unit_fwd(const float* c_prev, const void* gates, const void* cont,
+                   float* c_out, void* h_out, float* act, int64_t n, int H,
+                   hipStream_t stream) {
+ hipLaunchKernelGGL(( lstm_unit_fwd_kernel), dim3(nb(n)), dim3(256), 0, stream, 
+      c_prev, (const u16*)gates, (const u16*)cont, c_out, (u16*)h_out, act,
+      n, H);
+}
+
+void lstm_unit_bwd(const float* c_prev, const float* c_out, const float* act,
+                   const void* cont, const float* dc_next, const void* dh,
+                   float* dc_prev, void* dgates, int64_t n, int H,
+                   hipStream_t stream) {
+ hipLaunchKernelGGL(( lstm_unit_bwd_kernel), dim3(nb(n)), dim3(256), 0, stream, 
+      c_prev, c_out, act, (const u16*)cont, dc_next, (const u16*)dh, dc_prev,
+      (u16*)dgates, n, H);
+}
+
+void embed_fwd(const float* idx, const void* w, void* y, int64_t n, int E,
+               int V, hipStream_t stream) {
+ hipLaunchKernelGGL(( embed_fwd_kernel), dim3(nb(n * E)), dim3(256), 0, stream, 
+      idx, (const u16*)w, (u16*)y, n, E, V);
+}
+
+void embed_bwd(const float* idx, const void* dy, float* dw, int64_t n, int E,
+               int V, hipStream_t stream) {
+ hipLaunchKernelGGL(( embed_bwd_kernel), dim3(nb(n * E)), dim3(256), 0, stream, 
+      idx, (const u16*)dy, dw, n, E, V);
+}
+
+void softmax_loss_fwd(const void* x, const float* label, float* prob,
+                      float* loss, int* count, int64_t nrows, int C,
+                      int ignore, bool has_ignore, hipStream_t stream) {
+  int waves_per_block = 4;
+  int64_t blocks = (nrows + waves_per_block - 1) / waves_per_block;
+  unsigned grid = (unsigned)hmin<int64_t>(blocks, (int64_t)1 << 30);
+ hipLaunchKernelGGL(( softmax_loss_fwd_kernel), dim3(grid), dim3(waves_per_block * 64), 0, stream, 
+      (const u16*)x, label, prob, loss, count, nrows, C, ignore,
+      has_ignore ? 1 : 0);
+}
+
+void softmax_loss_bwd(const float* prob, const float* label, void* dx,
+                      float scale, int64_t nrows, int C, int ignore,
+                      bool has_ignore, hipStream_t stream) {
+ hipLaunchKernelGGL(( softmax_loss_bwd_kernel), dim3(nb(nrows * C)), dim3(256), 0, stream, 
+      prob, label, (u16*)dx, scale, nrows, C, ignore, has_ignore ? 1 : 0);
+}
+
+}  // namespace cosamd

@@ -1,0 +1,281 @@
+// Torch-facing bindings for the gfx950 kernel library.  Shape logic lives in
+// Python (ops/gpu.py); this layer validates tensors and launches kernels on
+// the current HIP stream.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <cstdint>
+
+typedef struct ihipStream_t* hipStream_t;
+
+namespace cosamd {
+
+void gemm_bf16(const void* A, const void* B, void* C, const float* bias,
+               int M, int N, int K, int lda, int ldb, int ldc,
+               bool trans_a, bool trans_b, int store_mode, int splitk,
+               bool relu, float alpha, hipStream_t stream);
+void im2col_nhwc(const void* x, void* col, int N, int H, int W, int C,
+                 int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
+                 int dil, int Kpad, int c0, int Ct, hipStream_t stream);
+void col2im_nhwc(const void* dcol, void* dx, int N, int H, int W, int C,
+                 int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
+                 int dil, int Kpad, int c0, int Ct, hipStream_t stream);
+void maxpool_fwd(const void* x, void* y, int* idx, int N, int H, int W, int C,
+                 int P, int Q, int kh, int kw, int sh, int sw, int ph, int pw,
+                 hipStream_t stream);
+void maxpool_bwd(const void* dy, const int* idx, void* dx, int N, int H,
+                 int W, int C, int P, int Q, int kh, int kw, int sh, int sw,
+                 int ph, int pw, hipStream_t stream);
+void avgpool_fwd(const void* x, void* y, int N, int H, int W, int C,
+                 int P, int Q, int kh, int kw, int sh, int sw, int ph, int pw,
+                 hipStream_t stream);
+void avgpool_bwd(const void* dy, void* dx, int N, int H, int W, int C,
+                 int P, int Q, int kh, int kw, int sh, int sw, int ph, int pw,
+                 hipStream_t stream);
+void lrn_fwd(const void* x, void* y, float* scale, int64_t npix, int C,
+             int local_size, float alpha, float beta, float k,
+             hipStream_t stream);
+void lrn_bwd(const void* x, const void* y, const float* scale, const void* dy,
+             void* dx, int64_t npix, int C, int local_size, float alpha,
+             float beta, hipStream_t stream);
+void relu_fwd(const void* x, void* y, float slope, int64_t n,
+              hipStream_t stream);
+void relu_bwd(const void* y, const void* dy, void* dx, float slope, int64_t n,
+              hipStream_t stream);
+void dropout_fwd(const void* x, void* y, void* mask, float ratio,
+                 uint64_t seed, int64_t n, hipStream_t stream);
+void mul_bf16(const void* a, const void* b, void* y, int64_t n,
+              hipStream_t stream);
+void sgd_update(float* p, const float* g, float* v, float lr, float mu,
+                float wd, int64_t n, hipStream_t stream);
+void colsum(const void* in, float* out, int64_t rows, int cols, int ld,
+            hipStream_t stream);
+void lstm_unit_fwd(const float* c_prev, const void* gates, const void* cont,
+                   float* c_out, void* h_out, float* act, int64_t n, int H,
+                   hipStream_t stream);
+void lstm_unit_bwd(const float* c_prev, const float* c_out, const float* act,
+                   const void* cont, const float* dc_next, const void* dh,
+                   float* dc_prev, void* dgates, int64_t n, int H,
+                   hipStream_t stream);
+void embed_fwd(const float* idx, const void* w, void* y, int64_t n, int E,
+               int V, hipStream_t stream);
+void embed_bwd(const float* idx, const void* dy, float* dw, int64_t n, int E,
+               int V, hipStream_t stream);
+void softmax_loss_fwd(const void* x, const float* label, float* prob,
+                      float* loss, int* count, int64_t nrows, int C,
+                      int ignore, bool has_ignore, hipStream_t stream);
+void softmax_loss_bwd(const float* prob, const float* label, void* dx,
+                      float scale, int64_t nrows, int C, int ignore,
+                      bool has_ignore, hipStream_t stream);
+
+}  // namespace cosamd
+
+namespace {
+
+using at::Tensor;
+
+hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+#define CHECK_BF16(t) \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16")
+#define CHECK_F32(t) \
+  TORCH_CHECK((t).scalar_type() == at::kFloat, #t " must be fp32")
+#define CHECK_CUDA(t) TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
+
+void py_gemm(Tensor A, Tensor B, Tensor C, c10::optional<Tensor> bias,
+             int64_t M, int64_t N, int64_t K, int64_t lda, int64_t ldb,
+             int64_t ldc, bool trans_a, bool trans_b, int64_t store_mode,
+             int64_t splitk, bool relu, double alpha) {
+  CHECK_CUDA(A); CHECK_BF16(A); CHECK_BF16(B);
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    CHECK_F32(bias.value());
+    bptr = bias.value().data_ptr<float>();
+  }
+  cosamd::gemm_bf16(
+      A.data_ptr(), B.data_ptr(), C.data_ptr(), bptr, (int)M, (int)N, (int)K, (int)lda, (int)ldb, (int)ldc,
+      trans_a, trans_b, (int)store_mode, (int)splitk, relu, (float)alpha,
+      cur_stream());
+}
+
+void py_im2col(Tensor x, Tensor col, int64_t N, int64_t H, int64_t W,
+               int64_t C, int64_t P, int64_t Q, int64_t R, int64_t S,
+               int64_t sh, int64_t sw, int64_t ph, int64_t pw, int64_t dil,
+               int64_t Kpad, int64_t c0, int64_t Ct) {
+  CHECK_CUDA(x); CHECK_BF16(x); CHECK_BF16(col);
+  cosamd::im2col_nhwc(x.data_ptr(), col.data_ptr(), N, H, W, C, P, Q, R, S,
+                      sh, sw, ph, pw, dil, Kpad, c0, Ct, cur_stream());
+}
+
+void py_col2im(Tensor dcol, Tensor dx, int64_t N, int64_t H, int64_t W,
+               int64_t C, int64_t P, int64_t Q, int64_t R, int64_t S,
+               int64_t sh, int64_t sw, int64_t ph, int64_t pw, int64_t dil,
+               int64_t Kpad, int64_t c0, int64_t Ct) {
+  CHECK_CUDA(dcol); CHECK_BF16(dcol); CHECK_BF16(dx);
+  cosamd::col2im_nhwc(dcol.data_ptr(), dx.data_ptr(), N, H, W, C, P, Q, R, S,
+                      sh, sw, ph, pw, dil, Kpad, c0, Ct, cur_stream());
+}
+
+void py_maxpool_fwd(Tensor x, Tensor y, Tensor idx, int64_t N, int64_t H,
+                    int64_t W, int64_t C, int64_t P, int64_t Q, int64_t kh,
+                    int64_t kw, int64_t sh, int64_t sw, int64_t ph,
+                    int64_t pw) {
+  CHECK_CUDA(x); CHECK_BF16(x);
+  cosamd::maxpool_fwd(x.data_ptr(), y.data_ptr(), idx.data_ptr<int>(), N, H,
+                      W, C, P, Q, kh, kw, sh, sw, ph, pw, cur_stream());
+}
+
+void py_maxpool_bwd(Tensor dy, Tensor idx, Tensor dx, int64_t N, int64_t H,
+                    int64_t W, int64_t C, int64_t P, int64_t Q, int64_t kh,
+                    int64_t kw, int64_t sh, int64_t sw, int64_t ph,
+                    int64_t pw) {
+  cosamd::maxpool_bwd(dy.data_ptr(), idx.data_ptr<int>(), dx.data_ptr(), N, H,
+                      W, C, P, Q, kh, kw, sh, sw, ph, pw, cur_stream());
+}
+
+void py_avgpool_fwd(Tensor x, Tensor y, int64_t N, int64_t H, int64_t W,
+                    int64_t C, int64_t P, int64_t Q, int64_t kh, int64_t kw,
+                    int64_t sh, int64_t sw, int64_t ph, int64_t pw) {
+  cosamd::avgpool_fwd(x.data_ptr(), y.data_ptr(), N, H, W, C, P, Q, kh, kw,
+                      sh, sw, ph, pw, cur_stream());
+}
+
+void py_avgpool_bwd(Tensor dy, Tensor dx, int64_t N, int64_t H, int64_t W,
+                    int64_t C, int64_t P, int64_t Q, int64_t kh, int64_t kw,
+                    int64_t sh, int64_t sw, int64_t ph, int64_t pw) {
+  cosamd::avgpool_bwd(dy.data_ptr(), dx.data_ptr(), N, H, W, C, P, Q, kh, kw,
+                      sh, sw, ph, pw, cur_stream());
+}
+
+void py_lrn_fwd(Tensor x, Tensor y, Tensor scale, int64_t npix, int64_t C,
+                int64_t local_size, double alpha, double beta, double k) {
+  CHECK_F32(scale);
+  cosamd::lrn_fwd(x.data_ptr(), y.data_ptr(), scale.data_ptr<float>(), npix,
+                  C, local_size, alpha, beta, k, cur_stream());
+}
+
+void py_lrn_bwd(Tensor x, Tensor y, Tensor scale, Tensor dy, Tensor dx,
+                int64_t npix, int64_t C, int64_t local_size, double alpha,
+                double beta) {
+  cosamd::lrn_bwd(x.data_ptr(), y.data_ptr(), scale.data_ptr<float>(),
+                  dy.data_ptr(), dx.data_ptr(), npix, C, local_size, alpha,
+                  beta, cur_stream());
+}
+
+void py_relu_fwd(Tensor x, Tensor y, double slope) {
+  CHECK_BF16(x);
+  cosamd::relu_fwd(x.data_ptr(), y.data_ptr(), (float)slope, x.numel(),
+                   cur_stream());
+}
+
+void py_relu_bwd(Tensor y, Tensor dy, Tensor dx, double slope) {
+  cosamd::relu_bwd(y.data_ptr(), dy.data_ptr(), dx.data_ptr(), (float)slope,
+                   y.numel(), cur_stream());
+}
+
+void py_dropout_fwd(Tensor x, Tensor y, Tensor mask, double ratio,
+                    int64_t seed) {
+  cosamd::dropout_fwd(x.data_ptr(), y.data_ptr(), mask.data_ptr(), ratio,
+                      (uint64_t)seed, x.numel(), cur_stream());
+}
+
+void py_mul(Tensor a, Tensor b, Tensor y) {
+  cosamd::mul_bf16(a.data_ptr(), b.data_ptr(), y.data_ptr(), a.numel(),
+                   cur_stream());
+}
+
+void py_sgd_update(Tensor p, Tensor g, Tensor v, double lr, double mu,
+                   double wd) {
+  CHECK_F32(p); CHECK_F32(g); CHECK_F32(v);
+  cosamd::sgd_update(p.data_ptr<float>(), g.data_ptr<float>(),
+                     v.data_ptr<float>(), lr, mu, wd, p.numel(),
+                     cur_stream());
+}
+
+void py_colsum(Tensor in, Tensor out, int64_t rows, int64_t cols,
+               int64_t ld) {
+  CHECK_F32(out);
+  cosamd::colsum(in.data_ptr(), out.data_ptr<float>(), rows, cols, ld,
+                 cur_stream());
+}
+
+void py_lstm_unit_fwd(Tensor c_prev, Tensor gates, Tensor cont, Tensor c_out,
+                      Tensor h_out, Tensor act) {
+  CHECK_F32(c_prev); CHECK_BF16(gates);
+  int64_t n = c_prev.numel();
+  int H = c_prev.size(-1);
+  cosamd::lstm_unit_fwd(c_prev.data_ptr<float>(), gates.data_ptr(),
+                        cont.data_ptr(), c_out.data_ptr<float>(),
+                        h_out.data_ptr(), act.data_ptr<float>(), n, H,
+                        cur_stream());
+}
+
+void py_lstm_unit_bwd(Tensor c_prev, Tensor c_out, Tensor act, Tensor cont,
+                      Tensor dc_next, Tensor dh, Tensor dc_prev,
+                      Tensor dgates) {
+  int64_t n = c_prev.numel();
+  int H = c_prev.size(-1);
+  cosamd::lstm_unit_bwd(c_prev.data_ptr<float>(), c_out.data_ptr<float>(),
+                        act.data_ptr<float>(), cont.data_ptr(),
+                        dc_next.data_ptr<float>(), dh.data_ptr(),
+                        dc_prev.data_ptr<float>(), dgates.data_ptr(), n, H,
+                        cur_stream());
+}
+
+void py_embed_fwd(Tensor idx, Tensor w, Tensor y, int64_t V) {
+  CHECK_F32(idx); CHECK_BF16(w);
+  cosamd::embed_fwd(idx.data_ptr<float>(), w.data_ptr(), y.data_ptr(),
+                    idx.numel(), w.size(1), V, cur_stream());
+}
+
+void py_embed_bwd(Tensor idx, Tensor dy, Tensor dw, int64_t V) {
+  CHECK_F32(dw);
+  cosamd::embed_bwd(idx.data_ptr<float>(), dy.data_ptr(),
+                    dw.data_ptr<float>(), idx.numel(), dw.size(1), V,
+                    cur_stream());
+}
+
+void py_softmax_loss_fwd(Tensor x, Tensor label, Tensor prob, Tensor loss,
+                         Tensor count, int64_t ignore, bool has_ignore) {
+  CHECK_BF16(x); CHECK_F32(label); CHECK_F32(prob);
+  cosamd::softmax_loss_fwd(x.data_ptr(), label.data_ptr<float>(),
+                           prob.data_ptr<float>(), loss.data_ptr<float>(),
+                           count.data_ptr<int>(), x.size(0), x.size(1),
+                           (int)ignore, has_ignore, cur_stream());
+}
+
+void py_softmax_loss_bwd(Tensor prob, Tensor label, Tensor dx, double scale,
+                         int64_t ignore, bool has_ignore) {
+  cosamd::softmax_loss_bwd(prob.data_ptr<float>(), label.data_ptr<float>(),
+                           dx.data_ptr(), scale, prob.size(0), prob.size(1),
+                           (int)ignore, has_ignore, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gemm", &py_gemm);
+  m.def("im2col", &py_im2col);
+  m.def("col2im", &py_col2im);
+  m.def("maxpool_fwd", &py_maxpool_fwd);
+  m.def("maxpool_bwd", &py_maxpool_bwd);
+  m.def("avgpool_fwd", &py_avgpool_fwd);
+  m.def("avgpool_bwd", &py_avgpool_bwd);
+  m.def("lrn_fwd", &py_lrn_fwd);
+  m.def("lrn_bwd", &py_lrn_bwd);
+  m.def("relu_fwd", &py_relu_fwd);
+  m.def("relu_bwd", &py_relu_bwd);
+  m.def("dropout_fwd", &py_dropout_fwd);
+  m.def("mul", &py_mul);
+  m.def("sgd_update", &py_sgd_update);
+  m.def("colsum", &py_colsum);
+  m.def("lstm_unit_fwd", &py_lstm_unit_fwd);
+  m.def("lstm_unit_bwd", &py_lstm_unit_bwd);
+  m.def("embed_fwd", &py_embed_fwd);
+  m.def("embed_bwd", &py_embed_bwd);
+  m.def("softmax_loss_fwd", &py_softmax_loss_fwd);
+  m.def("softmax_loss_bwd", &py_softmax_loss_bwd);
+}

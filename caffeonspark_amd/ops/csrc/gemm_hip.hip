@@ -1,0 +1,252 @@
+#include "hip/hip_runtime.h"
+// bf16 MFMA GEMM for gfx950 (CDNA4) — the compute core of conv (im2col) and
+// inner-product layers, forward and backward.
+//
+// C[M,N] = A' @ B' with A',B' presented as [rows][K] panels:
+//   TRANS_A=false: A global layout [M][K] (lda = row stride)
+//   TRANS_A=true : A global layout [K][M] (lda = row stride of the K-major
+//                  matrix) — staged transposed into LDS
+// (same for B with N rows).  All four combinations are used by the layer
+// catalog (fc fwd = NT-direct/direct, fc dx = direct/trans, dw = trans/trans).
+//
+// Structure: 128x128 tile, BK=32, 4 waves (2x2), 16x16x32 bf16 MFMA with
+// fp32 accumulation; interior aligned tiles stage A/B via
+// __builtin_amdgcn_global_load_lds (16 B/lane direct-to-LDS DMA); edge or
+// unaligned tiles take a guarded staging path.  XCD-aware block swizzle for
+// L2 locality; optional split-K with fp32 atomic reduction (STORE_MODE 2)
+// for the skinny dw GEMMs.  Epilogue fuses per-column bias add + ReLU.
+//
+// Replaces the reference's cuBLAS calls inside conv/ip layers (SURVEY.md
+// §3.6 rows "Convolution fwd/bwd", "InnerProduct") with a hand-written
+// CDNA4 kernel per the rebuild's north star.
+
+#include "common.h"
+
+namespace cosamd {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef unsigned short short8v __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+constexpr int BM = 128, BN = 128, BK = 32;
+constexpr int NTHREADS = 256;
+
+// ---------------------------------------------------------------- staging
+
+// direct ([rows][K]) guarded staging: vectorized 8-wide along K when the
+// 16-byte slot is fully in-bounds, scalar otherwise; zero-fills the rest.
+__device__ __forceinline__ void stage_direct_guarded(
+    bf16* lds_, const bf16* g_, int row0, int rows, int ld,
+    int k0, int kend, int tid) {
+  auto* lds = reinterpret_cast<unsigned short*>(lds_);
+  auto* g = reinterpret_cast<const unsigned short*>(g_);
+  // 4096 elements, 512 slots of 8; 256 threads -> 2 slots each
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    int slot = tid + it * NTHREADS;
+    int row = slot >> 2;            // 4 slots of 8 per 32-wide row
+    int kk = (slot & 3) * 8;
+    int grow = row0 + row;
+    int gk = k0 + kk;
+    unsigned short* dst = lds + row * BK + kk;
+    if (grow < rows && gk + 8 <= kend && ((ld | gk) % 8 == 0)) {
+      *reinterpret_cast<short8v*>(dst) =
+          *reinterpret_cast<const short8v*>(g + (int64_t)grow * ld + gk);
+    } else if (grow < rows) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dst[j] = (gk + j < kend) ? g[(int64_t)grow * ld + gk + j] : 0;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dst[j] = 0;
+    }
+  }
+}
+
+// transposed ([K][rows]) staging: vector load 8 contiguous along rows from
+// one K-row, scatter into LDS columns.
+__device__ __forceinline__ void stage_trans_guarded(
+    bf16* lds_, const bf16* g_, int row0, int rows, int ld,
+    int k0, int kend, int tid) {
+  auto* lds = reinterpret_cast<unsigned short*>(lds_);
+  auto* g = reinterpret_cast<const unsigned short*>(g_);
+  // 512 slots: slot -> (k, r0): 32 k-rows x 16 slots of 8 rows
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    int slot = tid + it * NTHREADS;
+    int kk = slot >> 4;             // 0..31
+    int r0 = (slot & 15) * 8;
+    int gk = k0 + kk;
+    unsigned short vals[8];
+    if (gk < kend && row0 + r0 + 8 <= rows && (ld | (row0 + r0)) % 8 == 0) {
+      short8v v = *reinterpret_cast<const short8v*>(
+          g + (int64_t)gk * ld + row0 + r0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[j] = v[j];
+    } else if (gk < kend) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        vals[j] = (row0 + r0 + j < rows)
+                      ? g[(int64_t)gk * ld + row0 + r0 + j] : 0;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[j] = 0;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) lds[(r0 + j) * BK + kk] = vals[j];
+  }
+}
+
+// fast path: interior tile, aligned — direct-to-LDS DMA, 16 B per lane.
+__device__ __forceinline__ void stage_direct_fast(
+    bf16* lds, const bf16* g, int row0, int ld, int k0, int tid) {
+  int wave = tid >> 6, lane = tid & 63;
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    int chunk = wave * 2 + it;           // 8 chunks of 512 elements
+    int idx = chunk * 512 + lane * 8;    // linear element index in tile
+    int row = idx >> 5;                  // /32
+    int kk = idx & 31;
+    auto* gp = (const __attribute__((address_space(1))) unsigned int*)(
+        g + (int64_t)(row0 + row) * ld + k0 + kk);
+    auto* lp = (__attribute__((address_space(3))) unsigned int*)(
+        lds + chunk * 512);
+    __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+  }
+}
+
+// ------------------------------------------------------------------ kernel
+
+template <bool TRANS_A, bool TRANS_B, int STORE_MODE>
+__global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    void* __restrict__ C, const float* __restrict__ bias,
+    int M, int N, int K, int lda, int ldb, int ldc,
+    int ksplit, int relu, float alpha) {
+  __shared__ bf16 As[BM * BK];
+  __shared__ bf16 Bs[BN * BK];
+
+  int mblocks = (M + BM - 1) / BM;
+  int nblocks = (N + BN - 1) / BN;
+  int bid = xcd_swizzle(blockIdx.x, mblocks * nblocks);
+  int bm = bid / nblocks, bn = bid % nblocks;
+  int tile_m = bm * BM, tile_n = bn * BN;
+
+  int k_begin = blockIdx.y * ksplit;
+  int k_end = min(K, k_begin + ksplit);
+
+  int tid = threadIdx.x;
+  int wave = tid >> 6, lane = tid & 63;
+  int wm = wave >> 1, wn = wave & 1;     // 2x2 wave grid, 64x64 each
+  int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+
+  f32x4 acc[4][4] = {};
+
+  bool a_fast = !TRANS_A && (tile_m + BM <= M) && (lda % 8 == 0) &&
+                ((k_end - k_begin) % BK == 0) && (k_begin % 8 == 0);
+  bool b_fast = !TRANS_B && (tile_n + BN <= N) && (ldb % 8 == 0) &&
+                ((k_end - k_begin) % BK == 0) && (k_begin % 8 == 0);
+
+  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
+    if (a_fast) {
+      stage_direct_fast(As, A, tile_m, lda, k0, tid);
+    } else if (TRANS_A) {
+      stage_trans_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
+    } else {
+      stage_direct_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
+    }
+    if (b_fast) {
+      stage_direct_fast(Bs, B, tile_n, ldb, k0, tid);
+    } else if (TRANS_B) {
+      stage_trans_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
+    } else {
+      stage_direct_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
+    }
+    __syncthreads();
+
+    bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      afrag[f] = *reinterpret_cast<const bf16x8*>(
+          As + (wm * 64 + f * 16 + lrow) * BK + lk8);
+      bfrag[f] = *reinterpret_cast<const bf16x8*>(
+          Bs + (wn * 64 + f * 16 + lrow) * BK + lk8);
+    }
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
+
+    // second half of the K-tile? no: 16x16x32 consumes all BK=32 at once
+    __syncthreads();
+  }
+
+  // ------------------------------------------------------------- epilogue
+  int crow0 = tile_m + wm * 64 + ((lane >> 4) << 2);
+  int ccol0 = tile_n + wn * 64 + (lane & 15);
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int col = ccol0 + fn * 16;
+      if (col >= N) continue;
+      float badd = (bias != nullptr) ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow0 + fm * 16 + r;
+        if (row >= M) continue;
+        float v = acc[fm][fn][r] * alpha + badd;
+        if (relu && v < 0.f) v = 0.f;
+        int64_t off = (int64_t)row * ldc + col;
+        if (STORE_MODE == 0) {
+          reinterpret_cast<bf16*>(C)[off] = f2bf(v);
+        } else if (STORE_MODE == 1) {
+          reinterpret_cast<float*>(C)[off] = v;
+        } else {
+          atomicAdd(reinterpret_cast<float*>(C) + off, v);
+        }
+      }
+    }
+  }
+}
+
+// ------------------------------------------------------------------- host
+
+void gemm_bf16(const void* A_, const void* B_, void* C, const float* bias,
+               int M, int N, int K, int lda, int ldb, int ldc,
+               bool trans_a, bool trans_b, int store_mode, int splitk,
+               bool relu, float alpha, hipStream_t stream) {
+  const bf16* A = reinterpret_cast<const bf16*>(A_);
+  const bf16* B = reinterpret_cast<const bf16*>(B_);
+  int mblocks = (M + BM - 1) / BM, nblocks = (N + BN - 1) / BN;
+  splitk = max(1, splitk);
+  int ksplit = (K + splitk - 1) / splitk;
+  ksplit = ((ksplit + BK - 1) / BK) * BK;  // keep split boundaries BK-aligned
+  int zblocks = (K + ksplit - 1) / ksplit;
+  if (zblocks > 1 && store_mode != 2)
+    throw std::runtime_error("gemm: split-K requires atomic store mode");
+  dim3 grid(mblocks * nblocks, zblocks);
+  dim3 block(NTHREADS);
+
+#define COS_GEMM_CASE(TA, TB, SM)                                          \
+ hipLaunchKernelGGL(( gemm_kernel<TA, TB, SM>), dim3(grid), dim3(block), 0, stream,                       \
+      A, B, C, bias, M, N, K, lda, ldb, ldc, ksplit, relu ? 1 : 0, alpha)
+
+#define COS_GEMM_SM(TA, TB)                                                 \
+  do {                                                                      \
+    if (store_mode == 0)      COS_GEMM_CASE(TA, TB, 0);                     \
+    else if (store_mode == 1) COS_GEMM_CASE(TA, TB, 1);                     \
+    else                      COS_GEMM_CASE(TA, TB, 2);                     \
+  } while (0)
+
+  if (!trans_a && !trans_b)      COS_GEMM_SM(false, false);
+  else if (!trans_a && trans_b)  COS_GEMM_SM(false, true);
+  else if (trans_a && !trans_b)  COS_GEMM_SM(true, false);
+  else                           COS_GEMM_SM(true, true);
+#undef COS_GEMM_SM
+#undef COS_GEMM_CASE
+}
+
+}  // namespace cosamd

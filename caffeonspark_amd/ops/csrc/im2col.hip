@@ -1,0 +1,121 @@
+// NHWC im2col / col2im for the implicit-GEMM convolution path.
+//
+// col[(n*P+p)*Q+q][(r*S+s)*C + c] = x[n][h][w][c],  h = p*sh - ph + r, etc.
+// The K (column) dimension is padded to a multiple of 8 (Kpad) so the GEMM's
+// 16-byte direct-to-LDS staging path stays aligned; pad columns are zeroed.
+// NHWC makes every (r,s) slice a C-contiguous copy — coalesced on both ends.
+// col2im is the gather formulation (no atomics): each input pixel sums the
+// col entries that reference it.
+
+#include "common.h"
+
+namespace cosamd {
+
+typedef unsigned short u16;
+typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
+
+__global__ void im2col_nhwc_kernel(
+    const u16* __restrict__ x, u16* __restrict__ col,
+    int N, int H, int W, int C, int P, int Q,
+    int R, int S, int sh, int sw, int ph, int pw,
+    int dil, int Kpad, int c0, int Ct, int64_t total_rs) {
+  // one thread per (npq, r, s) copying C elements
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total_rs; i += (int64_t)gridDim.x * blockDim.x) {
+    int rs = i % (R * S);
+    int64_t npq = i / (R * S);
+    int q = npq % Q;
+    int p = (npq / Q) % P;
+    int n = npq / ((int64_t)P * Q);
+    int r = rs / S, s = rs % S;
+    int h = p * sh - ph + r * dil;
+    int w = q * sw - pw + s * dil;
+    u16* dst = col + npq * Kpad + rs * Ct;
+    if (h >= 0 && h < H && w >= 0 && w < W) {
+      const u16* src = x + (((int64_t)n * H + h) * W + w) * C + c0;
+      int c = 0;
+      for (; c + 8 <= Ct; c += 8)
+        *reinterpret_cast<u16x8*>(dst + c) =
+            *reinterpret_cast<const u16x8*>(src + c);
+      for (; c < Ct; ++c) dst[c] = src[c];
+    } else {
+      int c = 0;
+      for (; c + 8 <= Ct; c += 8)
+        *reinterpret_cast<u16x8*>(dst + c) = u16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      for (; c < Ct; ++c) dst[c] = 0;
+    }
+  }
+}
+
+__global__ void col_zero_pad_kernel(u16* __restrict__ col, int Kcol, int Kpad,
+                                    int64_t rows) {
+  int padw = Kpad - Kcol;
+  int64_t total = rows * padw;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t row = i / padw;
+    col[row * Kpad + Kcol + (i % padw)] = 0;
+  }
+}
+
+// dcol -> dx (gather, fp32 accumulate, bf16 out)
+__global__ void col2im_nhwc_kernel(
+    const u16* __restrict__ dcol, u16* __restrict__ dx,
+    int N, int H, int W, int C, int P, int Q,
+    int R, int S, int sh, int sw, int ph, int pw,
+    int dil, int Kpad, int c0, int Ct, int64_t total) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int c = i % Ct;
+    int w = (i / Ct) % W;
+    int h = (i / ((int64_t)Ct * W)) % H;
+    int n = i / ((int64_t)Ct * W * H);
+    float acc = 0.f;
+    for (int r = 0; r < R; ++r) {
+      int hp = h + ph - r * dil;
+      if (hp < 0 || hp % sh) continue;
+      int p = hp / sh;
+      if (p >= P) continue;
+      for (int s = 0; s < S; ++s) {
+        int wp = w + pw - s * dil;
+        if (wp < 0 || wp % sw) continue;
+        int q = wp / sw;
+        if (q >= Q) continue;
+        int64_t npq = ((int64_t)n * P + p) * Q + q;
+        u16 v = dcol[npq * Kpad + (r * S + s) * Ct + c];
+        acc += bf2f(*reinterpret_cast<const bf16*>(&v));
+      }
+    }
+    bf16 out = f2bf(acc);
+    dx[(((int64_t)n * H + h) * W + w) * C + c0 + c] =
+        *reinterpret_cast<u16*>(&out);
+  }
+}
+
+void im2col_nhwc(const void* x, void* col, int N, int H, int W, int C,
+                 int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
+                 int dil, int Kpad, int c0, int Ct, hipStream_t stream) {
+  int64_t total = (int64_t)N * P * Q * R * S;
+  int blocks = hmin<int64_t>(4096, (total + 255) / 256);
+  im2col_nhwc_kernel<<<blocks, 256, 0, stream>>>(
+      (const u16*)x, (u16*)col, N, H, W, C, P, Q, R, S, sh, sw, ph, pw,
+      dil, Kpad, c0, Ct, total);
+  int Kcol = R * S * Ct;
+  if (Kpad > Kcol) {
+    int64_t rows = (int64_t)N * P * Q;
+    int b2 = hmin<int64_t>(2048, (rows * (Kpad - Kcol) + 255) / 256);
+    col_zero_pad_kernel<<<b2, 256, 0, stream>>>((u16*)col, Kcol, Kpad, rows);
+  }
+}
+
+void col2im_nhwc(const void* dcol, void* dx, int N, int H, int W, int C,
+                 int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
+                 int dil, int Kpad, int c0, int Ct, hipStream_t stream) {
+  int64_t total = (int64_t)N * H * W * Ct;
+  int blocks = hmin<int64_t>(8192, (total + 255) / 256);
+  col2im_nhwc_kernel<<<blocks, 256, 0, stream>>>(
+      (const u16*)dcol, (u16*)dx, N, H, W, C, P, Q, R, S, sh, sw, ph, pw,
+      dil, Kpad, c0, Ct, total);
+}
+
+}  // namespace cosamd

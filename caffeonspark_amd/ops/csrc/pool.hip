@@ -1,0 +1,166 @@
+// NHWC pooling (MAX with argmax, AVE with caffe divisor semantics).
+// Caffe semantics (SURVEY.md §3.6 "Pooling"): ceil-mode output dims; AVE
+// divides by the window clipped at the *padded* boundary; MAX backward
+// routes dy through the recorded argmax (gather formulation, no atomics).
+
+#include "common.h"
+
+namespace cosamd {
+
+typedef unsigned short u16;
+
+__global__ void maxpool_fwd_kernel(
+    const u16* __restrict__ x, u16* __restrict__ y, int* __restrict__ idx,
+    int N, int H, int W, int C, int P, int Q,
+    int kh, int kw, int sh, int sw, int ph, int pw, int64_t total) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int c = i % C;
+    int q = (i / C) % Q;
+    int p = (i / ((int64_t)C * Q)) % P;
+    int n = i / ((int64_t)C * Q * P);
+    int h0 = p * sh - ph, w0 = q * sw - pw;
+    float best = -3.4e38f;
+    int best_hw = 0;
+    for (int r = 0; r < kh; ++r) {
+      int h = h0 + r;
+      if (h < 0 || h >= H) continue;
+      for (int s = 0; s < kw; ++s) {
+        int w = w0 + s;
+        if (w < 0 || w >= W) continue;
+        u16 v = x[(((int64_t)n * H + h) * W + w) * C + c];
+        float f = bf2f(*reinterpret_cast<const bf16*>(&v));
+        if (f > best) { best = f; best_hw = h * W + w; }
+      }
+    }
+    bf16 out = f2bf(best);
+    y[i] = *reinterpret_cast<u16*>(&out);
+    idx[i] = best_hw;
+  }
+}
+
+__global__ void maxpool_bwd_kernel(
+    const u16* __restrict__ dy, const int* __restrict__ idx,
+    u16* __restrict__ dx,
+    int N, int H, int W, int C, int P, int Q,
+    int kh, int kw, int sh, int sw, int ph, int pw, int64_t total) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int c = i % C;
+    int w = (i / C) % W;
+    int h = (i / ((int64_t)C * W)) % H;
+    int n = i / ((int64_t)C * W * H);
+    int hw = h * W + w;
+    float acc = 0.f;
+    // windows (p,q) that can contain (h,w)
+    int plo = max(0, (h + ph - kh + sh) / sh), phi = min(P - 1, (h + ph) / sh);
+    int qlo = max(0, (w + pw - kw + sw) / sw), qhi = min(Q - 1, (w + pw) / sw);
+    for (int p = plo; p <= phi; ++p)
+      for (int q = qlo; q <= qhi; ++q) {
+        int64_t o = (((int64_t)n * P + p) * Q + q) * C + c;
+        if (idx[o] == hw) {
+          u16 v = dy[o];
+          acc += bf2f(*reinterpret_cast<const bf16*>(&v));
+        }
+      }
+    bf16 out = f2bf(acc);
+    dx[i] = *reinterpret_cast<u16*>(&out);
+  }
+}
+
+__global__ void avgpool_fwd_kernel(
+    const u16* __restrict__ x, u16* __restrict__ y,
+    int N, int H, int W, int C, int P, int Q,
+    int kh, int kw, int sh, int sw, int ph, int pw, int64_t total) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int c = i % C;
+    int q = (i / C) % Q;
+    int p = (i / ((int64_t)C * Q)) % P;
+    int n = i / ((int64_t)C * Q * P);
+    int h0 = p * sh - ph, w0 = q * sw - pw;
+    // caffe: divisor = window clipped at the padded boundary
+    int hend = min(h0 + kh, H + ph), wend = min(w0 + kw, W + pw);
+    int pool_size = (hend - h0) * (wend - w0);
+    float acc = 0.f;
+    for (int h = max(h0, 0); h < min(hend, H); ++h)
+      for (int w = max(w0, 0); w < min(wend, W); ++w) {
+        u16 v = x[(((int64_t)n * H + h) * W + w) * C + c];
+        acc += bf2f(*reinterpret_cast<const bf16*>(&v));
+      }
+    bf16 out = f2bf(acc / pool_size);
+    y[i] = *reinterpret_cast<u16*>(&out);
+  }
+}
+
+__global__ void avgpool_bwd_kernel(
+    const u16* __restrict__ dy, u16* __restrict__ dx,
+    int N, int H, int W, int C, int P, int Q,
+    int kh, int kw, int sh, int sw, int ph, int pw, int64_t total) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int c = i % C;
+    int w = (i / C) % W;
+    int h = (i / ((int64_t)C * W)) % H;
+    int n = i / ((int64_t)C * W * H);
+    float acc = 0.f;
+    int plo = max(0, (h + ph - kh + sh) / sh), phi = min(P - 1, (h + ph) / sh);
+    int qlo = max(0, (w + pw - kw + sw) / sw), qhi = min(Q - 1, (w + pw) / sw);
+    for (int p = plo; p <= phi; ++p) {
+      int h0 = p * sh - ph;
+      int hend = min(h0 + kh, H + ph);
+      for (int q = qlo; q <= qhi; ++q) {
+        int w0 = q * sw - pw;
+        int wend = min(w0 + kw, W + pw);
+        int pool_size = (hend - h0) * (wend - w0);
+        int64_t o = (((int64_t)n * P + p) * Q + q) * C + c;
+        u16 v = dy[o];
+        acc += bf2f(*reinterpret_cast<const bf16*>(&v)) / pool_size;
+      }
+    }
+    bf16 out = f2bf(acc);
+    dx[i] = *reinterpret_cast<u16*>(&out);
+  }
+}
+
+static int nblocks_for(int64_t total) {
+  return (int)hmin<int64_t>(8192, (total + 255) / 256);
+}
+
+void maxpool_fwd(const void* x, void* y, int* idx, int N, int H, int W, int C,
+                 int P, int Q, int kh, int kw, int sh, int sw, int ph, int pw,
+                 hipStream_t stream) {
+  int64_t total = (int64_t)N * P * Q * C;
+  maxpool_fwd_kernel<<<nblocks_for(total), 256, 0, stream>>>(
+      (const u16*)x, (u16*)y, idx, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw,
+      total);
+}
+
+void maxpool_bwd(const void* dy, const int* idx, void* dx, int N, int H,
+                 int W, int C, int P, int Q, int kh, int kw, int sh, int sw,
+                 int ph, int pw, hipStream_t stream) {
+  int64_t total = (int64_t)N * H * W * C;
+  maxpool_bwd_kernel<<<nblocks_for(total), 256, 0, stream>>>(
+      (const u16*)dy, idx, (u16*)dx, N, H, W, C, P, Q, kh, kw, sh, sw,
+      ph, pw, total);
+}
+
+void avgpool_fwd(const void* x, void* y, int N, int H, int W, int C,
+                 int P, int Q, int kh, int kw, int sh, int sw, int ph, int pw,
+                 hipStream_t stream) {
+  int64_t total = (int64_t)N * P * Q * C;
+  avgpool_fwd_kernel<<<nblocks_for(total), 256, 0, stream>>>(
+      (const u16*)x, (u16*)y, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw,
+      total);
+}
+
+void avgpool_bwd(const void* dy, void* dx, int N, int H, int W, int C,
+                 int P, int Q, int kh, int kw, int sh, int sw, int ph, int pw,
+                 hipStream_t stream) {
+  int64_t total = (int64_t)N * H * W * C;
+  avgpool_bwd_kernel<<<nblocks_for(total), 256, 0, stream>>>(
+      (const u16*)dy, (u16*)dx, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw,
+      total);
+}
+
+}  // namespace cosamd

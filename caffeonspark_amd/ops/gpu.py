@@ -1,0 +1,418 @@
+"""GPU op implementations: shape/layout glue around the gfx950 kernels.
+
+Layout strategy (MI355X-first): conv-stack activations live in NHWC
+(torch channels_last) so the implicit-GEMM epilogue writes coalesced rows
+and LRN's channel window is contiguous; weights are repacked [K,R,S,C/g]
+bf16 per forward (small).  The GEMM K dimension is padded to a multiple of
+8 so the 16-byte direct-to-LDS staging path stays aligned.  Compute dtype
+on GPU is bf16 with fp32 accumulation (fp32 master weights live in the
+solver); torch is used only for small glue (casts, pads, permutes).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+
+from . import reference
+
+_ext = None
+
+
+def _pad8(k: int) -> int:
+    return (k + 7) // 8 * 8
+
+
+def _cl(x: torch.Tensor) -> torch.Tensor:
+    """channels_last contiguous view of a 4-D activation."""
+    return x.contiguous(memory_format=torch.channels_last)
+
+
+def _as_bf16(t: torch.Tensor) -> torch.Tensor:
+    return t if t.dtype == torch.bfloat16 else t.to(torch.bfloat16)
+
+
+def _check_bf16(x, name):
+    if x.dtype != torch.bfloat16:
+        raise RuntimeError(
+            f"GPU path computes in bf16; got {x.dtype} for {name}. "
+            "Build the net with dtype=torch.bfloat16 on GPU.")
+
+
+def _splitk_for(mb: int, nb: int, K: int) -> int:
+    """Pick a split-K factor that fills the 256-CU chip (~2048 blocks)."""
+    tiles = mb * nb
+    if tiles >= 1024 or K < 1024:
+        return 1
+    return max(1, min(K // 256, 2048 // max(1, tiles)))
+
+
+def _gemm(A, B, C, bias, M, N, K, lda, ldb, ldc, ta, tb, store, splitk,
+          relu=False, alpha=1.0):
+    _ext.gemm(A, B, C, bias, M, N, K, lda, ldb, ldc, ta, tb, store, splitk,
+              relu, alpha)
+
+
+# ----------------------------------------------------------------- conv2d
+
+def _conv_out_dim(h, k, s, p, dil):
+    eff = (k - 1) * dil + 1
+    return (h + 2 * p - eff) // s + 1
+
+
+def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None):
+    _check_bf16(x, "conv input")
+    sh, sw = stride
+    ph, pw = pad
+    dil = dilation[0]
+    N, C, H, W = x.shape
+    Kout, Cg, R, S = w.shape
+    P = _conv_out_dim(H, R, sh, ph, dil)
+    Q = _conv_out_dim(W, S, sw, pw, dil)
+    G = groups
+    Kg = Kout // G
+    Kcol = R * S * Cg
+    Kpad = _pad8(Kcol)
+
+    xl = _cl(x)
+    # weight repack: [K, Cg, R, S] -> bf16 [K, R, S, Cg] padded to Kpad
+    wr = _as_bf16(w).permute(0, 2, 3, 1).contiguous().reshape(Kout, Kcol)
+    if Kpad != Kcol:
+        wr = torch.nn.functional.pad(wr, (0, Kpad - Kcol))
+    bias_f = b.float().contiguous() if b is not None else None
+
+    y = torch.empty((N, Kout, P, Q), dtype=torch.bfloat16, device=x.device) \
+        .contiguous(memory_format=torch.channels_last)
+    NPQ = N * P * Q
+    y2 = y.permute(0, 2, 3, 1).reshape(NPQ, Kout)  # NHWC flat alias (view)
+    col = torch.empty((G, NPQ, Kpad), dtype=torch.bfloat16, device=x.device)
+    for g in range(G):
+        _ext.im2col(xl, col[g], N, H, W, C, P, Q, R, S, sh, sw, ph, pw, dil,
+                    Kpad, g * Cg, Cg)
+        # C[npq, kout_g] — write into the column slice of NHWC y
+        _gemm(col[g], wr[g * Kg:(g + 1) * Kg], y2[:, g * Kg:],
+              bias_f[g * Kg:(g + 1) * Kg] if bias_f is not None else None,
+              NPQ, Kg, Kpad, Kpad, Kpad, Kout, False, False, 0, 1)
+    if ctx is not None:
+        ctx["col"] = col
+        ctx["shape"] = (N, C, H, W, P, Q, R, S, sh, sw, ph, pw, dil, G, Cg,
+                        Kg, Kpad, Kcol)
+        ctx["wr"] = wr
+    return y
+
+
+def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
+                    need_dx=True, need_dw=True, bias=True, ctx=None):
+    _check_bf16(dy, "conv dy")
+    if ctx is None or "col" not in ctx:
+        ctx = ctx if ctx is not None else {}
+        conv2d_forward(_cl(x), w, None, stride, pad, dilation, groups,
+                       ctx=ctx)
+    (N, C, H, W, P, Q, R, S, sh, sw, ph, pw, dil, G, Cg, Kg, Kpad,
+     Kcol) = ctx["shape"]
+    col, wr = ctx["col"], ctx["wr"]
+    Kout = Kg * G
+    NPQ = N * P * Q
+    dyl = _cl(dy)
+    dy2 = dyl.permute(0, 2, 3, 1).reshape(NPQ, Kout)
+
+    dx = dw = db = None
+    if need_dw:
+        dwp = torch.zeros((Kout, Kpad), dtype=torch.float32, device=dy.device)
+        for g in range(G):
+            mb, nb = (Kg + 127) // 128, (Kpad + 127) // 128
+            sk = _splitk_for(mb, nb, NPQ)
+            _gemm(dy2[:, g * Kg:], col[g], dwp[g * Kg:],
+                  None, Kg, Kpad, NPQ, Kout, Kpad, Kpad, True, True, 2, sk)
+        dw = dwp[:, :Kcol].reshape(Kout, R, S, Cg).permute(0, 3, 1, 2) \
+            .contiguous()
+    if bias:
+        db = torch.zeros(Kout, dtype=torch.float32, device=dy.device)
+        _ext.colsum(dy2, db, NPQ, Kout, Kout)
+    if need_dx:
+        dcol = torch.empty((NPQ, Kpad), dtype=torch.bfloat16,
+                           device=dy.device)
+        dx = torch.empty((N, C, H, W), dtype=torch.bfloat16,
+                         device=dy.device) \
+            .contiguous(memory_format=torch.channels_last)
+        for g in range(G):
+            # dcol[npq, kpad] = dy_g[npq, kg] @ w_g[kg, kpad]
+            _gemm(dy2[:, g * Kg:], wr[g * Kg:(g + 1) * Kg], dcol, None,
+                  NPQ, Kpad, Kg, Kout, Kpad, Kpad, False, True, 0, 1)
+            _ext.col2im(dcol, dx, N, H, W, C, P, Q, R, S, sh, sw, ph, pw,
+                        dil, Kpad, g * Cg, Cg)
+    return dx, dw, db
+
+
+# ----------------------------------------------------------------- linear
+
+def fc_forward(x, w, b):
+    _check_bf16(x, "fc input")
+    x = x.contiguous()
+    wb = _as_bf16(w).contiguous()
+    M, K = x.shape
+    Nout = wb.shape[0]
+    y = torch.empty((M, Nout), dtype=torch.bfloat16, device=x.device)
+    bias_f = b.float().contiguous() if b is not None else None
+    _gemm(x, wb, y, bias_f, M, Nout, K, K, K, Nout, False, False, 0, 1)
+    return y
+
+
+def fc_backward(x, w, dy, need_dx=True, bias=True):
+    _check_bf16(dy, "fc dy")
+    x = x.contiguous()
+    dy = dy.contiguous()
+    wb = _as_bf16(w).contiguous()
+    M, K = x.shape
+    Nout = wb.shape[0]
+    dx = dw = db = None
+    if need_dx:
+        dx = torch.empty((M, K), dtype=torch.bfloat16, device=x.device)
+        _gemm(dy, wb, dx, None, M, K, Nout, Nout, K, K, False, True, 0, 1)
+    dwp = torch.zeros((Nout, K), dtype=torch.float32, device=x.device)
+    mb, nb = (Nout + 127) // 128, (K + 127) // 128
+    _gemm(dy, x, dwp, None, Nout, K, M, Nout, K, K, True, True, 2,
+          _splitk_for(mb, nb, M))
+    dw = dwp
+    if bias:
+        db = torch.zeros(Nout, dtype=torch.float32, device=x.device)
+        _ext.colsum(dy, db, M, Nout, Nout)
+    return dx, dw, db
+
+
+# ------------------------------------------------------------- activations
+
+def relu_forward(x, negative_slope=0.0):
+    y = torch.empty_like(x)
+    _ext.relu_fwd(x, y, negative_slope)
+    return y
+
+
+def relu_backward(y, dy, negative_slope=0.0):
+    dy = dy if dy.dtype == torch.bfloat16 else dy.to(torch.bfloat16)
+    dx = torch.empty_like(y)
+    _ext.relu_bwd(y, dy.reshape(y.shape), dx, negative_slope)
+    return dx
+
+
+def dropout_forward(x, ratio, generator=None):
+    y = torch.empty_like(x)
+    mask = torch.empty_like(x)
+    seed = int(torch.randint(0, 2 ** 62, (1,)).item())
+    _ext.dropout_fwd(x, y, mask, ratio, seed)
+    return y, mask
+
+
+def dropout_backward(mask, dy):
+    dx = torch.empty_like(mask)
+    _ext.mul(mask, dy.to(mask.dtype), dx)
+    return dx
+
+
+# ---------------------------------------------------------------- pooling
+
+def _pool_out(h, k, s, p):
+    # caffe ceil-mode with clamp
+    o = int(math.ceil((h + 2 * p - k) / s)) + 1
+    if p > 0 and (o - 1) * s >= h + p:
+        o -= 1
+    return o
+
+
+def maxpool_forward(x, kernel, stride, pad):
+    _check_bf16(x, "maxpool input")
+    N, C, H, W = x.shape
+    kh, kw = kernel
+    sh, sw = stride
+    ph, pw = pad
+    P, Q = _pool_out(H, kh, sh, ph), _pool_out(W, kw, sw, pw)
+    xl = _cl(x)
+    y = torch.empty((N, C, P, Q), dtype=x.dtype, device=x.device) \
+        .contiguous(memory_format=torch.channels_last)
+    idx = torch.empty((N, P, Q, C), dtype=torch.int32, device=x.device)
+    _ext.maxpool_fwd(xl, y, idx, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw)
+    return y, (idx, (kh, kw, sh, sw, ph, pw), (N, C, H, W))
+
+
+def maxpool_backward(x_shape, idx_pack, dy):
+    idx, (kh, kw, sh, sw, ph, pw), (N, C, H, W) = idx_pack
+    P, Q = dy.shape[2], dy.shape[3]
+    dx = torch.empty((N, C, H, W), dtype=dy.dtype, device=dy.device) \
+        .contiguous(memory_format=torch.channels_last)
+    _ext.maxpool_bwd(_cl(dy), idx, dx, N, H, W, C, P, Q, kh, kw, sh, sw,
+                     ph, pw)
+    return dx
+
+
+def avgpool_forward(x, kernel, stride, pad):
+    N, C, H, W = x.shape
+    kh, kw = kernel
+    sh, sw = stride
+    ph, pw = pad
+    P, Q = _pool_out(H, kh, sh, ph), _pool_out(W, kw, sw, pw)
+    xl = _cl(x)
+    y = torch.empty((N, C, P, Q), dtype=x.dtype, device=x.device) \
+        .contiguous(memory_format=torch.channels_last)
+    _ext.avgpool_fwd(xl, y, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw)
+    return y
+
+
+def avgpool_backward(x, kernel, stride, pad, dy):
+    N, C, H, W = x.shape
+    kh, kw = kernel
+    sh, sw = stride
+    ph, pw = pad
+    P, Q = dy.shape[2], dy.shape[3]
+    dx = torch.empty((N, C, H, W), dtype=dy.dtype, device=dy.device) \
+        .contiguous(memory_format=torch.channels_last)
+    _ext.avgpool_bwd(_cl(dy), dx, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw)
+    return dx
+
+
+def global_avgpool_forward(x):
+    N, C, H, W = x.shape
+    xl = _cl(x)
+    y = torch.empty((N, C, 1, 1), dtype=x.dtype, device=x.device) \
+        .contiguous(memory_format=torch.channels_last)
+    _ext.avgpool_fwd(xl, y, N, H, W, C, 1, 1, H, W, 1, 1, 0, 0)
+    return y
+
+
+def global_avgpool_backward(x_shape, dy):
+    N, C, H, W = x_shape
+    dx = torch.empty((N, C, H, W), dtype=dy.dtype, device=dy.device) \
+        .contiguous(memory_format=torch.channels_last)
+    _ext.avgpool_bwd(_cl(dy), dx, N, H, W, C, 1, 1, H, W, 1, 1, 0, 0)
+    return dx
+
+
+# -------------------------------------------------------------------- LRN
+
+def lrn_forward(x, local_size, alpha, beta, k):
+    _check_bf16(x, "lrn input")
+    N, C, H, W = x.shape
+    xl = _cl(x)
+    y = torch.empty_like(xl)
+    scale = torch.empty((N, H, W, C), dtype=torch.float32, device=x.device)
+    _ext.lrn_fwd(xl, y, scale, N * H * W, C, local_size, alpha, beta, k)
+    return y, (scale, xl)
+
+
+def lrn_backward(x, y, scale_pack, dy, local_size, alpha, beta):
+    scale, xl = scale_pack
+    N, C, H, W = x.shape
+    dx = torch.empty_like(xl)
+    _ext.lrn_bwd(xl, _cl(y), scale, _cl(dy), dx, N * H * W, C, local_size,
+                 alpha, beta)
+    return dx
+
+
+# ----------------------------------------------------------- softmax loss
+
+def softmax_loss_forward(x, label, ignore_label, axis=1):
+    _check_bf16(x, "softmax loss input")
+    C = x.shape[axis]
+    if x.dim() > 2 or axis != 1:
+        x = x.movedim(axis, -1)
+    x2 = x.reshape(-1, C).contiguous()
+    lab = label.float().reshape(-1).contiguous()
+    prob = torch.empty_like(x2, dtype=torch.float32)
+    loss = torch.zeros(1, dtype=torch.float32, device=x.device)
+    count = torch.zeros(1, dtype=torch.int32, device=x.device)
+    _ext.softmax_loss_fwd(x2, lab, prob, loss, count,
+                          ignore_label if ignore_label is not None else 0,
+                          ignore_label is not None)
+    cnt = int(count.item()) if ignore_label is not None else lab.numel()
+    return loss.reshape(()), (prob, lab), cnt
+
+
+def softmax_loss_backward(prob_pack, label, ignore_label, scale, axis=1):
+    prob, lab = prob_pack
+    dx2 = torch.empty(prob.shape, dtype=torch.bfloat16, device=prob.device)
+    _ext.softmax_loss_bwd(prob, lab, dx2, scale,
+                          ignore_label if ignore_label is not None else 0,
+                          ignore_label is not None)
+    return dx2
+
+
+# ------------------------------------------------------------------ embed
+
+def embed_forward(idx, w, b=None):
+    wb = _as_bf16(w).contiguous()
+    idxf = idx.float().contiguous()
+    E = wb.shape[1]
+    y = torch.empty(list(idxf.shape) + [E], dtype=torch.bfloat16,
+                    device=w.device)
+    _ext.embed_fwd(idxf, wb, y, wb.shape[0])
+    if b is not None:
+        y = y + _as_bf16(b)
+    return y
+
+
+def embed_backward(idx, dy, vocab_size, bias=True):
+    e = dy.shape[-1]
+    dw = torch.zeros(vocab_size, e, dtype=torch.float32, device=dy.device)
+    _ext.embed_bwd(idx.float().contiguous(),
+                   _as_bf16(dy).contiguous(), dw, vocab_size)
+    db = dy.float().reshape(-1, e).sum(0) if bias else None
+    return dw, db
+
+
+# -------------------------------------------------------------- LSTM unit
+
+def lstm_unit_forward(c_prev, gates, cont):
+    gates_b = _as_bf16(gates).contiguous()
+    n, h4 = gates_b.shape
+    H = h4 // 4
+    cp = c_prev.float().contiguous()
+    cont_b = _as_bf16(cont).reshape(-1).contiguous()
+    c_out = torch.empty_like(cp)
+    h_out = torch.empty((n, H), dtype=torch.bfloat16, device=gates.device)
+    act = torch.empty((n, 4, H), dtype=torch.float32, device=gates.device)
+    _ext.lstm_unit_fwd(cp, gates_b, cont_b, c_out, h_out, act)
+    return c_out, h_out, (cp, c_out, act, cont_b)
+
+
+def lstm_unit_backward(c_prev, cache, dc_next, dh):
+    cp, c_out, act, cont_b = cache
+    n, _, H = act.shape
+    dc_prev = torch.empty_like(cp)
+    dgates = torch.empty((n, 4 * H), dtype=torch.bfloat16, device=act.device)
+    _ext.lstm_unit_bwd(cp, c_out, act, cont_b, dc_next.float().contiguous(),
+                       _as_bf16(dh).contiguous(), dc_prev, dgates)
+    return dc_prev, dgates
+
+
+# -------------------------------------------------------------- optimizer
+
+def sgd_update(param, grad, momentum_buf, lr, momentum, weight_decay):
+    _ext.sgd_update(param, grad.contiguous(), momentum_buf, lr, momentum,
+                    weight_decay)
+
+
+# --------------------------------------------------------------- registry
+
+GLUE_OPS = [
+    # small / test-time ops where torch (on HIP) is acceptable glue
+    "sigmoid_forward", "sigmoid_backward", "tanh_forward", "tanh_backward",
+    "softmax_forward", "softmax_backward", "accuracy", "bias_add",
+    "nesterov_update", "adam_update",
+]
+
+
+def register_all(ext_module):
+    global _ext
+    _ext = ext_module
+    from . import dispatcher
+    here = globals()
+    for name in list(here):
+        if name.startswith("_") or name in ("register_all", "reference",
+                                            "torch", "math"):
+            continue
+        fn = here[name]
+        if callable(fn) and not isinstance(fn, type):
+            dispatcher.register_gpu(name, fn)
+    for name in GLUE_OPS:
+        dispatcher.register_gpu(name, getattr(reference, name))

@@ -16,13 +16,13 @@ import torch.nn.functional as F
 
 # ----------------------------------------------------------------- convolution
 
-def conv2d_forward(x, w, b, stride, pad, dilation, groups):
+def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None):
     return F.conv2d(x, w, b, stride=stride, padding=pad,
                     dilation=dilation, groups=groups)
 
 
 def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
-                    need_dx=True, need_dw=True, bias=True):
+                    need_dx=True, need_dw=True, bias=True, ctx=None):
     dx = dw = db = None
     if need_dx:
         dx = torch.nn.grad.conv2d_input(list(x.shape), w, dy, stride=stride,

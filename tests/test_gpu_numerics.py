@@ -1,0 +1,330 @@
+"""gfx950 kernel numerics: every HIP op vs the plain-torch fp32 reference.
+
+Inputs are rounded to bf16 first, then the reference runs in fp32 on the
+rounded values, so tolerances only cover accumulation-order and output-
+rounding differences (guide G9: random asymmetric inputs catch transposed
+layouts)."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from caffeonspark_amd import ops
+from caffeonspark_amd.ops import reference
+
+
+def dev():
+    return torch.device("cuda:0")
+
+
+def bf(x):
+    return x.to(torch.bfloat16)
+
+
+def agree(got, want, rtol=0.02, atol=0.02):
+    torch.testing.assert_close(got.float().cpu(), want.float().cpu(),
+                               rtol=rtol, atol=atol)
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _seed():
+    torch.manual_seed(7)
+
+
+# ------------------------------------------------------------------- GEMM
+
+def _gemm_ref(a, b, ta, tb):
+    af = a.float().t() if ta else a.float()
+    bf_ = b.float().t() if tb else b.float()
+    return af @ bf_.t()
+
+
+@pytest.mark.parametrize("ta,tb", [(False, False), (False, True),
+                                   (True, False), (True, True)])
+@pytest.mark.parametrize("m,n,k", [(128, 128, 64), (256, 384, 512),
+                                   (100, 96, 363), (130, 70, 40)])
+def test_gemm_combos(ta, tb, m, n, k):
+    from caffeonspark_amd.ops import native
+    ext = native.load()
+    a_shape = (k, m) if ta else (m, k)
+    b_shape = (k, n) if tb else (n, k)
+    a = bf(torch.randn(a_shape)).to(dev())
+    b = bf(torch.randn(b_shape)).to(dev())
+    c = torch.empty(m, n, dtype=torch.bfloat16, device=dev())
+    ext.gemm(a, b, c, None, m, n, k, a.shape[1], b.shape[1], n,
+             ta, tb, 0, 1, False, 1.0)
+    torch.cuda.synchronize()
+    want = _gemm_ref(a.cpu(), b.cpu(), ta, tb)
+    agree(c, want, rtol=0.03, atol=0.05 * (k ** 0.5) * 0.02)
+
+
+def test_gemm_bias_relu():
+    from caffeonspark_amd.ops import native
+    ext = native.load()
+    m, n, k = 200, 130, 96
+    a = bf(torch.randn(m, k)).to(dev())
+    b = bf(torch.randn(n, k)).to(dev())
+    bias = torch.randn(n).to(dev())
+    c = torch.empty(m, n, dtype=torch.bfloat16, device=dev())
+    ext.gemm(a, b, c, bias, m, n, k, k, k, n, False, False, 0, 1, True, 1.0)
+    torch.cuda.synchronize()
+    want = torch.relu(_gemm_ref(a.cpu(), b.cpu(), False, False)
+                      + bias.cpu().float())
+    agree(c, want, rtol=0.03, atol=0.1)
+
+
+def test_gemm_splitk_atomic():
+    from caffeonspark_amd.ops import native
+    ext = native.load()
+    m, n, k = 96, 384, 8192
+    a = bf(torch.randn(k, m) * 0.1).to(dev())   # trans A
+    b = bf(torch.randn(k, n) * 0.1).to(dev())   # trans B
+    c = torch.zeros(m, n, dtype=torch.float32, device=dev())
+    ext.gemm(a, b, c, None, m, n, k, m, n, n, True, True, 2, 8, False, 1.0)
+    torch.cuda.synchronize()
+    want = a.cpu().float().t() @ b.cpu().float()
+    agree(c, want, rtol=0.03, atol=0.5)
+
+
+# ------------------------------------------------------------------- conv
+
+@pytest.mark.parametrize("case", [
+    dict(n=4, c=3, h=33, w=33, k=32, r=11, stride=4, pad=0, groups=1),
+    dict(n=2, c=16, h=15, w=15, k=32, r=5, stride=1, pad=2, groups=2),
+    dict(n=2, c=24, h=13, w=13, k=48, r=3, stride=1, pad=1, groups=1),
+])
+def test_conv_forward_backward(case):
+    n, c, h, w = case["n"], case["c"], case["h"], case["w"]
+    k, r, st, pd, g = (case["k"], case["r"], case["stride"], case["pad"],
+                       case["groups"])
+    x = bf(torch.randn(n, c, h, w)).to(dev())
+    wt = bf(torch.randn(k, c // g, r, r) * 0.1).to(dev()).float()
+    b = torch.randn(k).to(dev()).float()
+    ctx = {}
+    y = ops.conv2d_forward(x, wt, b, (st, st), (pd, pd), (1, 1), g, ctx=ctx)
+    x_cpu, w_cpu, b_cpu = x.float().cpu(), wt.float().cpu(), b.float().cpu()
+    want = reference.conv2d_forward(x_cpu, w_cpu, b_cpu, (st, st), (pd, pd),
+                                    (1, 1), g)
+    agree(y, want, rtol=0.05, atol=0.1)
+
+    dy = bf(torch.randn_like(y.float())).to(dev())
+    dx, dw, db = ops.conv2d_backward(x, wt, dy, (st, st), (pd, pd), (1, 1),
+                                     g, ctx=ctx)
+    rdx, rdw, rdb = reference.conv2d_backward(
+        x_cpu, w_cpu, dy.float().cpu(), (st, st), (pd, pd), (1, 1), g)
+    agree(dx, rdx, rtol=0.05, atol=0.15)
+    agree(dw, rdw, rtol=0.05, atol=0.3)
+    agree(db, rdb, rtol=0.05, atol=0.3)
+
+
+# --------------------------------------------------------------------- fc
+
+def test_fc_forward_backward():
+    m, k, n = 64, 500, 77
+    x = bf(torch.randn(m, k)).to(dev())
+    w = bf(torch.randn(n, k) * 0.05).to(dev()).float()
+    b = torch.randn(n).to(dev()).float()
+    y = ops.fc_forward(x, w.to(torch.bfloat16), b.to(torch.bfloat16))
+    want = reference.fc_forward(x.float().cpu(), w.cpu(), b.cpu())
+    agree(y, want, rtol=0.03, atol=0.2)
+
+    dy = bf(torch.randn(m, n)).to(dev())
+    dx, dw, db = ops.fc_backward(x, w.to(torch.bfloat16), dy)
+    rdx, rdw, rdb = reference.fc_backward(x.float().cpu(), w.cpu(),
+                                          dy.float().cpu())
+    agree(dx, rdx, rtol=0.03, atol=0.2)
+    agree(dw, rdw, rtol=0.03, atol=0.2)
+    agree(db, rdb, rtol=0.03, atol=0.2)
+
+
+# ------------------------------------------------------------ elementwise
+
+def test_relu_fwd_bwd():
+    x = bf(torch.randn(3, 17, 9, 5)).to(dev())
+    y = ops.relu_forward(x)
+    agree(y, reference.relu_forward(x.float().cpu()), rtol=0, atol=0)
+    dy = bf(torch.randn_like(x.float())).to(dev())
+    dx = ops.relu_backward(y, dy)
+    agree(dx, reference.relu_backward(y.float().cpu(), dy.float().cpu()))
+
+
+def test_dropout():
+    x = bf(torch.ones(100000)).to(dev())
+    y, mask = ops.dropout_forward(x, 0.4)
+    keep_frac = (mask.float() > 0).float().mean().item()
+    assert abs(keep_frac - 0.6) < 0.02
+    dy = bf(torch.randn_like(x.float())).to(dev())
+    dx = ops.dropout_backward(mask, dy)
+    agree(dx, dy.float().cpu() * mask.float().cpu())
+
+
+# ---------------------------------------------------------------- pooling
+
+def test_maxpool():
+    x = bf(torch.randn(2, 16, 13, 13)).to(dev())
+    y, pack = ops.maxpool_forward(x, (3, 3), (2, 2), (0, 0))
+    ry, _ = reference.maxpool_forward(x.float().cpu(), (3, 3), (2, 2), (0, 0))
+    agree(y, ry, rtol=0, atol=0)
+    dy = bf(torch.randn_like(y.float())).to(dev())
+    dx = ops.maxpool_backward(list(x.shape), pack, dy)
+    # reference backward via autograd
+    x2 = x.float().cpu().requires_grad_(True)
+    with torch.enable_grad():
+        yy = torch.nn.functional.max_pool2d(x2, 3, 2, 0, ceil_mode=True)
+    yy.backward(dy.float().cpu())
+    agree(dx, x2.grad)
+
+
+def test_avgpool_with_pad():
+    x = bf(torch.randn(2, 8, 14, 14)).to(dev())
+    y = ops.avgpool_forward(x, (3, 3), (2, 2), (1, 1))
+    ry = reference.avgpool_forward(x.float().cpu(), (3, 3), (2, 2), (1, 1))
+    agree(y, ry, rtol=0.01, atol=0.01)
+    dy = bf(torch.randn_like(y.float())).to(dev())
+    dx = ops.avgpool_backward(x, (3, 3), (2, 2), (1, 1), dy)
+    rdx = reference.avgpool_backward(x.float().cpu(), (3, 3), (2, 2), (1, 1),
+                                     dy.float().cpu())
+    agree(dx, rdx, rtol=0.02, atol=0.02)
+
+
+def test_global_avgpool():
+    x = bf(torch.randn(2, 32, 7, 7)).to(dev())
+    y = ops.global_avgpool_forward(x)
+    agree(y, reference.global_avgpool_forward(x.float().cpu()),
+          rtol=0.01, atol=0.01)
+
+
+# -------------------------------------------------------------------- LRN
+
+def test_lrn():
+    x = bf(torch.randn(2, 96, 9, 9)).to(dev())
+    y, scale = ops.lrn_forward(x, 5, 1e-4, 0.75, 1.0)
+    ry, rscale = reference.lrn_forward(x.float().cpu(), 5, 1e-4, 0.75, 1.0)
+    agree(y, ry, rtol=0.02, atol=0.02)
+    dy = bf(torch.randn_like(x.float())).to(dev())
+    dx = ops.lrn_backward(x, y, scale, dy, 5, 1e-4, 0.75)
+    rdx = reference.lrn_backward(x.float().cpu(), ry, rscale,
+                                 dy.float().cpu(), 5, 1e-4, 0.75)
+    agree(dx, rdx, rtol=0.03, atol=0.03)
+
+
+# ----------------------------------------------------------- softmax loss
+
+def test_softmax_loss():
+    n, c = 64, 1000
+    x = bf(torch.randn(n, c)).to(dev())
+    lab = torch.randint(0, c, (n,)).float().to(dev())
+    loss, pack, cnt = ops.softmax_loss_forward(x, lab, None)
+    rloss, rprob, rcnt = reference.softmax_loss_forward(
+        x.float().cpu(), lab.cpu(), None)
+    assert cnt == rcnt
+    assert abs(loss.item() - rloss.item()) / rloss.item() < 0.01
+    dx = ops.softmax_loss_backward(pack, lab, None, 1.0 / cnt)
+    rdx = reference.softmax_loss_backward(rprob, lab.cpu(), None, 1.0 / rcnt)
+    agree(dx, rdx, rtol=0.02, atol=1e-4)
+
+
+def test_softmax_loss_ignore():
+    n, c = 32, 11
+    x = bf(torch.randn(n, c)).to(dev())
+    lab = torch.randint(0, c, (n,)).float()
+    lab[::4] = -1
+    labd = lab.to(dev())
+    loss, pack, cnt = ops.softmax_loss_forward(x, labd, -1)
+    rloss, rprob, rcnt = reference.softmax_loss_forward(
+        x.float().cpu(), lab, -1)
+    assert cnt == rcnt
+    assert abs(loss.item() - rloss.item()) / max(rloss.item(), 1e-6) < 0.01
+
+
+# ------------------------------------------------------------------ embed
+
+def test_embed():
+    v, e = 50, 32
+    w = (torch.randn(v, e) * 0.1).to(dev())
+    idx = torch.randint(0, v, (7, 3)).float().to(dev())
+    y = ops.embed_forward(idx, w)
+    want = reference.embed_forward(idx.cpu(), w.to(torch.bfloat16).float().cpu())
+    agree(y, want, rtol=0.01, atol=0.01)
+    dy = bf(torch.randn(7, 3, e)).to(dev())
+    dw, db = ops.embed_backward(idx, dy, v)
+    rdw, rdb = reference.embed_backward(idx.cpu(), dy.float().cpu(), v)
+    agree(dw, rdw, rtol=0.02, atol=0.02)
+    agree(db, rdb, rtol=0.02, atol=0.02)
+
+
+# -------------------------------------------------------------- LSTM unit
+
+def test_lstm_unit():
+    n, h = 12, 20
+    c_prev = torch.randn(n, h).to(dev())
+    gates = bf(torch.randn(n, 4 * h)).to(dev())
+    cont = torch.ones(n).to(dev())
+    cont[::3] = 0
+    c, hh, cache = ops.lstm_unit_forward(c_prev, gates, cont)
+    rc, rh, rcache = reference.lstm_unit_forward(
+        c_prev.cpu(), gates.float().cpu(), cont.cpu())
+    agree(c, rc, rtol=0.02, atol=0.02)
+    agree(hh, rh, rtol=0.02, atol=0.02)
+    dc_next = torch.randn(n, h).to(dev())
+    dh = bf(torch.randn(n, h)).to(dev())
+    dcp, dg = ops.lstm_unit_backward(c_prev, cache, dc_next, dh)
+    rdcp, rdg = reference.lstm_unit_backward(c_prev.cpu(), rcache,
+                                             dc_next.cpu(),
+                                             dh.float().cpu())
+    agree(dcp, rdcp, rtol=0.03, atol=0.03)
+    agree(dg, rdg.reshape(n, 4 * h), rtol=0.03, atol=0.03)
+
+
+# -------------------------------------------------------------------- SGD
+
+def test_sgd_update():
+    n = 1003
+    p = torch.randn(n).to(dev())
+    g = torch.randn(n).to(dev())
+    v = torch.randn(n).to(dev())
+    p0, g0, v0 = p.cpu().clone(), g.cpu().clone(), v.cpu().clone()
+    ops.sgd_update(p, g, v, 0.1, 0.9, 0.005)
+    reference.sgd_update(p0, g0, v0, 0.1, 0.9, 0.005)
+    agree(p, p0, rtol=1e-5, atol=1e-6)
+    agree(v, v0, rtol=1e-5, atol=1e-6)
+
+
+# --------------------------------------------------- end-to-end net parity
+
+def test_lenet_step_parity():
+    """One LeNet fwd/bwd on GPU bf16 vs CPU fp32 with identical weights."""
+    import os
+
+    from caffeonspark_amd.core import net_from_prototxt
+    from caffeonspark_amd.proto import caffe_pb
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    proto = os.path.join(root, "caffeonspark_amd", "models",
+                         "lenet_memory_train_test.prototxt")
+    state = caffe_pb.NetState(phase=caffe_pb.Phase.TRAIN)
+    cpu_net = net_from_prototxt(proto, state=state, seed=3)
+    gpu_net = net_from_prototxt(proto, state=state, device=dev(),
+                                dtype=torch.bfloat16, seed=3)
+    # copy weights cpu -> gpu
+    for cl, gl in zip(cpu_net.layers, gpu_net.layers):
+        for cb, gb in zip(cl.blobs, gl.blobs):
+            gb.data.copy_(cb.data.to(gb.data.device))
+    x = torch.randn(32, 1, 28, 28)
+    y = torch.randint(0, 10, (32,)).float()
+    cpu_net.data_layers()[0].reset(x, y)
+    gpu_net.data_layers()[0].reset(x.to(dev(), torch.bfloat16), y.to(dev()))
+    closs = cpu_net.forward()
+    gloss = gpu_net.forward()
+    assert abs(closs - gloss) / max(abs(closs), 1e-6) < 0.05
+    cpu_net.backward()
+    gpu_net.backward()
+    for cl, gl in zip(cpu_net.layers, gpu_net.layers):
+        for cb, gb in zip(cl.blobs, gl.blobs):
+            if cb.diff is None:
+                continue
+            cg, gg = cb.diff.float(), gb.diff.float().cpu()
+            denom = cg.abs().max().clamp_min(1e-3)
+            rel = (cg - gg).abs().max() / denom
+            assert rel < 0.12, f"{cl.name} grad mismatch rel={rel:.3f}"
