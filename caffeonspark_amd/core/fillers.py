@@ -15,6 +15,12 @@ from ..proto import caffe_pb
 
 def fill(tensor: torch.Tensor, param: caffe_pb.FillerParameter,
          generator: torch.Generator = None) -> None:
+    if tensor.device.type != "cpu":
+        # fill on CPU (deterministic across devices), then copy
+        tmp = torch.empty(tensor.shape, dtype=torch.float32)
+        fill(tmp, param, generator)
+        tensor.copy_(tmp.to(tensor.dtype))
+        return
     ftype = param.type or "constant"
     if ftype == "constant":
         tensor.fill_(param.value)

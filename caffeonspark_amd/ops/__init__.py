@@ -20,21 +20,24 @@ from . import reference
 
 _native_mod = None
 _native_checked = False
+_native_err = None
 
 
 def native():
     """Return the loaded HIP extension module, or None on CPU-only hosts."""
-    global _native_mod, _native_checked
+    global _native_mod, _native_checked, _native_err
     if not _native_checked:
         _native_checked = True
         try:
-            from . import native as native_loader
+            from . import native_loader
             _native_mod = native_loader.load()
-        except Exception:
+        except Exception as e:
             _native_mod = None
-            if torch.cuda.is_available() and not os.environ.get(
-                    "COS_AMD_ALLOW_EAGER_FALLBACK"):
-                raise
+            _native_err = e
+    if _native_err is not None and torch.cuda.is_available() and \
+            not os.environ.get("COS_AMD_ALLOW_EAGER_FALLBACK"):
+        raise RuntimeError(
+            f"gfx950 HIP extension failed to load: {_native_err}")
     return _native_mod
 
 

@@ -45,8 +45,7 @@ def _gemm_ref(a, b, ta, tb):
 @pytest.mark.parametrize("m,n,k", [(128, 128, 64), (256, 384, 512),
                                    (100, 96, 363), (130, 70, 40)])
 def test_gemm_combos(ta, tb, m, n, k):
-    from caffeonspark_amd.ops import native
-    ext = native.load()
+    ext = ops.native()
     a_shape = (k, m) if ta else (m, k)
     b_shape = (k, n) if tb else (n, k)
     a = bf(torch.randn(a_shape)).to(dev())
@@ -60,8 +59,7 @@ def test_gemm_combos(ta, tb, m, n, k):
 
 
 def test_gemm_bias_relu():
-    from caffeonspark_amd.ops import native
-    ext = native.load()
+    ext = ops.native()
     m, n, k = 200, 130, 96
     a = bf(torch.randn(m, k)).to(dev())
     b = bf(torch.randn(n, k)).to(dev())
@@ -75,8 +73,7 @@ def test_gemm_bias_relu():
 
 
 def test_gemm_splitk_atomic():
-    from caffeonspark_amd.ops import native
-    ext = native.load()
+    ext = ops.native()
     m, n, k = 96, 384, 8192
     a = bf(torch.randn(k, m) * 0.1).to(dev())   # trans A
     b = bf(torch.randn(k, n) * 0.1).to(dev())   # trans B
