@@ -63,13 +63,24 @@ dispatcher = _Dispatcher()
 _TORCH_OK_ON_GPU = set()
 
 
+def _args_on_gpu(args) -> bool:
+    for a in args:
+        if isinstance(a, torch.Tensor):
+            if a.is_cuda:
+                return True
+        elif isinstance(a, (tuple, list)):
+            for b in a:
+                if isinstance(b, torch.Tensor) and b.is_cuda:
+                    return True
+    return False
+
+
 def _dispatch(name: str):
     ref_fn = getattr(reference, name)
 
     @functools.wraps(ref_fn)
     def wrapper(*args, **kwargs):
-        first = args[0] if args else None
-        on_gpu = isinstance(first, torch.Tensor) and first.is_cuda
+        on_gpu = _args_on_gpu(args)
         if on_gpu:
             native()  # force extension load (raises loudly if missing)
             fn = dispatcher.gpu_impls.get(name)
