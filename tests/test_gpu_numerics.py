@@ -322,6 +322,6 @@ def test_lenet_step_parity():
             if cb.diff is None:
                 continue
             cg, gg = cb.diff.float(), gb.diff.float().cpu()
-            denom = cg.abs().max().clamp_min(1e-3)
-            rel = (cg - gg).abs().max() / denom
-            assert rel < 0.12, f"{cl.name} grad mismatch rel={rel:.3f}"
+            # relative L2 error: robust to bf16 noise on single elements
+            rel = (cg - gg).norm() / cg.norm().clamp_min(1e-4)
+            assert rel < 0.08, f"{cl.name} grad mismatch relL2={rel:.3f}"
