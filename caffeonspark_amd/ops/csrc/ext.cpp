@@ -37,8 +37,8 @@ void lrn_fwd(const void* x, void* y, float* scale, int64_t npix, int C,
              int local_size, float alpha, float beta, float k,
              hipStream_t stream);
 void lrn_bwd(const void* x, const void* y, const float* scale, const void* dy,
-             void* dx, int64_t npix, int C, int local_size, float alpha,
-             float beta, hipStream_t stream);
+             void* dx, void* ratio, int64_t npix, int C, int local_size,
+             float alpha, float beta, hipStream_t stream);
 void relu_fwd(const void* x, void* y, float slope, int64_t n,
               hipStream_t stream);
 void relu_bwd(const void* y, const void* dy, void* dx, float slope, int64_t n,
@@ -158,11 +158,11 @@ void py_lrn_fwd(Tensor x, Tensor y, Tensor scale, int64_t npix, int64_t C,
 }
 
 void py_lrn_bwd(Tensor x, Tensor y, Tensor scale, Tensor dy, Tensor dx,
-                int64_t npix, int64_t C, int64_t local_size, double alpha,
-                double beta) {
+                Tensor ratio, int64_t npix, int64_t C, int64_t local_size,
+                double alpha, double beta) {
   cosamd::lrn_bwd(x.data_ptr(), y.data_ptr(), scale.data_ptr<float>(),
-                  dy.data_ptr(), dx.data_ptr(), npix, C, local_size, alpha,
-                  beta, cur_stream());
+                  dy.data_ptr(), dx.data_ptr(), ratio.data_ptr(), npix, C,
+                  local_size, alpha, beta, cur_stream());
 }
 
 void py_relu_fwd(Tensor x, Tensor y, double slope) {

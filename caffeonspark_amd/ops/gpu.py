@@ -304,8 +304,9 @@ def lrn_backward(x, y, scale_pack, dy, local_size, alpha, beta):
     scale, xl = scale_pack
     N, C, H, W = x.shape
     dx = torch.empty_like(xl)
-    _ext.lrn_bwd(xl, _cl(y), scale, _cl(dy), dx, N * H * W, C, local_size,
-                 alpha, beta)
+    ratio = torch.empty_like(xl)
+    _ext.lrn_bwd(xl, _cl(y), scale, _cl(dy), dx, ratio, N * H * W, C,
+                 local_size, alpha, beta)
     return dx
 
 
