@@ -301,14 +301,17 @@ def test_lenet_step_parity():
     proto = os.path.join(root, "caffeonspark_amd", "models",
                          "lenet_memory_train_test.prototxt")
     state = caffe_pb.NetState(phase=caffe_pb.Phase.TRAIN)
-    cpu_net = net_from_prototxt(proto, state=state, seed=3)
+    # CPU net also in bf16 so the comparison isolates kernel correctness
+    # (accumulation order), not fp32-vs-bf16 drift through the deep chain
+    cpu_net = net_from_prototxt(proto, state=state, seed=3,
+                                dtype=torch.bfloat16)
     gpu_net = net_from_prototxt(proto, state=state, device=dev(),
                                 dtype=torch.bfloat16, seed=3)
     # copy weights cpu -> gpu
     for cl, gl in zip(cpu_net.layers, gpu_net.layers):
         for cb, gb in zip(cl.blobs, gl.blobs):
             gb.data.copy_(cb.data.to(gb.data.device))
-    x = torch.randn(32, 1, 28, 28)
+    x = torch.randn(32, 1, 28, 28).to(torch.bfloat16)
     y = torch.randint(0, 10, (32,)).float()
     cpu_net.data_layers()[0].reset(x, y)
     gpu_net.data_layers()[0].reset(x.to(dev(), torch.bfloat16), y.to(dev()))
