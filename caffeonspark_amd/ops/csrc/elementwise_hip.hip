@@ -136,16 +136,32 @@ __global__ void sgd_update_kernel(float* __restrict__ p, const float* __restrict
 
 // ---------------------------------------------------------- column reduce
 // out[c] += sum_r in[r*ld + c]  (bias gradients; in bf16, out fp32)
+// Block = 256 threads as [4 row-groups x 64 cols]: coalesced 64-wide column
+// reads, deep grid.x row parallelism, LDS-reduced partials, one atomic per
+// (block, col).
 
 __global__ void colsum_kernel(const u16* __restrict__ in, float* __restrict__ out,
                               int64_t rows, int cols, int ld) {
-  // block handles 256 columns chunk; grid.y strides rows
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= cols) return;
-  float acc = 0.f;
-  for (int64_t r = blockIdx.y; r < rows; r += gridDim.y)
-    acc += ldbf(in + r * ld + c);
-  atomicAdd(out + c, acc);
+  __shared__ float part[4][64];
+  int lane_c = threadIdx.x & 63;
+  int rg = threadIdx.x >> 6;             // 0..3
+  for (int c0 = 0; c0 < cols; c0 += 64) {
+    int c = c0 + lane_c;
+    float acc = 0.f;
+    if (c < cols) {
+      for (int64_t r = (int64_t)blockIdx.x * 4 + rg; r < rows;
+           r += (int64_t)gridDim.x * 4)
+        acc += ldbf(in + r * ld + c);
+    }
+    part[rg][lane_c] = acc;
+    __syncthreads();
+    if (rg == 0 && c < cols) {
+      float s = part[0][lane_c] + part[1][lane_c] + part[2][lane_c] +
+                part[3][lane_c];
+      atomicAdd(out + c, s);
+    }
+    __syncthreads();
+  }
 }
 
 // --------------------------------------------------------------- LSTM unit
@@ -333,8 +349,9 @@ void sgd_update(float* p, const float* g, float* v, float lr, float mu,
 
 void colsum(const void* in, float* out, int64_t rows, int cols, int ld,
             hipStream_t stream) {
-  dim3 grid((cols + 255) / 256, (unsigned)hmin<int64_t>(rows, 64));
- hipLaunchKernelGGL(( colsum_kernel), dim3(grid), dim3(256), 0, stream, (const u16*)in, out, rows, cols, ld);
+  int blocks = (int)hmin<int64_t>((rows + 3) / 4, 2048);
+ hipLaunchKernelGGL(( colsum_kernel), dim3(blocks), dim3(256), 0, stream, (const u16*)in, out, rows, cols,
+                                            ld);
 }
 
 void lstm_unit_fwd(const float* c_prev, const void* gates, const void* cont,

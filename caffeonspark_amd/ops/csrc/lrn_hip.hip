@@ -1,10 +1,12 @@
 #include "hip/hip_runtime.h"
 // NHWC cross-channel LRN, fwd + bwd (AlexNet/GoogLeNet norm layers).
-// In NHWC the channel window is contiguous memory, so each thread walks a
-// sliding window with a running sum — one pass, no re-reads.
+// NHWC puts the channel window in contiguous memory, so one thread per
+// ELEMENT re-reads its small window with fully coalesced accesses (adjacent
+// lanes -> adjacent channels); the redundant window reads hit L1/L2.
 // scale = k + alpha/n * sum_win x^2 ;  y = x * scale^-beta
 // dx = dy*scale^-beta - (2*alpha*beta/n) * x * sum_win(dy*y/scale)
-// scale is stored fp32 for backward (matches the CPU reference exactly).
+// Backward is two passes: ratio = dy*y/scale materialized bf16, then the
+// windowed sum — 3x less traffic than recomputing ratio per window tap.
 
 #include "common.h"
 
@@ -20,80 +22,78 @@ __device__ __forceinline__ void st_bf(u16* p, float v) {
   *p = *reinterpret_cast<u16*>(&b);
 }
 
-// one thread per pixel: running-window over C
 __global__ void lrn_fwd_kernel(const u16* __restrict__ x,
                                u16* __restrict__ y,
                                float* __restrict__ scale,
-                               int64_t npix, int C, int half, float a_over_n,
+                               int64_t total, int C, int half, float a_over_n,
                                float beta, float k) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < npix; i += (int64_t)gridDim.x * blockDim.x) {
-    const u16* xp = x + i * C;
-    u16* yp = y + i * C;
-    float* sp = scale + i * C;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    int64_t base = i - c;
+    int lo = c - half > 0 ? c - half : 0;
+    int hi = c + half < C - 1 ? c + half : C - 1;
     float win = 0.f;
-    for (int c = 0; c <= half && c < C; ++c) {
-      float v = ld_bf(xp + c);
+    for (int cc = lo; cc <= hi; ++cc) {
+      float v = ld_bf(x + base + cc);
       win += v * v;
     }
-    for (int c = 0; c < C; ++c) {
-      float sc = k + a_over_n * win;
-      sp[c] = sc;
-      st_bf(yp + c, ld_bf(xp + c) * __powf(sc, -beta));
-      int add = c + half + 1, sub = c - half;
-      if (add < C) { float v = ld_bf(xp + add); win += v * v; }
-      if (sub >= 0) { float v = ld_bf(xp + sub); win -= v * v; }
-    }
+    float sc = k + a_over_n * win;
+    scale[i] = sc;
+    st_bf(y + i, ld_bf(x + i) * __powf(sc, -beta));
   }
 }
 
+__global__ void lrn_ratio_kernel(const u16* __restrict__ y,
+                                 const float* __restrict__ scale,
+                                 const u16* __restrict__ dy,
+                                 u16* __restrict__ ratio, int64_t total) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x)
+    st_bf(ratio + i, ld_bf(dy + i) * ld_bf(y + i) / scale[i]);
+}
+
 __global__ void lrn_bwd_kernel(const u16* __restrict__ x,
-                               const u16* __restrict__ y,
                                const float* __restrict__ scale,
                                const u16* __restrict__ dy,
+                               const u16* __restrict__ ratio,
                                u16* __restrict__ dx,
-                               int64_t npix, int C, int half, float a_over_n,
+                               int64_t total, int C, int half,
                                float beta, float ratio_coef) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < npix; i += (int64_t)gridDim.x * blockDim.x) {
-    const u16* xp = x + i * C;
-    const u16* yp = y + i * C;
-    const u16* dyp = dy + i * C;
-    const float* sp = scale + i * C;
-    u16* dxp = dx + i * C;
-    auto ratio = [&](int c) {
-      return ld_bf(dyp + c) * ld_bf(yp + c) / sp[c];
-    };
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    int64_t base = i - c;
+    int lo = c - half > 0 ? c - half : 0;
+    int hi = c + half < C - 1 ? c + half : C - 1;
     float win = 0.f;
-    for (int c = 0; c <= half && c < C; ++c) win += ratio(c);
-    for (int c = 0; c < C; ++c) {
-      float v = ld_bf(dyp + c) * __powf(sp[c], -beta)
-                - ratio_coef * ld_bf(xp + c) * win;
-      st_bf(dxp + c, v);
-      int add = c + half + 1, sub = c - half;
-      if (add < C) win += ratio(add);
-      if (sub >= 0) win -= ratio(sub);
-    }
+    for (int cc = lo; cc <= hi; ++cc) win += ld_bf(ratio + base + cc);
+    float v = ld_bf(dy + i) * __powf(scale[i], -beta)
+              - ratio_coef * ld_bf(x + i) * win;
+    st_bf(dx + i, v);
   }
 }
 
 void lrn_fwd(const void* x, void* y, float* scale, int64_t npix, int C,
              int local_size, float alpha, float beta, float k,
              hipStream_t stream) {
-  int blocks = (int)hmin<int64_t>(8192, (npix + 255) / 256);
+  int64_t total = npix * C;
+  int blocks = (int)hmin<int64_t>(4096, (total + 255) / 256);
  hipLaunchKernelGGL(( lrn_fwd_kernel), dim3(blocks), dim3(256), 0, stream, 
-      (const u16*)x, (u16*)y, scale, npix, C, local_size / 2,
+      (const u16*)x, (u16*)y, scale, total, C, local_size / 2,
       alpha / local_size, beta, k);
 }
 
 void lrn_bwd(const void* x, const void* y, const float* scale, const void* dy,
-             void* dx, int64_t npix, int C, int local_size, float alpha,
-             float beta, hipStream_t stream) {
-  int blocks = (int)hmin<int64_t>(8192, (npix + 255) / 256);
+             void* dx, void* ratio, int64_t npix, int C, int local_size,
+             float alpha, float beta, hipStream_t stream) {
+  int64_t total = npix * C;
+  int blocks = (int)hmin<int64_t>(4096, (total + 255) / 256);
+ hipLaunchKernelGGL(( lrn_ratio_kernel), dim3(blocks), dim3(256), 0, stream, 
+      (const u16*)y, scale, (const u16*)dy, (u16*)ratio, total);
  hipLaunchKernelGGL(( lrn_bwd_kernel), dim3(blocks), dim3(256), 0, stream, 
-      (const u16*)x, (const u16*)y, scale, (const u16*)dy, (u16*)dx,
-      npix, C, local_size / 2, alpha / local_size, beta,
-      2.f * alpha * beta / local_size);
+      (const u16*)x, scale, (const u16*)dy, (const u16*)ratio, (u16*)dx,
+      total, C, local_size / 2, beta, 2.f * alpha * beta / local_size);
 }
 
 }  // namespace cosamd
