@@ -183,7 +183,7 @@ class Net:
         return loss
 
     # --------------------------------------------------------------- backward
-    def backward(self) -> None:
+    def backward(self, on_layer_done=None) -> None:
         # clear activation diffs (param diffs persist for iter_size accum)
         for blob in self.blob_map.values():
             blob.diff = None
@@ -203,6 +203,8 @@ class Net:
                     t.ensure_diff()
             self.layers[i].backward(tops, self.layer_prop_down[i],
                                     self.layer_bottoms[i])
+            if on_layer_done is not None and self.layers[i].blobs:
+                on_layer_done(self.layers[i])
 
     def forward_backward(self) -> float:
         loss = self.forward()

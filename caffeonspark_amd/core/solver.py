@@ -53,6 +53,10 @@ class Solver:
             train_state.phase = caffe_pb.Phase.TRAIN
         self.net = Net(net_param, train_state, device=self.device,
                        dtype=self.dtype, seed=seed)
+        # arena must exist before test nets share weights (sharing points
+        # their blobs at the arena views)
+        self.params: List[Blob] = self.net.learnable_params()
+        self._build_arena()
         self.test_nets: List[Net] = []
         n_test = len(param.test_iter)
         for i in range(n_test):
@@ -67,12 +71,52 @@ class Solver:
             tn.share_trained_layers_with(self.net)
             self.test_nets.append(tn)
 
-        self.params: List[Blob] = self.net.learnable_params()
-        self.history = [torch.zeros_like(b.data) for b in self.params]
         self.history2 = [torch.zeros_like(b.data) for b in self.params] \
             if self.type in ("Adam", "AdaDelta") else []
         self._losses: List[float] = []
         self.smoothed_loss = 0.0
+
+    def _build_arena(self) -> None:
+        """Flat fp32 weight/grad/momentum arenas (the reference engine's
+        `Params` flat data_/diff_ arrays):
+        param blobs become views, so the optimizer runs one fused kernel
+        per (lr_mult, decay_mult) segment and the distributed sync
+        all-reduces contiguous slices without gather/scatter copies."""
+        total = sum(b.count for b in self.params)
+        dev = self.device
+        self.flat_w = torch.zeros(total, dtype=torch.float32, device=dev)
+        self.flat_g = torch.zeros(total, dtype=torch.float32, device=dev)
+        self.flat_m = torch.zeros(total, dtype=torch.float32, device=dev)
+        self.param_offsets: List[int] = []
+        off = 0
+        for b in self.params:
+            n = b.count
+            self.flat_w[off:off + n] = b.data.detach().reshape(-1)
+            b.data = self.flat_w.narrow(0, off, n).view(b.shape)
+            b.diff = self.flat_g.narrow(0, off, n).view(b.shape)
+            self.param_offsets.append(off)
+            off += n
+        self.history = [self.flat_m.narrow(0, o, b.count).view(b.shape)
+                        for o, b in zip(self.param_offsets, self.params)]
+        # contiguous segments sharing (lr_mult, decay_mult) for fused updates
+        self.segments = []  # (off, n, lr_mult, decay_mult)
+        for b, o in zip(self.params, self.param_offsets):
+            key = (b._lr_mult, b._decay_mult)
+            if self.segments and self.segments[-1][2:] == key and \
+                    self.segments[-1][0] + self.segments[-1][1] == o:
+                off0, n0, _, _ = self.segments[-1]
+                self.segments[-1] = (off0, n0 + b.count, *key)
+            else:
+                self.segments.append((o, b.count, *key))
+        # per-layer arena slices in reverse-layer order (backward completion
+        # order) for overlap-friendly bucket all-reduce
+        by_id = {id(b): o for b, o in zip(self.params, self.param_offsets)}
+        self.layer_slices = []  # (layer_name, [(off, count), ...])
+        for layer in self.net.layers:
+            sl = [(by_id[id(b)], b.count) for b in layer.blobs
+                  if id(b) in by_id]
+            if sl:
+                self.layer_slices.append((layer.name, sl))
 
     def _resolve_net_param(self, proto_dir: str) -> caffe_pb.NetParameter:
         p = self.param
@@ -128,9 +172,14 @@ class Solver:
             cb.on_start()
         self.net.zero_param_diffs()
         loss = 0.0
-        for _ in range(max(1, p.iter_size)):
-            loss += self.net.forward_backward()
-        loss /= max(1, p.iter_size)
+        iters = max(1, p.iter_size)
+        for it in range(iters):
+            loss += self.net.forward()
+            # per-layer completion hook on the LAST micro-batch only, so
+            # gradient all-reduce can overlap the rest of backward
+            cb = self._on_layer_backward if it == iters - 1 else None
+            self.net.backward(on_layer_done=cb)
+        loss /= iters
         for cb in self.callbacks:
             cb.on_gradients_ready()
         self.apply_update()
@@ -140,6 +189,12 @@ class Solver:
             print(f"[cos-amd] iter {self.iter} loss {self.smoothed_loss:.6f} "
                   f"lr {self.get_lr():.6g}", flush=True)
         return loss
+
+    def _on_layer_backward(self, layer) -> None:
+        for cb in self.callbacks:
+            fn = getattr(cb, "on_layer_backward", None)
+            if fn is not None:
+                fn(layer)
 
     def _update_smoothed_loss(self, loss: float) -> None:
         avg = max(1, self.param.average_loss)
@@ -152,27 +207,29 @@ class Solver:
     def apply_update(self) -> None:
         p = self.param
         rate = self.get_lr()
-        # normalize for iter_size
+        # normalize for iter_size (and solver_count, reference
+        # CaffeNet.cpp:620-625 gradient-scaling rule)
         scale = 1.0 / (max(1, p.iter_size) * self.solver_count)
         if scale != 1.0:
-            for b in self.params:
-                if b.diff is not None:
-                    b.diff.mul_(scale)
-        # gradient clipping (global L2 norm)
+            self.flat_g.mul_(scale)
+        # gradient clipping (global L2 norm, one fused reduction)
         clip = p.clip_gradients
         if clip > 0:
-            sumsq = 0.0
-            for b in self.params:
-                if b.diff is not None:
-                    sumsq += float(b.diff.float().pow(2).sum())
-            norm = math.sqrt(sumsq)
+            norm = float(self.flat_g.norm())
             if norm > clip:
-                cs = clip / norm
-                for b in self.params:
-                    if b.diff is not None:
-                        b.diff.mul_(cs)
+                self.flat_g.mul_(clip / norm)
         wd = p.weight_decay
         reg = p.regularization_type
+        if self.type == "SGD" and reg == "L2":
+            # fused per-segment update on the flat arenas
+            for (off, n, lrm, dm) in self.segments:
+                if lrm == 0:
+                    continue
+                ops.sgd_update(self.flat_w.narrow(0, off, n),
+                               self.flat_g.narrow(0, off, n),
+                               self.flat_m.narrow(0, off, n),
+                               rate * lrm, p.momentum, wd * dm)
+            return
         for i, b in enumerate(self.params):
             if b._lr_mult == 0 or b.diff is None:
                 continue
