@@ -1,13 +1,21 @@
-"""Synchronous data parallelism over RCCL (torch.distributed, backend
+"""Synchronous data parallelism over RCCL (torch.distributed backend
 "nccl" == RCCL on ROCm) — the MI355X-native replacement for the
 reference's two-tier P2PSync + SocketSync/RDMASync sharded parameter
-server (C1-C4/C9 collapse into one all-reduce per bucket
-over the 8-GPU xGMI mesh).
+server (collectives C1-C4/C9 collapse into bucketed
+all-reduce over the 8-GPU xGMI mesh).
 
-`DistributedSync` is a Solver callback (the same on_start /
-on_gradients_ready hook points the reference wires its syncs into,
-CaffeNet.cpp:592-654): on_gradients_ready all-reduces (avg) the param
-diffs in flat buckets sized for xGMI's per-link ring bandwidth.
+`DistributedSync` attaches to the Solver at the same hook points the
+reference wires its syncs into (on_start / on_gradients_ready,
+CaffeNet.cpp:592-654) plus a per-layer backward-completion hook the
+reference lacks: gradient buckets all-reduce asynchronously as soon as
+the backward pass finishes writing them, overlapping communication with
+the remaining backward compute (the reference does comm strictly around
+the step — SURVEY.md §5 notes this as a known ceiling to beat).
+
+Gradients live in the solver's flat fp32 arena, so a bucket is just a
+contiguous narrow() — no gather/scatter staging.  Averaging follows
+Caffe's solver_count convention: SUM all-reduce here, 1/world_size
+scaling inside Solver.apply_update.
 """
 
 from __future__ import annotations
@@ -18,7 +26,6 @@ from typing import List, Optional
 import torch
 import torch.distributed as dist
 
-from ..core.blob import Blob
 from ..core.solver import Callback, Solver
 
 
@@ -38,66 +45,91 @@ def init_distributed(backend: Optional[str] = None) -> int:
 
 
 class DistributedSync(Callback):
-    """Bucketed gradient all-reduce attached to a Solver.
+    """Bucketed, backward-overlapped gradient all-reduce.
 
-    Buckets default to 25 MB of fp32 grads — large enough to amortize
-    RCCL launch overhead on the 7x153 GB/s xGMI links, small enough to
-    pipeline several reduction rounds.  With `overlap=False` reduction
-    happens in one burst inside on_gradients_ready (still async on the
-    comm stream, synchronized before the optimizer runs).
+    Buckets are contiguous arena ranges assembled from per-layer param
+    slices in reverse layer order (backward completion order).  Default
+    25 MB fp32 buckets amortize RCCL launch overhead on the 7x153 GB/s
+    xGMI links while still pipelining several reduction rounds under the
+    tail of backward.
     """
 
     def __init__(self, solver: Solver, bucket_mb: float = 25.0):
         self.solver = solver
-        self.params: List[Blob] = solver.params
         self.world_size = dist.get_world_size() if dist.is_initialized() else 1
         solver.callbacks.append(self)
-        solver.rank = dist.get_rank() if dist.is_initialized() else 0
-        # flat fp32 bucket buffers, assigned in reverse order (grads become
-        # ready tail-first during backward)
-        self.buckets: List[List[Blob]] = []
+        if dist.is_initialized():
+            solver.rank = dist.get_rank()
+            solver.solver_count = self.world_size
         cap = int(bucket_mb * 1024 * 1024 / 4)
-        cur, cur_n = [], 0
-        for b in reversed(self.params):
-            cur.append(b)
-            cur_n += b.count
-            if cur_n >= cap:
-                self.buckets.append(cur)
-                cur, cur_n = [], 0
-        if cur:
-            self.buckets.append(cur)
-        self._flat = [torch.zeros(sum(b.count for b in bk), dtype=torch.float32,
-                                  device=solver.device) for bk in self.buckets]
 
+        # assemble buckets from reverse-layer-order arena slices
+        self.buckets: List[tuple] = []      # (lo, hi)
+        self.layer_bucket: dict = {}        # layer name -> bucket idx
+        pend_layers: List[str] = []
+        lo = hi = None
+        for lname, slices in reversed(solver.layer_slices):
+            s_lo = min(o for o, _ in slices)
+            s_hi = max(o + n for o, n in slices)
+            if lo is None:
+                lo, hi = s_lo, s_hi
+            else:
+                lo, hi = min(lo, s_lo), max(hi, s_hi)
+            pend_layers.append(lname)
+            if hi - lo >= cap:
+                idx = len(self.buckets)
+                self.buckets.append((lo, hi))
+                for l in pend_layers:
+                    self.layer_bucket[l] = idx
+                pend_layers, lo = [], None
+        if lo is not None:
+            idx = len(self.buckets)
+            self.buckets.append((lo, hi))
+            for l in pend_layers:
+                self.layer_bucket[l] = idx
+        # layers remaining per bucket before it can fire
+        self._layers_per_bucket = [0] * len(self.buckets)
+        for l, i in self.layer_bucket.items():
+            self._layers_per_bucket[i] += 1
+        self._reset()
+
+    def _reset(self):
+        self._remaining = list(self._layers_per_bucket)
+        self._works = []
+        self._fired = [False] * len(self.buckets)
+
+    def _fire(self, idx: int) -> None:
+        if self._fired[idx] or self.world_size <= 1:
+            return
+        self._fired[idx] = True
+        lo, hi = self.buckets[idx]
+        buf = self.solver.flat_g.narrow(0, lo, hi - lo)
+        self._works.append(dist.all_reduce(buf, op=dist.ReduceOp.SUM,
+                                           async_op=True))
+
+    # ---- Solver callback hooks -------------------------------------------
     def on_start(self) -> None:
         pass
 
-    def on_gradients_ready(self) -> None:
-        if self.world_size <= 1:
+    def on_layer_backward(self, layer) -> None:
+        idx = self.layer_bucket.get(layer.name)
+        if idx is None:
             return
-        works = []
-        for bk, flat in zip(self.buckets, self._flat):
-            off = 0
-            for b in bk:
-                d = b.ensure_diff()
-                flat[off:off + b.count].copy_(d.reshape(-1).float())
-                off += b.count
-            works.append((dist.all_reduce(flat, op=dist.ReduceOp.SUM,
-                                          async_op=True), bk, flat))
-        inv = 1.0 / self.world_size
-        for work, bk, flat in works:
-            work.wait()
-            off = 0
-            for b in bk:
-                b.diff.reshape(-1).copy_(flat[off:off + b.count] * inv)
-                off += b.count
+        self._remaining[idx] -= 1
+        if self._remaining[idx] <= 0:
+            self._fire(idx)
+
+    def on_gradients_ready(self) -> None:
+        for i in range(len(self.buckets)):
+            self._fire(i)
+        for w in self._works:
+            w.wait()
+        self._reset()
 
     def broadcast_params(self) -> None:
-        """Rank-0 weights to all ranks (reference: on_start weight
-        all-gather, C1)."""
+        """Rank-0 weights + momentum to all ranks (replaces the reference's
+        weight-shard all-gather, collective C1)."""
         if self.world_size <= 1:
             return
-        for b in self.params:
-            dist.broadcast(b.data, src=0)
-        for h in self.solver.history + self.solver.history2:
-            dist.broadcast(h, src=0)
+        dist.broadcast(self.solver.flat_w, src=0)
+        dist.broadcast(self.solver.flat_m, src=0)

@@ -1,0 +1,161 @@
+"""Multi-process data-parallel tests on CPU (gloo, world_size=2): the same
+distributed path the driver runs on 8 GPUs over RCCL, exercised here with
+the gloo backend (SURVEY.md §4 flags the reference's lack of a loopback
+multi-rank harness as a gap to fill)."""
+
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+
+def _find_port():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _worker(rank, ws, store_path, q):
+    import torch.distributed as dist
+
+    from caffeonspark_amd.core import solver_from_prototxt
+    from caffeonspark_amd.parallel import DistributedSync
+
+    dist.init_process_group("gloo", rank=rank, world_size=ws,
+                            init_method=f"file://{store_path}")
+    try:
+        s = solver_from_prototxt(
+            os.path.join(os.path.dirname(__file__), "..",
+                         "caffeonspark_amd", "models",
+                         "lenet_memory_solver.prototxt"))
+        sync = DistributedSync(s, bucket_mb=0.5)
+        sync.broadcast_params()
+        g = torch.Generator().manual_seed(100 + rank)
+        for it in range(5):
+            x = torch.randn(64, 1, 28, 28, generator=g)
+            y = torch.randint(0, 10, (64,), generator=g).float()
+            s.net.data_layers()[0].reset(x, y)
+            s._step_one()
+        q.put((rank, s.flat_w.clone(), s.flat_g.clone()))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ddp_two_ranks_identical_params():
+    """After synchronized steps on different data, all ranks must hold
+    bit-identical parameters."""
+    ws = 2
+    import tempfile
+    store = tempfile.mktemp(prefix="cosamd_ddp_")
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker, args=(r, ws, store, q))
+             for r in range(ws)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(ws):
+        rank, w, gr = q.get(timeout=300)
+        results[rank] = (w, gr)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    w0, g0 = results[0]
+    w1, g1 = results[1]
+    torch.testing.assert_close(w0, w1, rtol=0, atol=0)
+    torch.testing.assert_close(g0, g1, rtol=1e-6, atol=1e-7)
+
+
+def _seeded_solver():
+    from caffeonspark_amd.core.solver import Solver
+    from caffeonspark_amd.proto import caffe_pb, text_format
+
+    root = os.path.join(os.path.dirname(__file__), "..")
+    sp = text_format.parse_file(
+        os.path.join(root, "caffeonspark_amd", "models",
+                     "lenet_memory_solver.prototxt"),
+        caffe_pb.SolverParameter)
+    sp.random_seed = 7
+    sp.display = 0
+    return Solver(sp, proto_dir=os.path.join(root, "caffeonspark_amd",
+                                             "models"))
+
+
+def _worker_equiv(rank, ws, store_path, q, data):
+    import torch.distributed as dist
+
+    from caffeonspark_amd.parallel import DistributedSync
+
+    dist.init_process_group("gloo", rank=rank, world_size=ws,
+                            init_method=f"file://{store_path}")
+    try:
+        s = _seeded_solver()
+        sync = DistributedSync(s)
+        sync.broadcast_params()
+        x, y = data
+        half = x.shape[0] // ws
+        s.net.data_layers()[0].reset(x[rank * half:(rank + 1) * half],
+                                     y[rank * half:(rank + 1) * half])
+        s._step_one()
+        q.put((rank, s.flat_w.clone()))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ddp_matches_single_process():
+    """2-rank DP on split batch == single process on the full batch
+    (gradient averaging equivalence, equal per-rank batch sizes)."""
+    torch.manual_seed(5)
+    x = torch.randn(128, 1, 28, 28)
+    y = torch.randint(0, 10, (128,)).float()
+
+    ws = 2
+    import tempfile
+    store = tempfile.mktemp(prefix="cosamd_ddp_")
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_equiv,
+                         args=(r, ws, store, q, (x, y))) for r in range(ws)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(ws):
+        rank, w = q.get(timeout=300)
+        results[rank] = w
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    torch.testing.assert_close(results[0], results[1], rtol=0, atol=0)
+
+    # single-process run over the whole batch from the same seeded init:
+    # DP-averaged update must match the full-batch update (both losses are
+    # per-sample means over equal halves)
+    single = _seeded_solver()
+    single.net.data_layers()[0].reset(x, y)
+    single._step_one()
+    torch.testing.assert_close(results[0], single.flat_w,
+                               rtol=1e-4, atol=1e-6)
+
+
+def test_bucket_assembly_covers_all_params():
+    from caffeonspark_amd.core import solver_from_prototxt
+    from caffeonspark_amd.parallel.ddp import DistributedSync
+
+    s = solver_from_prototxt(
+        os.path.join(os.path.dirname(__file__), "..",
+                     "caffeonspark_amd", "models",
+                     "lenet_memory_solver.prototxt"))
+    sync = DistributedSync(s, bucket_mb=0.25)
+    covered = set()
+    for lo, hi in sync.buckets:
+        covered.update(range(lo, hi, 1))
+    total = int(s.flat_g.numel())
+    # every param element inside some bucket
+    for b, off in zip(s.params, s.param_offsets):
+        assert off in covered and (off + b.count - 1) in covered
+    assert len(sync.layer_bucket) == len(s.layer_slices)
