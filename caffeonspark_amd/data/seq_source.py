@@ -1,0 +1,39 @@
+"""SequenceFile image source (reference: SeqImageDataSource.scala)."""
+
+from __future__ import annotations
+
+import glob
+import os
+from typing import Iterator
+
+from ..proto import caffe_pb
+from .image_source import ImageDataSource, ImageSample
+from .seqfile import SequenceFileReader
+
+
+class SeqImageDataSource(ImageDataSource):
+    def init(self) -> None:
+        path = self.source_path
+        if path.startswith("file:"):
+            path = path[5:]
+        if os.path.isdir(path):
+            self.files = sorted(
+                f for f in glob.glob(os.path.join(path, "*"))
+                if not os.path.basename(f).startswith((".", "_")))
+        else:
+            self.files = [path]
+
+    def sample_iter(self, rank: int = 0, world: int = 1,
+                    epochs: int = -1) -> Iterator[ImageSample]:
+        epoch = 0
+        while epochs < 0 or epoch < epochs:
+            i = 0
+            for f in self.files:
+                for key, raw in SequenceFileReader(f).items():
+                    if i % world == rank:
+                        d = caffe_pb.Datum.FromString(raw)
+                        yield ImageSample(key.decode(), d.label, d.channels,
+                                          d.height, d.width, bool(d.encoded),
+                                          bytes(d.data))
+                    i += 1
+            epoch += 1

@@ -1,0 +1,204 @@
+"""End-to-end pipeline: LMDB/SeqFile/DataFrame sources -> processor ->
+solver -> test aggregation, mirroring the reference's own integration gates
+(InterleaveTest accuracy>0.8, SourceTest round-trips — SURVEY.md §4)."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from caffeonspark_amd.api import CaffeOnSpark, Config
+from caffeonspark_amd.data.lmdb_io import LmdbWriter
+from caffeonspark_amd.data.processor import CaffeProcessor
+from caffeonspark_amd.proto import caffe_pb, text_format
+from caffeonspark_amd.tools.seq_value import datum_from_array
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def make_synthetic_lmdb(path, n, seed):
+    """10-class separable 1x28x28 uint8 images: class-c has a bright
+    vertical stripe at column 2c."""
+    rng = np.random.RandomState(seed)
+    items = []
+    for i in range(n):
+        label = rng.randint(0, 10)
+        img = rng.randint(80, 150, size=(1, 28, 28)).astype(np.uint8)
+        img[0, :, label * 2] = 250
+        d = datum_from_array(img, label)
+        items.append((f"{i:08d}".encode(), d.SerializeToString()))
+    LmdbWriter(path).write(items)
+
+
+LENET_NET = """
+name: "LeNet"
+layer {{
+  name: "data" type: "MemoryData" top: "data" top: "label"
+  include {{ phase: TRAIN }}
+  source_class: "com.yahoo.ml.caffe.LMDB"
+  memory_data_param {{ source: "{train}" batch_size: 64
+                       channels: 1 height: 28 width: 28 }}
+  transform_param {{ scale: 0.00390625 mean_value: 128 }}
+}}
+layer {{
+  name: "data" type: "MemoryData" top: "data" top: "label"
+  include {{ phase: TEST }}
+  source_class: "com.yahoo.ml.caffe.LMDB"
+  memory_data_param {{ source: "{test}" batch_size: 100
+                       channels: 1 height: 28 width: 28 }}
+  transform_param {{ scale: 0.00390625 mean_value: 128 }}
+}}
+layer {{ name: "conv1" type: "Convolution" bottom: "data" top: "conv1"
+  param {{ lr_mult: 1 }} param {{ lr_mult: 2 }}
+  convolution_param {{ num_output: 20 kernel_size: 5
+    weight_filler {{ type: "xavier" }} }} }}
+layer {{ name: "pool1" type: "Pooling" bottom: "conv1" top: "pool1"
+  pooling_param {{ pool: MAX kernel_size: 2 stride: 2 }} }}
+layer {{ name: "ip1" type: "InnerProduct" bottom: "pool1" top: "ip1"
+  param {{ lr_mult: 1 }} param {{ lr_mult: 2 }}
+  inner_product_param {{ num_output: 100
+    weight_filler {{ type: "xavier" }} }} }}
+layer {{ name: "relu1" type: "ReLU" bottom: "ip1" top: "ip1" }}
+layer {{ name: "ip2" type: "InnerProduct" bottom: "ip1" top: "ip2"
+  param {{ lr_mult: 1 }} param {{ lr_mult: 2 }}
+  inner_product_param {{ num_output: 10
+    weight_filler {{ type: "xavier" }} }} }}
+layer {{ name: "accuracy" type: "Accuracy" bottom: "ip2" bottom: "label"
+  top: "accuracy" include {{ phase: TEST }} }}
+layer {{ name: "loss" type: "SoftmaxWithLoss" bottom: "ip2" bottom: "label"
+  top: "loss" }}
+"""
+
+SOLVER = """
+net: "{net}"
+test_iter: 2
+test_interval: {test_interval}
+base_lr: 0.01
+momentum: 0.9
+weight_decay: 0.0005
+lr_policy: "fixed"
+display: 0
+max_iter: {max_iter}
+snapshot: 0
+snapshot_prefix: "{prefix}"
+random_seed: 11
+"""
+
+
+@pytest.fixture(scope="module")
+def workdir(tmp_path_factory):
+    d = tmp_path_factory.mktemp("e2e")
+    make_synthetic_lmdb(str(d / "train_lmdb"), 1200, seed=1)
+    make_synthetic_lmdb(str(d / "test_lmdb"), 200, seed=2)
+    net_file = d / "lenet.prototxt"
+    net_file.write_text(LENET_NET.format(train=str(d / "train_lmdb"),
+                                         test=str(d / "test_lmdb")))
+    return d
+
+
+def _solver_file(workdir, name, test_interval=0, max_iter=120):
+    f = workdir / name
+    f.write_text(SOLVER.format(net=str(workdir / "lenet.prototxt"),
+                               test_interval=test_interval,
+                               max_iter=max_iter,
+                               prefix=str(workdir / "lenet")))
+    return str(f)
+
+
+def test_train_and_test_via_facade(workdir):
+    """BASELINE config 1: LeNet via memory source, CPU — the reference's
+    plumbing-correctness gate (accuracy > 0.8)."""
+    CaffeProcessor.reset_instance()
+    os.chdir(workdir)
+    conf = Config(["-conf", _solver_file(workdir, "solver.prototxt"),
+                   "-train", "-label", "label"])
+    cos = CaffeOnSpark(conf)
+    cos.train()
+    # snapshot written by train (snapshot_after_train)
+    snaps = [f for f in os.listdir(workdir) if f.endswith(".caffemodel")]
+    assert snaps, "no snapshot written"
+
+    CaffeProcessor.reset_instance()
+    conf2 = Config(["-conf", _solver_file(workdir, "solver2.prototxt"),
+                    "-test", "-weights",
+                    str(workdir / sorted(snaps)[-1])])
+    cos2 = CaffeOnSpark(conf2)
+    result = cos2.test(max_samples=200)
+    assert "accuracy" in result
+    acc = result["accuracy"][0]
+    assert acc > 0.8, f"accuracy {acc}"
+    CaffeProcessor.reset_instance()
+
+
+def test_train_with_validation(workdir):
+    CaffeProcessor.reset_instance()
+    os.chdir(workdir)
+    conf = Config(["-conf",
+                   _solver_file(workdir, "solver3.prototxt",
+                                test_interval=50, max_iter=110),
+                   "-train"])
+    cos = CaffeOnSpark(conf)
+    results = cos.train_with_validation()
+    assert len(results) >= 1
+    assert "accuracy" in results[-1]
+    CaffeProcessor.reset_instance()
+
+
+def test_features_extraction(workdir):
+    CaffeProcessor.reset_instance()
+    os.chdir(workdir)
+    out = str(workdir / "features.json")
+    conf = Config(["-conf", _solver_file(workdir, "solver4.prototxt"),
+                   "-features", "ip2", "-label", "label",
+                   "-output", out, "-outputFormat", "json"])
+    cos = CaffeOnSpark(conf)
+    df = cos.features(max_samples=100)
+    assert "SampleID" in df.columns and "ip2" in df.columns
+    assert len(df) >= 100
+    assert len(df["ip2"][0]) == 10
+    assert os.path.exists(out)
+    CaffeProcessor.reset_instance()
+
+
+def test_seqfile_source_roundtrip(workdir, tmp_path):
+    """SourceTest analog: LMDB -> SequenceFile -> read back."""
+    from caffeonspark_amd.data.seqfile import SequenceFileReader
+    from caffeonspark_amd.tools.converters import lmdb2sequence
+
+    seq = str(tmp_path / "train.seq")
+    n = lmdb2sequence(str(workdir / "train_lmdb"), seq)
+    assert n == 1200
+    r = SequenceFileReader(seq)
+    items = list(r.items())
+    assert len(items) == 1200
+    d = caffe_pb.Datum.FromString(items[0][1])
+    assert (d.channels, d.height, d.width) == (1, 28, 28)
+
+
+def test_dataframe_source(workdir, tmp_path):
+    """LMDB -> parquet -> ImageDataFrame source feeds a batch."""
+    from caffeonspark_amd.data.source import get_source
+    from caffeonspark_amd.tools.converters import lmdb2dataframe
+
+    pq_file = str(tmp_path / "train.parquet")
+    lmdb2dataframe(str(workdir / "train_lmdb"), pq_file)
+
+    net = text_format.parse_file(str(workdir / "lenet.prototxt"),
+                                 caffe_pb.NetParameter)
+    lp = net.layer[0]
+    lp.source_class = "com.yahoo.ml.caffe.ImageDataFrame"
+    lp.memory_data_param.source = pq_file
+
+    class FakeConf:
+        net_param = net
+        seed = 1
+
+    src = get_source(FakeConf(), True)
+    src.init()
+    it = src.sample_iter(0, 1, epochs=1)
+    for _ in range(70):
+        src.offer(next(it))
+    batch = src.next_batch(torch.device("cpu"), torch.float32)
+    assert batch[0].shape == (64, 1, 28, 28)
+    assert batch[1].shape == (64,)
