@@ -24,7 +24,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--batch", type=int, default=256)
+    ap.add_argument("--batch", type=int, default=0)
     ap.add_argument("--model", type=str, default="alexnet")
     args = ap.parse_args()
 
@@ -43,6 +43,9 @@ def main():
         dev = torch.device("cpu")
         dtype = torch.float32
 
+    if args.batch == 0:
+        args.batch = {"alexnet": 256, "googlenet": 128, "cifar10_quick": 100,
+                      "lrcn": 32}.get(args.model, 256)
     root = os.path.dirname(os.path.abspath(__file__))
     solver = solver_from_prototxt(
         os.path.join(root, "caffeonspark_amd", "models",
@@ -59,11 +62,26 @@ def main():
     n = args.batch
     g = torch.Generator().manual_seed(1234 + rank)
     dl = solver.net.data_layers()[0]
-    c, h, w = dl.channels, dl.height, dl.width
-    x = torch.randn(n, c, h, w, generator=g).to(dev, dtype)
-    y = torch.randint(0, 1000, (n,), generator=g).float().to(dev)
-    dl.batch_size = n
-    dl.reset(x, y)
+    from caffeonspark_amd.core.layers.data import CoSDataLayer
+    if isinstance(dl, CoSDataLayer):
+        # LRCN: image + time-major caption tops
+        T = int(dl.tops_cfg[2].channels)
+        V = 8801
+        x = torch.randn(n, 3, 227, 227, generator=g).to(dev, dtype)
+        label = torch.zeros(n, 1).to(dev)
+        cont = torch.ones(T, n)
+        cont[0] = 0
+        inp = torch.randint(0, V, (T, n), generator=g).float()
+        tgt = torch.randint(0, V, (T, n), generator=g).float()
+        dl.batch_size = n
+        dl.reset([x, label, cont.to(dev, dtype), inp.to(dev), tgt.to(dev)])
+    else:
+        classes = {"cifar10_quick": 10}.get(args.model, 1000)
+        c, h, w = dl.channels, dl.height, dl.width
+        x = torch.randn(n, c, h, w, generator=g).to(dev, dtype)
+        y = torch.randint(0, classes, (n,), generator=g).float().to(dev)
+        dl.batch_size = n
+        dl.reset(x, y)
 
     def barrier_sync():
         if ws > 1:
@@ -102,8 +120,10 @@ def main():
             "vs_baseline": None,
             "dtype": "bf16" if use_gpu else "fp32",
             "data": "synthetic",
-            "config": {"model": "alexnet(bvlc_reference/CaffeNet)",
-                       "global_batch": n * ws, "seq_len": None,
+            "config": {"model": {"alexnet": "alexnet(bvlc_reference/CaffeNet)"
+                                 }.get(args.model, args.model),
+                       "global_batch": n * ws,
+                       "seq_len": 21 if args.model == "lrcn" else None,
                        "parallelism": f"dp{ws}"},
         }), flush=True)
     if ws > 1:

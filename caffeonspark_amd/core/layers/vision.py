@@ -351,7 +351,8 @@ class AccuracyLayer(Layer):
 
     def forward(self, bottom, top):
         hits, count = ops.accuracy(bottom[0].data, bottom[1].data,
-                                   self.top_k, self.ignore_label)
+                                   self.top_k, self.ignore_label,
+                                   axis=self.param.accuracy_param.axis)
         top[0].data = (hits / max(1, count)).reshape(())
         return 0.0
 

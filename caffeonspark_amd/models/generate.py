@@ -199,6 +199,79 @@ def googlenet(batch=128) -> pb.NetParameter:
     return net
 
 
+def lrcn(batch=32, caption_len=21, vocab=8801) -> pb.NetParameter:
+    """LRCN captioning net (factored 2-layer variant): CaffeNet conv stack
+    + word Embed + 2x LSTM with the image feature as lstm2's static input,
+    time-major captions — the reference's lrcn_cos.prototxt family."""
+    net = pb.NetParameter(name="lrcn_caffenet_to_lstm")
+    for phase in (pb.Phase.TRAIN, pb.Phase.TEST):
+        lp = layer(net, "data", "CoSData", [],
+                   ["data", "label", "cont_sentence", "input_sentence",
+                    "target_sentence"], phase=phase)
+        lp.source_class = "com.yahoo.ml.caffe.DataFrameSource"
+        cp = pb.CoSDataParameter(source="coco_parquet", batch_size=batch)
+        cp.top.append(pb.CoSTopParameter(
+            name="data", type=pb.CoSTopType.ENCODED_IMAGE_WITH_DIM,
+            channels=3, height=256, width=256, out_channels=3,
+            out_height=227, out_width=227,
+            transform_param=pb.TransformationParameter(
+                mirror=True, crop_size=227,
+                mean_value=[104.0, 117.0, 123.0])))
+        cp.top.append(pb.CoSTopParameter(name="label",
+                                         type=pb.CoSTopType.INT))
+        for nm in ("cont_sentence", "input_sentence", "target_sentence"):
+            cp.top.append(pb.CoSTopParameter(
+                name=nm, type=pb.CoSTopType.INT_ARRAY, channels=caption_len,
+                sample_num_axes=1, transpose=True))
+        lp.cos_data_param = cp
+    x = conv(net, "conv1", "data", 96, 11, stride=4, std=0.01, w_lr=0)
+    x = maxpool(net, "pool1", x, 3, 2)
+    x = lrn(net, "norm1", x)
+    x = conv(net, "conv2", x, 256, 5, pad=2, group=2, bias=0.1, w_lr=0)
+    x = maxpool(net, "pool2", x, 3, 2)
+    x = lrn(net, "norm2", x)
+    x = conv(net, "conv3", x, 384, 3, pad=1, w_lr=0)
+    x = conv(net, "conv4", x, 384, 3, pad=1, group=2, bias=0.1, w_lr=0)
+    x = conv(net, "conv5", x, 256, 3, pad=1, group=2, bias=0.1, w_lr=0)
+    x = maxpool(net, "pool5", x, 3, 2)
+    x = fc(net, "fc6", x, 4096, dropout=0.5)
+    x = fc(net, "fc7", x, 4096, dropout=0.5)
+    x = fc(net, "fc8", x, 1000, std=0.01, bias=0.0, relu=False)
+    layer(net, "silence_label", "Silence", ["label"], [])
+    emb = layer(net, "embedding", "Embed", ["input_sentence"],
+                ["embedded_input_sentence"],
+                embed_param=pb.EmbedParameter(
+                    num_output=1000, input_dim=vocab, bias_term=False,
+                    weight_filler=_filler(type="uniform", min=-0.08,
+                                          max=0.08)))
+    emb.param = [pb.ParamSpec(lr_mult=1.0)]
+    rp = pb.RecurrentParameter(
+        num_output=1000,
+        weight_filler=_filler(type="uniform", min=-0.08, max=0.08),
+        bias_filler=_filler(type="constant"))
+    layer(net, "lstm1", "LSTM", ["embedded_input_sentence",
+                                 "cont_sentence"], ["lstm1"],
+          recurrent_param=rp)
+    layer(net, "lstm2", "LSTM", ["lstm1", "cont_sentence", "fc8"],
+          ["lstm2"], recurrent_param=rp.clone())
+    pred = layer(net, "predict", "InnerProduct", ["lstm2"], ["predict"],
+                 inner_product_param=pb.InnerProductParameter(
+                     num_output=vocab, axis=2,
+                     weight_filler=_filler(type="uniform", min=-0.08,
+                                           max=0.08),
+                     bias_filler=_filler(type="constant")))
+    pred.param = lr_params()
+    lp = layer(net, "cross_entropy_loss", "SoftmaxWithLoss",
+               ["predict", "target_sentence"], ["cross_entropy_loss"],
+               loss_param=pb.LossParameter(ignore_label=-1),
+               softmax_param=pb.SoftmaxParameter(axis=2))
+    lp.loss_weight = [20.0]
+    layer(net, "accuracy", "Accuracy", ["predict", "target_sentence"],
+          ["accuracy"], phase=pb.Phase.TEST,
+          accuracy_param=pb.AccuracyParameter(axis=2, ignore_label=-1))
+    return net
+
+
 def solver(net_file, **kw) -> pb.SolverParameter:
     sp = pb.SolverParameter(net=net_file, **kw)
     return sp
@@ -209,6 +282,7 @@ def main():
         "alexnet_train_test.prototxt": alexnet(),
         "cifar10_quick_train_test.prototxt": cifar10_quick(),
         "googlenet_train_test.prototxt": googlenet(),
+        "lrcn_train_test.prototxt": lrcn(),
     }
     for fname, net in jobs.items():
         with open(os.path.join(HERE, fname), "w") as fh:
@@ -234,6 +308,13 @@ def main():
             power=0.5, display=40, max_iter=2400000, momentum=0.9,
             weight_decay=0.0002, snapshot=0, snapshot_prefix="googlenet",
             solver_mode=pb.SolverMode.GPU),
+        "lrcn_solver.prototxt": solver(
+            "caffeonspark_amd/models/lrcn_train_test.prototxt",
+            test_iter=[0], test_interval=0, base_lr=0.01, lr_policy="step",
+            gamma=0.5, stepsize=20000, display=1, max_iter=110000,
+            momentum=0.9, weight_decay=0.0, snapshot=0,
+            snapshot_prefix="lrcn", clip_gradients=10.0, average_loss=100,
+            random_seed=1701, solver_mode=pb.SolverMode.GPU),
     }
     for fname, sp in solvers.items():
         with open(os.path.join(HERE, fname), "w") as fh:

@@ -317,6 +317,7 @@ def softmax_loss_forward(x, label, ignore_label, axis=1):
     C = x.shape[axis]
     if x.dim() > 2 or axis != 1:
         x = x.movedim(axis, -1)
+    moved_shape = tuple(x.shape)
     x2 = x.reshape(-1, C).contiguous()
     lab = label.float().reshape(-1).contiguous()
     prob = torch.empty_like(x2, dtype=torch.float32)
@@ -326,16 +327,20 @@ def softmax_loss_forward(x, label, ignore_label, axis=1):
                           ignore_label if ignore_label is not None else 0,
                           ignore_label is not None)
     cnt = int(count.item()) if ignore_label is not None else lab.numel()
-    return loss.reshape(()), (prob, lab), cnt
+    return loss.reshape(()), (prob, lab, moved_shape, axis), cnt
 
 
 def softmax_loss_backward(prob_pack, label, ignore_label, scale, axis=1):
-    prob, lab = prob_pack
+    prob, lab, moved_shape, ax = prob_pack
     dx2 = torch.empty(prob.shape, dtype=torch.bfloat16, device=prob.device)
     _ext.softmax_loss_bwd(prob, lab, dx2, scale,
                           ignore_label if ignore_label is not None else 0,
                           ignore_label is not None)
-    return dx2
+    # moved_shape is x.shape AFTER movedim(axis,-1); restore original order
+    dx = dx2.reshape(moved_shape)
+    if len(moved_shape) > 2 or ax != 1:
+        dx = dx.movedim(-1, ax)
+    return dx.contiguous()
 
 
 # ------------------------------------------------------------------ embed

@@ -161,15 +161,12 @@ def softmax_backward(y, dy, axis=1):
 
 
 def softmax_loss_forward(x, label, ignore_label: Optional[int], axis=1):
-    """Returns (loss_sum, prob, valid_count). x: [N, C, ...], label [N, ...]."""
+    """Returns (loss_sum, prob, valid_count). x: [..., C at axis, ...],
+    label matches x with the class axis removed."""
     xf = x.float()
     logp = F.log_softmax(xf, dim=axis)
     prob = logp.exp()
     lab = label.long()
-    if axis != 1:
-        raise NotImplementedError("softmax loss axis != 1")
-    flat_logp = logp.transpose(1, -1) if logp.dim() > 2 else logp
-    # gather per-sample log-prob
     lp = logp.movedim(axis, -1)
     lab_flat = lab.reshape(-1)
     lp_flat = lp.reshape(-1, lp.shape[-1])
@@ -188,17 +185,20 @@ def softmax_loss_forward(x, label, ignore_label: Optional[int], axis=1):
 
 def softmax_loss_backward(prob, label, ignore_label: Optional[int], scale, axis=1):
     lab = label.long()
-    dx = prob.clone()
-    lab_md = lab.reshape(lab.shape[0], 1, *lab.shape[1:]) if lab.dim() < dx.dim() \
-        else lab.unsqueeze(axis)
+    # operate with the class axis last, then restore
+    p = prob.movedim(axis, -1)
+    shape = p.shape
+    p2 = p.reshape(-1, shape[-1]).clone()
+    lab_flat = lab.reshape(-1, 1)
     if ignore_label is not None:
-        valid = lab_md != ignore_label
-        safe = torch.where(valid, lab_md, torch.zeros_like(lab_md))
-        dx.scatter_add_(axis, safe, -torch.ones_like(safe, dtype=dx.dtype))
-        dx = dx * valid.to(dx.dtype)
+        valid = lab_flat != ignore_label
+        safe = torch.where(valid, lab_flat, torch.zeros_like(lab_flat))
+        p2.scatter_add_(1, safe, -torch.ones_like(safe, dtype=p2.dtype))
+        p2 = p2 * valid.to(p2.dtype)
     else:
-        dx.scatter_add_(axis, lab_md, -torch.ones_like(lab_md, dtype=dx.dtype))
-    return dx * scale
+        p2.scatter_add_(1, lab_flat,
+                        -torch.ones_like(lab_flat, dtype=p2.dtype))
+    return (p2.reshape(shape).movedim(-1, axis) * scale)
 
 
 # --------------------------------------------------------------------- dropout
@@ -272,8 +272,11 @@ def lstm_unit_backward(c_prev, cache, dc_next, dh):
 
 # -------------------------------------------------------------------- concat &c
 
-def accuracy(x, label, top_k=1, ignore_label: Optional[int] = None):
+def accuracy(x, label, top_k=1, ignore_label: Optional[int] = None,
+             axis: int = -1):
     lab = label.long().reshape(-1)
+    if x.dim() > 2 and axis not in (-1, x.dim() - 1):
+        x = x.movedim(axis, -1)
     scores = x.reshape(lab.shape[0], -1)
     topk = scores.topk(min(top_k, scores.shape[1]), dim=1).indices
     hit = (topk == lab.unsqueeze(1)).any(dim=1)
