@@ -1,0 +1,131 @@
+"""HDF5 snapshot format, vocab/coco tools, mini-cluster CLI pieces."""
+
+import json
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from caffeonspark_amd.proto import caffe_pb, text_format
+from caffeonspark_amd.utils import hdf5
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_hdf5_roundtrip_nested(tmp_path):
+    root = hdf5.H5Group()
+    root.attrs["iter"] = 7
+    root.attrs["learned_net"] = "abc.h5"
+    g = hdf5.H5Group()
+    root["data"] = g
+    g["layer.with.dots"] = hdf5.H5Group()
+    g["layer.with.dots"]["0"] = np.random.randn(3, 4, 5).astype(np.float32)
+    f = str(tmp_path / "t.h5")
+    hdf5.save(f, root)
+    r = hdf5.load(f)
+    assert r.attrs["iter"] == 7
+    np.testing.assert_allclose(r["data"]["layer.with.dots"]["0"],
+                               g["layer.with.dots"]["0"])
+
+
+def test_hdf5_solver_snapshot_restore(tmp_path):
+    """cifar10_quick-style HDF5 snapshot (snapshot_format: HDF5) round-trip
+    through the Solver."""
+    from caffeonspark_amd.core.solver import Solver
+
+    sp = text_format.parse_file(
+        os.path.join(ROOT, "caffeonspark_amd", "models",
+                     "lenet_memory_solver.prototxt"),
+        caffe_pb.SolverParameter)
+    sp.snapshot_format = caffe_pb.SnapshotFormat.HDF5
+    sp.snapshot_prefix = str(tmp_path / "lenet")
+    sp.random_seed = 4
+    sp.display = 0
+    s = Solver(sp, proto_dir=os.path.join(ROOT, "caffeonspark_amd",
+                                          "models"))
+    x = torch.randn(64, 1, 28, 28)
+    y = torch.randint(0, 10, (64,)).float()
+    s.net.data_layers()[0].reset(x, y)
+    s.step(3)
+    model = s.snapshot()
+    assert model.endswith(".caffemodel.h5")
+    state = s.snapshot_filename("state")
+    assert os.path.exists(state)
+
+    s2 = Solver(sp, proto_dir=os.path.join(ROOT, "caffeonspark_amd",
+                                           "models"))
+    s2.restore(state)
+    assert s2.iter == 3
+    torch.testing.assert_close(s2.flat_w, s.flat_w)
+    torch.testing.assert_close(s2.flat_m, s.flat_m)
+
+
+def test_vocab_embed():
+    from caffeonspark_amd.tools.vocab import Vocab
+
+    v = Vocab.build(["a cat sat", "a dog sat", "a cat ran"], size=10)
+    assert len(v) <= 10
+    ids = v.embed("a cat flew", 6)
+    assert len(ids) == 6
+    assert ids[-3] == 0 or 0 in ids      # EOS present
+    assert ids[-1] == -1                 # padded with ignore label
+    # unk maps to the unk id
+    assert ids[2] == v.index["<unk>"]
+
+
+def test_coco_pipeline(tmp_path):
+    from caffeonspark_amd.tools.coco import coco_to_dataframe, embed_captions
+
+    # synthesize a tiny COCO-style dataset
+    img_dir = tmp_path / "imgs"
+    img_dir.mkdir()
+    from PIL import Image
+    for i in range(3):
+        Image.new("RGB", (32, 32), (i * 40, 100, 50)).save(
+            str(img_dir / f"im{i}.jpg"))
+    doc = {
+        "images": [{"id": i, "file_name": f"im{i}.jpg"} for i in range(3)],
+        "annotations": [
+            {"id": 10 + i, "image_id": i, "caption": f"a photo number {i}"}
+            for i in range(3)],
+    }
+    cj = tmp_path / "captions.json"
+    cj.write_text(json.dumps(doc))
+    df1 = str(tmp_path / "capdf.parquet")
+    n = coco_to_dataframe(str(cj), str(img_dir), df1)
+    assert n == 3
+    df2 = str(tmp_path / "embedded.parquet")
+    n2 = embed_captions(df1, str(tmp_path / "vocab.json"), df2,
+                        caption_length=8, vocab_size=50)
+    assert n2 == 3
+    import pyarrow.parquet as pq
+    t = pq.read_table(df2)
+    assert set(t.schema.names) >= {"data", "label", "input_sentence",
+                                   "target_sentence", "cont_sentence"}
+    row_inp = t.column("input_sentence")[0].as_py()
+    assert len(row_inp) == 9 and row_inp[0] == 0
+
+
+def test_mini_cluster_single(tmp_path):
+    """mini_cluster with -cluster 1 runs the train CLI end to end."""
+    from tests.test_e2e_pipeline import LENET_NET, SOLVER, \
+        make_synthetic_lmdb
+
+    make_synthetic_lmdb(str(tmp_path / "train_lmdb"), 400, 3)
+    make_synthetic_lmdb(str(tmp_path / "test_lmdb"), 100, 4)
+    (tmp_path / "net.prototxt").write_text(
+        LENET_NET.format(train=str(tmp_path / "train_lmdb"),
+                         test=str(tmp_path / "test_lmdb")))
+    (tmp_path / "solver.prototxt").write_text(
+        SOLVER.format(net=str(tmp_path / "net.prototxt"), test_interval=0,
+                      max_iter=10, prefix=str(tmp_path / "mc")))
+    from caffeonspark_amd.data.processor import CaffeProcessor
+    from caffeonspark_amd.tools import mini_cluster
+
+    CaffeProcessor.reset_instance()
+    mini_cluster.main(["-cluster", "1", "-conf",
+                       str(tmp_path / "solver.prototxt"), "-train"])
+    snaps = [f for f in os.listdir(tmp_path) if f.endswith(".caffemodel")]
+    assert snaps
+    CaffeProcessor.reset_instance()
