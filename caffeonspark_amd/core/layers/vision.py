@@ -68,7 +68,8 @@ class ConvolutionLayer(Layer):
         self._ctx = {}
         top[0].data = ops.conv2d_forward(
             x, w, b, (self.sh, self.sw), (self.ph, self.pw),
-            (self.dil, self.dil), self.groups, ctx=self._ctx)
+            (self.dil, self.dil), self.groups, ctx=self._ctx,
+            relu=getattr(self, "_fuse_relu", False))
         return 0.0
 
     def backward(self, top, propagate_down, bottom):
@@ -120,7 +121,7 @@ class InnerProductLayer(Layer):
         if self.transpose:
             w = w.t()
         b = self.cast(self.blobs[1].data) if self.bias_term else None
-        y = ops.fc_forward(x, w, b)
+        y = ops.fc_forward(x, w, b, relu=getattr(self, "_fuse_relu", False))
         out_shape = list(bottom[0].shape[:self.axis]) + [self.num_output]
         top[0].data = y.reshape(out_shape)
         return 0.0
@@ -150,6 +151,9 @@ class ReLULayer(Layer):
         self.slope = self.param.relu_param.negative_slope
 
     def forward(self, bottom, top):
+        if getattr(self, "_fused_upstream", False):
+            top[0].data = bottom[0].data  # producer already applied ReLU
+            return 0.0
         top[0].data = ops.relu_forward(bottom[0].data, self.slope)
         return 0.0
 

@@ -150,6 +150,26 @@ class Net:
                 w = layer.loss_weight(ti)
                 if w != 0.0:
                     self._loss_tops.append((idx, ti, w))
+        self._fuse_relu_peephole()
+
+    def _fuse_relu_peephole(self) -> None:
+        """Fuse Conv/IP + in-place ReLU pairs on GPU: the GEMM epilogue
+        applies the rectification, the ReLU layer's forward becomes a
+        no-op (backward is unchanged — it masks by top data, which the
+        fused producer already rectified)."""
+        if self.device.type != "cuda":
+            return
+        for a, b in zip(self.layers, self.layers[1:]):
+            if a.param.type not in ("Convolution", "InnerProduct"):
+                continue
+            if b.param.type != "ReLU" or getattr(b, "slope", 0.0) != 0.0:
+                continue
+            if list(b.param.bottom) != list(b.param.top):
+                continue
+            if not a.param.top or b.param.bottom[0] != a.param.top[0]:
+                continue
+            a._fuse_relu = True
+            b._fused_upstream = True
 
     def _prefeed_placeholder(self, layer) -> None:
         from .layers.data import CoSDataLayer, MemoryDataLayer

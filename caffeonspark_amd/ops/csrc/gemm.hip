@@ -66,15 +66,16 @@ __device__ __forceinline__ void stage_direct_guarded(
 // one K-row, scatter into LDS columns.
 __device__ __forceinline__ void stage_trans_guarded(
     bf16* lds_, const bf16* g_, int row0, int rows, int ld,
-    int k0, int kend, int tid) {
+    int k0, int kend, int tid, int nthreads = NTHREADS,
+    int tile_rows = BM) {
   auto* lds = reinterpret_cast<unsigned short*>(lds_);
   auto* g = reinterpret_cast<const unsigned short*>(g_);
-  // 512 slots: slot -> (k, r0): 32 k-rows x 16 slots of 8 rows
-#pragma unroll
-  for (int it = 0; it < 2; ++it) {
-    int slot = tid + it * NTHREADS;
-    int kk = slot >> 4;             // 0..31
-    int r0 = (slot & 15) * 8;
+  // slots: (k, r0): 32 k-rows x tile_rows/8 slots of 8 rows
+  int rslots = tile_rows >> 3;
+  int nslots = 32 * rslots;
+  for (int slot = tid; slot < nslots; slot += nthreads) {
+    int kk = slot / rslots;
+    int r0 = (slot % rslots) * 8;
     int gk = k0 + kk;
     unsigned short vals[8];
     if (gk < kend && row0 + r0 + 8 <= rows && (ld | (row0 + r0)) % 8 == 0) {
@@ -94,6 +95,74 @@ __device__ __forceinline__ void stage_trans_guarded(
 #pragma unroll
     for (int j = 0; j < 8; ++j) lds[(r0 + j) * BK + kk] = vals[j];
   }
+}
+
+// --------------------------------------------------- wide-N dw kernel
+// Specialized for the skinny trans/trans weight-gradient GEMMs
+// (dw[Kout][Kcol] = dy^T @ col, K = N*P*Q huge): BN=256 halves the
+// A-operand re-read factor vs the 128x128 kernel; fp32 atomic split-K.
+
+constexpr int WBN = 256;
+constexpr int WNT = 512;  // 8 waves: 2 (M) x 4 (N)
+
+__global__ __launch_bounds__(WNT, 1) void gemm_tt_wide_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    float* __restrict__ C, int M, int N, int K, int lda, int ldb, int ldc,
+    int ksplit, float alpha) {
+  __shared__ bf16 As[BM * BK];
+  __shared__ bf16 Bs[WBN * BK];
+
+  int mblocks = (M + BM - 1) / BM;
+  int nblocks = (N + WBN - 1) / WBN;
+  int bid = xcd_swizzle(blockIdx.x, mblocks * nblocks);
+  int bm = bid / nblocks, bn = bid % nblocks;
+  int tile_m = bm * BM, tile_n = bn * WBN;
+  int k_begin = blockIdx.y * ksplit;
+  int k_end = min(K, k_begin + ksplit);
+
+  int tid = threadIdx.x;
+  int wave = tid >> 6, lane = tid & 63;
+  int wm = wave >> 2, wn = wave & 3;     // 2x4 wave grid, 64x64 each
+  int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+
+  f32x4 acc[4][4] = {};
+
+  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
+    stage_trans_guarded(As, A, tile_m, M, lda, k0, k_end, tid, WNT, BM);
+    stage_trans_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid, WNT, WBN);
+    __syncthreads();
+    bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      afrag[f] = *reinterpret_cast<const bf16x8*>(
+          As + (wm * 64 + f * 16 + lrow) * BK + lk8);
+      bfrag[f] = *reinterpret_cast<const bf16x8*>(
+          Bs + (wn * 64 + f * 16 + lrow) * BK + lk8);
+    }
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
+    __syncthreads();
+  }
+
+  int crow0 = tile_m + wm * 64 + ((lane >> 4) << 2);
+  int ccol0 = tile_n + wn * 64 + (lane & 15);
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int col = ccol0 + fn * 16;
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow0 + fm * 16 + r;
+        if (row >= M) continue;
+        atomicAdd(C + (int64_t)row * ldc + col, acc[fm][fn][r] * alpha);
+      }
+    }
 }
 
 // fast path: interior tile, aligned — direct-to-LDS DMA, 16 B per lane.
@@ -240,6 +309,14 @@ void gemm_bf16(const void* A_, const void* B_, void* C, const float* bias,
     else                      COS_GEMM_CASE(TA, TB, 2);                     \
   } while (0)
 
+  if (trans_a && trans_b && store_mode == 2 && N > 128) {
+    int nb_w = (N + WBN - 1) / WBN;
+    dim3 gw(mblocks * nb_w, zblocks);
+    gemm_tt_wide_kernel<<<gw, WNT, 0, stream>>>(
+        A, B, reinterpret_cast<float*>(C), M, N, K, lda, ldb, ldc, ksplit,
+        alpha);
+    return;
+  }
   if (!trans_a && !trans_b)      COS_GEMM_SM(false, false);
   else if (!trans_a && trans_b)  COS_GEMM_SM(false, true);
   else if (trans_a && !trans_b)  COS_GEMM_SM(true, false);

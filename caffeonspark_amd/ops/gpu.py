@@ -62,7 +62,8 @@ def _conv_out_dim(h, k, s, p, dil):
     return (h + 2 * p - eff) // s + 1
 
 
-def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None):
+def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
+                   relu=False):
     _check_bf16(x, "conv input")
     sh, sw = stride
     ph, pw = pad
@@ -94,7 +95,8 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None):
         # C[npq, kout_g] — write into the column slice of NHWC y
         _gemm(col[g], wr[g * Kg:(g + 1) * Kg], y2[:, g * Kg:],
               bias_f[g * Kg:(g + 1) * Kg] if bias_f is not None else None,
-              NPQ, Kg, Kpad, Kpad, Kpad, Kout, False, False, 0, 1)
+              NPQ, Kg, Kpad, Kpad, Kpad, Kout, False, False, 0, 1,
+              relu=relu)
     if ctx is not None:
         ctx["col"] = col
         ctx["shape"] = (N, C, H, W, P, Q, R, S, sh, sw, ph, pw, dil, G, Cg,
@@ -148,7 +150,7 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
 
 # ----------------------------------------------------------------- linear
 
-def fc_forward(x, w, b):
+def fc_forward(x, w, b, relu=False):
     _check_bf16(x, "fc input")
     x = x.contiguous()
     wb = _as_bf16(w).contiguous()
@@ -156,7 +158,8 @@ def fc_forward(x, w, b):
     Nout = wb.shape[0]
     y = torch.empty((M, Nout), dtype=torch.bfloat16, device=x.device)
     bias_f = b.float().contiguous() if b is not None else None
-    _gemm(x, wb, y, bias_f, M, Nout, K, K, K, Nout, False, False, 0, 1)
+    _gemm(x, wb, y, bias_f, M, Nout, K, K, K, Nout, False, False, 0, 1,
+          relu=relu)
     return y
 
 

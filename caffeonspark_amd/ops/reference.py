@@ -16,9 +16,11 @@ import torch.nn.functional as F
 
 # ----------------------------------------------------------------- convolution
 
-def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None):
-    return F.conv2d(x, w, b, stride=stride, padding=pad,
-                    dilation=dilation, groups=groups)
+def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
+                   relu=False):
+    y = F.conv2d(x, w, b, stride=stride, padding=pad,
+                 dilation=dilation, groups=groups)
+    return F.relu(y) if relu else y
 
 
 def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
@@ -39,9 +41,10 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
 
 # ------------------------------------------------------------- inner product
 
-def fc_forward(x, w, b):
+def fc_forward(x, w, b, relu=False):
     """x: [M, K], w: [N, K] (caffe layout), b: [N] or None."""
-    return F.linear(x, w, b)
+    y = F.linear(x, w, b)
+    return F.relu(y) if relu else y
 
 
 def fc_backward(x, w, dy, need_dx=True, bias=True):
