@@ -122,12 +122,18 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
 
     dx = dw = db = None
     if need_dw:
+        # dw[kout][kpad] = sum_npq dy[npq][kout] * col[npq][kpad].
+        # Both operands are K(=npq)-major; transposing them once (cheap
+        # torch copy kernels) turns the GEMM into the fast NT direct/direct
+        # form with global_load_lds staging instead of LDS scatter staging.
+        dyT = dyl.permute(1, 0, 2, 3).reshape(Kout, NPQ).contiguous()
         dwp = torch.zeros((Kout, Kpad), dtype=torch.float32, device=dy.device)
         for g in range(G):
+            colT = col[g].t().contiguous()
             mb, nb = (Kg + 127) // 128, (Kpad + 127) // 128
             sk = _splitk_for(mb, nb, NPQ)
-            _gemm(dy2[:, g * Kg:], col[g], dwp[g * Kg:],
-                  None, Kg, Kpad, NPQ, Kout, Kpad, Kpad, True, True, 2, sk)
+            _gemm(dyT[g * Kg:(g + 1) * Kg], colT, dwp[g * Kg:],
+                  None, Kg, Kpad, NPQ, NPQ, NPQ, Kpad, False, False, 2, sk)
         dw = dwp[:, :Kcol].reshape(Kout, R, S, Cg).permute(0, 3, 1, 2) \
             .contiguous()
     if bias:
@@ -140,9 +146,11 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                          device=dy.device) \
             .contiguous(memory_format=torch.channels_last)
         for g in range(G):
-            # dcol[npq, kpad] = dy_g[npq, kg] @ w_g[kg, kpad]
-            _gemm(dy2[:, g * Kg:], wr[g * Kg:(g + 1) * Kg], dcol, None,
-                  NPQ, Kpad, Kg, Kout, Kpad, Kpad, False, True, 0, 1)
+            # dcol[npq, kpad] = dy_g[npq, kg] @ w_g[kg, kpad]: transpose the
+            # (small) packed weights so B is [kpad][kg] = NT direct form
+            wrT = wr[g * Kg:(g + 1) * Kg].t().contiguous()
+            _gemm(dy2[:, g * Kg:], wrT, dcol, None,
+                  NPQ, Kpad, Kg, Kout, Kg, Kpad, False, False, 0, 1)
             _ext.col2im(dcol, dx, N, H, W, C, P, Q, R, S, sh, sw, ph, pw,
                         dil, Kpad, g * Cg, Cg)
     return dx, dw, db
@@ -172,11 +180,17 @@ def fc_backward(x, w, dy, need_dx=True, bias=True):
     Nout = wb.shape[0]
     dx = dw = db = None
     if need_dx:
+        # dx = dy @ w: transpose w once -> NT direct/direct fast staging
+        wT = wb.t().contiguous()
         dx = torch.empty((M, K), dtype=torch.bfloat16, device=x.device)
-        _gemm(dy, wb, dx, None, M, K, Nout, Nout, K, K, False, True, 0, 1)
+        _gemm(dy, wT, dx, None, M, K, Nout, Nout, Nout, K, False, False,
+              0, 1)
+    # dw = dy^T @ x: transpose both (small vs the GEMM) -> NT fast form
+    dyT = dy.t().contiguous()
+    xT = x.t().contiguous()
     dwp = torch.zeros((Nout, K), dtype=torch.float32, device=x.device)
     mb, nb = (Nout + 127) // 128, (K + 127) // 128
-    _gemm(dy, x, dwp, None, Nout, K, M, Nout, K, K, True, True, 2,
+    _gemm(dyT, xT, dwp, None, Nout, K, M, M, M, K, False, False, 2,
           _splitk_for(mb, nb, M))
     dw = dwp
     if bias:
