@@ -134,6 +134,44 @@ __global__ void sgd_update_kernel(float* __restrict__ p, const float* __restrict
   }
 }
 
+// ------------------------------------------------------------ transpose
+// Tiled 2D bf16 transpose [R][C] -> [C][R]: 64x64 tiles staged through
+// LDS (row pad 66 u16 -> conflict-free transposed reads), coalesced
+// 128-byte accesses on both sides.  Used to turn K-major backward GEMM
+// operands into the fast NT direct form.
+
+__global__ void transpose_bf16_kernel(const u16* __restrict__ in,
+                                      u16* __restrict__ out,
+                                      int R, int C, int tiles_c) {
+  __shared__ u16 lds[64][66];
+  int tile = blockIdx.x;
+  int r0 = (tile / tiles_c) * 64;
+  int c0 = (tile % tiles_c) * 64;
+  int tc = threadIdx.x & 63;
+  int tr = threadIdx.x >> 6;           // 0..3
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    int r = tr + i * 4;
+    int gr = r0 + r, gc = c0 + tc;
+    lds[r][tc] = (gr < R && gc < C) ? in[(int64_t)gr * C + gc] : 0;
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    int cl = tr + i * 4;               // output row = original col
+    int gc = c0 + cl, gr = r0 + tc;    // output col = original row
+    if (gc < C && gr < R)
+      out[(int64_t)gc * R + gr] = lds[tc][cl];
+  }
+}
+
+void transpose_bf16(const void* in, void* out, int64_t R, int64_t C,
+                    hipStream_t stream) {
+  int tiles_r = (int)((R + 63) / 64), tiles_c = (int)((C + 63) / 64);
+ hipLaunchKernelGGL(( transpose_bf16_kernel), dim3(tiles_r * tiles_c), dim3(256), 0, stream, 
+      (const u16*)in, (u16*)out, (int)R, (int)C, tiles_c);
+}
+
 // ---------------------------------------------------------- column reduce
 // out[c] += sum_r in[r*ld + c]  (bias gradients; in bf16, out fp32)
 // Block = 256 threads as [4 row-groups x 64 cols]: coalesced 64-wide column

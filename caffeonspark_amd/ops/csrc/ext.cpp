@@ -51,6 +51,8 @@ void sgd_update(float* p, const float* g, float* v, float lr, float mu,
                 float wd, int64_t n, hipStream_t stream);
 void colsum(const void* in, float* out, int64_t rows, int cols, int ld,
             hipStream_t stream);
+void transpose_bf16(const void* in, void* out, int64_t R, int64_t C,
+                    hipStream_t stream);
 void lstm_unit_fwd(const float* c_prev, const void* gates, const void* cont,
                    float* c_out, void* h_out, float* act, int64_t n, int H,
                    hipStream_t stream);
@@ -195,6 +197,11 @@ void py_sgd_update(Tensor p, Tensor g, Tensor v, double lr, double mu,
                      cur_stream());
 }
 
+void py_transpose(Tensor in, Tensor out, int64_t R, int64_t C) {
+  CHECK_BF16(in); CHECK_BF16(out);
+  cosamd::transpose_bf16(in.data_ptr(), out.data_ptr(), R, C, cur_stream());
+}
+
 void py_colsum(Tensor in, Tensor out, int64_t rows, int64_t cols,
                int64_t ld) {
   CHECK_F32(out);
@@ -272,6 +279,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mul", &py_mul);
   m.def("sgd_update", &py_sgd_update);
   m.def("colsum", &py_colsum);
+  m.def("transpose", &py_transpose);
   m.def("lstm_unit_fwd", &py_lstm_unit_fwd);
   m.def("lstm_unit_bwd", &py_lstm_unit_bwd);
   m.def("embed_fwd", &py_embed_fwd);

@@ -49,6 +49,14 @@ def _splitk_for(mb: int, nb: int, K: int) -> int:
     return max(1, min(K // 256, 2048 // max(1, tiles)))
 
 
+def _transpose(t: torch.Tensor) -> torch.Tensor:
+    """Fast tiled transpose of a 2-D contiguous bf16 tensor."""
+    R, C = t.shape
+    out = torch.empty((C, R), dtype=t.dtype, device=t.device)
+    _ext.transpose(t, out, R, C)
+    return out
+
+
 def _gemm(A, B, C, bias, M, N, K, lda, ldb, ldc, ta, tb, store, splitk,
           relu=False, alpha=1.0):
     _ext.gemm(A, B, C, bias, M, N, K, lda, ldb, ldc, ta, tb, store, splitk,
@@ -126,10 +134,10 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         # Both operands are K(=npq)-major; transposing them once (cheap
         # torch copy kernels) turns the GEMM into the fast NT direct/direct
         # form with global_load_lds staging instead of LDS scatter staging.
-        dyT = dyl.permute(1, 0, 2, 3).reshape(Kout, NPQ).contiguous()
+        dyT = _transpose(dy2)  # [Kout][NPQ]
         dwp = torch.zeros((Kout, Kpad), dtype=torch.float32, device=dy.device)
         for g in range(G):
-            colT = col[g].t().contiguous()
+            colT = _transpose(col[g])
             mb, nb = (Kg + 127) // 128, (Kpad + 127) // 128
             sk = _splitk_for(mb, nb, NPQ)
             _gemm(dyT[g * Kg:(g + 1) * Kg], colT, dwp[g * Kg:],
@@ -148,7 +156,7 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         for g in range(G):
             # dcol[npq, kpad] = dy_g[npq, kg] @ w_g[kg, kpad]: transpose the
             # (small) packed weights so B is [kpad][kg] = NT direct form
-            wrT = wr[g * Kg:(g + 1) * Kg].t().contiguous()
+            wrT = _transpose(wr[g * Kg:(g + 1) * Kg].contiguous())
             _gemm(dy2[:, g * Kg:], wrT, dcol, None,
                   NPQ, Kpad, Kg, Kout, Kg, Kpad, False, False, 0, 1)
             _ext.col2im(dcol, dx, N, H, W, C, P, Q, R, S, sh, sw, ph, pw,
@@ -181,13 +189,13 @@ def fc_backward(x, w, dy, need_dx=True, bias=True):
     dx = dw = db = None
     if need_dx:
         # dx = dy @ w: transpose w once -> NT direct/direct fast staging
-        wT = wb.t().contiguous()
+        wT = _transpose(wb)
         dx = torch.empty((M, K), dtype=torch.bfloat16, device=x.device)
         _gemm(dy, wT, dx, None, M, K, Nout, Nout, Nout, K, False, False,
               0, 1)
     # dw = dy^T @ x: transpose both (small vs the GEMM) -> NT fast form
-    dyT = dy.t().contiguous()
-    xT = x.t().contiguous()
+    dyT = _transpose(dy)
+    xT = _transpose(x)
     dwp = torch.zeros((Nout, K), dtype=torch.float32, device=x.device)
     mb, nb = (Nout + 127) // 128, (K + 127) // 128
     _gemm(dyT, xT, dwp, None, Nout, K, M, M, M, K, False, False, 2,
