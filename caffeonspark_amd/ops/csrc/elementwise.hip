@@ -146,21 +146,47 @@ __global__ void transpose_bf16_kernel(const u16* __restrict__ in,
   int tile = blockIdx.x;
   int r0 = (tile / tiles_c) * 64;
   int c0 = (tile % tiles_c) * 64;
-  int tc = threadIdx.x & 63;
-  int tr = threadIdx.x >> 6;           // 0..3
+  int t = threadIdx.x;
+  int lc8 = (t & 7) * 8;               // 8-wide column slot
+  int lr = t >> 3;                     // 0..31
+  bool interior = (r0 + 64 <= R) && (c0 + 64 <= C);
 #pragma unroll
-  for (int i = 0; i < 16; ++i) {
-    int r = tr + i * 4;
-    int gr = r0 + r, gc = c0 + tc;
-    lds[r][tc] = (gr < R && gc < C) ? in[(int64_t)gr * C + gc] : 0;
+  for (int half = 0; half < 2; ++half) {
+    int r = lr + half * 32;
+    int gr = r0 + r;
+    if (interior) {
+      u16x8 v = *reinterpret_cast<const u16x8*>(
+          in + (int64_t)gr * C + c0 + lc8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds[r][lc8 + j] = v[j];
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int gc = c0 + lc8 + j;
+        lds[r][lc8 + j] = (gr < R && gc < C)
+                              ? in[(int64_t)gr * C + gc] : 0;
+      }
+    }
   }
   __syncthreads();
+  int cl = t >> 3;                      // output row = original col
+  int lr8 = (t & 7) * 8;                // 8-wide along original rows
 #pragma unroll
-  for (int i = 0; i < 16; ++i) {
-    int cl = tr + i * 4;               // output row = original col
-    int gc = c0 + cl, gr = r0 + tc;    // output col = original row
-    if (gc < C && gr < R)
-      out[(int64_t)gc * R + gr] = lds[tc][cl];
+  for (int half = 0; half < 2; ++half) {
+    int c = cl + half * 32;
+    int gc = c0 + c;
+    if (interior) {
+      u16x8 v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = lds[lr8 + j][c];
+      *reinterpret_cast<u16x8*>(out + (int64_t)gc * R + r0 + lr8) = v;
+    } else if (gc < C) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int gr = r0 + lr8 + j;
+        if (gr < R) out[(int64_t)gc * R + gr] = lds[lr8 + j][c];
+      }
+    }
   }
 }
 

@@ -14,7 +14,8 @@ namespace cosamd {
 void gemm_bf16(const void* A, const void* B, void* C, const float* bias,
                int M, int N, int K, int lda, int ldb, int ldc,
                bool trans_a, bool trans_b, int store_mode, int splitk,
-               bool relu, float alpha, hipStream_t stream);
+               bool relu, float alpha, int m_alloc, int n_alloc,
+               hipStream_t stream);
 void im2col_nhwc(const void* x, void* col, int N, int H, int W, int C,
                  int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
                  int dil, int Kpad, int c0, int Ct, hipStream_t stream);
@@ -90,7 +91,8 @@ hipStream_t cur_stream() {
 void py_gemm(Tensor A, Tensor B, Tensor C, c10::optional<Tensor> bias,
              int64_t M, int64_t N, int64_t K, int64_t lda, int64_t ldb,
              int64_t ldc, bool trans_a, bool trans_b, int64_t store_mode,
-             int64_t splitk, bool relu, double alpha) {
+             int64_t splitk, bool relu, double alpha, int64_t m_alloc,
+             int64_t n_alloc) {
   CHECK_CUDA(A); CHECK_BF16(A); CHECK_BF16(B);
   const float* bptr = nullptr;
   if (bias.has_value()) {
@@ -100,7 +102,7 @@ void py_gemm(Tensor A, Tensor B, Tensor C, c10::optional<Tensor> bias,
   cosamd::gemm_bf16(
       A.data_ptr(), B.data_ptr(), C.data_ptr(), bptr, (int)M, (int)N, (int)K, (int)lda, (int)ldb, (int)ldc,
       trans_a, trans_b, (int)store_mode, (int)splitk, relu, (float)alpha,
-      cur_stream());
+      (int)m_alloc, (int)n_alloc, cur_stream());
 }
 
 void py_im2col(Tensor x, Tensor col, int64_t N, int64_t H, int64_t W,

@@ -191,7 +191,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     int M, int N, int K, int lda, int ldb, int ldc,
-    int ksplit, int relu, float alpha) {
+    int ksplit, int relu, float alpha, int m_alloc, int n_alloc) {
   __shared__ bf16 As[BM * BK];
   __shared__ bf16 Bs[BN * BK];
 
@@ -211,9 +211,12 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
 
   f32x4 acc[4][4] = {};
 
-  bool a_fast = !TRANS_A && (tile_m + BM <= M) && (lda % 8 == 0) &&
+  // fast path only needs the STAGED rows to be readable: callers may
+  // over-allocate operand buffers (m_alloc/n_alloc >= M/N) so edge tiles
+  // stage junk rows whose outputs the epilogue guards discard
+  bool a_fast = !TRANS_A && (tile_m + BM <= m_alloc) && (lda % 8 == 0) &&
                 ((k_end - k_begin) % BK == 0) && (k_begin % 8 == 0);
-  bool b_fast = !TRANS_B && (tile_n + BN <= N) && (ldb % 8 == 0) &&
+  bool b_fast = !TRANS_B && (tile_n + BN <= n_alloc) && (ldb % 8 == 0) &&
                 ((k_end - k_begin) % BK == 0) && (k_begin % 8 == 0);
 
   for (int k0 = k_begin; k0 < k_end; k0 += BK) {
@@ -286,7 +289,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
 void gemm_bf16(const void* A_, const void* B_, void* C, const float* bias,
                int M, int N, int K, int lda, int ldb, int ldc,
                bool trans_a, bool trans_b, int store_mode, int splitk,
-               bool relu, float alpha, hipStream_t stream) {
+               bool relu, float alpha, int m_alloc, int n_alloc,
+               hipStream_t stream) {
+  if (m_alloc < M) m_alloc = M;
+  if (n_alloc < N) n_alloc = N;
   const bf16* A = reinterpret_cast<const bf16*>(A_);
   const bf16* B = reinterpret_cast<const bf16*>(B_);
   int mblocks = (M + BM - 1) / BM, nblocks = (N + BN - 1) / BN;
@@ -301,7 +307,8 @@ void gemm_bf16(const void* A_, const void* B_, void* C, const float* bias,
 
 #define COS_GEMM_CASE(TA, TB, SM)                                          \
  hipLaunchKernelGGL(( gemm_kernel<TA, TB, SM>), dim3(grid), dim3(block), 0, stream,                       \
-      A, B, C, bias, M, N, K, lda, ldb, ldc, ksplit, relu ? 1 : 0, alpha)
+      A, B, C, bias, M, N, K, lda, ldb, ldc, ksplit, relu ? 1 : 0, alpha,    \
+      m_alloc, n_alloc)
 
 #define COS_GEMM_SM(TA, TB)                                                 \
   do {                                                                      \

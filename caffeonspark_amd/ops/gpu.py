@@ -49,18 +49,28 @@ def _splitk_for(mb: int, nb: int, K: int) -> int:
     return max(1, min(K // 256, 2048 // max(1, tiles)))
 
 
+def _pad128(n: int) -> int:
+    return (n + 127) // 128 * 128
+
+
 def _transpose(t: torch.Tensor) -> torch.Tensor:
-    """Fast tiled transpose of a 2-D contiguous bf16 tensor."""
+    """Fast tiled transpose of a 2-D contiguous bf16 tensor.  Rows are
+    over-allocated to a multiple of 128 so the GEMM's direct-to-LDS fast
+    path can stage edge tiles (junk rows are never read back)."""
     R, C = t.shape
-    out = torch.empty((C, R), dtype=t.dtype, device=t.device)
+    buf = torch.empty((_pad128(C), R), dtype=t.dtype, device=t.device)
+    out = buf[:C]
     _ext.transpose(t, out, R, C)
     return out
 
 
 def _gemm(A, B, C, bias, M, N, K, lda, ldb, ldc, ta, tb, store, splitk,
-          relu=False, alpha=1.0):
+          relu=False, alpha=1.0, ma=0, na=0):
+    # staging-allocation bounds: number of rows safely readable past M/N
+    ma = ma or (A.shape[0] if not ta else M)
+    na = na or (B.shape[0] if not tb else N)
     _ext.gemm(A, B, C, bias, M, N, K, lda, ldb, ldc, ta, tb, store, splitk,
-              relu, alpha)
+              relu, alpha, ma, na)
 
 
 # ----------------------------------------------------------------- conv2d
@@ -87,9 +97,11 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
 
     xl = _cl(x)
     # weight repack: [K, Cg, R, S] -> bf16 [K, R, S, Cg] padded to Kpad
-    wr = _as_bf16(w).permute(0, 2, 3, 1).contiguous().reshape(Kout, Kcol)
-    if Kpad != Kcol:
-        wr = torch.nn.functional.pad(wr, (0, Kpad - Kcol))
+    # columns and 128-aligned rows (GEMM fast-staging bound)
+    wrb = torch.zeros((_pad128(Kout), Kpad), dtype=torch.bfloat16,
+                      device=x.device)
+    wrb[:Kout, :Kcol] = _as_bf16(w).permute(0, 2, 3, 1).reshape(Kout, Kcol)
+    wr = wrb[:Kout]
     bias_f = b.float().contiguous() if b is not None else None
 
     y = torch.empty((N, Kout, P, Q), dtype=torch.bfloat16, device=x.device) \
@@ -104,7 +116,7 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
         _gemm(col[g], wr[g * Kg:(g + 1) * Kg], y2[:, g * Kg:],
               bias_f[g * Kg:(g + 1) * Kg] if bias_f is not None else None,
               NPQ, Kg, Kpad, Kpad, Kpad, Kout, False, False, 0, 1,
-              relu=relu)
+              relu=relu, na=wrb.shape[0] - g * Kg)
     if ctx is not None:
         ctx["col"] = col
         ctx["shape"] = (N, C, H, W, P, Q, R, S, sh, sw, ph, pw, dil, G, Cg,
@@ -134,14 +146,15 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         # Both operands are K(=npq)-major; transposing them once (cheap
         # torch copy kernels) turns the GEMM into the fast NT direct/direct
         # form with global_load_lds staging instead of LDS scatter staging.
-        dyT = _transpose(dy2)  # [Kout][NPQ]
+        dyT = _transpose(dy2)  # [Kout][NPQ], rows over-allocated
         dwp = torch.zeros((Kout, Kpad), dtype=torch.float32, device=dy.device)
         for g in range(G):
             colT = _transpose(col[g])
             mb, nb = (Kg + 127) // 128, (Kpad + 127) // 128
             sk = _splitk_for(mb, nb, NPQ)
             _gemm(dyT[g * Kg:(g + 1) * Kg], colT, dwp[g * Kg:],
-                  None, Kg, Kpad, NPQ, NPQ, NPQ, Kpad, False, False, 2, sk)
+                  None, Kg, Kpad, NPQ, NPQ, NPQ, Kpad, False, False, 2, sk,
+                  ma=_pad128(Kout) - g * Kg, na=_pad128(Kpad))
         dw = dwp[:, :Kcol].reshape(Kout, R, S, Cg).permute(0, 3, 1, 2) \
             .contiguous()
     if bias:
@@ -158,7 +171,8 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
             # (small) packed weights so B is [kpad][kg] = NT direct form
             wrT = _transpose(wr[g * Kg:(g + 1) * Kg].contiguous())
             _gemm(dy2[:, g * Kg:], wrT, dcol, None,
-                  NPQ, Kpad, Kg, Kout, Kg, Kpad, False, False, 0, 1)
+                  NPQ, Kpad, Kg, Kout, Kg, Kpad, False, False, 0, 1,
+                  na=_pad128(Kpad))
             _ext.col2im(dcol, dx, N, H, W, C, P, Q, R, S, sh, sw, ph, pw,
                         dil, Kpad, g * Cg, Cg)
     return dx, dw, db
@@ -169,13 +183,21 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
 def fc_forward(x, w, b, relu=False):
     _check_bf16(x, "fc input")
     x = x.contiguous()
-    wb = _as_bf16(w).contiguous()
     M, K = x.shape
-    Nout = wb.shape[0]
+    Nout = w.shape[0]
+    if Nout % 128:
+        wb = torch.zeros((_pad128(Nout), K), dtype=torch.bfloat16,
+                         device=x.device)
+        wb[:Nout] = _as_bf16(w)
+        na = wb.shape[0]
+        wb = wb[:Nout]
+    else:
+        wb = _as_bf16(w).contiguous()
+        na = Nout
     y = torch.empty((M, Nout), dtype=torch.bfloat16, device=x.device)
     bias_f = b.float().contiguous() if b is not None else None
     _gemm(x, wb, y, bias_f, M, Nout, K, K, K, Nout, False, False, 0, 1,
-          relu=relu)
+          relu=relu, na=na)
     return y
 
 
@@ -192,14 +214,14 @@ def fc_backward(x, w, dy, need_dx=True, bias=True):
         wT = _transpose(wb)
         dx = torch.empty((M, K), dtype=torch.bfloat16, device=x.device)
         _gemm(dy, wT, dx, None, M, K, Nout, Nout, Nout, K, False, False,
-              0, 1)
+              0, 1, na=_pad128(K))
     # dw = dy^T @ x: transpose both (small vs the GEMM) -> NT fast form
     dyT = _transpose(dy)
     xT = _transpose(x)
     dwp = torch.zeros((Nout, K), dtype=torch.float32, device=x.device)
     mb, nb = (Nout + 127) // 128, (K + 127) // 128
     _gemm(dyT, xT, dwp, None, Nout, K, M, M, M, K, False, False, 2,
-          _splitk_for(mb, nb, M))
+          _splitk_for(mb, nb, M), ma=_pad128(Nout), na=_pad128(K))
     dw = dwp
     if bias:
         db = torch.zeros(Nout, dtype=torch.float32, device=x.device)
