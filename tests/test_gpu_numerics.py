@@ -52,7 +52,7 @@ def test_gemm_combos(ta, tb, m, n, k):
     b = bf(torch.randn(b_shape)).to(dev())
     c = torch.empty(m, n, dtype=torch.bfloat16, device=dev())
     ext.gemm(a, b, c, None, m, n, k, a.shape[1], b.shape[1], n,
-             ta, tb, 0, 1, False, 1.0)
+             ta, tb, 0, 1, False, 1.0, m, n)
     torch.cuda.synchronize()
     want = _gemm_ref(a.cpu(), b.cpu(), ta, tb)
     agree(c, want, rtol=0.03, atol=0.05 * (k ** 0.5) * 0.02)
@@ -65,7 +65,8 @@ def test_gemm_bias_relu():
     b = bf(torch.randn(n, k)).to(dev())
     bias = torch.randn(n).to(dev())
     c = torch.empty(m, n, dtype=torch.bfloat16, device=dev())
-    ext.gemm(a, b, c, bias, m, n, k, k, k, n, False, False, 0, 1, True, 1.0)
+    ext.gemm(a, b, c, bias, m, n, k, k, k, n, False, False, 0, 1, True,
+             1.0, m, n)
     torch.cuda.synchronize()
     want = torch.relu(_gemm_ref(a.cpu(), b.cpu(), False, False)
                       + bias.cpu().float())
@@ -78,7 +79,8 @@ def test_gemm_splitk_atomic():
     a = bf(torch.randn(k, m) * 0.1).to(dev())   # trans A
     b = bf(torch.randn(k, n) * 0.1).to(dev())   # trans B
     c = torch.zeros(m, n, dtype=torch.float32, device=dev())
-    ext.gemm(a, b, c, None, m, n, k, m, n, n, True, True, 2, 8, False, 1.0)
+    ext.gemm(a, b, c, None, m, n, k, m, n, n, True, True, 2, 8, False,
+             1.0, m, n)
     torch.cuda.synchronize()
     want = a.cpu().float().t() @ b.cpu().float()
     agree(c, want, rtol=0.03, atol=0.5)
