@@ -104,8 +104,8 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
     wr = wrb[:Kout]
     bias_f = b.float().contiguous() if b is not None else None
 
-    y = torch.empty((N, Kout, P, Q), dtype=torch.bfloat16, device=x.device) \
-        .contiguous(memory_format=torch.channels_last)
+    y = torch.empty((N, Kout, P, Q), dtype=torch.bfloat16, device=x.device,
+                    memory_format=torch.channels_last)
     NPQ = N * P * Q
     y2 = y.permute(0, 2, 3, 1).reshape(NPQ, Kout)  # NHWC flat alias (view)
     col = torch.empty((G, NPQ, Kpad), dtype=torch.bfloat16, device=x.device)
@@ -164,8 +164,8 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         dcol = torch.empty((NPQ, Kpad), dtype=torch.bfloat16,
                            device=dy.device)
         dx = torch.empty((N, C, H, W), dtype=torch.bfloat16,
-                         device=dy.device) \
-            .contiguous(memory_format=torch.channels_last)
+                         device=dy.device,
+                    memory_format=torch.channels_last)
         for g in range(G):
             # dcol[npq, kpad] = dy_g[npq, kg] @ w_g[kg, kpad]: transpose the
             # (small) packed weights so B is [kpad][kg] = NT direct form
@@ -276,8 +276,8 @@ def maxpool_forward(x, kernel, stride, pad):
     ph, pw = pad
     P, Q = _pool_out(H, kh, sh, ph), _pool_out(W, kw, sw, pw)
     xl = _cl(x)
-    y = torch.empty((N, C, P, Q), dtype=x.dtype, device=x.device) \
-        .contiguous(memory_format=torch.channels_last)
+    y = torch.empty((N, C, P, Q), dtype=x.dtype, device=x.device,
+                    memory_format=torch.channels_last)
     idx = torch.empty((N, P, Q, C), dtype=torch.int32, device=x.device)
     _ext.maxpool_fwd(xl, y, idx, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw)
     return y, (idx, (kh, kw, sh, sw, ph, pw), (N, C, H, W))
@@ -286,8 +286,8 @@ def maxpool_forward(x, kernel, stride, pad):
 def maxpool_backward(x_shape, idx_pack, dy):
     idx, (kh, kw, sh, sw, ph, pw), (N, C, H, W) = idx_pack
     P, Q = dy.shape[2], dy.shape[3]
-    dx = torch.empty((N, C, H, W), dtype=dy.dtype, device=dy.device) \
-        .contiguous(memory_format=torch.channels_last)
+    dx = torch.empty((N, C, H, W), dtype=dy.dtype, device=dy.device,
+                    memory_format=torch.channels_last)
     _ext.maxpool_bwd(_cl(dy), idx, dx, N, H, W, C, P, Q, kh, kw, sh, sw,
                      ph, pw)
     return dx
@@ -300,8 +300,8 @@ def avgpool_forward(x, kernel, stride, pad):
     ph, pw = pad
     P, Q = _pool_out(H, kh, sh, ph), _pool_out(W, kw, sw, pw)
     xl = _cl(x)
-    y = torch.empty((N, C, P, Q), dtype=x.dtype, device=x.device) \
-        .contiguous(memory_format=torch.channels_last)
+    y = torch.empty((N, C, P, Q), dtype=x.dtype, device=x.device,
+                    memory_format=torch.channels_last)
     _ext.avgpool_fwd(xl, y, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw)
     return y
 
@@ -312,8 +312,8 @@ def avgpool_backward(x, kernel, stride, pad, dy):
     sh, sw = stride
     ph, pw = pad
     P, Q = dy.shape[2], dy.shape[3]
-    dx = torch.empty((N, C, H, W), dtype=dy.dtype, device=dy.device) \
-        .contiguous(memory_format=torch.channels_last)
+    dx = torch.empty((N, C, H, W), dtype=dy.dtype, device=dy.device,
+                    memory_format=torch.channels_last)
     _ext.avgpool_bwd(_cl(dy), dx, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw)
     return dx
 
@@ -321,16 +321,16 @@ def avgpool_backward(x, kernel, stride, pad, dy):
 def global_avgpool_forward(x):
     N, C, H, W = x.shape
     xl = _cl(x)
-    y = torch.empty((N, C, 1, 1), dtype=x.dtype, device=x.device) \
-        .contiguous(memory_format=torch.channels_last)
+    y = torch.empty((N, C, 1, 1), dtype=x.dtype, device=x.device,
+                    memory_format=torch.channels_last)
     _ext.avgpool_fwd(xl, y, N, H, W, C, 1, 1, H, W, 1, 1, 0, 0)
     return y
 
 
 def global_avgpool_backward(x_shape, dy):
     N, C, H, W = x_shape
-    dx = torch.empty((N, C, H, W), dtype=dy.dtype, device=dy.device) \
-        .contiguous(memory_format=torch.channels_last)
+    dx = torch.empty((N, C, H, W), dtype=dy.dtype, device=dy.device,
+                    memory_format=torch.channels_last)
     _ext.avgpool_bwd(_cl(dy), dx, N, H, W, C, 1, 1, H, W, 1, 1, 0, 0)
     return dx
 
