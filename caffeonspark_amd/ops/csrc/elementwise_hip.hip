@@ -134,6 +134,31 @@ __global__ void sgd_update_kernel(float* __restrict__ p, const float* __restrict
   }
 }
 
+// ------------------------------------------------------- bias+act+cast
+// Finalize pass for split-K forward GEMMs: bf16(relu(acc + bias[col])).
+
+__global__ void bias_act_cast_kernel(const float* __restrict__ in,
+                                     const float* __restrict__ bias,
+                                     u16* __restrict__ out, int64_t rows,
+                                     int cols, int relu) {
+  int64_t total = rows * cols;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    float v = in[i];
+    if (bias != nullptr) v += bias[i % cols];
+    if (relu && v < 0.f) v = 0.f;
+    stbf(out + i, v);
+  }
+}
+
+void bias_act_cast(const float* in, const float* bias, void* out,
+                   int64_t rows, int cols, bool relu, hipStream_t stream) {
+  int64_t total = rows * cols;
+  int blocks = (int)hmin<int64_t>(4096, (total + 255) / 256);
+ hipLaunchKernelGGL(( bias_act_cast_kernel), dim3(blocks), dim3(256), 0, stream, 
+      in, bias, (u16*)out, rows, cols, relu ? 1 : 0);
+}
+
 // ------------------------------------------------------------ transpose
 // Tiled 2D bf16 transpose [R][C] -> [C][R]: 64x64 tiles staged through
 // LDS (row pad 66 u16 -> conflict-free transposed reads), coalesced

@@ -54,6 +54,8 @@ void colsum(const void* in, float* out, int64_t rows, int cols, int ld,
             hipStream_t stream);
 void transpose_bf16(const void* in, void* out, int64_t R, int64_t C,
                     hipStream_t stream);
+void bias_act_cast(const float* in, const float* bias, void* out,
+                   int64_t rows, int cols, bool relu, hipStream_t stream);
 void lstm_unit_fwd(const float* c_prev, const void* gates, const void* cont,
                    float* c_out, void* h_out, float* act, int64_t n, int H,
                    hipStream_t stream);
@@ -199,6 +201,15 @@ void py_sgd_update(Tensor p, Tensor g, Tensor v, double lr, double mu,
                      cur_stream());
 }
 
+void py_bias_act_cast(Tensor in, c10::optional<Tensor> bias, Tensor out,
+                      bool relu) {
+  CHECK_F32(in);
+  const float* bptr = bias.has_value() ? bias.value().data_ptr<float>()
+                                       : nullptr;
+  cosamd::bias_act_cast(in.data_ptr<float>(), bptr, out.data_ptr(),
+                        in.size(0), in.size(1), relu, cur_stream());
+}
+
 void py_transpose(Tensor in, Tensor out, int64_t R, int64_t C) {
   CHECK_BF16(in); CHECK_BF16(out);
   cosamd::transpose_bf16(in.data_ptr(), out.data_ptr(), R, C, cur_stream());
@@ -282,6 +293,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_update", &py_sgd_update);
   m.def("colsum", &py_colsum);
   m.def("transpose", &py_transpose);
+  m.def("bias_act_cast", &py_bias_act_cast);
   m.def("lstm_unit_fwd", &py_lstm_unit_fwd);
   m.def("lstm_unit_bwd", &py_lstm_unit_bwd);
   m.def("embed_fwd", &py_embed_fwd);

@@ -16,32 +16,44 @@ typedef unsigned short u16;
 typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
 
 // small-C variant (e.g. conv1, C=3): one thread per 8 output columns —
-// coalesced 16B writes; the gather reads hit L2/L3 (input is small).
+// coalesced 16B writes; a per-block LDS lookup table kills the per-element
+// div/mod for k -> (dh, dw, c); gather reads hit L2/L3 (input is small).
 __global__ void im2col_smallc_kernel(
     const u16* __restrict__ x, u16* __restrict__ col,
     int N, int H, int W, int C, int P, int Q,
     int R, int S, int sh, int sw, int ph, int pw,
     int dil, int Kpad, int c0, int Ct, int64_t total8) {
+  __shared__ int lut[1024];  // (dh<<20)|(dw<<10)|c, or -1 for pad columns
   int Kcol = R * S * Ct;
+  for (int k = threadIdx.x; k < Kpad; k += blockDim.x) {
+    if (k < Kcol) {
+      int rs = k / Ct, c = k % Ct;
+      lut[k] = (((rs / S) * dil) << 20) | (((rs % S) * dil) << 10) | c;
+    } else {
+      lut[k] = -1;
+    }
+  }
+  __syncthreads();
+  int kslots = Kpad / 8;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i < total8; i += (int64_t)gridDim.x * blockDim.x) {
-    int64_t npq = i / (Kpad / 8);
-    int k0 = (int)(i % (Kpad / 8)) * 8;
+    int64_t npq = i / kslots;
+    int k0 = (int)(i % kslots) * 8;
     int q = npq % Q;
     int p = (npq / Q) % P;
     int n = npq / ((int64_t)P * Q);
     int h0 = p * sh - ph, w0 = q * sw - pw;
+    const u16* xbase = x + (int64_t)n * H * W * C + c0;
     u16x8 out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      int k = k0 + j;
+      int e = lut[k0 + j];
       u16 v = 0;
-      if (k < Kcol) {
-        int rs = k / Ct, c = k % Ct;
-        int h = h0 + (rs / S) * dil;
-        int w = w0 + (rs % S) * dil;
+      if (e >= 0) {
+        int h = h0 + (e >> 20);
+        int w = w0 + ((e >> 10) & 1023);
         if (h >= 0 && h < H && w >= 0 && w < W)
-          v = x[(((int64_t)n * H + h) * W + w) * C + c0 + c];
+          v = xbase[((int64_t)h * W + w) * C + (e & 1023)];
       }
       out[j] = v;
     }
@@ -130,7 +142,7 @@ __global__ void col2im_nhwc_kernel(
 void im2col_nhwc(const void* x, void* col, int N, int H, int W, int C,
                  int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
                  int dil, int Kpad, int c0, int Ct, hipStream_t stream) {
-  if (Ct < 8) {
+  if (Ct < 8 && Kpad <= 1024) {
     int64_t total8 = (int64_t)N * P * Q * (Kpad / 8);
     int b = hmin<int64_t>(8192, (total8 + 255) / 256);
    hipLaunchKernelGGL(( im2col_smallc_kernel), dim3(b), dim3(256), 0, stream, 

@@ -108,17 +108,30 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
                     memory_format=torch.channels_last)
     NPQ = N * P * Q
     y2 = y.permute(0, 2, 3, 1).reshape(NPQ, Kout)  # NHWC flat alias (view)
-    col = torch.empty((G, NPQ, Kpad), dtype=torch.bfloat16, device=x.device)
-    for g in range(G):
-        _ext.im2col(xl, col[g], N, H, W, C, P, Q, R, S, sh, sw, ph, pw, dil,
-                    Kpad, g * Cg, Cg)
-        # C[npq, kout_g] — write into the column slice of NHWC y
-        _gemm(col[g], wr[g * Kg:(g + 1) * Kg], y2[:, g * Kg:],
-              bias_f[g * Kg:(g + 1) * Kg] if bias_f is not None else None,
-              NPQ, Kg, Kpad, Kpad, Kpad, Kout, False, False, 0, 1,
-              relu=relu, na=wrb.shape[0] - g * Kg)
+    # 1x1/stride-1 conv: im2col is the identity in NHWC — GEMM straight
+    # off the input (GoogLeNet's many 1x1 convs skip the col buffer)
+    is_1x1 = (R == S == 1 and sh == sw == 1 and ph == pw == 0 and
+              dil == 1 and G == 1 and C % 8 == 0)
+    if is_1x1:
+        x2 = xl.permute(0, 2, 3, 1).reshape(NPQ, C)
+        _gemm(x2, wr, y2, bias_f, NPQ, Kout, C, C, C, Kout,
+              False, False, 0, 1, relu=relu, na=wrb.shape[0])
+        col = None
+    else:
+        col = torch.empty((G, NPQ, Kpad), dtype=torch.bfloat16,
+                          device=x.device)
+        for g in range(G):
+            _ext.im2col(xl, col[g], N, H, W, C, P, Q, R, S, sh, sw, ph, pw,
+                        dil, Kpad, g * Cg, Cg)
+            # C[npq, kout_g] — write into the column slice of NHWC y
+            _gemm(col[g], wr[g * Kg:(g + 1) * Kg], y2[:, g * Kg:],
+                  bias_f[g * Kg:(g + 1) * Kg] if bias_f is not None else None,
+                  NPQ, Kg, Kpad, Kpad, Kpad, Kout, False, False, 0, 1,
+                  relu=relu, na=wrb.shape[0] - g * Kg)
     if ctx is not None:
         ctx["col"] = col
+        ctx["xl"] = xl
+        ctx["is_1x1"] = is_1x1
         ctx["shape"] = (N, C, H, W, P, Q, R, S, sh, sw, ph, pw, dil, G, Cg,
                         Kg, Kpad, Kcol)
         ctx["wr"] = wr
@@ -135,10 +148,13 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
     (N, C, H, W, P, Q, R, S, sh, sw, ph, pw, dil, G, Cg, Kg, Kpad,
      Kcol) = ctx["shape"]
     col, wr = ctx["col"], ctx["wr"]
+    is_1x1 = ctx.get("is_1x1", False)
     Kout = Kg * G
     NPQ = N * P * Q
     dyl = _cl(dy)
     dy2 = dyl.permute(0, 2, 3, 1).reshape(NPQ, Kout)
+    if is_1x1:
+        x2 = ctx["xl"].permute(0, 2, 3, 1).reshape(NPQ, C)
 
     dx = dw = db = None
     if need_dw:
@@ -149,7 +165,7 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         dyT = _transpose(dy2)  # [Kout][NPQ], rows over-allocated
         dwp = torch.zeros((Kout, Kpad), dtype=torch.float32, device=dy.device)
         for g in range(G):
-            colT = _transpose(col[g])
+            colT = _transpose(x2 if is_1x1 else col[g])
             mb, nb = (Kg + 127) // 128, (Kpad + 127) // 128
             sk = _splitk_for(mb, nb, NPQ)
             _gemm(dyT[g * Kg:(g + 1) * Kg], colT, dwp[g * Kg:],
@@ -161,11 +177,18 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         db = torch.zeros(Kout, dtype=torch.float32, device=dy.device)
         _ext.colsum(dy2, db, NPQ, Kout, Kout)
     if need_dx:
-        dcol = torch.empty((NPQ, Kpad), dtype=torch.bfloat16,
-                           device=dy.device)
         dx = torch.empty((N, C, H, W), dtype=torch.bfloat16,
                          device=dy.device,
-                    memory_format=torch.channels_last)
+                         memory_format=torch.channels_last)
+        if is_1x1:
+            # col2im is the identity: write dx's NHWC alias directly
+            dx2 = dx.permute(0, 2, 3, 1).reshape(NPQ, C)
+            wrT = _transpose(wr.contiguous())
+            _gemm(dy2, wrT, dx2, None, NPQ, C, Kout, Kout, Kout, C,
+                  False, False, 0, 1, na=_pad128(C))
+            return dx, dw, db
+        dcol = torch.empty((NPQ, Kpad), dtype=torch.bfloat16,
+                           device=dy.device)
         for g in range(G):
             # dcol[npq, kpad] = dy_g[npq, kg] @ w_g[kg, kpad]: transpose the
             # (small) packed weights so B is [kpad][kg] = NT direct form
@@ -196,8 +219,18 @@ def fc_forward(x, w, b, relu=False):
         na = Nout
     y = torch.empty((M, Nout), dtype=torch.bfloat16, device=x.device)
     bias_f = b.float().contiguous() if b is not None else None
-    _gemm(x, wb, y, bias_f, M, Nout, K, K, K, Nout, False, False, 0, 1,
-          relu=relu, na=na)
+    mb, nb = (M + 127) // 128, (Nout + 127) // 128
+    sk = _splitk_for(mb, nb, K)
+    if sk > 1:
+        # underfilled grid (e.g. fc6: 2x32 tiles): split K with fp32
+        # atomics, then one fused bias+ReLU+cast pass
+        wsp = torch.zeros((M, Nout), dtype=torch.float32, device=x.device)
+        _gemm(x, wb, wsp, None, M, Nout, K, K, K, Nout, False, False, 2,
+              sk, na=na)
+        _ext.bias_act_cast(wsp, bias_f, y, relu)
+    else:
+        _gemm(x, wb, y, bias_f, M, Nout, K, K, K, Nout, False, False, 0, 1,
+              relu=relu, na=na)
     return y
 
 
@@ -213,8 +246,16 @@ def fc_backward(x, w, dy, need_dx=True, bias=True):
         # dx = dy @ w: transpose w once -> NT direct/direct fast staging
         wT = _transpose(wb)
         dx = torch.empty((M, K), dtype=torch.bfloat16, device=x.device)
-        _gemm(dy, wT, dx, None, M, K, Nout, Nout, Nout, K, False, False,
-              0, 1, na=_pad128(K))
+        mb, nb = (M + 127) // 128, (K + 127) // 128
+        sk = _splitk_for(mb, nb, Nout)
+        if sk > 1:
+            wsp = torch.zeros((M, K), dtype=torch.float32, device=x.device)
+            _gemm(dy, wT, wsp, None, M, K, Nout, Nout, Nout, K, False,
+                  False, 2, sk, na=_pad128(K))
+            _ext.bias_act_cast(wsp, None, dx, False)
+        else:
+            _gemm(dy, wT, dx, None, M, K, Nout, Nout, Nout, K, False,
+                  False, 0, 1, na=_pad128(K))
     # dw = dy^T @ x: transpose both (small vs the GEMM) -> NT fast form
     dyT = _transpose(dy)
     xT = _transpose(x)
