@@ -22,12 +22,12 @@ void im2col_nhwc(const void* x, void* col, int N, int H, int W, int C,
 void col2im_nhwc(const void* dcol, void* dx, int N, int H, int W, int C,
                  int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
                  int dil, int Kpad, int c0, int Ct, hipStream_t stream);
-void maxpool_fwd(const void* x, void* y, int* idx, int N, int H, int W, int C,
-                 int P, int Q, int kh, int kw, int sh, int sw, int ph, int pw,
-                 hipStream_t stream);
-void maxpool_bwd(const void* dy, const int* idx, void* dx, int N, int H,
-                 int W, int C, int P, int Q, int kh, int kw, int sh, int sw,
-                 int ph, int pw, hipStream_t stream);
+void maxpool_fwd(const void* x, void* y, void* idx, bool idx16, int N,
+                 int H, int W, int C, int P, int Q, int kh, int kw, int sh,
+                 int sw, int ph, int pw, hipStream_t stream);
+void maxpool_bwd(const void* dy, const void* idx, bool idx16, void* dx,
+                 int N, int H, int W, int C, int P, int Q, int kh, int kw,
+                 int sh, int sw, int ph, int pw, hipStream_t stream);
 void avgpool_fwd(const void* x, void* y, int N, int H, int W, int C,
                  int P, int Q, int kh, int kw, int sh, int sw, int ph, int pw,
                  hipStream_t stream);
@@ -130,7 +130,8 @@ void py_maxpool_fwd(Tensor x, Tensor y, Tensor idx, int64_t N, int64_t H,
                     int64_t kw, int64_t sh, int64_t sw, int64_t ph,
                     int64_t pw) {
   CHECK_CUDA(x); CHECK_BF16(x);
-  cosamd::maxpool_fwd(x.data_ptr(), y.data_ptr(), idx.data_ptr<int>(), N, H,
+  cosamd::maxpool_fwd(x.data_ptr(), y.data_ptr(), idx.data_ptr(),
+                      idx.scalar_type() == at::kShort, N, H,
                       W, C, P, Q, kh, kw, sh, sw, ph, pw, cur_stream());
 }
 
@@ -138,7 +139,8 @@ void py_maxpool_bwd(Tensor dy, Tensor idx, Tensor dx, int64_t N, int64_t H,
                     int64_t W, int64_t C, int64_t P, int64_t Q, int64_t kh,
                     int64_t kw, int64_t sh, int64_t sw, int64_t ph,
                     int64_t pw) {
-  cosamd::maxpool_bwd(dy.data_ptr(), idx.data_ptr<int>(), dx.data_ptr(), N, H,
+  cosamd::maxpool_bwd(dy.data_ptr(), idx.data_ptr(),
+                      idx.scalar_type() == at::kShort, dx.data_ptr(), N, H,
                       W, C, P, Q, kh, kw, sh, sw, ph, pw, cur_stream());
 }
 

@@ -64,16 +64,17 @@ __global__ void im2col_nhwc_kernel(
     const u16* __restrict__ x, u16* __restrict__ col,
     int N, int H, int W, int C, int P, int Q,
     int R, int S, int sh, int sw, int ph, int pw,
-    int dil, int Kpad, int c0, int Ct, int64_t total_rs) {
-  // one thread per (npq, r, s) copying C elements
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total_rs; i += (int64_t)gridDim.x * blockDim.x) {
-    int rs = i % (R * S);
-    int64_t npq = i / (R * S);
+    int dil, int Kpad, int c0, int Ct, int total_npq) {
+  // one thread per npq, one grid.y slice per (r,s): all index math is
+  // 32-bit and the expensive div/mod runs once per thread
+  int rs = blockIdx.y;
+  int r = rs / S, s = rs % S;
+  for (int npq = blockIdx.x * blockDim.x + threadIdx.x;
+       npq < total_npq; npq += gridDim.x * blockDim.x) {
     int q = npq % Q;
-    int p = (npq / Q) % P;
-    int n = npq / ((int64_t)P * Q);
-    int r = rs / S, s = rs % S;
+    int pq = npq / Q;
+    int p = pq % P;
+    int n = pq / P;
     int h = p * sh - ph + r * dil;
     int w = q * sw - pw + s * dil;
     u16* dst = col + npq * Kpad + rs * Ct;
@@ -110,12 +111,15 @@ __global__ void col2im_nhwc_kernel(
     int N, int H, int W, int C, int P, int Q,
     int R, int S, int sh, int sw, int ph, int pw,
     int dil, int Kpad, int c0, int Ct, int64_t total) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total; i8 += (int64_t)gridDim.x * blockDim.x) {
+    unsigned i = (unsigned)i8;
     int c = i % Ct;
-    int w = (i / Ct) % W;
-    int h = (i / ((int64_t)Ct * W)) % H;
-    int n = i / ((int64_t)Ct * W * H);
+    unsigned iw = i / Ct;
+    int w = iw % W;
+    unsigned ih = iw / W;
+    int h = ih % H;
+    int n = ih / H;
     float acc = 0.f;
     for (int r = 0; r < R; ++r) {
       int hp = h + ph - r * dil;
@@ -149,11 +153,11 @@ void im2col_nhwc(const void* x, void* col, int N, int H, int W, int C,
         dil, Kpad, c0, Ct, total8);
     return;  // pad columns are zero-filled inline
   }
-  int64_t total = (int64_t)N * P * Q * R * S;
-  int blocks = hmin<int64_t>(4096, (total + 255) / 256);
-  im2col_nhwc_kernel<<<blocks, 256, 0, stream>>>(
+  int total_npq = N * P * Q;
+  dim3 grid(hmin<int>(4096, (total_npq + 255) / 256), R * S);
+  im2col_nhwc_kernel<<<grid, 256, 0, stream>>>(
       (const u16*)x, (u16*)col, N, H, W, C, P, Q, R, S, sh, sw, ph, pw,
-      dil, Kpad, c0, Ct, total);
+      dil, Kpad, c0, Ct, total_npq);
   int Kcol = R * S * Ct;
   if (Kpad > Kcol) {
     int64_t rows = (int64_t)N * P * Q;

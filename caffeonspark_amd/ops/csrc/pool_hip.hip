@@ -10,8 +10,9 @@ namespace cosamd {
 
 typedef unsigned short u16;
 
+template <typename IDX>
 __global__ void maxpool_fwd_kernel(
-    const u16* __restrict__ x, u16* __restrict__ y, int* __restrict__ idx,
+    const u16* __restrict__ x, u16* __restrict__ y, IDX* __restrict__ idx,
     int N, int H, int W, int C, int P, int Q,
     int kh, int kw, int sh, int sw, int ph, int pw, int64_t total) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -36,12 +37,13 @@ __global__ void maxpool_fwd_kernel(
     }
     bf16 out = f2bf(best);
     y[i] = *reinterpret_cast<u16*>(&out);
-    idx[i] = best_hw;
+    idx[i] = (IDX)best_hw;
   }
 }
 
+template <typename IDX>
 __global__ void maxpool_bwd_kernel(
-    const u16* __restrict__ dy, const int* __restrict__ idx,
+    const u16* __restrict__ dy, const IDX* __restrict__ idx,
     u16* __restrict__ dx,
     int N, int H, int W, int C, int P, int Q,
     int kh, int kw, int sh, int sw, int ph, int pw, int64_t total) {
@@ -59,7 +61,7 @@ __global__ void maxpool_bwd_kernel(
     for (int p = plo; p <= phi; ++p)
       for (int q = qlo; q <= qhi; ++q) {
         int64_t o = (((int64_t)n * P + p) * Q + q) * C + c;
-        if (idx[o] == hw) {
+        if ((int)idx[o] == hw) {
           u16 v = dy[o];
           acc += bf2f(*reinterpret_cast<const bf16*>(&v));
         }
@@ -128,22 +130,32 @@ static int nblocks_for(int64_t total) {
   return (int)hmin<int64_t>(8192, (total + 255) / 256);
 }
 
-void maxpool_fwd(const void* x, void* y, int* idx, int N, int H, int W, int C,
-                 int P, int Q, int kh, int kw, int sh, int sw, int ph, int pw,
-                 hipStream_t stream) {
+void maxpool_fwd(const void* x, void* y, void* idx, bool idx16, int N,
+                 int H, int W, int C, int P, int Q, int kh, int kw, int sh,
+                 int sw, int ph, int pw, hipStream_t stream) {
   int64_t total = (int64_t)N * P * Q * C;
- hipLaunchKernelGGL(( maxpool_fwd_kernel), dim3(nblocks_for(total)), dim3(256), 0, stream, 
-      (const u16*)x, (u16*)y, idx, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw,
-      total);
+  if (idx16)
+   hipLaunchKernelGGL(( maxpool_fwd_kernel<short>), dim3(nblocks_for(total)), dim3(256), 0, stream, 
+        (const u16*)x, (u16*)y, (short*)idx, N, H, W, C, P, Q, kh, kw, sh,
+        sw, ph, pw, total);
+  else
+   hipLaunchKernelGGL(( maxpool_fwd_kernel<int>), dim3(nblocks_for(total)), dim3(256), 0, stream, 
+        (const u16*)x, (u16*)y, (int*)idx, N, H, W, C, P, Q, kh, kw, sh,
+        sw, ph, pw, total);
 }
 
-void maxpool_bwd(const void* dy, const int* idx, void* dx, int N, int H,
-                 int W, int C, int P, int Q, int kh, int kw, int sh, int sw,
-                 int ph, int pw, hipStream_t stream) {
+void maxpool_bwd(const void* dy, const void* idx, bool idx16, void* dx,
+                 int N, int H, int W, int C, int P, int Q, int kh, int kw,
+                 int sh, int sw, int ph, int pw, hipStream_t stream) {
   int64_t total = (int64_t)N * H * W * C;
- hipLaunchKernelGGL(( maxpool_bwd_kernel), dim3(nblocks_for(total)), dim3(256), 0, stream, 
-      (const u16*)dy, idx, (u16*)dx, N, H, W, C, P, Q, kh, kw, sh, sw,
-      ph, pw, total);
+  if (idx16)
+   hipLaunchKernelGGL(( maxpool_bwd_kernel<short>), dim3(nblocks_for(total)), dim3(256), 0, stream, 
+        (const u16*)dy, (const short*)idx, (u16*)dx, N, H, W, C, P, Q, kh,
+        kw, sh, sw, ph, pw, total);
+  else
+   hipLaunchKernelGGL(( maxpool_bwd_kernel<int>), dim3(nblocks_for(total)), dim3(256), 0, stream, 
+        (const u16*)dy, (const int*)idx, (u16*)dx, N, H, W, C, P, Q, kh,
+        kw, sh, sw, ph, pw, total);
 }
 
 void avgpool_fwd(const void* x, void* y, int N, int H, int W, int C,
