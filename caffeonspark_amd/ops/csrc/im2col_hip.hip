@@ -65,17 +65,19 @@ __global__ void im2col_nhwc_kernel(
     const u16* __restrict__ x, u16* __restrict__ col,
     int N, int H, int W, int C, int P, int Q,
     int R, int S, int sh, int sw, int ph, int pw,
-    int dil, int Kpad, int c0, int Ct, int total_npq) {
-  // one thread per npq, one grid.y slice per (r,s): all index math is
-  // 32-bit and the expensive div/mod runs once per thread
-  int rs = blockIdx.y;
-  int r = rs / S, s = rs % S;
-  for (int npq = blockIdx.x * blockDim.x + threadIdx.x;
-       npq < total_npq; npq += gridDim.x * blockDim.x) {
+    int dil, int Kpad, int c0, int Ct, unsigned total_rs) {
+  // one thread per (npq, r, s): adjacent threads share npq and walk rs,
+  // so writes are contiguous along the col row; 32-bit index math
+  unsigned RS = R * S;
+  for (unsigned i = blockIdx.x * blockDim.x + threadIdx.x;
+       i < total_rs; i += gridDim.x * blockDim.x) {
+    int rs = i % RS;
+    unsigned npq = i / RS;
     int q = npq % Q;
-    int pq = npq / Q;
+    unsigned pq = npq / Q;
     int p = pq % P;
     int n = pq / P;
+    int r = rs / S, s = rs % S;
     int h = p * sh - ph + r * dil;
     int w = q * sw - pw + s * dil;
     u16* dst = col + npq * Kpad + rs * Ct;
@@ -154,11 +156,11 @@ void im2col_nhwc(const void* x, void* col, int N, int H, int W, int C,
         dil, Kpad, c0, Ct, total8);
     return;  // pad columns are zero-filled inline
   }
-  int total_npq = N * P * Q;
-  dim3 grid(hmin<int>(4096, (total_npq + 255) / 256), R * S);
- hipLaunchKernelGGL(( im2col_nhwc_kernel), dim3(grid), dim3(256), 0, stream, 
+  int64_t total = (int64_t)N * P * Q * R * S;
+  int blocks = hmin<int64_t>(4096, (total + 255) / 256);
+ hipLaunchKernelGGL(( im2col_nhwc_kernel), dim3(blocks), dim3(256), 0, stream, 
       (const u16*)x, (u16*)col, N, H, W, C, P, Q, R, S, sh, sw, ph, pw,
-      dil, Kpad, c0, Ct, total_npq);
+      dil, Kpad, c0, Ct, (unsigned)total);
   int Kcol = R * S * Ct;
   if (Kpad > Kcol) {
     int64_t rows = (int64_t)N * P * Q;
