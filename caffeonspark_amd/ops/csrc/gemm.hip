@@ -30,6 +30,18 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 constexpr int BM = 128, BN = 128, BK = 32;
 constexpr int NTHREADS = 256;
 
+// LDS chunk swizzle: a row's four 16-byte k-chunks are stored rotated by
+// XOR with ((row>>1)&3).  Without it, a fragment ds_read_b128's lane
+// groups (16 lanes = 16 rows, same k-chunk) land on only 2 of the 8
+// bank-quads (addr = row*64B + chunk*16B, 16*row mod 32 banks ∈ {0,16}),
+// an 8-way conflict; the swizzle spreads them across all 8 quads (2-way,
+// which CDNA4 serves conflict-free).  LDS stays linear so the
+// global_load_lds fast path just pre-swizzles the per-lane SOURCE address
+// (cdna_hip_programming.md §5: swizzle must be both-sides-or-neither).
+__device__ __forceinline__ int swz_chunk(int row, int chunk) {
+  return chunk ^ ((row >> 1) & 3);
+}
+
 // ---------------------------------------------------------------- staging
 
 // direct ([rows][K]) guarded staging: vectorized 8-wide along K when the
@@ -47,7 +59,7 @@ __device__ __forceinline__ void stage_direct_guarded(
     int kk = (slot & 3) * 8;
     int grow = row0 + row;
     int gk = k0 + kk;
-    unsigned short* dst = lds + row * BK + kk;
+    unsigned short* dst = lds + row * BK + (swz_chunk(row, slot & 3) << 3);
     if (grow < rows && gk + 8 <= kend && ((ld | gk) % 8 == 0)) {
       *reinterpret_cast<short8v*>(dst) =
           *reinterpret_cast<const short8v*>(g + (int64_t)grow * ld + gk);
@@ -93,7 +105,11 @@ __device__ __forceinline__ void stage_trans_guarded(
       for (int j = 0; j < 8; ++j) vals[j] = 0;
     }
 #pragma unroll
-    for (int j = 0; j < 8; ++j) lds[(r0 + j) * BK + kk] = vals[j];
+    for (int j = 0; j < 8; ++j) {
+      int row = r0 + j;
+      int dst_k = (swz_chunk(row, kk >> 3) << 3) | (kk & 7);
+      lds[row * BK + dst_k] = vals[j];
+    }
   }
 }
 
@@ -134,10 +150,12 @@ __global__ __launch_bounds__(WNT, 1) void gemm_tt_wide_kernel(
     bf16x8 afrag[4], bfrag[4];
 #pragma unroll
     for (int f = 0; f < 4; ++f) {
+      int ra = wm * 64 + f * 16 + lrow;
+      int rb = wn * 64 + f * 16 + lrow;
       afrag[f] = *reinterpret_cast<const bf16x8*>(
-          As + (wm * 64 + f * 16 + lrow) * BK + lk8);
+          As + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
       bfrag[f] = *reinterpret_cast<const bf16x8*>(
-          Bs + (wn * 64 + f * 16 + lrow) * BK + lk8);
+          Bs + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
     }
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm)
@@ -174,7 +192,7 @@ __device__ __forceinline__ void stage_direct_fast(
     int chunk = wave * 2 + it;           // 8 chunks of 512 elements
     int idx = chunk * 512 + lane * 8;    // linear element index in tile
     int row = idx >> 5;                  // /32
-    int kk = idx & 31;
+    int kk = swz_chunk(row, (idx & 31) >> 3) << 3;  // pre-swizzled source
     auto* gp = (const __attribute__((address_space(1))) unsigned int*)(
         g + (int64_t)(row0 + row) * ld + k0 + kk);
     auto* lp = (__attribute__((address_space(3))) unsigned int*)(
@@ -238,10 +256,12 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
     bf16x8 afrag[4], bfrag[4];
 #pragma unroll
     for (int f = 0; f < 4; ++f) {
+      int ra = wm * 64 + f * 16 + lrow;
+      int rb = wn * 64 + f * 16 + lrow;
       afrag[f] = *reinterpret_cast<const bf16x8*>(
-          As + (wm * 64 + f * 16 + lrow) * BK + lk8);
+          As + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
       bfrag[f] = *reinterpret_cast<const bf16x8*>(
-          Bs + (wn * 64 + f * 16 + lrow) * BK + lk8);
+          Bs + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
     }
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm)
