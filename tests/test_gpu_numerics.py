@@ -330,3 +330,126 @@ def test_lenet_step_parity():
             # relative L2 error: robust to bf16 noise on single elements
             rel = (cg - gg).norm() / cg.norm().clamp_min(1e-4)
             assert rel < 0.08, f"{cl.name} grad mismatch relL2={rel:.3f}"
+
+
+def test_inception_block_parity():
+    """GoogLeNet inception wiring (1x1 bypass + concat) GPU vs CPU."""
+    import os
+
+    from caffeonspark_amd.core import Net
+    from caffeonspark_amd.proto import caffe_pb, text_format
+
+    text = """
+    layer { name: "data" type: "DummyData" top: "data"
+            dummy_data_param { shape { dim: 2 dim: 32 dim: 9 dim: 9 }
+                               data_filler { type: "gaussian" std: 1.0 } } }
+    layer { name: "b1" type: "Convolution" bottom: "data" top: "b1"
+            convolution_param { num_output: 16 kernel_size: 1
+              weight_filler { type: "xavier" } } }
+    layer { name: "relu_b1" type: "ReLU" bottom: "b1" top: "b1" }
+    layer { name: "b3r" type: "Convolution" bottom: "data" top: "b3r"
+            convolution_param { num_output: 8 kernel_size: 1
+              weight_filler { type: "xavier" } } }
+    layer { name: "b3" type: "Convolution" bottom: "b3r" top: "b3"
+            convolution_param { num_output: 16 kernel_size: 3 pad: 1
+              weight_filler { type: "xavier" } } }
+    layer { name: "pool" type: "Pooling" bottom: "data" top: "pool"
+            pooling_param { pool: MAX kernel_size: 3 stride: 1 pad: 1 } }
+    layer { name: "bp" type: "Convolution" bottom: "pool" top: "bp"
+            convolution_param { num_output: 8 kernel_size: 1
+              weight_filler { type: "xavier" } } }
+    layer { name: "cat" type: "Concat" bottom: "b1" bottom: "b3"
+            bottom: "bp" top: "cat" concat_param { axis: 1 } }
+    layer { name: "gap" type: "Pooling" bottom: "cat" top: "gap"
+            pooling_param { pool: AVE global_pooling: true } }
+    layer { name: "fc" type: "InnerProduct" bottom: "gap" top: "fc"
+            inner_product_param { num_output: 5
+              weight_filler { type: "xavier" } } }
+    layer { name: "loss" type: "SoftmaxWithLoss" bottom: "fc" bottom: "lab"
+            top: "loss" }
+    layer { name: "lab" type: "DummyData" top: "lab"
+            dummy_data_param { shape { dim: 2 } } }
+    """
+    # reorder: lab before loss
+    param = text_format.parse(text, caffe_pb.NetParameter)
+    layers = list(param.layer)
+    lab = [l for l in layers if l.name == "lab"][0]
+    layers.remove(lab)
+    layers.insert(0, lab)
+    param.layer = layers
+    state = caffe_pb.NetState(phase=caffe_pb.Phase.TRAIN)
+    cpu = Net(param, state, seed=9, dtype=torch.bfloat16)
+    gpu = Net(param, state, seed=9, device=dev(), dtype=torch.bfloat16)
+    for cl, gl in zip(cpu.layers, gpu.layers):
+        for cb, gb in zip(cl.blobs, gl.blobs):
+            gb.data.copy_(cb.data.to(gb.data.device))
+    closs = cpu.forward()
+    gloss = gpu.forward()
+    assert abs(closs - gloss) / max(abs(closs), 1e-5) < 0.05
+    cpu.backward()
+    gpu.backward()
+    for cl, gl in zip(cpu.layers, gpu.layers):
+        for cb, gb in zip(cl.blobs, gl.blobs):
+            if cb.diff is None:
+                continue
+            rel = (cb.diff.float() - gb.diff.float().cpu()).norm() / \
+                cb.diff.float().norm().clamp_min(1e-4)
+            assert rel < 0.1, f"{cl.name} relL2={rel:.3f}"
+
+
+def test_lstm_layer_gpu_parity():
+    """Full LSTM layer (time-major, cont gating, static input) GPU vs CPU."""
+    from caffeonspark_amd.core.layers.base import create_layer
+    from caffeonspark_amd.core.blob import Blob
+    from caffeonspark_amd.proto import caffe_pb, text_format
+
+    text = '''name: "l" type: "LSTM" bottom: "x" bottom: "cont"
+        top: "h" recurrent_param { num_output: 16
+          weight_filler { type: "uniform" min: -0.1 max: 0.1 } }'''
+    T, N, D = 5, 4, 12
+
+    class FN:
+        phase = caffe_pb.Phase.TRAIN
+        dtype = torch.float32
+        device = torch.device("cpu")
+        generator = torch.Generator().manual_seed(2)
+
+    class FNG:
+        phase = caffe_pb.Phase.TRAIN
+        dtype = torch.bfloat16
+        device = dev()
+        generator = torch.Generator().manual_seed(2)
+
+    lp = text_format.parse(text, caffe_pb.LayerParameter)
+    cl = create_layer(lp, FN())
+    gl = create_layer(lp, FNG())
+    x = torch.randn(T, N, D)
+    cont = torch.ones(T, N)
+    cont[0] = 0
+    cont[3, 2] = 0
+
+    def run(layer, xx, cc, dtype, device):
+        bx, bc = Blob(xx.shape), Blob(cc.shape)
+        bx.data = xx.to(device, dtype)
+        bc.data = cc.to(device, dtype)
+        top = [Blob([0])]
+        layer.setup([bx, bc], top)
+        return layer, [bx, bc], top
+
+    cl, cbot, ctop = run(cl, x, cont, torch.float32, torch.device("cpu"))
+    gl, gbot, gtop = run(gl, x, cont, torch.bfloat16, dev())
+    for cb, gb in zip(cl.blobs, gl.blobs):
+        gb.data.copy_(cb.data.to(gb.data.device))
+    cl.forward(cbot, ctop)
+    gl.forward(gbot, gtop)
+    agree(gtop[0].data, ctop[0].data, rtol=0.05, atol=0.03)
+    dy = torch.randn(T, N, 16)
+    ctop[0].diff = dy
+    gtop[0].diff = dy.to(dev(), torch.bfloat16)
+    cl.backward(ctop, [True, False], cbot)
+    gl.backward(gtop, [True, False], gbot)
+    agree(gbot[0].diff, cbot[0].diff, rtol=0.08, atol=0.05)
+    for i, (cb, gb) in enumerate(zip(cl.blobs, gl.blobs)):
+        rel = (cb.diff.float() - gb.diff.float().cpu()).norm() / \
+            cb.diff.float().norm().clamp_min(1e-4)
+        assert rel < 0.1, f"param {i} relL2={rel:.3f}"
