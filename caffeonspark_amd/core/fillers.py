@@ -15,8 +15,10 @@ from ..proto import caffe_pb
 
 def fill(tensor: torch.Tensor, param: caffe_pb.FillerParameter,
          generator: torch.Generator = None) -> None:
-    if tensor.device.type != "cpu":
-        # fill on CPU (deterministic across devices), then copy
+    if tensor.device.type != "cpu" or tensor.dtype != torch.float32:
+        # always draw in fp32 on CPU (identical streams across devices AND
+        # dtypes — normal_ on a bf16 tensor draws a different sequence),
+        # then cast
         tmp = torch.empty(tensor.shape, dtype=torch.float32)
         fill(tmp, param, generator)
         tensor.copy_(tmp.to(tensor.dtype))

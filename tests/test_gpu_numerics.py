@@ -92,6 +92,8 @@ def test_gemm_splitk_atomic():
     dict(n=4, c=3, h=33, w=33, k=32, r=11, stride=4, pad=0, groups=1),
     dict(n=2, c=16, h=15, w=15, k=32, r=5, stride=1, pad=2, groups=2),
     dict(n=2, c=24, h=13, w=13, k=48, r=3, stride=1, pad=1, groups=1),
+    dict(n=2, c=32, h=9, w=9, k=16, r=1, stride=1, pad=0, groups=1),
+    dict(n=2, c=32, h=8, w=8, k=16, r=1, stride=1, pad=0, groups=1),
 ])
 def test_conv_forward_backward(case):
     n, c, h, w = case["n"], case["c"], case["h"], case["w"]
