@@ -45,7 +45,7 @@ def main():
 
     if args.batch == 0:
         args.batch = {"alexnet": 256, "googlenet": 128, "cifar10_quick": 100,
-                      "lrcn": 32}.get(args.model, 256)
+                      "lrcn": 64}.get(args.model, 256)
     root = os.path.dirname(os.path.abspath(__file__))
     solver = solver_from_prototxt(
         os.path.join(root, "caffeonspark_amd", "models",
