@@ -210,11 +210,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
     void* __restrict__ C, const float* __restrict__ bias,
     int M, int N, int K, int lda, int ldb, int ldc,
     int ksplit, int relu, float alpha, int m_alloc, int n_alloc) {
-  // double-buffered LDS: stage tile t+1 while the MFMAs consume tile t —
-  // one barrier per K-tile and the staging latency hides under compute
-  // (cdna_hip_programming.md §5.5 T3 "minimum 2-phase" recipe)
-  __shared__ bf16 As[2][BM * BK];
-  __shared__ bf16 Bs[2][BN * BK];
+  __shared__ bf16 As[BM * BK];
+  __shared__ bf16 Bs[BN * BK];
 
   int mblocks = (M + BM - 1) / BM;
   int nblocks = (N + BN - 1) / BN;
@@ -240,30 +237,22 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
   bool b_fast = !TRANS_B && (tile_n + BN <= n_alloc) && (ldb % 8 == 0) &&
                 ((k_end - k_begin) % BK == 0) && (k_begin % 8 == 0);
 
-#define COS_STAGE(buf, k0)                                                  \
-  do {                                                                      \
-    if (a_fast) {                                                           \
-      stage_direct_fast(As[buf], A, tile_m, lda, (k0), tid);                \
-    } else if (TRANS_A) {                                                   \
-      stage_trans_guarded(As[buf], A, tile_m, M, lda, (k0), k_end, tid);    \
-    } else {                                                                \
-      stage_direct_guarded(As[buf], A, tile_m, M, lda, (k0), k_end, tid);   \
-    }                                                                       \
-    if (b_fast) {                                                           \
-      stage_direct_fast(Bs[buf], B, tile_n, ldb, (k0), tid);                \
-    } else if (TRANS_B) {                                                   \
-      stage_trans_guarded(Bs[buf], B, tile_n, N, ldb, (k0), k_end, tid);    \
-    } else {                                                                \
-      stage_direct_guarded(Bs[buf], B, tile_n, N, ldb, (k0), k_end, tid);   \
-    }                                                                       \
-  } while (0)
-
-  COS_STAGE(0, k_begin);
-  __syncthreads();
-  int cur = 0;
   for (int k0 = k_begin; k0 < k_end; k0 += BK) {
-    if (k0 + BK < k_end)
-      COS_STAGE(cur ^ 1, k0 + BK);  // prefetch next tile into the idle buf
+    if (a_fast) {
+      stage_direct_fast(As, A, tile_m, lda, k0, tid);
+    } else if (TRANS_A) {
+      stage_trans_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
+    } else {
+      stage_direct_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
+    }
+    if (b_fast) {
+      stage_direct_fast(Bs, B, tile_n, ldb, k0, tid);
+    } else if (TRANS_B) {
+      stage_trans_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
+    } else {
+      stage_direct_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
+    }
+    __syncthreads();
 
     bf16x8 afrag[4], bfrag[4];
 #pragma unroll
@@ -271,23 +260,20 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
       int ra = wm * 64 + f * 16 + lrow;
       int rb = wn * 64 + f * 16 + lrow;
       afrag[f] = *reinterpret_cast<const bf16x8*>(
-          As[cur] + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
+          As + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
       bfrag[f] = *reinterpret_cast<const bf16x8*>(
-          Bs[cur] + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
+          Bs + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
     }
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
       for (int fn = 0; fn < 4; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
-    // barrier implies vmcnt(0): next buffer's loads have landed
+
+    // second half of the K-tile? no: 16x16x32 consumes all BK=32 at once
     __syncthreads();
-    cur ^= 1;
   }
-#undef COS_STAGE
 
   // ------------------------------------------------------------- epilogue
   int crow0 = tile_m + wm * 64 + ((lane >> 4) << 2);
