@@ -52,9 +52,11 @@ class ImageDataSource(DataSource):
         return self._batch
 
     def _decode(self, s: ImageSample) -> np.ndarray:
+        want_resize = getattr(self.conf, "resize", False)
         if s.encoded:
             resize = None
-            if self.height and self.width:
+            if self.height and self.width and (
+                    want_resize or not self.transformer.param.crop_size):
                 resize = (self.height, self.width)
             return decode_image(s.data, color=(self.channels == 3),
                                 resize_hw=resize)

@@ -49,8 +49,8 @@ def test_ddp_two_ranks_identical_params():
     """After synchronized steps on different data, all ranks must hold
     bit-identical parameters."""
     ws = 2
-    import tempfile
-    store = tempfile.mktemp(prefix="cosamd_ddp_")
+    import tempfile, uuid
+    store = tempfile.mktemp(prefix=f"cosamd_ddp_{uuid.uuid4().hex}_")
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_worker, args=(r, ws, store, q))
@@ -114,8 +114,8 @@ def test_ddp_matches_single_process():
     y = torch.randint(0, 10, (128,)).float()
 
     ws = 2
-    import tempfile
-    store = tempfile.mktemp(prefix="cosamd_ddp_")
+    import tempfile, uuid
+    store = tempfile.mktemp(prefix=f"cosamd_ddp_{uuid.uuid4().hex}_")
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_worker_equiv,

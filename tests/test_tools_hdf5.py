@@ -129,3 +129,25 @@ def test_mini_cluster_single(tmp_path):
     snaps = [f for f in os.listdir(tmp_path) if f.endswith(".caffemodel")]
     assert snaps
     CaffeProcessor.reset_instance()
+
+
+def test_pycaffe_shim(tmp_path):
+    """pycaffe-style inference API over our Net."""
+    import caffeonspark_amd.pycaffe as caffe
+
+    proto = tmp_path / "deploy.prototxt"
+    proto.write_text("""
+    layer { name: "data" type: "Input" top: "data"
+            input_param { shape { dim: 2 dim: 4 } } }
+    layer { name: "ip" type: "InnerProduct" bottom: "data" top: "ip"
+            inner_product_param { num_output: 3
+              weight_filler { type: "xavier" } } }
+    layer { name: "prob" type: "Softmax" bottom: "ip" top: "prob" }
+    """)
+    net = caffe.Net(str(proto), None, caffe.TEST)
+    x = np.random.randn(2, 4).astype(np.float32)
+    out = net.forward(data=x)
+    assert "prob" in out
+    np.testing.assert_allclose(out["prob"].sum(axis=1), [1.0, 1.0],
+                               rtol=1e-4)
+    assert net.params["ip"][0].shape == (3, 4)
