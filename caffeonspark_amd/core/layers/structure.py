@@ -27,7 +27,12 @@ class ConcatLayer(Layer):
         pieces = torch.split(top[0].diff, sizes, dim=self.axis)
         for b, pd, piece in zip(bottom, propagate_down, pieces):
             if pd:
-                self.acc_blob_diff(b, piece.contiguous(), False)
+                if piece.dim() == 4:
+                    piece = piece.contiguous(
+                        memory_format=torch.channels_last)
+                else:
+                    piece = piece.contiguous()
+                self.acc_blob_diff(b, piece, False)
 
 
 @register_layer("Slice")
@@ -50,7 +55,10 @@ class SliceLayer(Layer):
         pieces = torch.split(bottom[0].data, self._sizes(bottom[0].data),
                              dim=self.axis)
         for t, piece in zip(top, pieces):
-            t.data = piece.contiguous()
+            if piece.dim() == 4:
+                t.data = piece.contiguous(memory_format=torch.channels_last)
+            else:
+                t.data = piece.contiguous()
         return 0.0
 
     def backward(self, top, propagate_down, bottom):

@@ -437,7 +437,11 @@ void sgd_update(float* p, const float* g, float* v, float lr, float mu,
 
 void colsum(const void* in, float* out, int64_t rows, int cols, int ld,
             hipStream_t stream) {
-  int blocks = (int)hmin<int64_t>((rows + 3) / 4, 2048);
+  // scale the grid to the actual element count so small reductions don't
+  // dispatch thousands of idle blocks (then atomic-sum their zeros)
+  int64_t work_blocks = (rows * hmax<int64_t>(cols, 1)) / (256 * 8) + 1;
+  int blocks = (int)hmin<int64_t>(hmin<int64_t>((rows + 3) / 4, 2048),
+                                  work_blocks);
   colsum_kernel<<<blocks, 256, 0, stream>>>((const u16*)in, out, rows, cols,
                                             ld);
 }

@@ -131,7 +131,7 @@ void py_maxpool_fwd(Tensor x, Tensor y, Tensor idx, int64_t N, int64_t H,
                     int64_t pw) {
   CHECK_CUDA(x); CHECK_BF16(x);
   cosamd::maxpool_fwd(x.data_ptr(), y.data_ptr(), idx.data_ptr(),
-                      idx.scalar_type() == at::kShort, N, H,
+                      idx.scalar_type() == at::kChar, N, H,
                       W, C, P, Q, kh, kw, sh, sw, ph, pw, cur_stream());
 }
 
@@ -140,7 +140,7 @@ void py_maxpool_bwd(Tensor dy, Tensor idx, Tensor dx, int64_t N, int64_t H,
                     int64_t kw, int64_t sh, int64_t sw, int64_t ph,
                     int64_t pw) {
   cosamd::maxpool_bwd(dy.data_ptr(), idx.data_ptr(),
-                      idx.scalar_type() == at::kShort, dx.data_ptr(), N, H,
+                      idx.scalar_type() == at::kChar, dx.data_ptr(), N, H,
                       W, C, P, Q, kh, kw, sh, sw, ph, pw, cur_stream());
 }
 

@@ -23,7 +23,7 @@ __global__ void maxpool_fwd_kernel(
     int n = i / ((int64_t)C * Q * P);
     int h0 = p * sh - ph, w0 = q * sw - pw;
     float best = -3.4e38f;
-    int best_hw = 0;
+    int best_rs = 0;  // window-relative argmax (fits int8 for k<=11)
     for (int r = 0; r < kh; ++r) {
       int h = h0 + r;
       if (h < 0 || h >= H) continue;
@@ -32,12 +32,12 @@ __global__ void maxpool_fwd_kernel(
         if (w < 0 || w >= W) continue;
         u16 v = x[(((int64_t)n * H + h) * W + w) * C + c];
         float f = bf2f(*reinterpret_cast<const bf16*>(&v));
-        if (f > best) { best = f; best_hw = h * W + w; }
+        if (f > best) { best = f; best_rs = r * kw + s; }
       }
     }
     bf16 out = f2bf(best);
     y[i] = *reinterpret_cast<u16*>(&out);
-    idx[i] = (IDX)best_hw;
+    idx[i] = (IDX)best_rs;
   }
 }
 
@@ -53,19 +53,20 @@ __global__ void maxpool_bwd_kernel(
     int w = (i / C) % W;
     int h = (i / ((int64_t)C * W)) % H;
     int n = i / ((int64_t)C * W * H);
-    int hw = h * W + w;
     float acc = 0.f;
-    // windows (p,q) that can contain (h,w)
+    // windows (p,q) that can contain (h,w); match window-relative argmax
     int plo = max(0, (h + ph - kh + sh) / sh), phi = min(P - 1, (h + ph) / sh);
     int qlo = max(0, (w + pw - kw + sw) / sw), qhi = min(Q - 1, (w + pw) / sw);
-    for (int p = plo; p <= phi; ++p)
+    for (int p = plo; p <= phi; ++p) {
+      int rr = h - (p * sh - ph);
       for (int q = qlo; q <= qhi; ++q) {
         int64_t o = (((int64_t)n * P + p) * Q + q) * C + c;
-        if ((int)idx[o] == hw) {
+        if ((int)idx[o] == rr * kw + (w - (q * sw - pw))) {
           u16 v = dy[o];
           acc += bf2f(*reinterpret_cast<const bf16*>(&v));
         }
       }
+    }
     bf16 out = f2bf(acc);
     dx[i] = *reinterpret_cast<u16*>(&out);
   }
@@ -135,8 +136,8 @@ void maxpool_fwd(const void* x, void* y, void* idx, bool idx16, int N,
                  int sw, int ph, int pw, hipStream_t stream) {
   int64_t total = (int64_t)N * P * Q * C;
   if (idx16)
-   hipLaunchKernelGGL(( maxpool_fwd_kernel<short>), dim3(nblocks_for(total)), dim3(256), 0, stream, 
-        (const u16*)x, (u16*)y, (short*)idx, N, H, W, C, P, Q, kh, kw, sh,
+   hipLaunchKernelGGL(( maxpool_fwd_kernel<signed char>), dim3(nblocks_for(total)), dim3(256), 0, stream, 
+        (const u16*)x, (u16*)y, (signed char*)idx, N, H, W, C, P, Q, kh, kw, sh,
         sw, ph, pw, total);
   else
    hipLaunchKernelGGL(( maxpool_fwd_kernel<int>), dim3(nblocks_for(total)), dim3(256), 0, stream, 
@@ -149,8 +150,8 @@ void maxpool_bwd(const void* dy, const void* idx, bool idx16, void* dx,
                  int sh, int sw, int ph, int pw, hipStream_t stream) {
   int64_t total = (int64_t)N * H * W * C;
   if (idx16)
-   hipLaunchKernelGGL(( maxpool_bwd_kernel<short>), dim3(nblocks_for(total)), dim3(256), 0, stream, 
-        (const u16*)dy, (const short*)idx, (u16*)dx, N, H, W, C, P, Q, kh,
+   hipLaunchKernelGGL(( maxpool_bwd_kernel<signed char>), dim3(nblocks_for(total)), dim3(256), 0, stream, 
+        (const u16*)dy, (const signed char*)idx, (u16*)dx, N, H, W, C, P, Q, kh,
         kw, sh, sw, ph, pw, total);
   else
    hipLaunchKernelGGL(( maxpool_bwd_kernel<int>), dim3(nblocks_for(total)), dim3(256), 0, stream, 

@@ -319,7 +319,7 @@ def maxpool_forward(x, kernel, stride, pad):
     xl = _cl(x)
     y = torch.empty((N, C, P, Q), dtype=x.dtype, device=x.device,
                     memory_format=torch.channels_last)
-    idt = torch.int16 if H * W < 32768 else torch.int32
+    idt = torch.int8 if kh * kw < 128 else torch.int32
     idx = torch.empty((N, P, Q, C), dtype=idt, device=x.device)
     _ext.maxpool_fwd(xl, y, idx, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw)
     return y, (idx, (kh, kw, sh, sw, ph, pw), (N, C, H, W))
