@@ -202,3 +202,59 @@ def test_dataframe_source(workdir, tmp_path):
     batch = src.next_batch(torch.device("cpu"), torch.float32)
     assert batch[0].shape == (64, 1, 28, 28)
     assert batch[1].shape == (64,)
+
+
+def test_cos_dataframe_training(workdir, tmp_path):
+    """lenet_cos: CoSData layer fed by DataFrameSource (parquet), the
+    reference's lenet_cos_train_test.prototxt path."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from caffeonspark_amd.proto import text_format
+
+    rng = np.random.RandomState(7)
+    ids, datas, labels = [], [], []
+    for i in range(800):
+        label = rng.randint(0, 10)
+        img = rng.randint(80, 150, size=(1, 28, 28)).astype(np.uint8)
+        img[0, :, label * 2] = 250
+        ids.append(str(i))
+        datas.append(img.tobytes())
+        labels.append(label)
+    pq_file = str(tmp_path / "mnist.parquet")
+    pq.write_table(pa.table({"id": ids, "data": datas, "label": labels}),
+                   pq_file)
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    net = text_format.parse_file(
+        os.path.join(root, "caffeonspark_amd", "models",
+                     "lenet_cos_train_test.prototxt"), caffe_pb.NetParameter)
+    for lp in net.layer:
+        if lp.type == "CoSData":
+            lp.cos_data_param.source = pq_file
+    net_file = tmp_path / "lenet_cos.prototxt"
+    net_file.write_text(text_format.dumps(net))
+    solver_file = tmp_path / "solver_cos.prototxt"
+    solver_file.write_text(f"""
+net: "{net_file}"
+test_iter: 0
+test_interval: 0
+base_lr: 0.01
+momentum: 0.9
+lr_policy: "fixed"
+display: 0
+max_iter: 60
+snapshot: 0
+snapshot_prefix: "{tmp_path}/lenet_cos"
+random_seed: 3
+""")
+    CaffeProcessor.reset_instance()
+    os.chdir(tmp_path)
+    conf = Config(["-conf", str(solver_file), "-train"])
+    cos = CaffeOnSpark(conf)
+    cos.train()
+    proc_loss = None
+    # training ran to max_iter and snapshotted
+    snaps = [f for f in os.listdir(tmp_path) if f.endswith(".caffemodel")]
+    assert snaps
+    CaffeProcessor.reset_instance()
