@@ -63,6 +63,13 @@ class LSTMLayer(Layer):
             xs = ops.fc_forward(bottom[2].data.reshape(N, -1), self.weight(3),
                                 None)
             xg = xg + xs.unsqueeze(0)
+        seq_fwd = ops.gpu_op("lstm_seq_forward") if x.is_cuda else None
+        if seq_fwd is not None:
+            # whole recurrence driven from C++ (3 kernel launches/step)
+            y, seq_cache = seq_fwd(xg, w_hc, cont)
+            top[0].data = y
+            self._cache = ("seq", x, cont, seq_cache)
+            return 0.0
         h_prev = torch.zeros(N, H, dtype=x.dtype, device=x.device)
         c_prev = torch.zeros(N, H, dtype=torch.float32, device=x.device)
         hs, caches, h_prevs, c_prevs = [], [], [], []
@@ -83,6 +90,10 @@ class LSTMLayer(Layer):
         return 0.0
 
     def backward(self, top, propagate_down, bottom):
+        if self._cache and isinstance(self._cache[0], str) \
+                and self._cache[0] == "seq":
+            self._backward_seq(top, propagate_down, bottom)
+            return
         x, cont, caches, h_prevs, c_prevs, hs = self._cache
         T, N = x.shape[0], x.shape[1]
         H = self.h
@@ -122,6 +133,35 @@ class LSTMLayer(Layer):
                 self.acc_blob_diff(bottom[2],
                                    dxs.reshape(bottom[2].data.shape).to(x.dtype),
                                    False)
+
+
+    def _backward_seq(self, top, propagate_down, bottom):
+        _, x, cont, seq_cache = self._cache
+        T, N = x.shape[0], x.shape[1]
+        H = self.h
+        dy = top[0].diff
+        seq_bwd = ops.gpu_op("lstm_seq_backward")
+        dxg, dw_hc = seq_bwd(dy.reshape(T, N, H), self.weight(2), seq_cache)
+        dxg_flat = dxg.reshape(T * N, 4 * H)
+        x_flat = x.reshape(T * N, -1)
+        # shared tail: input GEMM grads == a linear layer's backward
+        dx, dw_xc, db = ops.fc_backward(
+            x_flat, self.weight(0), dxg_flat,
+            need_dx=propagate_down[0], bias=True)
+        if self.blobs[0]._lr_mult != 0:
+            self.acc_param_diff(0, dw_xc)
+            self.acc_param_diff(1, db)
+            self.acc_param_diff(2, dw_hc)
+        if propagate_down[0]:
+            self.acc_blob_diff(bottom[0], dx.reshape(x.shape), False)
+        if self.static and self.blobs[3]._lr_mult != 0:
+            xs = bottom[2].data.reshape(N, -1)
+            d_static = dxg.float().sum(dim=0)  # [N, 4H]
+            self.acc_param_diff(3, d_static.t() @ xs.float())
+            if propagate_down[2]:
+                dxs = (d_static @ self.weight(3).float()).to(x.dtype)
+                self.acc_blob_diff(bottom[2],
+                                   dxs.reshape(bottom[2].data.shape), False)
 
 
 @register_layer("Embed")

@@ -130,3 +130,9 @@ bias_add = _dispatch("bias_add")
 sgd_update = _dispatch("sgd_update")
 nesterov_update = _dispatch("nesterov_update")
 adam_update = _dispatch("adam_update")
+
+
+def gpu_op(name):
+    """GPU-only op (no CPU reference), or None if unavailable."""
+    native()
+    return dispatcher.gpu_impls.get(name)

@@ -67,6 +67,14 @@ void embed_fwd(const float* idx, const void* w, void* y, int64_t n, int E,
                int V, hipStream_t stream);
 void embed_bwd(const float* idx, const void* dy, float* dw, int64_t n, int E,
                int V, hipStream_t stream);
+void lstm_seq_fwd(const void* xg, const void* w_hc, const void* cont,
+                  void* h, float* c, float* act, void* h_in, void* hg,
+                  int T, int N, int H, int n_alloc_whc, hipStream_t stream);
+void lstm_seq_bwd(const void* dy, const void* w_hcT, const void* cont,
+                  const void* h, const float* c, const float* act,
+                  void* dxg, void* dh_rec, float* dc_a, float* dc_b,
+                  int T, int N, int H, int n_alloc_whcT,
+                  hipStream_t stream);
 void softmax_loss_fwd(const void* x, const float* label, float* prob,
                       float* loss, int* count, int64_t nrows, int C,
                       int ignore, bool has_ignore, hipStream_t stream);
@@ -260,6 +268,29 @@ void py_embed_bwd(Tensor idx, Tensor dy, Tensor dw, int64_t V) {
                     cur_stream());
 }
 
+void py_lstm_seq_fwd(Tensor xg, Tensor w_hc, Tensor cont, Tensor h,
+                     Tensor c, Tensor act, Tensor h_in, Tensor hg,
+                     int64_t T, int64_t N, int64_t H, int64_t n_alloc) {
+  CHECK_BF16(xg); CHECK_BF16(w_hc); CHECK_F32(c); CHECK_F32(act);
+  cosamd::lstm_seq_fwd(xg.data_ptr(), w_hc.data_ptr(), cont.data_ptr(),
+                       h.data_ptr(), c.data_ptr<float>(),
+                       act.data_ptr<float>(), h_in.data_ptr(),
+                       hg.data_ptr(), T, N, H, n_alloc, cur_stream());
+}
+
+void py_lstm_seq_bwd(Tensor dy, Tensor w_hcT, Tensor cont, Tensor h,
+                     Tensor c, Tensor act, Tensor dxg, Tensor dh_rec,
+                     Tensor dc_a, Tensor dc_b, int64_t T, int64_t N,
+                     int64_t H, int64_t n_alloc) {
+  CHECK_BF16(dy); CHECK_BF16(w_hcT);
+  cosamd::lstm_seq_bwd(dy.data_ptr(), w_hcT.data_ptr(), cont.data_ptr(),
+                       h.data_ptr(), c.data_ptr<float>(),
+                       act.data_ptr<float>(), dxg.data_ptr(),
+                       dh_rec.data_ptr(), dc_a.data_ptr<float>(),
+                       dc_b.data_ptr<float>(), T, N, H, n_alloc,
+                       cur_stream());
+}
+
 void py_softmax_loss_fwd(Tensor x, Tensor label, Tensor prob, Tensor loss,
                          Tensor count, int64_t ignore, bool has_ignore) {
   CHECK_BF16(x); CHECK_F32(label); CHECK_F32(prob);
@@ -296,6 +327,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum", &py_colsum);
   m.def("transpose", &py_transpose);
   m.def("bias_act_cast", &py_bias_act_cast);
+  m.def("lstm_seq_fwd", &py_lstm_seq_fwd);
+  m.def("lstm_seq_bwd", &py_lstm_seq_bwd);
   m.def("lstm_unit_fwd", &py_lstm_unit_fwd);
   m.def("lstm_unit_bwd", &py_lstm_unit_bwd);
   m.def("embed_fwd", &py_embed_fwd);

@@ -480,6 +480,53 @@ def lstm_unit_backward(c_prev, cache, dc_next, dh):
     return dc_prev, dgates
 
 
+def lstm_seq_forward(xg, w_hc, cont):
+    """Whole-sequence LSTM forward driven from C++ (3 launches/step).
+    xg: [T,N,4H] pre-activation input gates (x @ W_xc + b [+ static]),
+    cont: [T,N].  Returns (h [T,N,H] bf16, cache)."""
+    T, N, H4 = xg.shape
+    H = H4 // 4
+    dev_ = xg.device
+    xg = _as_bf16(xg).contiguous()
+    cont_b = _as_bf16(cont).contiguous()
+    whc = torch.zeros((_pad128(H4), H), dtype=torch.bfloat16, device=dev_)
+    whc[:H4] = _as_bf16(w_hc)
+    h = torch.empty((T, N, H), dtype=torch.bfloat16, device=dev_)
+    c = torch.empty((T, N, H), dtype=torch.float32, device=dev_)
+    act = torch.empty((T, N, H4), dtype=torch.float32, device=dev_)
+    h_in = torch.empty((T, N, H), dtype=torch.bfloat16, device=dev_)
+    hg = torch.empty((N, H4), dtype=torch.bfloat16, device=dev_)
+    _ext.lstm_seq_fwd(xg, whc[:H4], cont_b, h, c, act, h_in, hg,
+                      T, N, H, whc.shape[0])
+    return h, (cont_b, h, c, act, h_in, T, N, H)
+
+
+def lstm_seq_backward(dy, w_hc, cache):
+    """Returns (dxg [T,N,4H] bf16, dw_hc fp32)."""
+    cont_b, h, c, act, h_in, T, N, H = cache
+    H4 = 4 * H
+    dev_ = dy.device
+    dy = _as_bf16(dy).contiguous()
+    whcT = _transpose(_as_bf16(w_hc).contiguous())      # [H][4H], padded rows
+    dxg = torch.empty((T, N, H4), dtype=torch.bfloat16, device=dev_)
+    dh_rec = torch.empty((N, H), dtype=torch.bfloat16, device=dev_)
+    dc_a = torch.empty((N, H), dtype=torch.float32, device=dev_)
+    dc_b = torch.empty((N, H), dtype=torch.float32, device=dev_)
+    _ext.lstm_seq_bwd(dy, whcT, cont_b, h, c, act, dxg, dh_rec, dc_a, dc_b,
+                      T, N, H, _pad128(H4))
+    # dw_hc = dgates_all^T @ h_in_all — one big NT GEMM
+    dxg_flat = dxg.reshape(T * N, H4)
+    h_in_flat = h_in.reshape(T * N, H)
+    dgT = _transpose(dxg_flat)
+    hiT = _transpose(h_in_flat)
+    dwhc = torch.zeros((H4, H), dtype=torch.float32, device=dev_)
+    mb, nb = (H4 + 127) // 128, (H + 127) // 128
+    _gemm(dgT, hiT, dwhc, None, H4, H, T * N, T * N, T * N, H,
+          False, False, 2, _splitk_for(mb, nb, T * N),
+          ma=_pad128(H4), na=_pad128(H))
+    return dxg, dwhc
+
+
 # -------------------------------------------------------------- optimizer
 
 def sgd_update(param, grad, momentum_buf, lr, momentum, weight_decay):
