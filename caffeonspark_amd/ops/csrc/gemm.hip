@@ -48,13 +48,13 @@ __device__ __forceinline__ int swz_chunk(int row, int chunk) {
 // 16-byte slot is fully in-bounds, scalar otherwise; zero-fills the rest.
 __device__ __forceinline__ void stage_direct_guarded(
     bf16* lds_, const bf16* g_, int row0, int rows, int ld,
-    int k0, int kend, int tid, int nthreads = NTHREADS,
-    int tile_rows = BM) {
+    int k0, int kend, int tid) {
   auto* lds = reinterpret_cast<unsigned short*>(lds_);
   auto* g = reinterpret_cast<const unsigned short*>(g_);
-  // tile_rows*4 slots of 8 elements each
-  int nslots = tile_rows * 4;
-  for (int slot = tid; slot < nslots; slot += nthreads) {
+  // 4096 elements, 512 slots of 8; 256 threads -> 2 slots each
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    int slot = tid + it * NTHREADS;
     int row = slot >> 2;            // 4 slots of 8 per 32-wide row
     int kk = (slot & 3) * 8;
     int grow = row0 + row;
@@ -70,124 +70,6 @@ __device__ __forceinline__ void stage_direct_guarded(
     } else {
 #pragma unroll
       for (int j = 0; j < 8; ++j) dst[j] = 0;
-    }
-  }
-}
-
-// fast DMA staging for arbitrary tile size (chunks of 1 KiB per wave-issue)
-__device__ __forceinline__ void stage_direct_fast2(
-    bf16* lds, const bf16* g, int row0, int ld, int k0, int tid,
-    int nthreads, int tile_rows) {
-  int wave = tid >> 6, lane = tid & 63;
-  int nchunks = tile_rows / 16;          // 1 KiB = 16 rows of 32 bf16
-  int waves = nthreads >> 6;
-  for (int chunk = wave; chunk < nchunks; chunk += waves) {
-    int idx = chunk * 512 + lane * 8;
-    int row = idx >> 5;
-    int kk = swz_chunk(row, (idx & 31) >> 3) << 3;
-    auto* gp = (const __attribute__((address_space(1))) unsigned int*)(
-        g + (int64_t)(row0 + row) * ld + k0 + kk);
-    auto* lp = (__attribute__((address_space(3))) unsigned int*)(
-        lds + chunk * 512);
-    __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
-  }
-}
-
-// ------------------------------------------------ 256x256 big-tile kernel
-// For large-N direct/direct GEMMs (fc layers, conv dcol/dw, vocab
-// projection): doubles arithmetic intensity per staged byte vs the 128
-// tile.  8 waves as 2(M)x4(N), per-wave 128x64 output (8x4 fragments).
-
-constexpr int BM2 = 256, BN2 = 256, NT2 = 512;
-
-template <int STORE_MODE>
-__global__ __launch_bounds__(NT2, 2) void gemm256_kernel(
-    const bf16* __restrict__ A, const bf16* __restrict__ B,
-    void* __restrict__ C, const float* __restrict__ bias,
-    int M, int N, int K, int lda, int ldb, int ldc,
-    int ksplit, int relu, float alpha, int m_alloc, int n_alloc) {
-  __shared__ bf16 As[BM2 * BK];
-  __shared__ bf16 Bs[BN2 * BK];
-
-  int mblocks = (M + BM2 - 1) / BM2;
-  int nblocks = (N + BN2 - 1) / BN2;
-  int bid = xcd_swizzle(blockIdx.x, mblocks * nblocks);
-  int bm = bid / nblocks, bn = bid % nblocks;
-  int tile_m = bm * BM2, tile_n = bn * BN2;
-  int k_begin = blockIdx.y * ksplit;
-  int k_end = min(K, k_begin + ksplit);
-
-  int tid = threadIdx.x;
-  int wave = tid >> 6, lane = tid & 63;
-  int wm = wave >> 2, wn = wave & 3;     // 2x4 waves, 128x64 each
-  int lrow = lane & 15, lk8 = (lane >> 4) * 8;
-
-  f32x4 acc[8][4] = {};
-
-  bool a_fast = (tile_m + BM2 <= m_alloc) && (lda % 8 == 0) &&
-                ((k_end - k_begin) % BK == 0) && (k_begin % 8 == 0);
-  bool b_fast = (tile_n + BN2 <= n_alloc) && (ldb % 8 == 0) &&
-                ((k_end - k_begin) % BK == 0) && (k_begin % 8 == 0);
-
-  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
-    if (a_fast)
-      stage_direct_fast2(As, A, tile_m, lda, k0, tid, NT2, BM2);
-    else
-      stage_direct_guarded(As, A, tile_m, M, lda, k0, k_end, tid, NT2, BM2);
-    if (b_fast)
-      stage_direct_fast2(Bs, B, tile_n, ldb, k0, tid, NT2, BN2);
-    else
-      stage_direct_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid, NT2, BN2);
-    __syncthreads();
-
-    bf16x8 afrag[8], bfrag[4];
-#pragma unroll
-    for (int f = 0; f < 8; ++f) {
-      int ra = wm * 128 + f * 16 + lrow;
-      afrag[f] = *reinterpret_cast<const bf16x8*>(
-          As + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
-    }
-#pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      int rb = wn * 64 + f * 16 + lrow;
-      bfrag[f] = *reinterpret_cast<const bf16x8*>(
-          Bs + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
-    }
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int fm = 0; fm < 8; ++fm)
-#pragma unroll
-      for (int fn = 0; fn < 4; ++fn)
-        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
-    __syncthreads();
-  }
-
-  int crow0 = tile_m + wm * 128 + ((lane >> 4) << 2);
-  int ccol0 = tile_n + wn * 64 + (lane & 15);
-#pragma unroll
-  for (int fm = 0; fm < 8; ++fm) {
-#pragma unroll
-    for (int fn = 0; fn < 4; ++fn) {
-      int col = ccol0 + fn * 16;
-      if (col >= N) continue;
-      float badd = (bias != nullptr) ? bias[col] : 0.f;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int row = crow0 + fm * 16 + r;
-        if (row >= M) continue;
-        float v = acc[fm][fn][r] * alpha + badd;
-        if (relu && v < 0.f) v = 0.f;
-        int64_t off = (int64_t)row * ldc + col;
-        if (STORE_MODE == 0) {
-          reinterpret_cast<bf16*>(C)[off] = f2bf(v);
-        } else if (STORE_MODE == 1) {
-          reinterpret_cast<float*>(C)[off] = v;
-        } else {
-          atomicAdd(reinterpret_cast<float*>(C) + off, v);
-        }
-      }
     }
   }
 }
@@ -454,23 +336,6 @@ void gemm_bf16(const void* A_, const void* B_, void* C, const float* bias,
     else                      COS_GEMM_CASE(TA, TB, 2);                     \
   } while (0)
 
-  if (!trans_a && !trans_b && N >= 512 && M >= 192) {
-    int mb2 = (M + BM2 - 1) / BM2, nb2 = (N + BN2 - 1) / BN2;
-    dim3 g2(mb2 * nb2, zblocks);
-    if (store_mode == 0)
-      gemm256_kernel<0><<<g2, NT2, 0, stream>>>(
-          A, B, C, bias, M, N, K, lda, ldb, ldc, ksplit, relu ? 1 : 0,
-          alpha, m_alloc, n_alloc);
-    else if (store_mode == 1)
-      gemm256_kernel<1><<<g2, NT2, 0, stream>>>(
-          A, B, C, bias, M, N, K, lda, ldb, ldc, ksplit, relu ? 1 : 0,
-          alpha, m_alloc, n_alloc);
-    else
-      gemm256_kernel<2><<<g2, NT2, 0, stream>>>(
-          A, B, C, bias, M, N, K, lda, ldb, ldc, ksplit, relu ? 1 : 0,
-          alpha, m_alloc, n_alloc);
-    return;
-  }
   if (trans_a && trans_b && store_mode == 2 && N > 128) {
     int nb_w = (N + WBN - 1) / WBN;
     dim3 gw(mblocks * nb_w, zblocks);
