@@ -12,6 +12,7 @@ solver); torch is used only for small glue (casts, pads, permutes).
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional, Tuple
 
 import torch
@@ -162,15 +163,22 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         # Both operands are K(=npq)-major; transposing them once (cheap
         # torch copy kernels) turns the GEMM into the fast NT direct/direct
         # form with global_load_lds staging instead of LDS scatter staging.
-        dyT = _transpose(dy2)  # [Kout][NPQ], rows over-allocated
         dwp = torch.zeros((Kout, Kpad), dtype=torch.float32, device=dy.device)
+        use_tt = bool(int(os.environ.get("COS_DW_TT", "0")))
+        dyT = None if use_tt else _transpose(dy2)
         for g in range(G):
-            colT = _transpose(x2 if is_1x1 else col[g])
             mb, nb = (Kg + 127) // 128, (Kpad + 127) // 128
             sk = _splitk_for(mb, nb, NPQ)
-            _gemm(dyT[g * Kg:(g + 1) * Kg], colT, dwp[g * Kg:],
-                  None, Kg, Kpad, NPQ, NPQ, NPQ, Kpad, False, False, 2, sk,
-                  ma=_pad128(Kout) - g * Kg, na=_pad128(Kpad))
+            if use_tt:
+                src = x2 if is_1x1 else col[g]
+                _gemm(dy2[:, g * Kg:], src, dwp[g * Kg:], None,
+                      Kg, Kpad, NPQ, Kout, Kpad if not is_1x1 else C,
+                      Kpad, True, True, 2, sk)
+            else:
+                colT = _transpose(x2 if is_1x1 else col[g])
+                _gemm(dyT[g * Kg:(g + 1) * Kg], colT, dwp[g * Kg:],
+                      None, Kg, Kpad, NPQ, NPQ, NPQ, Kpad, False, False,
+                      2, sk, ma=_pad128(Kout) - g * Kg, na=_pad128(Kpad))
         dw = dwp[:, :Kcol].reshape(Kout, R, S, Cg).permute(0, 3, 1, 2) \
             .contiguous()
     if bias:
