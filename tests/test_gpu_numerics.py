@@ -43,7 +43,8 @@ def _gemm_ref(a, b, ta, tb):
 @pytest.mark.parametrize("ta,tb", [(False, False), (False, True),
                                    (True, False), (True, True)])
 @pytest.mark.parametrize("m,n,k", [(128, 128, 64), (256, 384, 512),
-                                   (100, 96, 363), (130, 70, 40)])
+                                   (100, 96, 363), (130, 70, 40),
+                                   (512, 1024, 768)])
 def test_gemm_combos(ta, tb, m, n, k):
     ext = ops.native()
     a_shape = (k, m) if ta else (m, k)
