@@ -77,7 +77,7 @@ class Blob:
         proto.shape = caffe_pb.BlobShape(dim=[int(d) for d in self.data.shape])
         t = self.data.detach().cpu().contiguous()
         if t.dtype == torch.float32:
-            proto.data = t.view(-1).tolist()
+            proto.data = t.view(-1).numpy()  # codec fast-paths ndarrays
         else:
             proto.raw_data = t.numpy().tobytes()
             proto.raw_dtype = str(t.dtype).replace("torch.", "")
@@ -89,7 +89,7 @@ class Blob:
     def _tensor_to_proto(t: torch.Tensor) -> caffe_pb.BlobProto:
         proto = caffe_pb.BlobProto()
         proto.shape = caffe_pb.BlobShape(dim=[int(d) for d in t.shape])
-        proto.data = t.detach().cpu().float().view(-1).tolist()
+        proto.data = t.detach().cpu().float().view(-1).numpy()
         return proto
 
     @staticmethod
@@ -116,9 +116,15 @@ class Blob:
                 t = t.view(torch.bfloat16)
             src = t.reshape(shape)
         elif proto.has_field("double_data"):
-            src = torch.tensor(proto.double_data, dtype=torch.float64).reshape(shape)
+            import numpy as np
+            src = torch.from_numpy(
+                np.asarray(proto.double_data, dtype=np.float64).copy()
+            ).reshape(shape)
         else:
-            src = torch.tensor(proto.data, dtype=torch.float32).reshape(shape)
+            import numpy as np
+            src = torch.from_numpy(
+                np.asarray(proto.data, dtype=np.float32).copy()
+            ).reshape(shape)
         self.data = src.to(device=device, dtype=dtype)
         if proto.has_field("diff"):
             self.diff = torch.tensor(proto.diff, dtype=torch.float32).reshape(

@@ -14,12 +14,15 @@ __global__ void maxpool_fwd_kernel(
     const u16* __restrict__ x, u16* __restrict__ y, IDX* __restrict__ idx,
     int N, int H, int W, int C, int P, int Q,
     int kh, int kw, int sh, int sw, int ph, int pw, int64_t total) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total; i8 += (int64_t)gridDim.x * blockDim.x) {
+    unsigned i = (unsigned)i8;
     int c = i % C;
-    int q = (i / C) % Q;
-    int p = (i / ((int64_t)C * Q)) % P;
-    int n = i / ((int64_t)C * Q * P);
+    unsigned iq = i / C;
+    int q = iq % Q;
+    unsigned ip = iq / Q;
+    int p = ip % P;
+    int n = ip / P;
     int h0 = p * sh - ph, w0 = q * sw - pw;
     float best = -3.4e38f;
     int best_rs = 0;  // window-relative argmax (fits int8 for k<=11)
@@ -35,8 +38,8 @@ __global__ void maxpool_fwd_kernel(
       }
     }
     bf16 out = f2bf(best);
-    y[i] = *reinterpret_cast<u16*>(&out);
-    idx[i] = (IDX)best_rs;
+    y[i8] = *reinterpret_cast<u16*>(&out);
+    idx[i8] = (IDX)best_rs;
   }
 }
 
@@ -46,12 +49,15 @@ __global__ void maxpool_bwd_kernel(
     u16* __restrict__ dx,
     int N, int H, int W, int C, int P, int Q,
     int kh, int kw, int sh, int sw, int ph, int pw, int64_t total) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total; i8 += (int64_t)gridDim.x * blockDim.x) {
+    unsigned i = (unsigned)i8;
     int c = i % C;
-    int w = (i / C) % W;
-    int h = (i / ((int64_t)C * W)) % H;
-    int n = i / ((int64_t)C * W * H);
+    unsigned iw = i / C;
+    int w = iw % W;
+    unsigned ih = iw / W;
+    int h = ih % H;
+    int n = ih / H;
     float acc = 0.f;
     // windows (p,q) that can contain (h,w); match window-relative argmax
     int plo = max(0, (h + ph - kh + sh) / sh), phi = min(P - 1, (h + ph) / sh);
@@ -67,7 +73,7 @@ __global__ void maxpool_bwd_kernel(
       }
     }
     bf16 out = f2bf(acc);
-    dx[i] = *reinterpret_cast<u16*>(&out);
+    dx[i8] = *reinterpret_cast<u16*>(&out);
   }
 }
 
@@ -75,12 +81,15 @@ __global__ void avgpool_fwd_kernel(
     const u16* __restrict__ x, u16* __restrict__ y,
     int N, int H, int W, int C, int P, int Q,
     int kh, int kw, int sh, int sw, int ph, int pw, int64_t total) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total; i8 += (int64_t)gridDim.x * blockDim.x) {
+    unsigned i = (unsigned)i8;
     int c = i % C;
-    int q = (i / C) % Q;
-    int p = (i / ((int64_t)C * Q)) % P;
-    int n = i / ((int64_t)C * Q * P);
+    unsigned iq = i / C;
+    int q = iq % Q;
+    unsigned ip = iq / Q;
+    int p = ip % P;
+    int n = ip / P;
     int h0 = p * sh - ph, w0 = q * sw - pw;
     // caffe: divisor = window clipped at the padded boundary
     int hend = min(h0 + kh, H + ph), wend = min(w0 + kw, W + pw);
@@ -92,7 +101,7 @@ __global__ void avgpool_fwd_kernel(
         acc += bf2f(*reinterpret_cast<const bf16*>(&v));
       }
     bf16 out = f2bf(acc / pool_size);
-    y[i] = *reinterpret_cast<u16*>(&out);
+    y[i8] = *reinterpret_cast<u16*>(&out);
   }
 }
 
@@ -100,12 +109,15 @@ __global__ void avgpool_bwd_kernel(
     const u16* __restrict__ dy, u16* __restrict__ dx,
     int N, int H, int W, int C, int P, int Q,
     int kh, int kw, int sh, int sw, int ph, int pw, int64_t total) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total; i8 += (int64_t)gridDim.x * blockDim.x) {
+    unsigned i = (unsigned)i8;
     int c = i % C;
-    int w = (i / C) % W;
-    int h = (i / ((int64_t)C * W)) % H;
-    int n = i / ((int64_t)C * W * H);
+    unsigned iw = i / C;
+    int w = iw % W;
+    unsigned ih = iw / W;
+    int h = ih % H;
+    int n = ih / H;
     float acc = 0.f;
     int plo = max(0, (h + ph - kh + sh) / sh), phi = min(P - 1, (h + ph) / sh);
     int qlo = max(0, (w + pw - kw + sw) / sw), qhi = min(Q - 1, (w + pw) / sw);
@@ -122,7 +134,7 @@ __global__ void avgpool_bwd_kernel(
       }
     }
     bf16 out = f2bf(acc);
-    dx[i] = *reinterpret_cast<u16*>(&out);
+    dx[i8] = *reinterpret_cast<u16*>(&out);
   }
 }
 

@@ -148,8 +148,9 @@ class Message(metaclass=MessageMeta):
         f = type(self)._by_name.get(name)
         if f is None:
             raise AttributeError(f"{type(self).__name__} has no field {name!r}")
-        if f.repeated and not isinstance(value, list):
-            value = list(value)
+        if f.repeated and not isinstance(value, list) and \
+                not (hasattr(value, "dtype") and hasattr(value, "tobytes")):
+            value = list(value)  # ndarrays pass through for the fast path
         self._values[name] = value
 
     def has_field(self, name: str) -> bool:
@@ -198,12 +199,20 @@ class Message(metaclass=MessageMeta):
                 continue
             v = self._values[name]
             if f.repeated:
-                if not v:
+                if not (len(v) if hasattr(v, "__len__") else True):
                     continue
                 if f.packed:
-                    payload = bytearray()
-                    for item in v:
-                        _encode_scalar(payload, f, item)
+                    if f.type in ("float", "double"):
+                        # numpy fast path: packed floats are exactly the
+                        # little-endian array bytes (snapshots carry
+                        # millions of weights)
+                        import numpy as _np
+                        dt = "<f4" if f.type == "float" else "<f8"
+                        payload = _np.asarray(v, dtype=dt).tobytes()
+                    else:
+                        payload = bytearray()
+                        for item in v:
+                            _encode_scalar(payload, f, item)
                     _encode_varint(buf, (f.number << 3) | WT_LEN)
                     _encode_varint(buf, len(payload))
                     buf += payload
@@ -239,6 +248,19 @@ class Message(metaclass=MessageMeta):
                 # packed repeated scalars
                 length, pos = _decode_varint(data, pos)
                 end = pos + length
+                if f.type in ("float", "double"):
+                    import numpy as _np
+                    dt = "<f4" if f.type == "float" else "<f8"
+                    arr = _np.frombuffer(data[pos:end], dtype=dt)
+                    cur = self._values.get(f.name)
+                    if cur is None or (hasattr(cur, "__len__")
+                                       and len(cur) == 0):
+                        self._values[f.name] = arr  # ndarray: list-like
+                    else:
+                        self._values[f.name] = _np.concatenate(
+                            [_np.asarray(cur, dtype=dt), arr])
+                    pos = end
+                    continue
                 lst = getattr(self, f.name)
                 while pos < end:
                     val, pos = _decode_scalar(data, pos, f)
