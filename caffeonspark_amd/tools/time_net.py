@@ -40,7 +40,7 @@ def time_net(solver_path: str, iters: int = 10, batch: int = 0,
         dl.batch_size = n
         dl.reset(torch.randn(n, dl.channels, dl.height, dl.width)
                  .to(device, dtype),
-                 torch.randint(0, 1000, (n,)).float().to(device))
+                 torch.randint(0, 10, (n,)).float().to(device))
 
     def sync():
         if device.type == "cuda":
