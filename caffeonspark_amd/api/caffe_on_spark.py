@@ -107,7 +107,6 @@ class CaffeOnSpark:
             else proc.solver.net
         dl = net.data_layers()[0]
         rows = []
-        stop = object()
 
         def feed():
             self._feed(proc, 0, epochs=1, limit=max_samples)

@@ -14,7 +14,6 @@ from __future__ import annotations
 
 import glob
 import os
-import queue
 from typing import Iterator, List, Optional
 
 import numpy as np

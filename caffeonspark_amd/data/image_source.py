@@ -8,7 +8,6 @@ zero-copy to the MemoryData layer.
 
 from __future__ import annotations
 
-import queue
 from typing import List, Optional
 
 import numpy as np

@@ -12,7 +12,7 @@ from __future__ import annotations
 
 import os
 import struct
-from typing import Iterator, List, Tuple
+from typing import Iterator, Tuple
 
 _SYNC_INTERVAL = 2000
 _CLASS = b"org.apache.hadoop.io.BytesWritable"

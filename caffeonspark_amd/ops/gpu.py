@@ -13,8 +13,6 @@ from __future__ import annotations
 
 import math
 import os
-from typing import Optional, Tuple
-
 import torch
 
 from . import reference

@@ -8,7 +8,7 @@ semantics follow upstream Caffe (the engine the reference drives through
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 import torch.nn.functional as F

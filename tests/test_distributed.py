@@ -10,15 +10,6 @@ import torch
 import torch.multiprocessing as mp
 
 
-def _find_port():
-    import socket
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    port = s.getsockname()[1]
-    s.close()
-    return port
-
-
 def _worker(rank, ws, store_path, q):
     import torch.distributed as dist
 

@@ -253,7 +253,6 @@ random_seed: 3
     conf = Config(["-conf", str(solver_file), "-train"])
     cos = CaffeOnSpark(conf)
     cos.train()
-    proc_loss = None
     # training ran to max_iter and snapshotted
     snaps = [f for f in os.listdir(tmp_path) if f.endswith(".caffemodel")]
     assert snaps
