@@ -210,8 +210,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
     void* __restrict__ C, const float* __restrict__ bias,
     int M, int N, int K, int lda, int ldb, int ldc,
     int ksplit, int relu, float alpha, int m_alloc, int n_alloc) {
-  __shared__ bf16 As[BM * BK];
-  __shared__ bf16 Bs[BN * BK];
+  __shared__ bf16 Asb[2][BM * BK];
+  __shared__ bf16 Bsb[2][BN * BK];
 
   int mblocks = (M + BM - 1) / BM;
   int nblocks = (N + BN - 1) / BN;
@@ -237,42 +237,91 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
   bool b_fast = !TRANS_B && (tile_n + BN <= n_alloc) && (ldb % 8 == 0) &&
                 ((k_end - k_begin) % BK == 0) && (k_begin % 8 == 0);
 
-  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
-    if (a_fast) {
-      stage_direct_fast(As, A, tile_m, lda, k0, tid);
-    } else if (TRANS_A) {
-      stage_trans_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
-    } else {
-      stage_direct_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
+  if (a_fast && b_fast && (k_end - k_begin) >= 2 * BK) {
+    // Pipelined double-buffered path (interior aligned tiles): the next
+    // K-tile's direct-to-LDS DMA is issued BEFORE computing the current
+    // tile, and the pre-barrier wait is a COUNTED `s_waitcnt vmcnt(4)`
+    // (the 4 loads just issued stay in flight across the barrier) rather
+    // than __syncthreads' full drain — cdna_hip_programming.md §5.5
+    // T3/T4: "never drain vmcnt to 0 in the main loop".
+    int cur = 0;
+    stage_direct_fast(Asb[0], A, tile_m, lda, k_begin, tid);
+    stage_direct_fast(Bsb[0], B, tile_n, ldb, k_begin, tid);
+    for (int k0 = k_begin; k0 < k_end; k0 += BK) {
+      if (k0 + BK < k_end) {
+        stage_direct_fast(Asb[cur ^ 1], A, tile_m, lda, k0 + BK, tid);
+        stage_direct_fast(Bsb[cur ^ 1], B, tile_n, ldb, k0 + BK, tid);
+      }
+      // wait for the CURRENT tile's DMA loads; the prefetched tile's 4
+      // (2 per operand per wave) remain outstanding
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_barrier();
+      const bf16* Asp = Asb[cur];
+      const bf16* Bsp = Bsb[cur];
+      bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        int ra = wm * 64 + f * 16 + lrow;
+        int rb = wn * 64 + f * 16 + lrow;
+        afrag[f] = *reinterpret_cast<const bf16x8*>(
+            Asp + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
+        bfrag[f] = *reinterpret_cast<const bf16x8*>(
+            Bsp + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      // all waves finish reading buf[cur] before the next iteration's
+      // stage overwrites it; the ds_reads completed before their MFMAs,
+      // so no counter drain is needed at this barrier
+      __builtin_amdgcn_s_barrier();
+      cur ^= 1;
     }
-    if (b_fast) {
-      stage_direct_fast(Bs, B, tile_n, ldb, k0, tid);
-    } else if (TRANS_B) {
-      stage_trans_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
-    } else {
-      stage_direct_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
-    }
-    __syncthreads();
+  } else {
+    bf16* As = Asb[0];
+    bf16* Bs = Bsb[0];
+    for (int k0 = k_begin; k0 < k_end; k0 += BK) {
+      if (a_fast) {
+        stage_direct_fast(As, A, tile_m, lda, k0, tid);
+      } else if (TRANS_A) {
+        stage_trans_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
+      } else {
+        stage_direct_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
+      }
+      if (b_fast) {
+        stage_direct_fast(Bs, B, tile_n, ldb, k0, tid);
+      } else if (TRANS_B) {
+        stage_trans_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
+      } else {
+        stage_direct_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
+      }
+      __syncthreads();
 
-    bf16x8 afrag[4], bfrag[4];
+      bf16x8 afrag[4], bfrag[4];
 #pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      int ra = wm * 64 + f * 16 + lrow;
-      int rb = wn * 64 + f * 16 + lrow;
-      afrag[f] = *reinterpret_cast<const bf16x8*>(
-          As + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
-      bfrag[f] = *reinterpret_cast<const bf16x8*>(
-          Bs + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
-    }
+      for (int f = 0; f < 4; ++f) {
+        int ra = wm * 64 + f * 16 + lrow;
+        int rb = wn * 64 + f * 16 + lrow;
+        afrag[f] = *reinterpret_cast<const bf16x8*>(
+            As + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
+        bfrag[f] = *reinterpret_cast<const bf16x8*>(
+            Bs + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
+      }
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
+      for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn)
-        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
+        for (int fn = 0; fn < 4; ++fn)
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
 
-    // second half of the K-tile? no: 16x16x32 consumes all BK=32 at once
-    __syncthreads();
+      __syncthreads();
+    }
   }
 
   // ------------------------------------------------------------- epilogue
