@@ -1,0 +1,83 @@
+"""Format-level unit tests: LMDB edge cases, SequenceFile sync markers,
+DataTransformer semantics."""
+
+import numpy as np
+import pytest
+import torch
+
+from caffeonspark_amd.data.lmdb_io import LmdbReader, LmdbWriter
+from caffeonspark_amd.data.seqfile import SequenceFileReader, \
+    SequenceFileWriter
+from caffeonspark_amd.data.transformer import DataTransformer
+from caffeonspark_amd.proto import caffe_pb
+
+
+def test_lmdb_empty(tmp_path):
+    p = str(tmp_path / "empty")
+    LmdbWriter(p).write([])
+    r = LmdbReader(p)
+    assert len(list(r.items())) == 0
+    assert len(r) == 0
+
+
+def test_lmdb_single_and_sorted(tmp_path):
+    p = str(tmp_path / "one")
+    items = [(b"zz", b"last"), (b"aa", b"first")]
+    LmdbWriter(p).write(items)
+    got = list(LmdbReader(p).items())
+    assert got == sorted(items)  # B+tree iterates in key order
+
+
+def test_lmdb_large_values_multilevel(tmp_path):
+    """Overflow pages + multi-level branch tree."""
+    rng = np.random.RandomState(3)
+    items = [(f"k{i:06d}".encode(), rng.bytes(6000)) for i in range(2000)]
+    p = str(tmp_path / "big")
+    LmdbWriter(p).write(items)
+    got = list(LmdbReader(p).items())
+    assert len(got) == 2000
+    assert got[1234] == sorted(items)[1234]
+
+
+def test_seqfile_sync_markers(tmp_path):
+    """Records spanning many sync intervals round-trip."""
+    p = str(tmp_path / "s.seq")
+    items = [(f"key{i}".encode(), bytes([i % 251]) * 300)
+             for i in range(100)]
+    with SequenceFileWriter(p) as w:
+        for k, v in items:
+            w.append(k, v)
+    r = SequenceFileReader(p)
+    assert r.key_class == "org.apache.hadoop.io.BytesWritable"
+    assert list(r.items()) == items
+
+
+def test_transformer_crop_and_mean():
+    tp = caffe_pb.TransformationParameter(
+        crop_size=4, mean_value=[10.0], scale=0.5)
+    xf = DataTransformer(tp, caffe_pb.Phase.TEST)  # center crop, no mirror
+    img = np.arange(36, dtype=np.uint8).reshape(6, 6, 1)
+    out = xf.transform_one(img)
+    assert out.shape == (1, 4, 4)
+    # center crop offset (6-4)//2 = 1; value = (pixel - 10) * 0.5
+    assert out[0, 0, 0] == pytest.approx((img[1, 1, 0] - 10.0) * 0.5)
+
+
+def test_transformer_mirror_deterministic_seed():
+    tp = caffe_pb.TransformationParameter(mirror=True)
+    a = DataTransformer(tp, caffe_pb.Phase.TRAIN, seed=3)
+    b = DataTransformer(tp, caffe_pb.Phase.TRAIN, seed=3)
+    img = np.random.RandomState(0).randint(0, 255, (5, 5, 3)).astype(
+        np.uint8)
+    for _ in range(5):
+        np.testing.assert_array_equal(a.transform_one(img),
+                                      b.transform_one(img))
+
+
+def test_transformer_batch_shape():
+    tp = caffe_pb.TransformationParameter()
+    xf = DataTransformer(tp, caffe_pb.Phase.TEST)
+    imgs = [np.zeros((8, 8, 3), dtype=np.uint8) for _ in range(4)]
+    out = xf.transform(imgs)
+    assert isinstance(out, torch.Tensor)
+    assert out.shape == (4, 3, 8, 8)
