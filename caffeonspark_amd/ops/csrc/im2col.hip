@@ -14,15 +14,15 @@ namespace cosamd {
 typedef unsigned short u16;
 typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
 
-// small-C variant (e.g. conv1, C=3): one thread per 8 output columns —
-// coalesced 16B writes; a per-block LDS lookup table kills the per-element
-// div/mod for k -> (dh, dw, c); gather reads hit L2/L3 (input is small).
+// column-major variant: one thread per 8 output columns — fully coalesced
+// 16-byte writes AND (for contiguous c-runs) coalesced reads; a per-block
+// LDS lookup table kills the per-element div/mod for k -> (dh, dw, c).
 __global__ void im2col_smallc_kernel(
     const u16* __restrict__ x, u16* __restrict__ col,
     int N, int H, int W, int C, int P, int Q,
     int R, int S, int sh, int sw, int ph, int pw,
     int dil, int Kpad, int c0, int Ct, int64_t total8) {
-  __shared__ int lut[1024];  // (dh<<20)|(dw<<10)|c, or -1 for pad columns
+  __shared__ int lut[2560];  // (dh<<20)|(dw<<10)|c, or -1 for pad columns
   int Kcol = R * S * Ct;
   for (int k = threadIdx.x; k < Kpad; k += blockDim.x) {
     if (k < Kcol) {
@@ -147,7 +147,7 @@ __global__ void col2im_nhwc_kernel(
 void im2col_nhwc(const void* x, void* col, int N, int H, int W, int C,
                  int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
                  int dil, int Kpad, int c0, int Ct, hipStream_t stream) {
-  if (Ct < 8 && Kpad <= 1024) {
+  if (Kpad <= 2560) {
     int64_t total8 = (int64_t)N * P * Q * (Kpad / 8);
     int b = hmin<int64_t>(8192, (total8 + 255) / 256);
     im2col_smallc_kernel<<<b, 256, 0, stream>>>(
