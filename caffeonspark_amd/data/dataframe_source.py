@@ -140,12 +140,15 @@ class CoSDataFrameSource(DataSource):
                 np.float32))
 
     def next_batch(self, device, dtype) -> Optional[List[torch.Tensor]]:
+        if getattr(self, "_drained", False):
+            return None
         rows = []
         while len(rows) < self._batch:
             item = self.take()
             if item is STOP_MARK:
                 if not rows:
                     return None
+                self._drained = True
                 break
             rows.append(item)
         out = []

@@ -69,12 +69,17 @@ class ImageDataSource(DataSource):
         return arr.reshape(c, h, w).transpose(1, 2, 0)
 
     def next_batch(self, device, dtype) -> Optional[List[torch.Tensor]]:
+        if getattr(self, "_drained", False):
+            return None
         images, labels = [], []
         while len(images) < self._batch:
             item = self.take()
             if item is STOP_MARK:
                 if not images:
                     return None
+                # partial final batch consumed the stop mark: remember so
+                # the NEXT call terminates instead of blocking forever
+                self._drained = True
                 break
             images.append(self._decode(item))
             labels.append(item.label)

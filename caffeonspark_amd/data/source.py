@@ -67,6 +67,7 @@ class DataSource:
 
     def reset_queue(self) -> None:
         self.queue = queue.Queue(maxsize=1024)
+        self._drained = False
 
     def next_batch(self, device, dtype) -> Optional[List[torch.Tensor]]:
         """Assemble one batch of tensors for the data layer (None at stop)."""

@@ -257,3 +257,20 @@ random_seed: 3
     snaps = [f for f in os.listdir(tmp_path) if f.endswith(".caffemodel")]
     assert snaps
     CaffeProcessor.reset_instance()
+
+
+def test_ml_pipeline_example(workdir):
+    """MyMLPipeline analog: DL features -> scikit-learn classifier."""
+    pytest.importorskip("sklearn")
+    import sys
+    sys.path.insert(0, os.path.join(ROOT, "examples"))
+    import ml_pipeline
+
+    CaffeProcessor.reset_instance()
+    snaps = [f for f in os.listdir(workdir) if f.endswith(".caffemodel")]
+    if not snaps:
+        pytest.skip("depends on test_train_and_test_via_facade snapshot")
+    ml_pipeline.main(["-conf", str(workdir / "solver.prototxt"),
+                      "-weights", str(workdir / sorted(snaps)[-1]),
+                      "-features", "ip1", "-label", "label"])
+    CaffeProcessor.reset_instance()
