@@ -108,6 +108,52 @@ __global__ void col_zero_pad_kernel(u16* __restrict__ col, int Kcol, int Kpad,
   }
 }
 
+// stride-1 vectorized col2im: one thread per 8 channels of one input
+// pixel — no divisions in the window walk (p = h+ph-r), 16-byte
+// coalesced dcol reads and dx writes.
+__global__ void col2im_s1_kernel(
+    const u16* __restrict__ dcol, u16* __restrict__ dx,
+    int N, int H, int W, int C, int P, int Q,
+    int R, int S, int ph, int pw, int Kpad, int c0, int Ct,
+    int64_t total8) {
+  int c8s = Ct / 8;
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total8; i8 += (int64_t)gridDim.x * blockDim.x) {
+    unsigned i = (unsigned)i8;
+    int cc = (i % c8s) * 8;
+    unsigned iw = i / c8s;
+    int w = iw % W;
+    unsigned ih = iw / W;
+    int h = ih % H;
+    int n = ih / H;
+    float acc[8] = {};
+    for (int r = 0; r < R; ++r) {
+      int p = h + ph - r;
+      if (p < 0 || p >= P) continue;
+      for (int s = 0; s < S; ++s) {
+        int q = w + pw - s;
+        if (q < 0 || q >= Q) continue;
+        int64_t npq = ((int64_t)n * P + p) * Q + q;
+        u16x8 v = *reinterpret_cast<const u16x8*>(
+            dcol + npq * Kpad + (r * S + s) * Ct + cc);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          u16 raw = v[j];
+          acc[j] += bf2f(*reinterpret_cast<const bf16*>(&raw));
+        }
+      }
+    }
+    u16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      bf16 b = f2bf(acc[j]);
+      out[j] = *reinterpret_cast<u16*>(&b);
+    }
+    *reinterpret_cast<u16x8*>(
+        dx + (((int64_t)n * H + h) * W + w) * C + c0 + cc) = out;
+  }
+}
+
 // dcol -> dx (gather, fp32 accumulate, bf16 out)
 __global__ void col2im_nhwc_kernel(
     const u16* __restrict__ dcol, u16* __restrict__ dx,
@@ -172,6 +218,14 @@ void im2col_nhwc(const void* x, void* col, int N, int H, int W, int C,
 void col2im_nhwc(const void* dcol, void* dx, int N, int H, int W, int C,
                  int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
                  int dil, int Kpad, int c0, int Ct, hipStream_t stream) {
+  if (sh == 1 && sw == 1 && dil == 1 && Ct % 8 == 0) {
+    int64_t total8 = (int64_t)N * H * W * (Ct / 8);
+    int b = hmin<int64_t>(8192, (total8 + 255) / 256);
+   hipLaunchKernelGGL(( col2im_s1_kernel), dim3(b), dim3(256), 0, stream, 
+        (const u16*)dcol, (u16*)dx, N, H, W, C, P, Q, R, S, ph, pw, Kpad,
+        c0, Ct, total8);
+    return;
+  }
   int64_t total = (int64_t)N * H * W * Ct;
   int blocks = hmin<int64_t>(8192, (total + 255) / 256);
  hipLaunchKernelGGL(( col2im_nhwc_kernel), dim3(blocks), dim3(256), 0, stream, 
