@@ -177,6 +177,89 @@ __global__ void maxpool_bwd_kernel(
   }
 }
 
+__global__ void avgpool_fwd8_kernel(
+    const u16* __restrict__ x, u16* __restrict__ y,
+    int N, int H, int W, int C, int P, int Q,
+    int kh, int kw, int sh, int sw, int ph, int pw, int64_t total8) {
+  int c8s = C / 8;
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total8; i8 += (int64_t)gridDim.x * blockDim.x) {
+    unsigned i = (unsigned)i8;
+    int cc = (i % c8s) * 8;
+    unsigned iq = i / c8s;
+    int q = iq % Q;
+    unsigned ip = iq / Q;
+    int p = ip % P;
+    int n = ip / P;
+    int h0 = p * sh - ph, w0 = q * sw - pw;
+    int hend = min(h0 + kh, H + ph), wend = min(w0 + kw, W + pw);
+    float inv = 1.f / ((hend - h0) * (wend - w0));
+    float acc[8] = {};
+    for (int h = max(h0, 0); h < min(hend, H); ++h)
+      for (int w = max(w0, 0); w < min(wend, W); ++w) {
+        u16x8 v = *reinterpret_cast<const u16x8*>(
+            x + (((int64_t)n * H + h) * W + w) * C + cc);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          u16 raw = v[j];
+          acc[j] += bf2f(*reinterpret_cast<const bf16*>(&raw));
+        }
+      }
+    u16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      bf16 b = f2bf(acc[j] * inv);
+      out[j] = *reinterpret_cast<u16*>(&b);
+    }
+    *reinterpret_cast<u16x8*>(
+        y + ((((int64_t)n * P + p) * Q + q) * C + cc)) = out;
+  }
+}
+
+__global__ void avgpool_bwd8_kernel(
+    const u16* __restrict__ dy, u16* __restrict__ dx,
+    int N, int H, int W, int C, int P, int Q,
+    int kh, int kw, int sh, int sw, int ph, int pw, int64_t total8) {
+  int c8s = C / 8;
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total8; i8 += (int64_t)gridDim.x * blockDim.x) {
+    unsigned i = (unsigned)i8;
+    int cc = (i % c8s) * 8;
+    unsigned iw = i / c8s;
+    int w = iw % W;
+    unsigned ih = iw / W;
+    int h = ih % H;
+    int n = ih / H;
+    float acc[8] = {};
+    int plo = max(0, (h + ph - kh + sh) / sh), phi = min(P - 1, (h + ph) / sh);
+    int qlo = max(0, (w + pw - kw + sw) / sw), qhi = min(Q - 1, (w + pw) / sw);
+    for (int p = plo; p <= phi; ++p) {
+      int h0 = p * sh - ph;
+      int hend = min(h0 + kh, H + ph);
+      for (int q = qlo; q <= qhi; ++q) {
+        int w0 = q * sw - pw;
+        int wend = min(w0 + kw, W + pw);
+        float inv = 1.f / ((hend - h0) * (wend - w0));
+        u16x8 v = *reinterpret_cast<const u16x8*>(
+            dy + ((((int64_t)n * P + p) * Q + q) * C + cc));
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          u16 raw = v[j];
+          acc[j] += bf2f(*reinterpret_cast<const bf16*>(&raw)) * inv;
+        }
+      }
+    }
+    u16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      bf16 b = f2bf(acc[j]);
+      out[j] = *reinterpret_cast<u16*>(&b);
+    }
+    *reinterpret_cast<u16x8*>(
+        dx + (((int64_t)n * H + h) * W + w) * C + cc) = out;
+  }
+}
+
 __global__ void avgpool_fwd_kernel(
     const u16* __restrict__ x, u16* __restrict__ y,
     int N, int H, int W, int C, int P, int Q,
@@ -284,6 +367,12 @@ void avgpool_fwd(const void* x, void* y, int N, int H, int W, int C,
                  int P, int Q, int kh, int kw, int sh, int sw, int ph, int pw,
                  hipStream_t stream) {
   int64_t total = (int64_t)N * P * Q * C;
+  if (C % 8 == 0) {
+    avgpool_fwd8_kernel<<<nblocks_for(total / 8), 256, 0, stream>>>(
+        (const u16*)x, (u16*)y, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw,
+        total / 8);
+    return;
+  }
   avgpool_fwd_kernel<<<nblocks_for(total), 256, 0, stream>>>(
       (const u16*)x, (u16*)y, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw,
       total);
@@ -293,6 +382,12 @@ void avgpool_bwd(const void* dy, void* dx, int N, int H, int W, int C,
                  int P, int Q, int kh, int kw, int sh, int sw, int ph, int pw,
                  hipStream_t stream) {
   int64_t total = (int64_t)N * H * W * C;
+  if (C % 8 == 0) {
+    avgpool_bwd8_kernel<<<nblocks_for(total / 8), 256, 0, stream>>>(
+        (const u16*)dy, (u16*)dx, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw,
+        total / 8);
+    return;
+  }
   avgpool_bwd_kernel<<<nblocks_for(total), 256, 0, stream>>>(
       (const u16*)dy, (u16*)dx, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw,
       total);
