@@ -47,7 +47,7 @@ def _splitk_for(mb: int, nb: int, K: int) -> int:
     tiles = mb * nb
     if tiles >= 1024 or K < 1024:
         return 1
-    target = int(os.environ.get("COS_SPLITK_TARGET", "896"))
+    target = int(os.environ.get("COS_SPLITK_TARGET", "448"))
     return max(1, min(K // 256, target // max(1, tiles)))
 
 
