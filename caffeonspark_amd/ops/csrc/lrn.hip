@@ -12,6 +12,7 @@
 namespace cosamd {
 
 typedef unsigned short u16;
+typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
 
 __device__ __forceinline__ float ld_bf(const u16* p) {
   return bf2f(*reinterpret_cast<const bf16*>(p));
@@ -26,30 +27,55 @@ __global__ void lrn_fwd_kernel(const u16* __restrict__ x,
                                float* __restrict__ scale,
                                int64_t total, int C, int half, float a_over_n,
                                float beta, float k) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total; i += (int64_t)gridDim.x * blockDim.x) {
-    int c = (int)(i % C);
-    int64_t base = i - c;
-    int lo = c - half > 0 ? c - half : 0;
-    int hi = c + half < C - 1 ? c + half : C - 1;
-    float win = 0.f;
-    for (int cc = lo; cc <= hi; ++cc) {
-      float v = ld_bf(x + base + cc);
-      win += v * v;
+  // one thread per 8 channels: vectorized loads, per-lane sliding window
+  int c8s = C / 8;
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total; i8 += (int64_t)gridDim.x * blockDim.x) {
+    int cc = (int)(i8 % c8s) * 8;
+    int64_t base = (i8 / c8s) * (int64_t)C;
+    float v[16];  // window neighborhood [cc-half, cc+7+half], half<=4
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      int c = cc - half + j;
+      v[j] = (c >= 0 && c < C && j < 8 + 2 * half)
+                 ? ld_bf(x + base + c) : 0.f;
     }
-    float sc = k + a_over_n * win;
-    scale[i] = sc;
-    st_bf(y + i, ld_bf(x + i) * __powf(sc, -beta));
+    u16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float win = 0.f;
+#pragma unroll
+      for (int t = 0; t <= 8; ++t)  // window size 2*half+1 <= 9
+        if (t <= 2 * half) win += v[j + t] * v[j + t];
+      float sc = k + a_over_n * win;
+      scale[base + cc + j] = sc;
+      bf16 b = f2bf(v[j + half] * __powf(sc, -beta));
+      out[j] = *reinterpret_cast<u16*>(&b);
+    }
+    *reinterpret_cast<u16x8*>(y + base + cc) = out;
   }
 }
 
 __global__ void lrn_ratio_kernel(const u16* __restrict__ y,
                                  const float* __restrict__ scale,
                                  const u16* __restrict__ dy,
-                                 u16* __restrict__ ratio, int64_t total) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total; i += (int64_t)gridDim.x * blockDim.x)
-    st_bf(ratio + i, ld_bf(dy + i) * ld_bf(y + i) / scale[i]);
+                                 u16* __restrict__ ratio, int64_t total8) {
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total8; i8 += (int64_t)gridDim.x * blockDim.x) {
+    int64_t i = i8 * 8;
+    u16x8 vy = *reinterpret_cast<const u16x8*>(y + i);
+    u16x8 vd = *reinterpret_cast<const u16x8*>(dy + i);
+    u16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      u16 ry = vy[j], rd = vd[j];
+      float f = bf2f(*reinterpret_cast<const bf16*>(&rd)) *
+                bf2f(*reinterpret_cast<const bf16*>(&ry)) / scale[i + j];
+      bf16 b = f2bf(f);
+      out[j] = *reinterpret_cast<u16*>(&b);
+    }
+    *reinterpret_cast<u16x8*>(ratio + i) = out;
+  }
 }
 
 __global__ void lrn_bwd_kernel(const u16* __restrict__ x,
@@ -57,42 +83,60 @@ __global__ void lrn_bwd_kernel(const u16* __restrict__ x,
                                const u16* __restrict__ dy,
                                const u16* __restrict__ ratio,
                                u16* __restrict__ dx,
-                               int64_t total, int C, int half,
+                               int64_t total8, int C, int half,
                                float beta, float ratio_coef) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < total; i += (int64_t)gridDim.x * blockDim.x) {
-    int c = (int)(i % C);
-    int64_t base = i - c;
-    int lo = c - half > 0 ? c - half : 0;
-    int hi = c + half < C - 1 ? c + half : C - 1;
-    float win = 0.f;
-    for (int cc = lo; cc <= hi; ++cc) win += ld_bf(ratio + base + cc);
-    float v = ld_bf(dy + i) * __powf(scale[i], -beta)
-              - ratio_coef * ld_bf(x + i) * win;
-    st_bf(dx + i, v);
+  int c8s = C / 8;
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total8; i8 += (int64_t)gridDim.x * blockDim.x) {
+    int cc = (int)(i8 % c8s) * 8;
+    int64_t base = (i8 / c8s) * (int64_t)C;
+    float r[16];
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      int c = cc - half + j;
+      r[j] = (c >= 0 && c < C && j < 8 + 2 * half)
+                 ? ld_bf(ratio + base + c) : 0.f;
+    }
+    u16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float win = 0.f;
+#pragma unroll
+      for (int t = 0; t <= 8; ++t)
+        if (t <= 2 * half) win += r[j + t];
+      int64_t i = base + cc + j;
+      float v = ld_bf(dy + i) * __powf(scale[i], -beta)
+                - ratio_coef * ld_bf(x + i) * win;
+      bf16 b = f2bf(v);
+      out[j] = *reinterpret_cast<u16*>(&b);
+    }
+    *reinterpret_cast<u16x8*>(dx + base + cc) = out;
   }
 }
 
 void lrn_fwd(const void* x, void* y, float* scale, int64_t npix, int C,
              int local_size, float alpha, float beta, float k,
              hipStream_t stream) {
-  int64_t total = npix * C;
-  int blocks = (int)hmin<int64_t>(4096, (total + 255) / 256);
+  // vectorized kernel requires C%8 and window radius <=4 (local_size<=9,
+  // true for AlexNet/GoogLeNet's local_size 5); callers with other
+  // configs run the CPU-parity fallback in ops/gpu.py
+  int64_t total8 = npix * (C / 8);
+  int blocks = (int)hmin<int64_t>(4096, (total8 + 255) / 256);
   lrn_fwd_kernel<<<blocks, 256, 0, stream>>>(
-      (const u16*)x, (u16*)y, scale, total, C, local_size / 2,
+      (const u16*)x, (u16*)y, scale, total8, C, local_size / 2,
       alpha / local_size, beta, k);
 }
 
 void lrn_bwd(const void* x, const void* y, const float* scale, const void* dy,
              void* dx, void* ratio, int64_t npix, int C, int local_size,
              float alpha, float beta, hipStream_t stream) {
-  int64_t total = npix * C;
-  int blocks = (int)hmin<int64_t>(4096, (total + 255) / 256);
+  int64_t total8 = npix * (C / 8);
+  int blocks = (int)hmin<int64_t>(4096, (total8 + 255) / 256);
   lrn_ratio_kernel<<<blocks, 256, 0, stream>>>(
-      (const u16*)y, scale, (const u16*)dy, (u16*)ratio, total);
+      (const u16*)y, scale, (const u16*)dy, (u16*)ratio, total8);
   lrn_bwd_kernel<<<blocks, 256, 0, stream>>>(
       (const u16*)x, scale, (const u16*)dy, (const u16*)ratio, (u16*)dx,
-      total, C, local_size / 2, beta, 2.f * alpha * beta / local_size);
+      total8, C, local_size / 2, beta, 2.f * alpha * beta / local_size);
 }
 
 }  // namespace cosamd

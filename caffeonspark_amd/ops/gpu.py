@@ -391,6 +391,9 @@ def global_avgpool_backward(x_shape, dy):
 def lrn_forward(x, local_size, alpha, beta, k):
     _check_bf16(x, "lrn input")
     N, C, H, W = x.shape
+    if C % 8 or local_size > 9:
+        # rare config the vectorized kernel doesn't cover: torch fallback
+        return reference.lrn_forward(x, local_size, alpha, beta, k)
     xl = _cl(x)
     y = torch.empty_like(xl)
     scale = torch.empty((N, H, W, C), dtype=torch.float32, device=x.device)
@@ -399,8 +402,11 @@ def lrn_forward(x, local_size, alpha, beta, k):
 
 
 def lrn_backward(x, y, scale_pack, dy, local_size, alpha, beta):
-    scale, xl = scale_pack
     N, C, H, W = x.shape
+    if C % 8 or local_size > 9:
+        return reference.lrn_backward(x, y, scale_pack, dy, local_size,
+                                      alpha, beta)
+    scale, xl = scale_pack
     dx = torch.empty_like(xl)
     ratio = torch.empty_like(xl)
     _ext.lrn_bwd(xl, _cl(y), scale, _cl(dy), dx, ratio, N * H * W, C,
