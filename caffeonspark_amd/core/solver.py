@@ -221,7 +221,12 @@ class Solver:
         wd = p.weight_decay
         reg = p.regularization_type
         if self.type == "SGD" and reg == "L2":
-            # fused per-segment update on the flat arenas
+            if self.device.type == "cuda":
+                # single fused kernel over all segments
+                fn = ops.gpu_op("sgd_update_multi_arena")
+                if fn is not None:
+                    fn(self, rate, p.momentum, wd)
+                    return
             for (off, n, lrm, dm) in self.segments:
                 if lrm == 0:
                     continue

@@ -222,6 +222,40 @@ void transpose_bf16(const void* in, void* out, int64_t R, int64_t C,
       (const u16*)in, (u16*)out, (int)R, (int)C, tiles_c);
 }
 
+// one launch for ALL (lr_mult, decay_mult) segments of the flat arena:
+// blocks are striped across segments proportionally to their size
+__global__ void sgd_update_multi_kernel(
+    float* __restrict__ p, const float* __restrict__ g,
+    float* __restrict__ v, const int64_t* __restrict__ seg_off,
+    const float* __restrict__ seg_lr, const float* __restrict__ seg_wd,
+    int nseg, float mu, int64_t total4) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total4; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t e0 = i * 4;
+    // find segment (nseg is small: linear scan)
+    int s = 0;
+    while (s + 1 < nseg && e0 >= seg_off[s + 1]) ++s;
+    float lr = seg_lr[s], wd = seg_wd[s];
+    if (lr == 0.f) continue;
+    int64_t end = seg_off[s + 1] < e0 + 4 ? seg_off[s + 1] : e0 + 4;
+    for (int64_t k = e0; k < end; ++k) {
+      float gg = g[k] + wd * p[k];
+      v[k] = mu * v[k] + lr * gg;
+      p[k] -= v[k];
+    }
+  }
+}
+
+void sgd_update_multi(float* p, const float* g, float* v,
+                      const int64_t* seg_off, const float* seg_lr,
+                      const float* seg_wd, int nseg, float mu,
+                      int64_t total, hipStream_t stream) {
+  int64_t total4 = (total + 3) / 4;
+  int blocks = (int)hmin<int64_t>(2048, (total4 + 255) / 256);
+  sgd_update_multi_kernel<<<blocks, 256, 0, stream>>>(
+      p, g, v, seg_off, seg_lr, seg_wd, nseg, mu, total4);
+}
+
 // ---------------------------------------------------------- column reduce
 // out[c] += sum_r in[r*ld + c]  (bias gradients; in bf16, out fp32)
 // Block = 256 threads as [4 row-groups x 64 cols]: coalesced 64-wide column

@@ -50,6 +50,10 @@ void mul_bf16(const void* a, const void* b, void* y, int64_t n,
               hipStream_t stream);
 void sgd_update(float* p, const float* g, float* v, float lr, float mu,
                 float wd, int64_t n, hipStream_t stream);
+void sgd_update_multi(float* p, const float* g, float* v,
+                      const int64_t* seg_off, const float* seg_lr,
+                      const float* seg_wd, int nseg, float mu,
+                      int64_t total, hipStream_t stream);
 void colsum(const void* in, float* out, int64_t rows, int cols, int ld,
             hipStream_t stream);
 void transpose_bf16(const void* in, void* out, int64_t R, int64_t C,
@@ -203,6 +207,18 @@ void py_mul(Tensor a, Tensor b, Tensor y) {
                    cur_stream());
 }
 
+void py_sgd_update_multi(Tensor p, Tensor g, Tensor v, Tensor seg_off,
+                         Tensor seg_lr, Tensor seg_wd, double mu) {
+  CHECK_F32(p); CHECK_F32(g); CHECK_F32(v);
+  cosamd::sgd_update_multi(p.data_ptr<float>(), g.data_ptr<float>(),
+                           v.data_ptr<float>(),
+                           seg_off.data_ptr<int64_t>(),
+                           seg_lr.data_ptr<float>(),
+                           seg_wd.data_ptr<float>(),
+                           (int)seg_lr.numel(), mu, p.numel(),
+                           cur_stream());
+}
+
 void py_sgd_update(Tensor p, Tensor g, Tensor v, double lr, double mu,
                    double wd) {
   CHECK_F32(p); CHECK_F32(g); CHECK_F32(v);
@@ -324,6 +340,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_fwd", &py_dropout_fwd);
   m.def("mul", &py_mul);
   m.def("sgd_update", &py_sgd_update);
+  m.def("sgd_update_multi", &py_sgd_update_multi);
   m.def("colsum", &py_colsum);
   m.def("transpose", &py_transpose);
   m.def("bias_act_cast", &py_bias_act_cast);

@@ -549,6 +549,25 @@ def sgd_update(param, grad, momentum_buf, lr, momentum, weight_decay):
                     weight_decay)
 
 
+def sgd_update_multi_arena(solver, rate, momentum, wd):
+    """One kernel over the whole flat arena, per-segment lr/decay."""
+    key = getattr(solver, "_seg_cache_key", None)
+    if key != (id(solver.flat_w), len(solver.segments)):
+        offs = [s[0] for s in solver.segments] + [int(solver.flat_w.numel())]
+        solver._seg_off = torch.tensor(offs, dtype=torch.int64,
+                                       device=solver.device)
+        solver._seg_lrm = torch.tensor([s[2] for s in solver.segments],
+                                       dtype=torch.float32,
+                                       device=solver.device)
+        solver._seg_dm = torch.tensor([s[3] for s in solver.segments],
+                                      dtype=torch.float32,
+                                      device=solver.device)
+        solver._seg_cache_key = (id(solver.flat_w), len(solver.segments))
+    _ext.sgd_update_multi(solver.flat_w, solver.flat_g, solver.flat_m,
+                          solver._seg_off, solver._seg_lrm * rate,
+                          solver._seg_dm * wd, momentum)
+
+
 # --------------------------------------------------------------- registry
 
 GLUE_OPS = [
