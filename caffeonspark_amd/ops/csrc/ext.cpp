@@ -16,6 +16,9 @@ void gemm_bf16(const void* A, const void* B, void* C, const float* bias,
                bool trans_a, bool trans_b, int store_mode, int splitk,
                bool relu, float alpha, int m_alloc, int n_alloc,
                hipStream_t stream);
+void im2col_t(const void* x, void* colT, int N, int H, int W, int C,
+              int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
+              int dil, int Kpad, int c0, int Cg, hipStream_t stream);
 void im2col_nhwc(const void* x, void* col, int N, int H, int W, int C,
                  int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
                  int dil, int Kpad, int c0, int Ct, hipStream_t stream);
@@ -328,6 +331,14 @@ void py_softmax_loss_bwd(Tensor prob, Tensor label, Tensor dx, double scale,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm", &py_gemm);
   m.def("im2col", &py_im2col);
+  m.def("im2col_t", [](Tensor x, Tensor colT, int64_t N, int64_t H,
+                       int64_t W, int64_t C, int64_t P, int64_t Q,
+                       int64_t R, int64_t S, int64_t sh, int64_t sw,
+                       int64_t ph, int64_t pw, int64_t dil, int64_t Kpad,
+                       int64_t c0, int64_t Cg) {
+    cosamd::im2col_t(x.data_ptr(), colT.data_ptr(), N, H, W, C, P, Q, R,
+                     S, sh, sw, ph, pw, dil, Kpad, c0, Cg, cur_stream());
+  });
   m.def("col2im", &py_col2im);
   m.def("maxpool_fwd", &py_maxpool_fwd);
   m.def("maxpool_bwd", &py_maxpool_bwd);

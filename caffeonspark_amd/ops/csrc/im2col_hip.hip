@@ -215,6 +215,58 @@ void im2col_nhwc(const void* x, void* col, int N, int H, int W, int C,
   }
 }
 
+// Transposed im2col for the dW backward GEMM's B operand: writes
+// colT[k][npq] (k = (r*S+s)*Cg + c, row stride NPQ) straight from the
+// NHWC input.  Replaces an explicit transpose of the forward col matrix:
+// the transpose reads + writes the full col matrix (the largest tensor in
+// a conv backward), while this reads only the R*S-times-smaller input
+// (L2-resident) and does the same coalesced writes.  Block = 32 npq-slots
+// x 8 k-rows; k tiles on blockIdx.x so the round-robin XCD dispatch gives
+// every XCD the same small input window to reuse.
+__global__ void im2col_t_kernel(
+    const u16* __restrict__ x, u16* __restrict__ out,
+    int N, int H, int W, int C, int P, int Q, int R, int S,
+    int sh, int sw, int ph, int pw, int dil, int c0, int Cg,
+    int64_t NPQ) {
+  int k = blockIdx.x * 8 + threadIdx.y;
+  int r = k / (S * Cg), rem = k % (S * Cg);
+  int s = rem / Cg, c = c0 + rem % Cg;
+  int64_t npq = (int64_t)blockIdx.y * 256 + threadIdx.x * 8;
+  if (npq >= NPQ) return;
+  int PQ = P * Q;
+  int n = (int)(npq / PQ), pq = (int)(npq % PQ);
+  int p = pq / Q, q = pq % Q;
+  u16x8 v;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    u16 val = 0;
+    if (npq + j < NPQ && r < R) {
+      int hi = p * sh - ph + r * dil;
+      int wi = q * sw - pw + s * dil;
+      if (hi >= 0 && hi < H && wi >= 0 && wi < W)
+        val = x[(((int64_t)n * H + hi) * W + wi) * C + c];
+    }
+    v[j] = val;
+    if (++q == Q) { q = 0; if (++p == P) { p = 0; ++n; } }
+  }
+  int64_t o = (int64_t)k * NPQ + npq;
+  if (npq + 8 <= NPQ && (o & 7) == 0)
+    *reinterpret_cast<u16x8*>(out + o) = v;
+  else
+    for (int j = 0; j < 8 && npq + j < NPQ; ++j) out[o + j] = v[j];
+}
+
+void im2col_t(const void* x, void* colT, int N, int H, int W, int C,
+              int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
+              int dil, int Kpad, int c0, int Cg, hipStream_t stream) {
+  int64_t NPQ = (int64_t)N * P * Q;
+  dim3 grid((Kpad + 7) / 8, (unsigned)((NPQ + 255) / 256));
+  dim3 block(32, 8);
+ hipLaunchKernelGGL(( im2col_t_kernel), dim3(grid), dim3(block), 0, stream, 
+      (const u16*)x, (u16*)colT, N, H, W, C, P, Q, R, S, sh, sw, ph, pw,
+      dil, c0, Cg, NPQ);
+}
+
 void col2im_nhwc(const void* dcol, void* dx, int N, int H, int W, int C,
                  int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
                  int dil, int Kpad, int c0, int Ct, hipStream_t stream) {

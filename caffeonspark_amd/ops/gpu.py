@@ -176,7 +176,17 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                       Kg, Kpad, NPQ, Kout, Kpad if not is_1x1 else C,
                       Kpad, True, True, 2, sk)
             else:
-                colT = _transpose(x2 if is_1x1 else col[g])
+                if is_1x1 or not int(os.environ.get("COS_DW_IM2COLT", "0")):
+                    colT = _transpose(x2 if is_1x1 else col[g])
+                else:
+                    # build colT[k][npq] straight from the input (skips a
+                    # pass over the col matrix but re-reads the input R*S
+                    # times across XCD L2s: measured slower on AlexNet —
+                    # 24.9k vs 27.1k img/s — so off by default)
+                    colT = torch.empty((_pad128(Kpad), NPQ),
+                                       dtype=torch.bfloat16, device=dy.device)
+                    _ext.im2col_t(ctx["xl"], colT, N, H, W, C, P, Q, R, S,
+                                  sh, sw, ph, pw, dil, Kpad, g * Cg, Cg)
                 _gemm(dyT[g * Kg:(g + 1) * Kg], colT, dwp[g * Kg:],
                       None, Kg, Kpad, NPQ, NPQ, NPQ, Kpad, False, False,
                       2, sk, ma=_pad128(Kout) - g * Kg, na=_pad128(Kpad))
