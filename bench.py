@@ -26,6 +26,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=0)
     ap.add_argument("--model", type=str, default="alexnet")
+    ap.add_argument("--graph", type=int, default=1,
+                    help="hipGraph-captured steps when single-process GPU")
     args = ap.parse_args()
 
     from caffeonspark_amd.core import solver_from_prototxt
@@ -89,12 +91,14 @@ def main():
         if use_gpu:
             torch.cuda.synchronize()
 
+    step = solver.graph_step if (args.graph and ws == 1 and use_gpu) \
+        else solver._step_one
     for _ in range(args.warmup):
-        solver._step_one()
+        step()
     barrier_sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        solver._step_one()
+        step()
     barrier_sync()
     elapsed = time.perf_counter() - t0
 

@@ -17,7 +17,8 @@ from ..proto import caffe_pb
 
 
 class Blob:
-    __slots__ = ("name", "data", "diff", "_lr_mult", "_decay_mult")
+    __slots__ = ("name", "data", "diff", "_lr_mult", "_decay_mult",
+                 "_loss_weight")
 
     def __init__(self, shape: Sequence[int] = (), *, name: str = "",
                  dtype: torch.dtype = torch.float32,

@@ -338,8 +338,10 @@ class SoftmaxWithLossLayer(Layer):
         if len(propagate_down) > 1 and propagate_down[1]:
             raise RuntimeError("SoftmaxWithLoss cannot backprop to labels")
         if propagate_down[0]:
-            lw = float(top[0].diff.reshape(-1)[0]) if top[0].diff is not None \
-                else 1.0
+            lw = getattr(top[0], "_loss_weight", None)
+            if lw is None:
+                lw = float(top[0].diff.reshape(-1)[0]) \
+                    if top[0].diff is not None else 1.0
             scale = lw / self._norm_denom(bottom[0].data)
             dx = ops.softmax_loss_backward(self._prob, bottom[1].data,
                                            self.ignore_label, scale, self.axis)

@@ -193,13 +193,16 @@ class Net:
             layer.reset(fed)
 
     # ---------------------------------------------------------------- forward
-    def forward(self) -> float:
+    def forward(self, read_loss: bool = True) -> float:
+        """read_loss=False skips the device->host loss readout (the only
+        sync in a forward), for hipGraph capture and sync-free stepping."""
         loss = 0.0
         for layer, bottoms, tops in zip(self.layers, self.layer_bottoms,
                                         self.layer_tops):
             layer.forward(bottoms, tops)
-        for (li, ti, w) in self._loss_tops:
-            loss += w * float(self.layer_tops[li][ti].data.float().sum())
+        if read_loss:
+            for (li, ti, w) in self._loss_tops:
+                loss += w * float(self.layer_tops[li][ti].data.float().sum())
         return loss
 
     # --------------------------------------------------------------- backward
@@ -211,6 +214,7 @@ class Net:
         for (li, ti, w) in self._loss_tops:
             top = self.layer_tops[li][ti]
             top.diff = torch.full_like(top.data, w, dtype=torch.float32)
+            top._loss_weight = w  # host-side copy: layers avoid a sync
         for i in range(len(self.layers) - 1, -1, -1):
             if not self.layer_need_backward[i]:
                 continue

@@ -166,29 +166,80 @@ class Solver:
             loss = self._step_one()
         return loss
 
-    def _step_one(self) -> float:
+    def _step_body(self, read_loss: bool = True) -> float:
+        """Device-side work of one iteration: zero grads, forward(s),
+        backward(s), fused update.  With read_loss=False this enqueues no
+        host sync and is hipGraph-capturable."""
         p = self.param
-        for cb in self.callbacks:
-            cb.on_start()
         self.net.zero_param_diffs()
         loss = 0.0
         iters = max(1, p.iter_size)
         for it in range(iters):
-            loss += self.net.forward()
+            loss += self.net.forward(read_loss=read_loss)
             # per-layer completion hook on the LAST micro-batch only, so
             # gradient all-reduce can overlap the rest of backward
             cb = self._on_layer_backward if it == iters - 1 else None
             self.net.backward(on_layer_done=cb)
-        loss /= iters
         for cb in self.callbacks:
-            cb.on_gradients_ready()
+            cb.on_gradients_ready()     # DDP: drain bucket all-reduces
         self.apply_update()
+        return loss / iters
+
+    def _step_one(self) -> float:
+        p = self.param
+        for cb in self.callbacks:
+            cb.on_start()
+        loss = self._step_body()
         self.iter += 1
         self._update_smoothed_loss(loss)
         if p.display and self.iter % p.display == 0 and self.rank == 0:
             print(f"[cos-amd] iter {self.iter} loss {self.smoothed_loss:.6f} "
                   f"lr {self.get_lr():.6g}", flush=True)
         return loss
+
+    # ------------------------------------------------------------ hipGraph
+    def graph_step(self) -> float:
+        """One iteration by hipGraph replay: the whole step (zero-grad,
+        forward, backward, fused SGD) is captured once and replayed with a
+        single launch, eliminating per-kernel submission gaps — the CDNA4
+        answer to launch-bound small-model steps (MI355X guide: capture
+        launch-bound inner loops in hipGraphs).
+
+        Falls back to the eager step when capture can't apply: CPU device,
+        distributed callbacks (RCCL hooks enqueue per-layer collectives),
+        gradient clipping (host-side norm read), or a display interval.
+        Recaptures when the lr policy changes the rate (kernel scalars are
+        frozen into the graph).  Assumes a device-resident data layer
+        (MemoryData/CoSData reset with GPU tensors)."""
+        p = self.param
+        if (self.device.type != "cuda" or self.callbacks
+                or p.clip_gradients > 0 or p.iter_size > 1 or p.display):
+            return self._step_one()
+        rate = self.get_lr()
+        if getattr(self, "_graph_rate", None) != rate:
+            try:
+                self._capture_graph(rate)
+            except RuntimeError:
+                self._graph_rate = None
+                return self._step_one()
+        self._graph.replay()
+        self.iter += 1
+        return self.smoothed_loss
+
+    def _capture_graph(self, rate: float) -> None:
+        torch.cuda.synchronize(self.device)
+        side = torch.cuda.Stream(self.device)
+        side.wait_stream(torch.cuda.current_stream(self.device))
+        with torch.cuda.stream(side):
+            for _ in range(2):       # allocator warmup (real, uncounted
+                self._step_body(read_loss=False)   # training steps)
+            self.iter += 2
+        torch.cuda.current_stream(self.device).wait_stream(side)
+        torch.cuda.synchronize(self.device)
+        self._graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph):
+            self._step_body(read_loss=False)
+        self._graph_rate = rate
 
     def _on_layer_backward(self, layer) -> None:
         for cb in self.callbacks:

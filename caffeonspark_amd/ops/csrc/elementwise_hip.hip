@@ -83,10 +83,18 @@ __device__ __forceinline__ unsigned hash_rng(uint64_t seed, uint64_t idx) {
   return (unsigned)(z >> 32);
 }
 
+// seed lives in device memory (advanced by seed_bump below) so a step
+// captured in a hipGraph draws a fresh mask on every replay
+__global__ void seed_bump_kernel(unsigned long long* s) {
+  *s = *s * 6364136223846793005ull + 1442695040888963407ull;
+}
+
 __global__ void dropout_fwd_kernel(const u16* __restrict__ x,
                                    u16* __restrict__ y, u16* __restrict__ mask,
-                                   float keep, float inv_keep, uint64_t seed,
+                                   float keep, float inv_keep,
+                                   const unsigned long long* __restrict__ seedp,
                                    int64_t n) {
+  uint64_t seed = *seedp;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i < n; i += (int64_t)gridDim.x * blockDim.x) {
     float u = (hash_rng(seed, i) >> 8) * (1.f / 16777216.f);
@@ -451,11 +459,15 @@ void relu_bwd(const void* y, const void* dy, void* dx, float slope, int64_t n,
       (const u16*)y, (const u16*)dy, (u16*)dx, slope, n8, n);
 }
 
+void seed_bump(void* s, hipStream_t stream) {
+ hipLaunchKernelGGL(( seed_bump_kernel), dim3(1), dim3(1), 0, stream, (unsigned long long*)s);
+}
+
 void dropout_fwd(const void* x, void* y, void* mask, float ratio,
-                 uint64_t seed, int64_t n, hipStream_t stream) {
+                 const void* seed, int64_t n, hipStream_t stream) {
   float keep = 1.f - ratio;
  hipLaunchKernelGGL(( dropout_fwd_kernel), dim3(nb(n)), dim3(256), 0, stream, 
-      (const u16*)x, (u16*)y, (u16*)mask, keep, 1.f / keep, seed, n);
+      (const u16*)x, (u16*)y, (u16*)mask, keep, 1.f / keep, (const unsigned long long*)seed, n);
 }
 
 void mul_bf16(const void* a, const void* b, void* y, int64_t n,

@@ -48,7 +48,8 @@ void relu_fwd(const void* x, void* y, float slope, int64_t n,
 void relu_bwd(const void* y, const void* dy, void* dx, float slope, int64_t n,
               hipStream_t stream);
 void dropout_fwd(const void* x, void* y, void* mask, float ratio,
-                 uint64_t seed, int64_t n, hipStream_t stream);
+                 const void* seed, int64_t n, hipStream_t stream);
+void seed_bump(void* s, hipStream_t stream);
 void mul_bf16(const void* a, const void* b, void* y, int64_t n,
               hipStream_t stream);
 void sgd_update(float* p, const float* g, float* v, float lr, float mu,
@@ -200,9 +201,9 @@ void py_relu_bwd(Tensor y, Tensor dy, Tensor dx, double slope) {
 }
 
 void py_dropout_fwd(Tensor x, Tensor y, Tensor mask, double ratio,
-                    int64_t seed) {
+                    Tensor seed) {
   cosamd::dropout_fwd(x.data_ptr(), y.data_ptr(), mask.data_ptr(), ratio,
-                      (uint64_t)seed, x.numel(), cur_stream());
+                      seed.data_ptr(), x.numel(), cur_stream());
 }
 
 void py_mul(Tensor a, Tensor b, Tensor y) {
@@ -349,6 +350,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_fwd", &py_relu_fwd);
   m.def("relu_bwd", &py_relu_bwd);
   m.def("dropout_fwd", &py_dropout_fwd);
+  m.def("seed_bump", [](Tensor s) {
+    cosamd::seed_bump(s.data_ptr(), cur_stream());
+  });
   m.def("mul", &py_mul);
   m.def("sgd_update", &py_sgd_update);
   m.def("sgd_update_multi", &py_sgd_update_multi);

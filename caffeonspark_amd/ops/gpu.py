@@ -304,10 +304,25 @@ def relu_backward(y, dy, negative_slope=0.0):
     return dx
 
 
+_dropout_seeds = {}
+
+
+def _dropout_seed(device):
+    """Per-device RNG seed in DEVICE memory, advanced on-GPU by seed_bump:
+    no host sync per call, and a hipGraph-captured step re-draws the mask
+    on every replay (a host kernel-arg seed would freeze it)."""
+    t = _dropout_seeds.get(device.index)
+    if t is None:
+        t = torch.randint(1, 2 ** 62, (1,), dtype=torch.int64, device=device)
+        _dropout_seeds[device.index] = t
+    return t
+
+
 def dropout_forward(x, ratio, generator=None):
     y = torch.empty_like(x)
     mask = torch.empty_like(x)
-    seed = int(torch.randint(0, 2 ** 62, (1,)).item())
+    seed = _dropout_seed(x.device)
+    _ext.seed_bump(seed)
     _ext.dropout_fwd(x, y, mask, ratio, seed)
     return y, mask
 

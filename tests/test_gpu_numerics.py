@@ -335,6 +335,42 @@ def test_lenet_step_parity():
             assert rel < 0.08, f"{cl.name} grad mismatch relL2={rel:.3f}"
 
 
+def test_graph_step_parity():
+    """hipGraph-captured solver steps == eager steps (same seed, same
+    data).  graph_step's capture performs 2 uncounted warmup steps, so
+    1 graph_step == 3 eager steps; 2 more replays -> 5 total."""
+    import os
+
+    from caffeonspark_amd.core import solver_from_prototxt
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    proto = os.path.join(root, "caffeonspark_amd", "models",
+                         "lenet_memory_solver.prototxt")
+    torch.manual_seed(7)
+    x = torch.randn(32, 1, 28, 28).to(dev(), torch.bfloat16)
+    y = torch.randint(0, 10, (32,)).float().to(dev())
+
+    def make():
+        s = solver_from_prototxt(proto, device=dev(), dtype=torch.bfloat16)
+        s.param.display = 0
+        dl = s.net.data_layers()[0]
+        dl.batch_size = 32
+        dl.reset(x, y)
+        return s
+
+    eager, graph = make(), make()
+    for _ in range(5):
+        eager._step_one()
+    graph.graph_step()            # capture: 2 warmup + 1 replay
+    graph.graph_step()
+    graph.graph_step()
+    assert graph.iter == 5
+    torch.cuda.synchronize()
+    rel = (eager.flat_w - graph.flat_w).norm() / \
+        eager.flat_w.norm().clamp_min(1e-4)
+    assert float(rel) < 1e-3, f"graph vs eager weight drift relL2={rel}"
+
+
 def test_inception_block_parity():
     """GoogLeNet inception wiring (1x1 bypass + concat) GPU vs CPU."""
     import os
