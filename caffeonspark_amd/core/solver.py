@@ -166,10 +166,12 @@ class Solver:
             loss = self._step_one()
         return loss
 
-    def _step_body(self, read_loss: bool = True) -> float:
+    def _step_body(self, read_loss: bool = True, rate=None) -> float:
         """Device-side work of one iteration: zero grads, forward(s),
         backward(s), fused update.  With read_loss=False this enqueues no
-        host sync and is hipGraph-capturable."""
+        host sync and is hipGraph-capturable; `rate` may be a device
+        scalar so a captured graph picks up lr-policy changes at replay
+        time instead of freezing the capture-time value."""
         p = self.param
         self.net.zero_param_diffs()
         loss = 0.0
@@ -182,7 +184,7 @@ class Solver:
             self.net.backward(on_layer_done=cb)
         for cb in self.callbacks:
             cb.on_gradients_ready()     # DDP: drain bucket all-reduces
-        self.apply_update()
+        self.apply_update(rate)
         return loss / iters
 
     def _step_one(self) -> float:
@@ -215,31 +217,39 @@ class Solver:
         if (self.device.type != "cuda" or self.callbacks
                 or p.clip_gradients > 0 or p.iter_size > 1 or p.display):
             return self._step_one()
-        rate = self.get_lr()
-        if getattr(self, "_graph_rate", None) != rate:
+        if self.type != "SGD" or p.regularization_type != "L2":
+            return self._step_one()    # fused-arena update path only
+        if getattr(self, "_graph", None) is None:
             try:
-                self._capture_graph(rate)
+                self._capture_graph()
             except RuntimeError:
-                self._graph_rate = None
+                self._graph = None
                 return self._step_one()
+        # lr-policy rate flows through a device scalar: one capture
+        # serves every iteration (poly/inv/step policies change the rate
+        # per step — recapturing each time would cost more than eager)
+        self._rate_dev.fill_(self.get_lr())
         self._graph.replay()
         self.iter += 1
         return self.smoothed_loss
 
-    def _capture_graph(self, rate: float) -> None:
+    def _capture_graph(self) -> None:
+        self._rate_dev = torch.zeros((), dtype=torch.float32,
+                                     device=self.device)
+        self._rate_dev.fill_(self.get_lr())
         torch.cuda.synchronize(self.device)
         side = torch.cuda.Stream(self.device)
         side.wait_stream(torch.cuda.current_stream(self.device))
         with torch.cuda.stream(side):
             for _ in range(2):       # allocator warmup (real, uncounted
-                self._step_body(read_loss=False)   # training steps)
+                self._step_body(read_loss=False,   # training steps)
+                                rate=self._rate_dev)
             self.iter += 2
         torch.cuda.current_stream(self.device).wait_stream(side)
         torch.cuda.synchronize(self.device)
         self._graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self._graph):
-            self._step_body(read_loss=False)
-        self._graph_rate = rate
+            self._step_body(read_loss=False, rate=self._rate_dev)
 
     def _on_layer_backward(self, layer) -> None:
         for cb in self.callbacks:
@@ -255,9 +265,10 @@ class Solver:
         self.smoothed_loss = sum(self._losses) / len(self._losses)
 
     # ----------------------------------------------------------------- update
-    def apply_update(self) -> None:
+    def apply_update(self, rate=None) -> None:
         p = self.param
-        rate = self.get_lr()
+        if rate is None:
+            rate = self.get_lr()
         # normalize for iter_size (and solver_count, reference
         # CaffeNet.cpp:620-625 gradient-scaling rule)
         scale = 1.0 / (max(1, p.iter_size) * self.solver_count)
