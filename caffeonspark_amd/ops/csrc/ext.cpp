@@ -16,6 +16,10 @@ void gemm_bf16(const void* A, const void* B, void* C, const float* bias,
                bool trans_a, bool trans_b, int store_mode, int splitk,
                bool relu, float alpha, int m_alloc, int n_alloc,
                hipStream_t stream);
+void wino_conv(const void* x, const float* w, const float* bias, void* y,
+               void* U, void* V, void* Mbuf, int N, int H, int W, int Cin,
+               int K, int wK, int wC, int u_rows_alloc, bool flip,
+               bool relu, hipStream_t stream);
 void im2col_t(const void* x, void* colT, int N, int H, int W, int C,
               int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
               int dil, int Kpad, int c0, int Cg, hipStream_t stream);
@@ -332,6 +336,17 @@ void py_softmax_loss_bwd(Tensor prob, Tensor label, Tensor dx, double scale,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm", &py_gemm);
   m.def("im2col", &py_im2col);
+  m.def("wino_conv", [](Tensor x, Tensor w, c10::optional<Tensor> bias,
+                        Tensor y, Tensor U, Tensor V, Tensor M, int64_t N,
+                        int64_t H, int64_t W, int64_t Cin, int64_t K,
+                        int64_t wK, int64_t wC, int64_t u_rows_alloc,
+                        bool flip, bool relu) {
+    cosamd::wino_conv(x.data_ptr(), w.data_ptr<float>(),
+                      bias ? bias->data_ptr<float>() : nullptr,
+                      y.data_ptr(), U.data_ptr(), V.data_ptr(),
+                      M.data_ptr(), N, H, W, Cin, K, wK, wC, u_rows_alloc,
+                      flip, relu, cur_stream());
+  });
   m.def("im2col_t", [](Tensor x, Tensor colT, int64_t N, int64_t H,
                        int64_t W, int64_t C, int64_t P, int64_t Q,
                        int64_t R, int64_t S, int64_t sh, int64_t sw,

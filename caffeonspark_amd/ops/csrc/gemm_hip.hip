@@ -209,7 +209,13 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     int M, int N, int K, int lda, int ldb, int ldc,
-    int ksplit, int relu, float alpha, int m_alloc, int n_alloc) {
+    int ksplit, int relu, float alpha, int m_alloc, int n_alloc,
+    int64_t sA, int64_t sB, int64_t sC) {
+  // batched operation (blockIdx.z = batch index, strides in elements for
+  // A/B and BYTES for C since its type depends on STORE_MODE)
+  A += (int64_t)blockIdx.z * sA;
+  B += (int64_t)blockIdx.z * sB;
+  C = (void*)((char*)C + (int64_t)blockIdx.z * sC);
   __shared__ bf16 Asb[2][BM * BK];
   __shared__ bf16 Bsb[2][BN * BK];
 
@@ -355,11 +361,12 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
 
 // ------------------------------------------------------------------- host
 
-void gemm_bf16(const void* A_, const void* B_, void* C, const float* bias,
-               int M, int N, int K, int lda, int ldb, int ldc,
-               bool trans_a, bool trans_b, int store_mode, int splitk,
-               bool relu, float alpha, int m_alloc, int n_alloc,
-               hipStream_t stream) {
+void gemm_bf16_batched(const void* A_, const void* B_, void* C,
+                       const float* bias, int M, int N, int K, int lda,
+                       int ldb, int ldc, bool trans_a, bool trans_b,
+                       int store_mode, int splitk, bool relu, float alpha,
+                       int m_alloc, int n_alloc, int batch, int64_t sA,
+                       int64_t sB, int64_t sC_bytes, hipStream_t stream) {
   if (m_alloc < M) m_alloc = M;
   if (n_alloc < N) n_alloc = N;
   const bf16* A = reinterpret_cast<const bf16*>(A_);
@@ -371,13 +378,13 @@ void gemm_bf16(const void* A_, const void* B_, void* C, const float* bias,
   int zblocks = (K + ksplit - 1) / ksplit;
   if (zblocks > 1 && store_mode != 2)
     throw std::runtime_error("gemm: split-K requires atomic store mode");
-  dim3 grid(mblocks * nblocks, zblocks);
+  dim3 grid(mblocks * nblocks, zblocks, batch);
   dim3 block(NTHREADS);
 
 #define COS_GEMM_CASE(TA, TB, SM)                                          \
  hipLaunchKernelGGL(( gemm_kernel<TA, TB, SM>), dim3(grid), dim3(block), 0, stream,                       \
       A, B, C, bias, M, N, K, lda, ldb, ldc, ksplit, relu ? 1 : 0, alpha,    \
-      m_alloc, n_alloc)
+      m_alloc, n_alloc, sA, sB, sC_bytes)
 
 #define COS_GEMM_SM(TA, TB)                                                 \
   do {                                                                      \
@@ -386,7 +393,7 @@ void gemm_bf16(const void* A_, const void* B_, void* C, const float* bias,
     else                      COS_GEMM_CASE(TA, TB, 2);                     \
   } while (0)
 
-  if (trans_a && trans_b && store_mode == 2 && N > 128) {
+  if (trans_a && trans_b && store_mode == 2 && N > 128 && batch == 1) {
     int nb_w = (N + WBN - 1) / WBN;
     dim3 gw(mblocks * nb_w, zblocks);
    hipLaunchKernelGGL(( gemm_tt_wide_kernel), dim3(gw), dim3(WNT), 0, stream, 
@@ -400,6 +407,16 @@ void gemm_bf16(const void* A_, const void* B_, void* C, const float* bias,
   else                           COS_GEMM_SM(true, true);
 #undef COS_GEMM_SM
 #undef COS_GEMM_CASE
+}
+
+void gemm_bf16(const void* A_, const void* B_, void* C, const float* bias,
+               int M, int N, int K, int lda, int ldb, int ldc,
+               bool trans_a, bool trans_b, int store_mode, int splitk,
+               bool relu, float alpha, int m_alloc, int n_alloc,
+               hipStream_t stream) {
+  gemm_bf16_batched(A_, B_, C, bias, M, N, K, lda, ldb, ldc, trans_a,
+                    trans_b, store_mode, splitk, relu, alpha, m_alloc,
+                    n_alloc, 1, 0, 0, 0, stream);
 }
 
 }  // namespace cosamd

@@ -1,0 +1,283 @@
+// Winograd F(2x2, 3x3) convolution for stride-1 pad-1 3x3 layers
+// (AlexNet conv3-5, GoogLeNet 3x3 towers): 16 MACs produce 4 outputs
+// instead of 36 — a 2.25x arithmetic reduction.  F(2,3) rather than
+// F(4,3) deliberately: F(4,3)'s inverse-transform coefficients reach
+// 8x8 and amplify bf16 quantization of the staged V/M matrices ~20x
+// (measured relL2 ~0.1 vs fp32 — outside training tolerance), while
+// F(2,3)'s coefficients are <=2 and stay within bf16 training noise.
+//
+// Pipeline per conv (all device, launched from the host fn at the end):
+//   U[16][K][C]  = G w Gt          (filter transform, fp32 weights in)
+//   V[16][T][C]  = Bt d B          (input transform, NHWC bf16, T tiles)
+//   M[16][T][K]  = V[i] @ U[i]^T   (ONE batched NT-direct GEMM, grid.z=16)
+//   y tiles      = At M A  + bias/ReLU   (output transform, NHWC bf16)
+// dgrad reuses the same machinery with the filter transform flipped 180
+// and K<->C swapped (U'[16][C][K]), applied to dy.
+//
+// Reference obligation: SURVEY.md §3.6 convolution row (the reference
+// uses im2col+cublasSgemm; Winograd is the "rebuild addition" named
+// there).  Transform matrices are the standard F(2,3) set (Lavin &
+// Gray, "Fast Algorithms for Convolutional Neural Networks").
+
+#include "common.h"
+
+namespace cosamd {
+
+typedef unsigned short u16;
+typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
+
+__device__ __forceinline__ float ldbf(const u16* p) {
+  return bf2f(*reinterpret_cast<const bf16*>(p));
+}
+__device__ __forceinline__ void stbf(u16* p, float v) {
+  bf16 b = f2bf(v);
+  *p = *reinterpret_cast<u16*>(&b);
+}
+
+void gemm_bf16_batched(const void* A_, const void* B_, void* C,
+                       const float* bias, int M, int N, int K, int lda,
+                       int ldb, int ldc, bool trans_a, bool trans_b,
+                       int store_mode, int splitk, bool relu, float alpha,
+                       int m_alloc, int n_alloc, int batch, int64_t sA,
+                       int64_t sB, int64_t sC_bytes, hipStream_t stream);
+
+// ---- 1-D transform helpers (applied twice for the 2-D transforms) ----
+
+// Bt (4x4) applied to a 4-vector
+__device__ __forceinline__ void wino_bt4(const float* in, float* out) {
+  out[0] = in[0] - in[2];
+  out[1] = in[1] + in[2];
+  out[2] = in[2] - in[1];
+  out[3] = in[1] - in[3];
+}
+
+// G (4x3) applied to a 3-vector
+__device__ __forceinline__ void wino_g3(const float* in, float* out) {
+  out[0] = in[0];
+  out[1] = 0.5f * (in[0] + in[1] + in[2]);
+  out[2] = 0.5f * (in[0] - in[1] + in[2]);
+  out[3] = in[2];
+}
+
+// At (2x4) applied to a 4-vector
+__device__ __forceinline__ void wino_at4(const float* in, float* out) {
+  out[0] = in[0] + in[1] + in[2];
+  out[1] = in[1] - in[2] - in[3];
+}
+
+// ------------------------------------------------------ filter transform
+// w fp32 [K][C][3][3] -> U bf16 [16][rows][cols]:
+//   fwd:   rows=K cols=C (FLIP=0)  — B operand of M = V @ U^T
+//   dgrad: rows=C cols=K (FLIP=1)  — filter rotated 180, K<->C swapped
+template <int FLIP>
+__global__ void wino_filter_kernel(const float* __restrict__ w,
+                                   u16* __restrict__ U, int K, int C,
+                                   int64_t rstride /* rows*cols */) {
+  int64_t id = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (id >= (int64_t)K * C) return;
+  // fastest index = cols of U for coalesced writes
+  int col = (int)(id % (FLIP ? K : C));
+  int row = (int)(id / (FLIP ? K : C));
+  int k = FLIP ? col : row, c = FLIP ? row : col;
+  const float* wp = w + ((int64_t)k * C + c) * 9;
+  float g[3][3];
+#pragma unroll
+  for (int r = 0; r < 3; ++r)
+#pragma unroll
+    for (int s = 0; s < 3; ++s)
+      g[r][s] = FLIP ? wp[(2 - r) * 3 + (2 - s)] : wp[r * 3 + s];
+  float t[4][3], u[4][4], tmp[4];
+#pragma unroll
+  for (int s = 0; s < 3; ++s) {     // G applied to columns
+    float colv[3] = {g[0][s], g[1][s], g[2][s]};
+    wino_g3(colv, tmp);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) t[r][s] = tmp[r];
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r)       // G applied to rows
+    wino_g3(t[r], u[r]);
+  int cols = FLIP ? K : C;
+#pragma unroll
+  for (int i = 0; i < 16; ++i)
+    stbf(U + (int64_t)i * rstride + (int64_t)row * cols + col,
+         u[i / 4][i % 4]);
+}
+
+// ------------------------------------------------------- input transform
+// NHWC bf16 x [N][H][W][Cin] -> V bf16 [16][T][Cin], T = N*th*tw tiles of
+// 2x2 outputs (4x4 inputs, stride 2, pad 1).  One thread = one tile x 8
+// channels; c8 is the fastest tid dimension so the 16-byte loads/stores
+// coalesce across the wavefront.
+__global__ void wino_input_kernel(const u16* __restrict__ x,
+                                  u16* __restrict__ V, int N, int H, int W,
+                                  int C, int th, int tw, int64_t T) {
+  int64_t id = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int nc8 = C / 8;
+  if (id >= T * nc8) return;
+  int c0 = (int)(id % nc8) * 8;
+  int64_t tile = id / nc8;
+  int txy = (int)(tile % (th * tw));
+  int n = (int)(tile / (th * tw));
+  int ty = txy / tw, tx = txy % tw;
+  int h0 = ty * 2 - 1, w0 = tx * 2 - 1;
+
+  float d[4][4][8];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int h = h0 + r;
+    bool hin = (h >= 0 && h < H);
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      int ww = w0 + s;
+      if (hin && ww >= 0 && ww < W) {
+        const u16x8 v = *reinterpret_cast<const u16x8*>(
+            x + (((int64_t)n * H + h) * W + ww) * C + c0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) d[r][s][j] = ldbf((const u16*)&v + j);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) d[r][s][j] = 0.f;
+      }
+    }
+  }
+  float out[4][4][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    float e[4][4];
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {   // Bt on columns
+      float colv[4] = {d[0][s][j], d[1][s][j], d[2][s][j], d[3][s][j]};
+      float outv[4];
+      wino_bt4(colv, outv);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) e[r][s] = outv[r];
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {   // Bt on rows
+      float outv[4];
+      wino_bt4(e[r], outv);
+#pragma unroll
+      for (int s = 0; s < 4; ++s) out[r][s][j] = outv[s];
+    }
+  }
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    u16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) stbf((u16*)&v + j, out[i / 4][i % 4][j]);
+    *reinterpret_cast<u16x8*>(V + ((int64_t)i * T + tile) * C + c0) = v;
+  }
+}
+
+// ------------------------------------------------------ output transform
+// M bf16 [16][T][K] -> y NHWC bf16 [N][P][Q][K] (+bias/ReLU), guarding
+// the ragged right/bottom tiles.  One thread = one tile x 8 filters.
+__global__ void wino_output_kernel(const u16* __restrict__ Mt,
+                                   u16* __restrict__ y,
+                                   const float* __restrict__ bias,
+                                   int N, int P, int Q, int K,
+                                   int th, int tw, int64_t T, int relu) {
+  int64_t id = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int nk8 = K / 8;
+  if (id >= T * nk8) return;
+  int k0 = (int)(id % nk8) * 8;
+  int64_t tile = id / nk8;
+  int txy = (int)(tile % (th * tw));
+  int n = (int)(tile / (th * tw));
+  int ty = txy / tw, tx = txy % tw;
+
+  float m[4][4][8];
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    const u16x8 v = *reinterpret_cast<const u16x8*>(
+        Mt + ((int64_t)i * T + tile) * K + k0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) m[i / 4][i % 4][j] = ldbf((const u16*)&v + j);
+  }
+  float b8[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) b8[j] = bias ? bias[k0 + j] : 0.f;
+
+  float o[2][2][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    float e[2][4];
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {   // At on columns
+      float colv[4] = {m[0][s][j], m[1][s][j], m[2][s][j], m[3][s][j]};
+      float outv[2];
+      wino_at4(colv, outv);
+      e[0][s] = outv[0];
+      e[1][s] = outv[1];
+    }
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {   // At on rows
+      float outv[2];
+      wino_at4(e[r], outv);
+#pragma unroll
+      for (int q = 0; q < 2; ++q) {
+        float v = outv[q] + b8[j];
+        if (relu && v < 0.f) v = 0.f;
+        o[r][q][j] = v;
+      }
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < 2; ++r) {
+    int p = ty * 2 + r;
+    if (p >= P) break;
+#pragma unroll
+    for (int q2 = 0; q2 < 2; ++q2) {
+      int q = tx * 2 + q2;
+      if (q >= Q) break;
+      u16x8 v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) stbf((u16*)&v + j, o[r][q2][j]);
+      *reinterpret_cast<u16x8*>(
+          y + (((int64_t)n * P + p) * Q + q) * K + k0) = v;
+    }
+  }
+}
+
+// ------------------------------------------------------------- host side
+
+// x NHWC bf16 [N][H][W][Cin] conv w fp32 [K][Cin][3][3] (s=1 p=1) -> y
+// NHWC bf16 [N][H][W][K].  U/V/M are caller-allocated workspaces:
+//   U bf16 [16][u_rows_alloc][u_cols],  V bf16 [16][T][Cin],
+//   M bf16 [16][T][K].
+// flip=true computes the data-gradient form (weights rotated, K<->C swap
+// — then "Cin" is the forward K and "K" is the forward C).
+void wino_conv(const void* x, const float* w, const float* bias, void* y,
+               void* U, void* V, void* Mbuf, int N, int H, int W, int Cin,
+               int K, int wK, int wC, int u_rows_alloc, bool flip,
+               bool relu, hipStream_t stream) {
+  int th = (H + 1) / 2, tw = (W + 1) / 2;
+  int64_t T = (int64_t)N * th * tw;
+  int64_t rstride = (int64_t)u_rows_alloc * (flip ? wK : wC);
+
+  int64_t nf = (int64_t)wK * wC;
+  int fb = (int)((nf + 255) / 256);
+  if (flip)
+    wino_filter_kernel<1><<<fb, 256, 0, stream>>>(w, (u16*)U, wK, wC,
+                                                  rstride);
+  else
+    wino_filter_kernel<0><<<fb, 256, 0, stream>>>(w, (u16*)U, wK, wC,
+                                                  rstride);
+
+  int64_t ni = T * (Cin / 8);
+  wino_input_kernel<<<(int)((ni + 255) / 256), 256, 0, stream>>>(
+      (const u16*)x, (u16*)V, N, H, W, Cin, th, tw, T);
+
+  gemm_bf16_batched(V, U, Mbuf, nullptr, (int)T, K, Cin, Cin, Cin, K,
+                    false, false, 0, 1, false, 1.f, (int)T, u_rows_alloc,
+                    16, T * Cin, rstride, T * K * 2 /* bf16 bytes */,
+                    stream);
+
+  int64_t no = T * (K / 8);
+  wino_output_kernel<<<(int)((no + 255) / 256), 256, 0, stream>>>(
+      (const u16*)Mbuf, (u16*)y, bias, N, H, W, K, th, tw, T,
+      relu ? 1 : 0);
+}
+
+}  // namespace cosamd
