@@ -116,7 +116,11 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
     # still uses the im2col form).
     if (R == S == 3 and sh == sw == 1 and ph == pw == 1 and dil == 1
             and G == 1 and C % 8 == 0 and Kout % 8 == 0
-            and int(os.environ.get("COS_WINOGRAD", "1"))):
+            and int(os.environ.get("COS_WINOGRAD", "0"))):
+        # measured on MI355X: ~1-5% slower than the tuned direct im2col
+        # GEMM on AlexNet/GoogLeNet (these shapes are staging-bound, so
+        # the 2.25x MAC cut loses to the V/M transform traffic); kept as
+        # an option for MFMA-bound shapes
         _wino_run(xl, w, b, y, N, P, Q, C, Kout, relu)
         if ctx is not None:
             ctx["col"] = None

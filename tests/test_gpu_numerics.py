@@ -335,6 +335,30 @@ def test_lenet_step_parity():
             assert rel < 0.08, f"{cl.name} grad mismatch relL2={rel:.3f}"
 
 
+def test_winograd_conv_parity(monkeypatch):
+    """F(2x2,3x3) Winograd path (COS_WINOGRAD=1) vs the direct im2col
+    path: fwd/dx/dw/db must agree within bf16 training noise."""
+    import os as _os
+    from caffeonspark_amd.ops import gpu as g
+
+    torch.manual_seed(5)
+    x = bf(torch.randn(2, 32, 13, 13)).to(dev())
+    wt = (torch.randn(64, 32, 3, 3) * 0.1).to(dev()).float()
+    b = torch.randn(64).to(dev()).float()
+    dy = bf(torch.randn(2, 64, 13, 13)).to(dev())
+    outs = {}
+    for flag in ("1", "0"):
+        monkeypatch.setenv("COS_WINOGRAD", flag)
+        ctx = {}
+        y = g.conv2d_forward(x, wt, b, (1, 1), (1, 1), (1, 1), 1, ctx=ctx)
+        assert ctx.get("wino", False) == (flag == "1")
+        outs[flag] = (y,) + g.conv2d_backward(x, wt, dy, (1, 1), (1, 1),
+                                              (1, 1), 1, ctx=ctx)
+    for name, a, c in zip(("y", "dx", "dw", "db"), outs["1"], outs["0"]):
+        rel = (a.float() - c.float()).norm() / c.float().norm().clamp_min(1e-5)
+        assert float(rel) < 0.02, f"winograd {name} relL2={float(rel):.4f}"
+
+
 def test_graph_step_parity():
     """hipGraph-captured solver steps == eager steps (same seed, same
     data).  graph_step's capture performs 2 uncounted warmup steps, so
