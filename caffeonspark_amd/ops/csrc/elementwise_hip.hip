@@ -233,24 +233,47 @@ void transpose_bf16(const void* in, void* out, int64_t R, int64_t C,
 
 // one launch for ALL (lr_mult, decay_mult) segments of the flat arena:
 // blocks are striped across segments proportionally to their size
+// segment table cached in LDS (a per-quad global scan serializes on
+// dependent loads: measured 241us -> bandwidth-bound after this)
+#define COS_SGD_MAX_SEG 512
 __global__ void sgd_update_multi_kernel(
     float* __restrict__ p, const float* __restrict__ g,
     float* __restrict__ v, const int64_t* __restrict__ seg_off,
     const float* __restrict__ seg_lr, const float* __restrict__ seg_wd,
     int nseg, float mu, int64_t total4) {
+  __shared__ int64_t soff[COS_SGD_MAX_SEG + 1];
+  __shared__ float slr[COS_SGD_MAX_SEG], swd[COS_SGD_MAX_SEG];
+  for (int i = threadIdx.x; i <= nseg; i += blockDim.x) {
+    soff[i] = seg_off[i];
+    if (i < nseg) { slr[i] = seg_lr[i]; swd[i] = seg_wd[i]; }
+  }
+  __syncthreads();
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i < total4; i += (int64_t)gridDim.x * blockDim.x) {
     int64_t e0 = i * 4;
-    // find segment (nseg is small: linear scan)
     int s = 0;
-    while (s + 1 < nseg && e0 >= seg_off[s + 1]) ++s;
-    float lr = seg_lr[s], wd = seg_wd[s];
+    while (s + 1 < nseg && e0 >= soff[s + 1]) ++s;
+    float lr = slr[s], wd = swd[s];
     if (lr == 0.f) continue;
-    int64_t end = seg_off[s + 1] < e0 + 4 ? seg_off[s + 1] : e0 + 4;
-    for (int64_t k = e0; k < end; ++k) {
-      float gg = g[k] + wd * p[k];
-      v[k] = mu * v[k] + lr * gg;
-      p[k] -= v[k];
+    if (soff[s + 1] >= e0 + 4) {         // interior quad: vector path
+      f32x4 pv = *reinterpret_cast<f32x4*>(p + e0);
+      f32x4 gv = *reinterpret_cast<const f32x4*>(g + e0);
+      f32x4 vv = *reinterpret_cast<f32x4*>(v + e0);
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float gg = gv[k] + wd * pv[k];
+        vv[k] = mu * vv[k] + lr * gg;
+        pv[k] -= vv[k];
+      }
+      *reinterpret_cast<f32x4*>(v + e0) = vv;
+      *reinterpret_cast<f32x4*>(p + e0) = pv;
+    } else {
+      int64_t end = soff[s + 1];
+      for (int64_t k = e0; k < end; ++k) {
+        float gg = g[k] + wd * p[k];
+        v[k] = mu * v[k] + lr * gg;
+        p[k] -= v[k];
+      }
     }
   }
 }
@@ -271,25 +294,43 @@ void sgd_update_multi(float* p, const float* g, float* v,
 // reads, deep grid.x row parallelism, LDS-reduced partials, one atomic per
 // (block, col).
 
+// 8-wide column reduce: each thread owns 8 consecutive bf16 columns
+// (one 16-byte load per row), 8 row-groups per block, LDS tree reduce.
 __global__ void colsum_kernel(const u16* __restrict__ in, float* __restrict__ out,
                               int64_t rows, int cols, int ld) {
-  __shared__ float part[4][64];
-  int lane_c = threadIdx.x & 63;
-  int rg = threadIdx.x >> 6;             // 0..3
-  for (int c0 = 0; c0 < cols; c0 += 64) {
-    int c = c0 + lane_c;
-    float acc = 0.f;
-    if (c < cols) {
-      for (int64_t r = (int64_t)blockIdx.x * 4 + rg; r < rows;
-           r += (int64_t)gridDim.x * 4)
-        acc += ldbf(in + r * ld + c);
+  __shared__ float part[8][256];
+  int c8 = threadIdx.x & 31;             // 32 col-groups x 8 = 256 cols
+  int rg = threadIdx.x >> 5;             // 0..7 row groups
+  for (int c0 = 0; c0 < cols; c0 += 256) {
+    int c = c0 + c8 * 8;
+    float acc[8] = {};
+    if (c + 8 <= cols) {
+      for (int64_t r = (int64_t)blockIdx.x * 8 + rg; r < rows;
+           r += (int64_t)gridDim.x * 8) {
+        u16x8 v = *reinterpret_cast<const u16x8*>(in + r * ld + c);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += ldbf((const u16*)&v + j);
+      }
+    } else if (c < cols) {
+      for (int64_t r = (int64_t)blockIdx.x * 8 + rg; r < rows;
+           r += (int64_t)gridDim.x * 8)
+        for (int j = 0; c + j < cols; ++j)
+          acc[j] += ldbf(in + r * ld + c + j);
     }
-    part[rg][lane_c] = acc;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) part[rg][c8 * 8 + j] = acc[j];
     __syncthreads();
-    if (rg == 0 && c < cols) {
-      float s = part[0][lane_c] + part[1][lane_c] + part[2][lane_c] +
-                part[3][lane_c];
-      atomicAdd(out + c, s);
+    if (threadIdx.x < 64) {              // tree-reduce 256 LDS columns
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        int col = threadIdx.x * 4 + t;
+        if (c0 + col < cols) {
+          float sum = 0.f;
+#pragma unroll
+          for (int g = 0; g < 8; ++g) sum += part[g][col];
+          atomicAdd(out + c0 + col, sum);
+        }
+      }
     }
     __syncthreads();
   }
@@ -487,7 +528,7 @@ void colsum(const void* in, float* out, int64_t rows, int cols, int ld,
   // scale the grid to the actual element count so small reductions don't
   // dispatch thousands of idle blocks (then atomic-sum their zeros)
   int64_t work_blocks = (rows * hmax<int64_t>(cols, 1)) / (256 * 8) + 1;
-  int blocks = (int)hmin<int64_t>(hmin<int64_t>((rows + 3) / 4, 2048),
+  int blocks = (int)hmin<int64_t>(hmin<int64_t>((rows + 7) / 8, 2048),
                                   work_blocks);
  hipLaunchKernelGGL(( colsum_kernel), dim3(blocks), dim3(256), 0, stream, (const u16*)in, out, rows, cols,
                                             ld);

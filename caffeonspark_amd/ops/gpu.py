@@ -633,6 +633,15 @@ def sgd_update(param, grad, momentum_buf, lr, momentum, weight_decay):
 
 def sgd_update_multi_arena(solver, rate, momentum, wd):
     """One kernel over the whole flat arena, per-segment lr/decay."""
+    if len(solver.segments) > 512:      # LDS segment-table capacity
+        for (off, n, lrm, dm) in solver.segments:
+            if lrm == 0:
+                continue
+            _ext.sgd_update(solver.flat_w.narrow(0, off, n),
+                            solver.flat_g.narrow(0, off, n),
+                            solver.flat_m.narrow(0, off, n),
+                            float(rate) * lrm, momentum, wd * dm)
+        return
     key = getattr(solver, "_seg_cache_key", None)
     if key != (id(solver.flat_w), len(solver.segments)):
         offs = [s[0] for s in solver.segments] + [int(solver.flat_w.numel())]
