@@ -294,42 +294,44 @@ void sgd_update_multi(float* p, const float* g, float* v,
 // reads, deep grid.x row parallelism, LDS-reduced partials, one atomic per
 // (block, col).
 
-// 8-wide column reduce: each thread owns 8 consecutive bf16 columns
-// (one 16-byte load per row), 8 row-groups per block, LDS tree reduce.
+// 8-wide column reduce with adaptive block shape: bx col-groups (8
+// bf16 columns = one 16-byte load each) x by row-groups, bx sized to
+// the matrix so narrow matrices (conv bias grads, cols 96-384) still
+// keep every lane loading 16 bytes.  LDS tree reduce, one atomicAdd
+// per column per block.
 __global__ void colsum_kernel(const u16* __restrict__ in, float* __restrict__ out,
-                              int64_t rows, int cols, int ld) {
-  __shared__ float part[8][256];
-  int c8 = threadIdx.x & 31;             // 32 col-groups x 8 = 256 cols
-  int rg = threadIdx.x >> 5;             // 0..7 row groups
-  for (int c0 = 0; c0 < cols; c0 += 256) {
+                              int64_t rows, int cols, int ld, int bx) {
+  __shared__ float part[2048];           // by * bx * 8 <= 256*8
+  int by = blockDim.x / bx;
+  int c8 = threadIdx.x % bx;
+  int rg = threadIdx.x / bx;             // 0..by-1
+  for (int c0 = 0; c0 < cols; c0 += bx * 8) {
     int c = c0 + c8 * 8;
     float acc[8] = {};
-    if (c + 8 <= cols) {
-      for (int64_t r = (int64_t)blockIdx.x * 8 + rg; r < rows;
-           r += (int64_t)gridDim.x * 8) {
-        u16x8 v = *reinterpret_cast<const u16x8*>(in + r * ld + c);
+    if (rg < by) {
+      if (c + 8 <= cols) {
+        for (int64_t r = (int64_t)blockIdx.x * by + rg; r < rows;
+             r += (int64_t)gridDim.x * by) {
+          u16x8 v = *reinterpret_cast<const u16x8*>(in + r * ld + c);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) acc[j] += ldbf((const u16*)&v + j);
-      }
-    } else if (c < cols) {
-      for (int64_t r = (int64_t)blockIdx.x * 8 + rg; r < rows;
-           r += (int64_t)gridDim.x * 8)
-        for (int j = 0; c + j < cols; ++j)
-          acc[j] += ldbf(in + r * ld + c + j);
-    }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) part[rg][c8 * 8 + j] = acc[j];
-    __syncthreads();
-    if (threadIdx.x < 64) {              // tree-reduce 256 LDS columns
-#pragma unroll
-      for (int t = 0; t < 4; ++t) {
-        int col = threadIdx.x * 4 + t;
-        if (c0 + col < cols) {
-          float sum = 0.f;
-#pragma unroll
-          for (int g = 0; g < 8; ++g) sum += part[g][col];
-          atomicAdd(out + c0 + col, sum);
+          for (int j = 0; j < 8; ++j) acc[j] += ldbf((const u16*)&v + j);
         }
+      } else if (c < cols) {
+        for (int64_t r = (int64_t)blockIdx.x * by + rg; r < rows;
+             r += (int64_t)gridDim.x * by)
+          for (int j = 0; c + j < cols; ++j)
+            acc[j] += ldbf(in + r * ld + c + j);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) part[rg * bx * 8 + c8 * 8 + j] = acc[j];
+    }
+    __syncthreads();
+    int pass_cols = bx * 8;
+    for (int col = threadIdx.x; col < pass_cols; col += blockDim.x) {
+      if (c0 + col < cols) {
+        float sum = 0.f;
+        for (int g = 0; g < by; ++g) sum += part[g * pass_cols + col];
+        atomicAdd(out + c0 + col, sum);
       }
     }
     __syncthreads();
@@ -527,11 +529,14 @@ void colsum(const void* in, float* out, int64_t rows, int cols, int ld,
             hipStream_t stream) {
   // scale the grid to the actual element count so small reductions don't
   // dispatch thousands of idle blocks (then atomic-sum their zeros)
+  int bx = hmin<int>(32, (cols + 7) / 8);
+  int nthreads = 256 - (256 % bx);       // whole row-groups only
+  int by = nthreads / bx;
   int64_t work_blocks = (rows * hmax<int64_t>(cols, 1)) / (256 * 8) + 1;
-  int blocks = (int)hmin<int64_t>(hmin<int64_t>((rows + 7) / 8, 2048),
+  int blocks = (int)hmin<int64_t>(hmin<int64_t>((rows + by - 1) / by, 2048),
                                   work_blocks);
- hipLaunchKernelGGL(( colsum_kernel), dim3(blocks), dim3(256), 0, stream, (const u16*)in, out, rows, cols,
-                                            ld);
+ hipLaunchKernelGGL(( colsum_kernel), dim3(blocks), dim3(nthreads), 0, stream, (const u16*)in, out, rows,
+                                                 cols, ld, bx);
 }
 
 void lstm_unit_fwd(const float* c_prev, const void* gates, const void* cont,
