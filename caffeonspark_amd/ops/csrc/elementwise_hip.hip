@@ -176,30 +176,33 @@ void bias_act_cast(const float* in, const float* bias, void* out,
 __global__ void transpose_bf16_kernel(const u16* __restrict__ in,
                                       u16* __restrict__ out,
                                       int R, int C, int tiles_c) {
-  __shared__ u16 lds[64][66];
+  // row-PAIRS packed as u32 in LDS: halves the LDS op count vs scalar
+  // b16 stores/loads (b16 LDS traffic runs at half rate)
+  __shared__ unsigned lds32[32][66];
   int tile = blockIdx.x;
   int r0 = (tile / tiles_c) * 64;
   int c0 = (tile % tiles_c) * 64;
   int t = threadIdx.x;
-  int lc8 = (t & 7) * 8;               // 8-wide column slot
-  int lr = t >> 3;                     // 0..31
   bool interior = (r0 + 64 <= R) && (c0 + 64 <= C);
+  int lc8 = (t & 7) * 8;               // 8-wide column slot
+  if (interior) {
+    int r2 = t >> 3;                   // row pair 0..31
+    u16x8 vlo = *reinterpret_cast<const u16x8*>(
+        in + (int64_t)(r0 + 2 * r2) * C + c0 + lc8);
+    u16x8 vhi = *reinterpret_cast<const u16x8*>(
+        in + (int64_t)(r0 + 2 * r2 + 1) * C + c0 + lc8);
 #pragma unroll
-  for (int half = 0; half < 2; ++half) {
-    int r = lr + half * 32;
-    int gr = r0 + r;
-    if (interior) {
-      u16x8 v = *reinterpret_cast<const u16x8*>(
-          in + (int64_t)gr * C + c0 + lc8);
+    for (int j = 0; j < 8; ++j)
+      lds32[r2][lc8 + j] = (unsigned)vlo[j] | ((unsigned)vhi[j] << 16);
+  } else {
+    int r2 = t >> 3;                   // each thread owns a full row PAIR
+    int grlo = r0 + 2 * r2, grhi = grlo + 1;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) lds[r][lc8 + j] = v[j];
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int gc = c0 + lc8 + j;
-        lds[r][lc8 + j] = (gr < R && gc < C)
-                              ? in[(int64_t)gr * C + gc] : 0;
-      }
+    for (int j = 0; j < 8; ++j) {
+      int gc = c0 + lc8 + j;
+      u16 lo = (grlo < R && gc < C) ? in[(int64_t)grlo * C + gc] : 0;
+      u16 hi = (grhi < R && gc < C) ? in[(int64_t)grhi * C + gc] : 0;
+      lds32[r2][lc8 + j] = (unsigned)lo | ((unsigned)hi << 16);
     }
   }
   __syncthreads();
@@ -212,13 +215,19 @@ __global__ void transpose_bf16_kernel(const u16* __restrict__ in,
     if (interior) {
       u16x8 v;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) v[j] = lds[lr8 + j][c];
+      for (int m = 0; m < 4; ++m) {
+        unsigned pair = lds32[(lr8 >> 1) + m][c];
+        v[2 * m] = (u16)(pair & 0xFFFF);
+        v[2 * m + 1] = (u16)(pair >> 16);
+      }
       *reinterpret_cast<u16x8*>(out + (int64_t)gc * R + r0 + lr8) = v;
     } else if (gc < C) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         int gr = r0 + lr8 + j;
-        if (gr < R) out[(int64_t)gc * R + gr] = lds[lr8 + j][c];
+        unsigned pair = lds32[(lr8 + j) >> 1][c];
+        u16 v = ((lr8 + j) & 1) ? (u16)(pair >> 16) : (u16)(pair & 0xFFFF);
+        if (gr < R) out[(int64_t)gc * R + gr] = v;
       }
     }
   }
