@@ -34,6 +34,10 @@ __global__ void im2col_smallc_kernel(
   }
   __syncthreads();
   int kslots = Kpad / 8;
+  // Ct%8==0 makes every 8-slot a contiguous channel run of ONE (r,s):
+  // one 16-byte gather instead of 8 scalar loads (conv2-5 of AlexNet,
+  // all GoogLeNet 3x3/5x5 towers; conv1's Ct=3 takes the scalar path)
+  bool runs8 = (Ct % 8 == 0);
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i < total8; i += (int64_t)gridDim.x * blockDim.x) {
     int64_t npq = i / kslots;
@@ -44,17 +48,33 @@ __global__ void im2col_smallc_kernel(
     int h0 = p * sh - ph, w0 = q * sw - pw;
     const u16* xbase = x + (int64_t)n * H * W * C + c0;
     u16x8 out;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int e = lut[k0 + j];
-      u16 v = 0;
+    int e = lut[k0];
+    if (runs8) {
       if (e >= 0) {
         int h = h0 + (e >> 20);
         int w = w0 + ((e >> 10) & 1023);
-        if (h >= 0 && h < H && w >= 0 && w < W)
-          v = xbase[((int64_t)h * W + w) * C + (e & 1023)];
+        if (h >= 0 && h < H && w >= 0 && w < W) {
+          out = *reinterpret_cast<const u16x8*>(
+              xbase + ((int64_t)h * W + w) * C + (e & 1023));
+        } else {
+          out = u16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        }
+      } else {
+        out = u16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
-      out[j] = v;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int ej = lut[k0 + j];
+        u16 v = 0;
+        if (ej >= 0) {
+          int h = h0 + (ej >> 20);
+          int w = w0 + ((ej >> 10) & 1023);
+          if (h >= 0 && h < H && w >= 0 && w < W)
+            v = xbase[((int64_t)h * W + w) * C + (ej & 1023)];
+        }
+        out[j] = v;
+      }
     }
     *reinterpret_cast<u16x8*>(col + npq * Kpad + k0) = out;
   }
