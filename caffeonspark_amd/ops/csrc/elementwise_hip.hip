@@ -176,41 +176,46 @@ void bias_act_cast(const float* in, const float* bias, void* out,
 __global__ void transpose_bf16_kernel(const u16* __restrict__ in,
                                       u16* __restrict__ out,
                                       int R, int C, int tiles_c) {
-  // row-PAIRS packed as u32 in LDS: halves the LDS op count vs scalar
-  // b16 stores/loads (b16 LDS traffic runs at half rate)
-  __shared__ unsigned lds32[32][66];
-  int tile = blockIdx.x;
-  int r0 = (tile / tiles_c) * 64;
+  // 128(rows) x 64(cols) tiles; row-PAIRS packed as u32 in LDS (halves
+  // LDS op count — b16 LDS traffic runs at half rate); write phase maps
+  // 16 threads per output row so each store instruction covers 256
+  // contiguous bytes; XCD-swizzled tile order for L2 row reuse.
+  __shared__ unsigned lds32[64][67];
+  int nblk = gridDim.x;
+  int tile = xcd_swizzle(blockIdx.x, nblk);
+  int r0 = (tile / tiles_c) * 128;
   int c0 = (tile % tiles_c) * 64;
   int t = threadIdx.x;
-  bool interior = (r0 + 64 <= R) && (c0 + 64 <= C);
+  bool interior = (r0 + 128 <= R) && (c0 + 64 <= C);
   int lc8 = (t & 7) * 8;               // 8-wide column slot
-  if (interior) {
-    int r2 = t >> 3;                   // row pair 0..31
-    u16x8 vlo = *reinterpret_cast<const u16x8*>(
-        in + (int64_t)(r0 + 2 * r2) * C + c0 + lc8);
-    u16x8 vhi = *reinterpret_cast<const u16x8*>(
-        in + (int64_t)(r0 + 2 * r2 + 1) * C + c0 + lc8);
 #pragma unroll
-    for (int j = 0; j < 8; ++j)
-      lds32[r2][lc8 + j] = (unsigned)vlo[j] | ((unsigned)vhi[j] << 16);
-  } else {
-    int r2 = t >> 3;                   // each thread owns a full row PAIR
+  for (int half = 0; half < 2; ++half) {
+    int r2 = (t >> 3) + half * 32;     // row pair 0..63
     int grlo = r0 + 2 * r2, grhi = grlo + 1;
+    if (interior) {
+      u16x8 vlo = *reinterpret_cast<const u16x8*>(
+          in + (int64_t)grlo * C + c0 + lc8);
+      u16x8 vhi = *reinterpret_cast<const u16x8*>(
+          in + (int64_t)grhi * C + c0 + lc8);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int gc = c0 + lc8 + j;
-      u16 lo = (grlo < R && gc < C) ? in[(int64_t)grlo * C + gc] : 0;
-      u16 hi = (grhi < R && gc < C) ? in[(int64_t)grhi * C + gc] : 0;
-      lds32[r2][lc8 + j] = (unsigned)lo | ((unsigned)hi << 16);
+      for (int j = 0; j < 8; ++j)
+        lds32[r2][lc8 + j] = (unsigned)vlo[j] | ((unsigned)vhi[j] << 16);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int gc = c0 + lc8 + j;
+        u16 lo = (grlo < R && gc < C) ? in[(int64_t)grlo * C + gc] : 0;
+        u16 hi = (grhi < R && gc < C) ? in[(int64_t)grhi * C + gc] : 0;
+        lds32[r2][lc8 + j] = (unsigned)lo | ((unsigned)hi << 16);
+      }
     }
   }
   __syncthreads();
-  int cl = t >> 3;                      // output row = original col
-  int lr8 = (t & 7) * 8;                // 8-wide along original rows
+  int cl = t >> 4;                      // 16 output rows (orig cols)/pass
+  int lr8 = (t & 15) * 8;               // 16 threads x 8 = 128 orig rows
 #pragma unroll
-  for (int half = 0; half < 2; ++half) {
-    int c = cl + half * 32;
+  for (int pass = 0; pass < 4; ++pass) {
+    int c = cl + pass * 16;
     int gc = c0 + c;
     if (interior) {
       u16x8 v;
@@ -235,7 +240,7 @@ __global__ void transpose_bf16_kernel(const u16* __restrict__ in,
 
 void transpose_bf16(const void* in, void* out, int64_t R, int64_t C,
                     hipStream_t stream) {
-  int tiles_r = (int)((R + 63) / 64), tiles_c = (int)((C + 63) / 64);
+  int tiles_r = (int)((R + 127) / 128), tiles_c = (int)((C + 63) / 64);
  hipLaunchKernelGGL(( transpose_bf16_kernel), dim3(tiles_r * tiles_c), dim3(256), 0, stream, 
       (const u16*)in, (u16*)out, (int)R, (int)C, tiles_c);
 }
