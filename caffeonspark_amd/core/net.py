@@ -197,6 +197,9 @@ class Net:
         """read_loss=False skips the device->host loss readout (the only
         sync in a forward), for hipGraph capture and sync-free stepping."""
         loss = 0.0
+        refresh = getattr(self, "_bf16_refresh", None)
+        if refresh is not None:
+            refresh()      # re-cast the fp32 master arena to bf16 once
         for layer, bottoms, tops in zip(self.layers, self.layer_bottoms,
                                         self.layer_tops):
             layer.forward(bottoms, tops)

@@ -69,6 +69,8 @@ class Solver:
             tn = Net(net_param, ts, device=self.device, dtype=self.dtype,
                      seed=seed)
             tn.share_trained_layers_with(self.net)
+            if hasattr(self.net, "_bf16_refresh"):
+                tn._bf16_refresh = self.net._bf16_refresh
             self.test_nets.append(tn)
 
         self.history2 = [torch.zeros_like(b.data) for b in self.params] \
@@ -98,6 +100,18 @@ class Solver:
             off += n
         self.history = [self.flat_m.narrow(0, o, b.count).view(b.shape)
                         for o, b in zip(self.param_offsets, self.params)]
+        if self.device.type == "cuda" and self.dtype == torch.bfloat16:
+            # bf16 shadow arena: ONE whole-arena cast per forward replaces
+            # per-layer fp32->bf16 weight casts (fc6 alone is 75 MB); the
+            # compute path reads these views via Blob.data._cos_bf16
+            self.flat_wb = torch.empty_like(self.flat_w,
+                                            dtype=torch.bfloat16)
+            self.flat_wb.copy_(self.flat_w)
+            for b, o in zip(self.params, self.param_offsets):
+                b.data._cos_bf16 = self.flat_wb.narrow(
+                    0, o, b.count).view(b.shape)
+            refresh = lambda: self.flat_wb.copy_(self.flat_w)  # noqa: E731
+            self.net._bf16_refresh = refresh
         # contiguous segments sharing (lr_mult, decay_mult) for fused updates
         self.segments = []  # (off, n, lr_mult, decay_mult)
         for b, o in zip(self.params, self.param_offsets):

@@ -30,7 +30,12 @@ def _cl(x: torch.Tensor) -> torch.Tensor:
 
 
 def _as_bf16(t: torch.Tensor) -> torch.Tensor:
-    return t if t.dtype == torch.bfloat16 else t.to(torch.bfloat16)
+    if t.dtype == torch.bfloat16:
+        return t
+    cached = getattr(t, "_cos_bf16", None)   # solver bf16 shadow arena
+    if cached is not None:
+        return cached
+    return t.to(torch.bfloat16)
 
 
 def _check_bf16(x, name):
@@ -99,9 +104,14 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
 
     xl = _cl(x)
     # weight repack: [K, Cg, R, S] -> bf16 [K, R, S, Cg] padded to Kpad
-    # columns and 128-aligned rows (GEMM fast-staging bound)
-    wrb = torch.zeros((_pad128(Kout), Kpad), dtype=torch.bfloat16,
-                      device=x.device)
+    # columns and 128-aligned rows (GEMM fast-staging bound); the padded
+    # buffer is cached on the weight tensor (pad region stays zero),
+    # only the permute-copy runs per step
+    wrb = getattr(w, "_cos_wrb", None)
+    if wrb is None or wrb.shape != (_pad128(Kout), Kpad):
+        wrb = torch.zeros((_pad128(Kout), Kpad), dtype=torch.bfloat16,
+                          device=x.device)
+        w._cos_wrb = wrb
     wrb[:Kout, :Kcol] = _as_bf16(w).permute(0, 2, 3, 1).reshape(Kout, Kcol)
     wr = wrb[:Kout]
     bias_f = b.float().contiguous() if b is not None else None
