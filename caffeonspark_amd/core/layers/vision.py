@@ -385,42 +385,34 @@ class BatchNormLayer(Layer):
         self._cache = None
 
     def forward(self, bottom, top):
-        x = bottom[0].data.float()
-        dims = [0] + list(range(2, x.dim()))
+        x = bottom[0].data
         if self.use_global:
-            sf = self.blobs[2].data.item()
+            sf = float(self.blobs[2].data.float())
             scale = 0 if sf == 0 else 1.0 / sf
-            mean = self.blobs[0].data * scale
-            var = self.blobs[1].data * scale
+            mean = (self.blobs[0].data.float() * scale).contiguous()
+            var = (self.blobs[1].data.float() * scale).contiguous()
+            y, inv_std = ops.bn_forward_infer(x, mean, var, self.eps)
         else:
-            mean = x.mean(dim=dims)
-            var = x.var(dim=dims, unbiased=False)
+            y, mean, var, inv_std = ops.bn_forward_train(x, self.eps)
             m = x.numel() / x.shape[1]
-            self.blobs[0].data.mul_(self.maf).add_(mean)
+            self.blobs[0].data.mul_(self.maf).add_(
+                mean.to(self.blobs[0].data.dtype))
             bias_corr = m / max(1.0, m - 1.0)
-            self.blobs[1].data.mul_(self.maf).add_(var * bias_corr)
+            self.blobs[1].data.mul_(self.maf).add_(
+                (var * bias_corr).to(self.blobs[1].data.dtype))
             self.blobs[2].data.mul_(self.maf).add_(1.0)
-        shape = [1, -1] + [1] * (x.dim() - 2)
-        inv_std = (var + self.eps).rsqrt()
-        xhat = (x - mean.reshape(shape)) * inv_std.reshape(shape)
-        self._cache = (xhat, inv_std)
-        top[0].data = xhat.to(self.dtype)
+        # caffe BN has no affine term, so the output IS xhat: backward
+        # reads it from the top blob (in-place safe)
+        self._cache = inv_std
+        top[0].data = y
         return 0.0
 
     def backward(self, top, propagate_down, bottom):
         if not propagate_down[0]:
             return
-        xhat, inv_std = self._cache
-        dy = top[0].diff.float()
-        shape = [1, -1] + [1] * (dy.dim() - 2)
-        if self.use_global:
-            dx = dy * inv_std.reshape(shape)
-        else:
-            dims = [0] + list(range(2, dy.dim()))
-            m = dy.numel() / dy.shape[1]
-            dmean = dy.mean(dim=dims).reshape(shape)
-            dvar = (dy * xhat).mean(dim=dims).reshape(shape)
-            dx = (dy - dmean - xhat * dvar) * inv_std.reshape(shape)
+        inv_std = self._cache
+        dx = ops.bn_backward(top[0].data, top[0].diff, inv_std,
+                             train=not self.use_global)
         self.acc_blob_diff(bottom[0], dx.to(bottom[0].data.dtype),
                            top[0] is bottom[0])
 

@@ -125,6 +125,9 @@ embed_forward = _dispatch("embed_forward")
 embed_backward = _dispatch("embed_backward")
 lstm_unit_forward = _dispatch("lstm_unit_forward")
 lstm_unit_backward = _dispatch("lstm_unit_backward")
+bn_forward_train = _dispatch("bn_forward_train")
+bn_forward_infer = _dispatch("bn_forward_infer")
+bn_backward = _dispatch("bn_backward")
 accuracy = _dispatch("accuracy")
 bias_add = _dispatch("bias_add")
 sgd_update = _dispatch("sgd_update")

@@ -20,6 +20,15 @@ void wino_conv(const void* x, const float* w, const float* bias, void* y,
                void* U, void* V, void* Mbuf, int N, int H, int W, int Cin,
                int K, int wK, int wC, int u_rows_alloc, bool flip,
                bool relu, hipStream_t stream);
+void bn_stats(const void* x, float* sum, float* sumsq, int64_t rows, int C,
+              hipStream_t stream);
+void bn_bwd_sums(const void* dy, const void* xhat, float* s1, float* s2,
+                 int64_t rows, int C, hipStream_t stream);
+void bn_norm(const void* x, void* y, const float* mean, const float* invstd,
+             int64_t rows, int C, hipStream_t stream);
+void bn_bwd(const void* xhat, const void* dy, void* dx,
+            const float* invstd, const float* s1, const float* s2,
+            float inv_m, int64_t rows, int C, hipStream_t stream);
 void im2col_t(const void* x, void* colT, int N, int H, int W, int C,
               int P, int Q, int R, int S, int sh, int sw, int ph, int pw,
               int dil, int Kpad, int c0, int Cg, hipStream_t stream);
@@ -336,6 +345,30 @@ void py_softmax_loss_bwd(Tensor prob, Tensor label, Tensor dx, double scale,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm", &py_gemm);
   m.def("im2col", &py_im2col);
+  m.def("bn_stats", [](Tensor x, Tensor sum, Tensor sq, int64_t rows,
+                       int64_t C) {
+    cosamd::bn_stats(x.data_ptr(), sum.data_ptr<float>(),
+                     sq.data_ptr<float>(), rows, C, cur_stream());
+  });
+  m.def("bn_bwd_sums", [](Tensor dy, Tensor xhat, Tensor s1, Tensor s2,
+                          int64_t rows, int64_t C) {
+    cosamd::bn_bwd_sums(dy.data_ptr(), xhat.data_ptr(),
+                        s1.data_ptr<float>(), s2.data_ptr<float>(), rows,
+                        C, cur_stream());
+  });
+  m.def("bn_norm", [](Tensor x, Tensor y, Tensor mean, Tensor invstd,
+                      int64_t rows, int64_t C) {
+    cosamd::bn_norm(x.data_ptr(), y.data_ptr(), mean.data_ptr<float>(),
+                    invstd.data_ptr<float>(), rows, C, cur_stream());
+  });
+  m.def("bn_bwd", [](Tensor xhat, Tensor dy, Tensor dx, Tensor invstd,
+                     Tensor s1, Tensor s2, double inv_m, int64_t rows,
+                     int64_t C) {
+    cosamd::bn_bwd(xhat.data_ptr(), dy.data_ptr(), dx.data_ptr(),
+                   invstd.data_ptr<float>(), s1.data_ptr<float>(),
+                   s2.data_ptr<float>(), (float)inv_m, rows, C,
+                   cur_stream());
+  });
   m.def("wino_conv", [](Tensor x, Tensor w, c10::optional<Tensor> bias,
                         Tensor y, Tensor U, Tensor V, Tensor M, int64_t N,
                         int64_t H, int64_t W, int64_t Cin, int64_t K,

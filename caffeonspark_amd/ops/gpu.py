@@ -634,6 +634,69 @@ def lstm_seq_backward(dy, w_hc, cache):
     return dxg, dwhc
 
 
+# -------------------------------------------------------------- batchnorm
+
+def _bn_eligible(x):
+    return x.dim() == 4 and x.shape[1] % 8 == 0 and \
+        x.dtype == torch.bfloat16
+
+
+def bn_forward_train(x, eps):
+    if not _bn_eligible(x):
+        return reference.bn_forward_train(x, eps)
+    xl = _cl(x)
+    N, C, H, W = x.shape
+    rows = N * H * W
+    x2 = xl.permute(0, 2, 3, 1).reshape(rows, C)
+    s0 = torch.zeros(C, dtype=torch.float32, device=x.device)
+    s1 = torch.zeros(C, dtype=torch.float32, device=x.device)
+    _ext.bn_stats(x2, s0, s1, rows, C)
+    mean = s0 / rows
+    var = (s1 / rows - mean * mean).clamp_min_(0)
+    invstd = (var + eps).rsqrt()
+    y = torch.empty((N, C, H, W), dtype=torch.bfloat16, device=x.device,
+                    memory_format=torch.channels_last)
+    _ext.bn_norm(x2, y.permute(0, 2, 3, 1).reshape(rows, C), mean, invstd,
+                 rows, C)
+    return y, mean, var, invstd
+
+
+def bn_forward_infer(x, mean, var, eps):
+    if not _bn_eligible(x):
+        return reference.bn_forward_infer(x, mean, var, eps)
+    xl = _cl(x)
+    N, C, H, W = x.shape
+    rows = N * H * W
+    invstd = (var.float() + eps).rsqrt().contiguous()
+    y = torch.empty((N, C, H, W), dtype=torch.bfloat16, device=x.device,
+                    memory_format=torch.channels_last)
+    _ext.bn_norm(xl.permute(0, 2, 3, 1).reshape(rows, C),
+                 y.permute(0, 2, 3, 1).reshape(rows, C),
+                 mean.float().contiguous(), invstd, rows, C)
+    return y, invstd
+
+
+def bn_backward(xhat, dy, invstd, train):
+    if not _bn_eligible(xhat):
+        return reference.bn_backward(xhat, dy, invstd, train)
+    xl = _cl(xhat)
+    dyl = _cl(dy.to(torch.bfloat16))
+    N, C, H, W = xhat.shape
+    rows = N * H * W
+    x2 = xl.permute(0, 2, 3, 1).reshape(rows, C)
+    dy2 = dyl.permute(0, 2, 3, 1).reshape(rows, C)
+    s1 = torch.zeros(C, dtype=torch.float32, device=xhat.device)
+    s2 = torch.zeros(C, dtype=torch.float32, device=xhat.device)
+    if train:
+        _ext.bn_bwd_sums(dy2, x2, s1, s2, rows, C)
+    dx = torch.empty((N, C, H, W), dtype=torch.bfloat16,
+                     device=xhat.device, memory_format=torch.channels_last)
+    _ext.bn_bwd(x2, dy2, dx.permute(0, 2, 3, 1).reshape(rows, C),
+                invstd.contiguous(), s1, s2,
+                (1.0 / rows) if train else 0.0, rows, C)
+    return dx
+
+
 # -------------------------------------------------------------- optimizer
 
 def sgd_update(param, grad, momentum_buf, lr, momentum, weight_decay):

@@ -326,3 +326,37 @@ def adam_update(param, grad, m, v, lr, beta1, beta2, eps, weight_decay, t):
     v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
     correction = (1 - beta2 ** t) ** 0.5 / (1 - beta1 ** t)
     param.addcdiv_(m, v.sqrt().add_(eps), value=-lr * correction)
+
+
+# -------------------------------------------------------------- batchnorm
+# caffe BatchNorm: normalization only (affine lives in the Scale layer),
+# so the output IS xhat — backward takes it instead of caching x.
+
+def bn_forward_train(x, eps):
+    xf = x.float()
+    dims = [0] + list(range(2, xf.dim()))
+    mean = xf.mean(dim=dims)
+    var = xf.var(dim=dims, unbiased=False)
+    invstd = (var + eps).rsqrt()
+    shape = [1, -1] + [1] * (xf.dim() - 2)
+    y = (xf - mean.reshape(shape)) * invstd.reshape(shape)
+    return y.to(x.dtype), mean, var, invstd
+
+
+def bn_forward_infer(x, mean, var, eps):
+    invstd = (var.float() + eps).rsqrt()
+    shape = [1, -1] + [1] * (x.dim() - 2)
+    y = (x.float() - mean.float().reshape(shape)) * invstd.reshape(shape)
+    return y.to(x.dtype), invstd
+
+
+def bn_backward(xhat, dy, invstd, train):
+    dyf, xh = dy.float(), xhat.float()
+    shape = [1, -1] + [1] * (dy.dim() - 2)
+    if not train:
+        return (dyf * invstd.reshape(shape)).to(xhat.dtype)
+    dims = [0] + list(range(2, dy.dim()))
+    dmean = dyf.mean(dim=dims).reshape(shape)
+    dvar = (dyf * xh).mean(dim=dims).reshape(shape)
+    dx = (dyf - dmean - xh * dvar) * invstd.reshape(shape)
+    return dx.to(xhat.dtype)

@@ -335,6 +335,35 @@ def test_lenet_step_parity():
             assert rel < 0.08, f"{cl.name} grad mismatch relL2={rel:.3f}"
 
 
+def test_batchnorm_parity():
+    """HIP NHWC BatchNorm (stats reduce + normalize + backward) vs the
+    fp32 torch reference, train and global-stats modes."""
+    from caffeonspark_amd.ops import gpu as g, reference as ref
+
+    torch.manual_seed(11)
+    x = bf(torch.randn(4, 32, 9, 9) * 2 + 0.5).to(dev())
+    xc = x.float().cpu()
+    y, mean, var, invstd = g.bn_forward_train(x, 1e-5)
+    ry, rmean, rvar, rinvstd = ref.bn_forward_train(xc, 1e-5)
+    agree(mean, rmean, rtol=0.02, atol=0.02)
+    agree(var, rvar, rtol=0.03, atol=0.03)
+    agree(y, ry, rtol=0.05, atol=0.05)
+    dy = bf(torch.randn_like(xc)).to(dev())
+    dx = g.bn_backward(y, dy, invstd, train=True)
+    rdx = ref.bn_backward(ry, dy.float().cpu(), rinvstd, train=True)
+    rel = (dx.float().cpu() - rdx.float()).norm() / rdx.float().norm()
+    assert float(rel) < 0.05, f"bn dx relL2={float(rel):.4f}"
+    # global-stats mode
+    gm = torch.randn(32).to(dev())
+    gv = torch.rand(32).to(dev()) + 0.5
+    y2, inv2 = g.bn_forward_infer(x, gm, gv, 1e-5)
+    ry2, rinv2 = ref.bn_forward_infer(xc, gm.cpu(), gv.cpu(), 1e-5)
+    agree(y2, ry2, rtol=0.05, atol=0.05)
+    dx2 = g.bn_backward(y2, dy, inv2, train=False)
+    rdx2 = ref.bn_backward(ry2, dy.float().cpu(), rinv2, train=False)
+    agree(dx2, rdx2, rtol=0.05, atol=0.05)
+
+
 def test_winograd_conv_parity(monkeypatch):
     """F(2x2,3x3) Winograd path (COS_WINOGRAD=1) vs the direct im2col
     path: fwd/dx/dw/db must agree within bf16 training noise."""
