@@ -261,10 +261,11 @@ __global__ void sgd_update_multi_kernel(
     if (i < nseg) { slr[i] = seg_lr[i]; swd[i] = seg_wd[i]; }
   }
   __syncthreads();
+  int s = 0;  // a thread's quads are monotonically increasing, so the
+              // segment cursor only moves forward: amortized O(nseg)
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i < total4; i += (int64_t)gridDim.x * blockDim.x) {
     int64_t e0 = i * 4;
-    int s = 0;
     while (s + 1 < nseg && e0 >= soff[s + 1]) ++s;
     float lr = slr[s], wd = swd[s];
     if (lr == 0.f) continue;
