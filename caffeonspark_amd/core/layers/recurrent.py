@@ -184,7 +184,9 @@ class EmbedLayer(Layer):
         # caffe embed output: bottom shape (with trailing singleton dims
         # dropped) + [E]
         base = [s for s in idx.shape]
-        while len(base) > 1 and base[-1] == 1:
+        # drop trailing singleton SPATIAL dims but never the batch axis
+        # (time-major [T, 1] single-image decode must stay 2-d)
+        while len(base) > 2 and base[-1] == 1:
             base = base[:-1]
         top[0].data = y.reshape(base + [self.e])
         return 0.0

@@ -42,6 +42,7 @@ def caption_images(conf: Config, image_files, max_len: int = 20):
     xf = DataTransformer(tp, caffe_pb.Phase.TEST)
 
     dl = net.data_layers()[0]
+    dl.batch_size = 1
     T = int(dl.tops_cfg[2].channels)
     results = []
     for path in image_files:

@@ -274,3 +274,32 @@ def test_ml_pipeline_example(workdir):
                       "-weights", str(workdir / sorted(snaps)[-1]),
                       "-features", "ip1", "-label", "label"])
     CaffeProcessor.reset_instance()
+
+
+def test_image_caption_example(tmp_path):
+    """ImageCaption.py analog: greedy LRCN decode over our Net (random
+    weights; checks the decode loop plumbing, not caption quality)."""
+    import sys
+
+    import numpy as np
+    sys.path.insert(0, os.path.join(ROOT, "examples"))
+    import image_caption
+
+    from caffeonspark_amd.api import Config
+
+    try:
+        from PIL import Image
+    except ImportError:
+        pytest.skip("PIL not available")
+    img = (np.random.RandomState(0).rand(64, 64, 3) * 255).astype("uint8")
+    p = str(tmp_path / "cat.jpg")
+    Image.fromarray(img).save(p)
+
+    solver = os.path.join(ROOT, "caffeonspark_amd", "models",
+                          "lrcn_solver.prototxt")
+    conf = Config(["-conf", solver])
+    results = image_caption.caption_images(conf, [p], max_len=2)
+    assert len(results) == 1
+    path, caption = results[0]
+    assert path == p
+    assert isinstance(caption, str)
