@@ -342,8 +342,18 @@ def test_gpu_end_to_end_training(tmp_path):
                               dtype=torch.bfloat16)
     s2.restore(str(tmp_path / states[-1]))
     assert s2.iter == 150
-    x = torch.randn(64, 1, 28, 28, dtype=torch.bfloat16, device="cuda")
-    y = torch.randint(0, 10, (64,), device="cuda").float()
+    # evaluate on a batch drawn from the training distribution (stripe at
+    # column 2*label), with the net's transform (x-128)/256 applied
+    rng = np.random.RandomState(9)
+    imgs, labels = [], []
+    for _ in range(64):
+        label = rng.randint(0, 10)
+        img = rng.randint(80, 150, size=(1, 28, 28)).astype(np.float32)
+        img[0, :, label * 2] = 250
+        imgs.append((img - 128.0) * 0.00390625)
+        labels.append(label)
+    x = torch.tensor(np.stack(imgs)).to("cuda", torch.bfloat16)
+    y = torch.tensor(labels, dtype=torch.float32, device="cuda")
     s2.net.data_layers()[0].reset(x, y)
     loss = s2.net.forward()
     assert loss < 1.0, f"restored net loss {loss} (random init would be ~2.3)"
