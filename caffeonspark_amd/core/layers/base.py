@@ -105,12 +105,18 @@ class Layer:
         top and bottom share the tensor, so += would double-count)."""
         if inplace:
             blob.diff = dx
-        else:
-            d = blob.ensure_diff()
-            if d.dtype != dx.dtype:
-                blob.diff = d.to(dx.dtype)
-                d = blob.diff
-            d.add_(dx)
+            return
+        d = blob.diff
+        if d is None or list(d.shape) != list(dx.shape):
+            # first consumer: adopt the gradient tensor outright — the
+            # zeros_like + add_ alternative costs two extra full passes
+            # over the activation per backward layer
+            blob.diff = dx
+            return
+        if d.dtype != dx.dtype:
+            blob.diff = d.to(dx.dtype)
+            d = blob.diff
+        d.add_(dx)
 
     def loss_weight(self, i: int) -> float:
         lw = self.param.loss_weight
