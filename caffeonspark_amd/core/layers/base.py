@@ -96,8 +96,16 @@ class Layer:
         return self.cast(self.blobs[i].data)
 
     def acc_param_diff(self, i: int, grad: torch.Tensor) -> None:
-        d = self.blobs[i].ensure_diff()
-        d.add_(grad.to(d.dtype))
+        b = self.blobs[i]
+        d = b.ensure_diff()
+        if getattr(b, "_grad_virgin", False):
+            # first write this step: overwrite — Net.zero_param_diffs
+            # marks instead of zeroing, saving a full pass over the
+            # gradient arena every iteration
+            d.copy_(grad.to(d.dtype))
+            b._grad_virgin = False
+        else:
+            d.add_(grad.to(d.dtype))
 
     @staticmethod
     def acc_blob_diff(blob: Blob, dx: torch.Tensor, inplace: bool) -> None:

@@ -249,8 +249,11 @@ class Net:
         return out
 
     def zero_param_diffs(self) -> None:
+        # mark rather than zero: the first acc_param_diff this step
+        # copies instead of accumulating (param regions never written by
+        # backward stay zero from the arena's init-time fill)
         for b in self.learnable_params():
-            b.zero_diff()
+            b._grad_virgin = True
 
     # ----------------------------------------------------------------- access
     def blob_by_name(self, name: str) -> Blob:
