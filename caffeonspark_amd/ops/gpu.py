@@ -25,8 +25,20 @@ def _pad8(k: int) -> int:
 
 
 def _cl(x: torch.Tensor) -> torch.Tensor:
-    """channels_last contiguous view of a 4-D activation."""
-    return x.contiguous(memory_format=torch.channels_last)
+    """channels_last contiguous form of a 4-D activation; the converted
+    copy is cached on the tensor (keyed by torch's in-place version
+    counter) so a statically-fed input converts once, not every step."""
+    if x.is_contiguous(memory_format=torch.channels_last):
+        return x
+    cached = getattr(x, "_cos_cl", None)
+    if cached is not None and cached[0] == x._version:
+        return cached[1]
+    y = x.contiguous(memory_format=torch.channels_last)
+    try:
+        x._cos_cl = (x._version, y)
+    except AttributeError:
+        pass
+    return y
 
 
 def _as_bf16(t: torch.Tensor) -> torch.Tensor:
