@@ -22,6 +22,19 @@ __device__ __forceinline__ void st_bf(u16* p, float v) {
   *p = *reinterpret_cast<u16*>(&b);
 }
 
+// sc^-beta: __powf costs ~10x a rsqrt, and caffe LRN is almost always
+// beta = 0.75 (sc^-3/4 = rsqrt(sc) * sqrt(rsqrt(sc))) or 0.5
+template <int BMODE>  // 0: generic powf, 1: beta=0.75, 2: beta=0.5
+__device__ __forceinline__ float pow_negbeta(float sc, float beta) {
+  if (BMODE == 1) {
+    float r = __frsqrt_rn(sc);
+    return r * __fsqrt_rn(r);
+  }
+  if (BMODE == 2) return __frsqrt_rn(sc);
+  return __powf(sc, -beta);
+}
+
+template <int BMODE>
 __global__ void lrn_fwd_kernel(const u16* __restrict__ x,
                                u16* __restrict__ y,
                                float* __restrict__ scale,
@@ -49,7 +62,7 @@ __global__ void lrn_fwd_kernel(const u16* __restrict__ x,
         if (t <= 2 * half) win += v[j + t] * v[j + t];
       float sc = k + a_over_n * win;
       scale[base + cc + j] = sc;
-      bf16 b = f2bf(v[j + half] * __powf(sc, -beta));
+      bf16 b = f2bf(v[j + half] * pow_negbeta<BMODE>(sc, beta));
       out[j] = *reinterpret_cast<u16*>(&b);
     }
     *reinterpret_cast<u16x8*>(y + base + cc) = out;
@@ -78,6 +91,7 @@ __global__ void lrn_ratio_kernel(const u16* __restrict__ y,
   }
 }
 
+template <int BMODE>
 __global__ void lrn_bwd_kernel(const u16* __restrict__ x,
                                const float* __restrict__ scale,
                                const u16* __restrict__ dy,
@@ -105,7 +119,7 @@ __global__ void lrn_bwd_kernel(const u16* __restrict__ x,
       for (int t = 0; t <= 8; ++t)
         if (t <= 2 * half) win += r[j + t];
       int64_t i = base + cc + j;
-      float v = ld_bf(dy + i) * __powf(scale[i], -beta)
+      float v = ld_bf(dy + i) * pow_negbeta<BMODE>(scale[i], beta)
                 - ratio_coef * ld_bf(x + i) * win;
       bf16 b = f2bf(v);
       out[j] = *reinterpret_cast<u16*>(&b);
@@ -122,9 +136,18 @@ void lrn_fwd(const void* x, void* y, float* scale, int64_t npix, int C,
   // configs run the CPU-parity fallback in ops/gpu.py
   int64_t total8 = npix * (C / 8);
   int blocks = (int)hmin<int64_t>(4096, (total8 + 255) / 256);
-  lrn_fwd_kernel<<<blocks, 256, 0, stream>>>(
-      (const u16*)x, (u16*)y, scale, total8, C, local_size / 2,
-      alpha / local_size, beta, k);
+  if (beta == 0.75f)
+    lrn_fwd_kernel<1><<<blocks, 256, 0, stream>>>(
+        (const u16*)x, (u16*)y, scale, total8, C, local_size / 2,
+        alpha / local_size, beta, k);
+  else if (beta == 0.5f)
+    lrn_fwd_kernel<2><<<blocks, 256, 0, stream>>>(
+        (const u16*)x, (u16*)y, scale, total8, C, local_size / 2,
+        alpha / local_size, beta, k);
+  else
+    lrn_fwd_kernel<0><<<blocks, 256, 0, stream>>>(
+        (const u16*)x, (u16*)y, scale, total8, C, local_size / 2,
+        alpha / local_size, beta, k);
 }
 
 void lrn_bwd(const void* x, const void* y, const float* scale, const void* dy,
@@ -134,9 +157,19 @@ void lrn_bwd(const void* x, const void* y, const float* scale, const void* dy,
   int blocks = (int)hmin<int64_t>(4096, (total8 + 255) / 256);
   lrn_ratio_kernel<<<blocks, 256, 0, stream>>>(
       (const u16*)y, scale, (const u16*)dy, (u16*)ratio, total8);
-  lrn_bwd_kernel<<<blocks, 256, 0, stream>>>(
-      (const u16*)x, scale, (const u16*)dy, (const u16*)ratio, (u16*)dx,
-      total8, C, local_size / 2, beta, 2.f * alpha * beta / local_size);
+  float rc = 2.f * alpha * beta / local_size;
+  if (beta == 0.75f)
+    lrn_bwd_kernel<1><<<blocks, 256, 0, stream>>>(
+        (const u16*)x, scale, (const u16*)dy, (const u16*)ratio, (u16*)dx,
+        total8, C, local_size / 2, beta, rc);
+  else if (beta == 0.5f)
+    lrn_bwd_kernel<2><<<blocks, 256, 0, stream>>>(
+        (const u16*)x, scale, (const u16*)dy, (const u16*)ratio, (u16*)dx,
+        total8, C, local_size / 2, beta, rc);
+  else
+    lrn_bwd_kernel<0><<<blocks, 256, 0, stream>>>(
+        (const u16*)x, scale, (const u16*)dy, (const u16*)ratio, (u16*)dx,
+        total8, C, local_size / 2, beta, rc);
 }
 
 }  // namespace cosamd
