@@ -89,12 +89,12 @@ void embed_fwd(const float* idx, const void* w, void* y, int64_t n, int E,
 void embed_bwd(const float* idx, const void* dy, float* dw, int64_t n, int E,
                int V, hipStream_t stream);
 void lstm_seq_fwd(const void* xg, const void* w_hc, const void* cont,
-                  void* h, float* c, float* act, void* h_in, void* hg,
+                  void* h, float* c, float* act, void* h_in, float* hg,
                   int T, int N, int H, int n_alloc_whc, hipStream_t stream);
 void lstm_seq_bwd(const void* dy, const void* w_hcT, const void* cont,
                   const void* h, const float* c, const float* act,
-                  void* dxg, void* dh_rec, float* dc_a, float* dc_b,
-                  int T, int N, int H, int n_alloc_whcT,
+                  void* dxg, void* dh_rec, float* dh_f, float* dc_a,
+                  float* dc_b, int T, int N, int H, int n_alloc_whcT,
                   hipStream_t stream);
 void softmax_loss_fwd(const void* x, const float* label, float* prob,
                       float* loss, int* count, int64_t nrows, int C,
@@ -305,23 +305,25 @@ void py_lstm_seq_fwd(Tensor xg, Tensor w_hc, Tensor cont, Tensor h,
                      Tensor c, Tensor act, Tensor h_in, Tensor hg,
                      int64_t T, int64_t N, int64_t H, int64_t n_alloc) {
   CHECK_BF16(xg); CHECK_BF16(w_hc); CHECK_F32(c); CHECK_F32(act);
+  CHECK_F32(hg);
   cosamd::lstm_seq_fwd(xg.data_ptr(), w_hc.data_ptr(), cont.data_ptr(),
                        h.data_ptr(), c.data_ptr<float>(),
                        act.data_ptr<float>(), h_in.data_ptr(),
-                       hg.data_ptr(), T, N, H, n_alloc, cur_stream());
+                       hg.data_ptr<float>(), T, N, H, n_alloc,
+                       cur_stream());
 }
 
 void py_lstm_seq_bwd(Tensor dy, Tensor w_hcT, Tensor cont, Tensor h,
                      Tensor c, Tensor act, Tensor dxg, Tensor dh_rec,
-                     Tensor dc_a, Tensor dc_b, int64_t T, int64_t N,
-                     int64_t H, int64_t n_alloc) {
-  CHECK_BF16(dy); CHECK_BF16(w_hcT);
+                     Tensor dh_f, Tensor dc_a, Tensor dc_b, int64_t T,
+                     int64_t N, int64_t H, int64_t n_alloc) {
+  CHECK_BF16(dy); CHECK_BF16(w_hcT); CHECK_F32(dh_f);
   cosamd::lstm_seq_bwd(dy.data_ptr(), w_hcT.data_ptr(), cont.data_ptr(),
                        h.data_ptr(), c.data_ptr<float>(),
                        act.data_ptr<float>(), dxg.data_ptr(),
-                       dh_rec.data_ptr(), dc_a.data_ptr<float>(),
-                       dc_b.data_ptr<float>(), T, N, H, n_alloc,
-                       cur_stream());
+                       dh_rec.data_ptr(), dh_f.data_ptr<float>(),
+                       dc_a.data_ptr<float>(), dc_b.data_ptr<float>(),
+                       T, N, H, n_alloc, cur_stream());
 }
 
 void py_softmax_loss_fwd(Tensor x, Tensor label, Tensor prob, Tensor loss,

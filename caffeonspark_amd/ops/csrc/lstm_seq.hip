@@ -36,10 +36,21 @@ __global__ void rowscale_kernel(const u16* __restrict__ x,
     stbf_(y + i, ldbf_(x + i) * ldbf_(s + i / H));
 }
 
-// fused unit with two pre-activation gate inputs (xg_t + hg_t)
+// h_in[n][h] = h_prev[n][h] * cont[n], fp32 source (split-K GEMM out)
+__global__ void rowscale_f32_kernel(const float* __restrict__ x,
+                                    const u16* __restrict__ s,
+                                    u16* __restrict__ y, int64_t n, int H) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += (int64_t)gridDim.x * blockDim.x)
+    stbf_(y + i, x[i] * ldbf_(s + i / H));
+}
+
+// fused unit with two pre-activation gate inputs (xg_t + hg_t); hg is
+// fp32 — it comes from the split-K atomic recurrent GEMM (the [N,4H]
+// hidden GEMM underfills the chip without split-K: 32 blocks on 256 CUs)
 __global__ void lstm_unit2_fwd_kernel(
     const float* __restrict__ c_prev, const u16* __restrict__ xg,
-    const u16* __restrict__ hg, const u16* __restrict__ cont,
+    const float* __restrict__ hg, const u16* __restrict__ cont,
     float* __restrict__ c_out, u16* __restrict__ h_out,
     float* __restrict__ act, int64_t n, int H) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -47,12 +58,11 @@ __global__ void lstm_unit2_fwd_kernel(
     int64_t row = i / H;
     int hh = i % H;
     int64_t g0 = row * 4 * H + hh;
-    float gi = 1.f / (1.f + __expf(-(ldbf_(xg + g0) + ldbf_(hg + g0))));
-    float gf = 1.f / (1.f + __expf(-(ldbf_(xg + g0 + H) +
-                                     ldbf_(hg + g0 + H))));
+    float gi = 1.f / (1.f + __expf(-(ldbf_(xg + g0) + hg[g0])));
+    float gf = 1.f / (1.f + __expf(-(ldbf_(xg + g0 + H) + hg[g0 + H])));
     float go = 1.f / (1.f + __expf(-(ldbf_(xg + g0 + 2 * H) +
-                                     ldbf_(hg + g0 + 2 * H))));
-    float gg = tanhf(ldbf_(xg + g0 + 3 * H) + ldbf_(hg + g0 + 3 * H));
+                                     hg[g0 + 2 * H])));
+    float gg = tanhf(ldbf_(xg + g0 + 3 * H) + hg[g0 + 3 * H]);
     float ct = ldbf_(cont + row);
     float c = gf * c_prev[i] * ct + gi * gg;
     c_out[i] = c;
@@ -101,10 +111,15 @@ static int nb_(int64_t total) {
 // h_in:[T,N,H] bf16 (gated prev hidden, kept for the weight-grad GEMM);
 // hg: scratch [N,4H] bf16.
 void lstm_seq_fwd(const void* xg, const void* w_hc, const void* cont,
-                  void* h, float* c, float* act, void* h_in, void* hg,
+                  void* h, float* c, float* act, void* h_in, float* hg,
                   int T, int N, int H, int n_alloc_whc,
                   hipStream_t stream) {
   int64_t nh = (int64_t)N * H;
+  // callers over-allocate h_in by 128 rows so per-step A slices keep the
+  // fast DMA staging (m_alloc = N + 128); split-K fills the chip for the
+  // skinny [N,4H] recurrent GEMM
+  int nb4h = (4 * H + 127) / 128;
+  int sk = (448 + nb4h - 1) / nb4h;
   COS_CHECK_HIP(hipMemsetAsync(h_in, 0, nh * 2, stream));  // h_-1 = 0
   COS_CHECK_HIP(hipMemsetAsync(c, 0, nh * 4, stream));     // reused as c_-1? no
   for (int t = 0; t < T; ++t) {
@@ -122,13 +137,15 @@ void lstm_seq_fwd(const void* xg, const void* w_hc, const void* cont,
       rowscale_kernel<<<nb_(nh), 256, 0, stream>>>(
           h_in_t, cont_t, h_in_t, nh, H);
     }
+    COS_CHECK_HIP(hipMemsetAsync(hg, 0, (int64_t)N * 4 * H * 4, stream));
     gemm_bf16(h_in_t, w_hc, hg, nullptr, N, 4 * H, H, H, H, 4 * H,
-              false, false, 0, 1, false, 1.0f, N, n_alloc_whc, stream);
+              false, false, 2, sk, false, 1.0f, N + 128, n_alloc_whc,
+              stream);
     const float* c_prev = (t > 0) ? c + (int64_t)(t - 1) * nh : nullptr;
     // t==0: c_prev unused since cont_0 should be 0; still need a valid
     // pointer — use c_t (will read garbage * cont ... safer: zero buffer)
     lstm_unit2_fwd_kernel<<<nb_(nh), 256, 0, stream>>>(
-        (t > 0) ? c_prev : c_t /* zeroed above */, xg_t, (const u16*)hg,
+        (t > 0) ? c_prev : c_t /* zeroed above */, xg_t, hg,
         cont_t, c_t, h_t, act_t, nh, H);
   }
 }
@@ -138,10 +155,12 @@ void lstm_seq_fwd(const void* xg, const void* w_hc, const void* cont,
 // bf16, dc buffers [N,H] f32 x2.
 void lstm_seq_bwd(const void* dy, const void* w_hcT, const void* cont,
                   const void* h, const float* c, const float* act,
-                  void* dxg, void* dh_rec, float* dc_a, float* dc_b,
-                  int T, int N, int H, int n_alloc_whcT,
+                  void* dxg, void* dh_rec, float* dh_f, float* dc_a,
+                  float* dc_b, int T, int N, int H, int n_alloc_whcT,
                   hipStream_t stream) {
   int64_t nh = (int64_t)N * H;
+  int nbh = (H + 127) / 128;
+  int skb = (448 + nbh - 1) / nbh;
   COS_CHECK_HIP(hipMemsetAsync(dc_a, 0, nh * 4, stream));
   float* dc_next = dc_a;
   float* dc_prev = dc_b;
@@ -162,10 +181,12 @@ void lstm_seq_bwd(const void* dy, const void* w_hcT, const void* cont,
         (t < T - 1) ? (const u16*)dh_rec : nullptr, dc_prev, dxg_t, nh, H);
     if (t > 0) {
       // dh_rec = (dgates @ w_hc) * cont_t  — the recurrent grad into h_{t-1}
-      gemm_bf16(dxg_t, w_hcT, dh_rec, nullptr, N, H, 4 * H, 4 * H, 4 * H,
-                H, false, false, 0, 1, false, 1.0f, N, n_alloc_whcT, stream);
-      rowscale_kernel<<<nb_(nh), 256, 0, stream>>>(
-          (const u16*)dh_rec, cont_t, (u16*)dh_rec, nh, H);
+      COS_CHECK_HIP(hipMemsetAsync(dh_f, 0, nh * 4, stream));
+      gemm_bf16(dxg_t, w_hcT, dh_f, nullptr, N, H, 4 * H, 4 * H, 4 * H,
+                H, false, false, 2, skb, false, 1.0f, N + 128,
+                n_alloc_whcT, stream);
+      rowscale_f32_kernel<<<nb_(nh), 256, 0, stream>>>(
+          dh_f, cont_t, (u16*)dh_rec, nh, H);
     }
     float* tmp = dc_next; dc_next = dc_prev; dc_prev = tmp;
   }

@@ -613,8 +613,11 @@ def lstm_seq_forward(xg, w_hc, cont):
     h = torch.empty((T, N, H), dtype=torch.bfloat16, device=dev_)
     c = torch.empty((T, N, H), dtype=torch.float32, device=dev_)
     act = torch.empty((T, N, H4), dtype=torch.float32, device=dev_)
-    h_in = torch.empty((T, N, H), dtype=torch.bfloat16, device=dev_)
-    hg = torch.empty((N, H4), dtype=torch.bfloat16, device=dev_)
+    # +128 rows so per-step A slices keep fast DMA staging (m_alloc)
+    h_in_store = torch.empty(((T * N + 128) * H,), dtype=torch.bfloat16,
+                             device=dev_)
+    h_in = h_in_store[:T * N * H].view(T, N, H)
+    hg = torch.empty((N, H4), dtype=torch.float32, device=dev_)
     _ext.lstm_seq_fwd(xg, whc[:H4], cont_b, h, c, act, h_in, hg,
                       T, N, H, whc.shape[0])
     return h, (cont_b, h, c, act, h_in, T, N, H)
@@ -627,12 +630,15 @@ def lstm_seq_backward(dy, w_hc, cache):
     dev_ = dy.device
     dy = _as_bf16(dy).contiguous()
     whcT = _transpose(_as_bf16(w_hc).contiguous())      # [H][4H], padded rows
-    dxg = torch.empty((T, N, H4), dtype=torch.bfloat16, device=dev_)
+    dxg_store = torch.empty(((T * N + 128) * H4,), dtype=torch.bfloat16,
+                            device=dev_)
+    dxg = dxg_store[:T * N * H4].view(T, N, H4)
     dh_rec = torch.empty((N, H), dtype=torch.bfloat16, device=dev_)
+    dh_f = torch.empty((N, H), dtype=torch.float32, device=dev_)
     dc_a = torch.empty((N, H), dtype=torch.float32, device=dev_)
     dc_b = torch.empty((N, H), dtype=torch.float32, device=dev_)
-    _ext.lstm_seq_bwd(dy, whcT, cont_b, h, c, act, dxg, dh_rec, dc_a, dc_b,
-                      T, N, H, _pad128(H4))
+    _ext.lstm_seq_bwd(dy, whcT, cont_b, h, c, act, dxg, dh_rec, dh_f,
+                      dc_a, dc_b, T, N, H, _pad128(H4))
     # dw_hc = dgates_all^T @ h_in_all — one big NT GEMM
     dxg_flat = dxg.reshape(T * N, H4)
     h_in_flat = h_in.reshape(T * N, H)
