@@ -172,6 +172,20 @@ def test_concat_eltwise_slice_grads():
     run_grad_check(layer, [torch.randn(3, 4), torch.randn(3, 4)])
 
 
+def test_power_flatten_reshape_split_grads():
+    layer = make_layer("""name: "p" type: "Power" bottom: "x" top: "y"
+        power_param { power: 2.0 scale: 0.5 shift: 1.0 }""")
+    run_grad_check(layer, [torch.rand(3, 4) + 0.5])
+    layer = make_layer("""name: "f" type: "Flatten" bottom: "x" top: "y"
+        flatten_param { axis: 1 }""")
+    run_grad_check(layer, [torch.randn(2, 3, 4, 5)])
+    layer = make_layer("""name: "r" type: "Reshape" bottom: "x" top: "y"
+        reshape_param { shape { dim: 0 dim: -1 dim: 2 } }""")
+    run_grad_check(layer, [torch.randn(2, 6, 2)])
+    layer = make_layer('name: "s" type: "Split" bottom: "x" top: "a" top: "b"')
+    run_grad_check(layer, [torch.randn(3, 4)])
+
+
 def test_batchnorm_scale_grads():
     layer = make_layer('name: "bn" type: "BatchNorm" bottom: "x" top: "y"')
     run_grad_check(layer, [torch.randn(4, 3, 5, 5)], rtol=5e-3, atol=5e-4)
