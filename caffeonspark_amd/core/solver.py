@@ -223,10 +223,12 @@ class Solver:
 
         Falls back to the eager step when capture can't apply: CPU device,
         distributed callbacks (RCCL hooks enqueue per-layer collectives),
-        gradient clipping (host-side norm read), or a display interval.
-        Recaptures when the lr policy changes the rate (kernel scalars are
-        frozen into the graph).  Assumes a device-resident data layer
-        (MemoryData/CoSData reset with GPU tensors)."""
+        gradient clipping (host-side norm read), a display interval, or a
+        non-SGD/L2 solver.  The lr-policy rate flows through a device
+        scalar refreshed before each replay, and the dropout RNG seed
+        lives in device memory, so one capture serves every iteration.
+        Assumes a device-resident data layer (MemoryData/CoSData reset
+        with GPU tensors)."""
         p = self.param
         if (self.device.type != "cuda" or self.callbacks
                 or p.clip_gradients > 0 or p.iter_size > 1 or p.display):
