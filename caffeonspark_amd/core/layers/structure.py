@@ -18,11 +18,39 @@ class ConcatLayer(Layer):
         self.axis = p.axis if p.has_field("axis") or not p.has_field("concat_dim") \
             else int(p.concat_dim)
 
+    def ensure_buf(self, N, P, Q, dtype, device):
+        """Fused mode: the shared output buffer the branch convs write
+        their channel windows into (allocated by the first producer each
+        step, consumed by forward())."""
+        buf = getattr(self, "_fused_buf", None)
+        if buf is None or buf.shape[0] != N or buf.shape[2] != P:
+            buf = torch.empty((N, self._fused_ctot, P, Q), dtype=dtype,
+                              device=device,
+                              memory_format=torch.channels_last)
+            self._fused_buf = buf
+        return buf
+
     def forward(self, bottom, top):
+        if getattr(self, "_fused", False) and self._fused_buf is not None:
+            # producers already wrote their slices: adopt, no copy
+            top[0].data = self._fused_buf
+            self._fused_buf = None      # next step reallocates
+            return 0.0
         top[0].data = torch.cat([b.data for b in bottom], dim=self.axis)
         return 0.0
 
     def backward(self, top, propagate_down, bottom):
+        if getattr(self, "_fused", False):
+            # hand each branch a channel-slice VIEW — the in-place ReLU
+            # backward consumes it stride-aware, so no slicing copy
+            dy = top[0].diff
+            off = 0
+            for b, pd in zip(bottom, propagate_down):
+                k = b.data.shape[1]
+                if pd:
+                    self.acc_blob_diff(b, dy[:, off:off + k], False)
+                off += k
+            return
         sizes = [b.data.shape[self.axis] for b in bottom]
         pieces = torch.split(top[0].diff, sizes, dim=self.axis)
         for b, pd, piece in zip(bottom, propagate_down, pieces):

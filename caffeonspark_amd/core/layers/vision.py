@@ -66,10 +66,20 @@ class ConvolutionLayer(Layer):
         w = self.weight(0)
         b = self.cast(self.blobs[1].data) if self.bias_term else None
         self._ctx = {}
+        kw = {}
+        co = getattr(self, "_concat_out", None)
+        if co is not None and x.is_cuda:
+            concat_l, c_off = co
+            eff_h = (self.kh - 1) * self.dil + 1
+            eff_w = (self.kw - 1) * self.dil + 1
+            P = (x.shape[2] + 2 * self.ph - eff_h) // self.sh + 1
+            Q = (x.shape[3] + 2 * self.pw - eff_w) // self.sw + 1
+            kw["out_into"] = (concat_l.ensure_buf(
+                x.shape[0], P, Q, torch.bfloat16, x.device), c_off)
         top[0].data = ops.conv2d_forward(
             x, w, b, (self.sh, self.sw), (self.ph, self.pw),
             (self.dil, self.dil), self.groups, ctx=self._ctx,
-            relu=getattr(self, "_fuse_relu", False))
+            relu=getattr(self, "_fuse_relu", False), **kw)
         return 0.0
 
     def backward(self, top, propagate_down, bottom):

@@ -11,6 +11,8 @@ Split layers.
 
 from __future__ import annotations
 
+import os
+
 from typing import Dict, List, Optional
 
 import torch
@@ -151,6 +153,7 @@ class Net:
                 if w != 0.0:
                     self._loss_tops.append((idx, ti, w))
         self._fuse_relu_peephole()
+        self._fuse_concat_peephole()
 
     def _fuse_relu_peephole(self) -> None:
         """Fuse Conv/IP + in-place ReLU pairs on GPU: the GEMM epilogue
@@ -170,6 +173,47 @@ class Net:
                 continue
             a._fuse_relu = True
             b._fused_upstream = True
+
+    def _fuse_concat_peephole(self) -> None:
+        """Inception fusion (GPU): when every input of a channel Concat
+        is a Convolution top consumed only by that concat (after the
+        in-place-ReLU fusion), the branch convs write straight into
+        channel windows of the concat's output buffer (GEMM ldc = total
+        channels) and backward hands out channel-slice views — the
+        concat's forward copies and backward slicing disappear."""
+        if self.device.type != "cuda" or \
+                not int(os.environ.get("COS_CONCAT_FUSE", "1")):
+            return
+        # consumer counts by blob name (in-place layers don't count:
+        # they pass the blob through)
+        consumers: dict = {}
+        for lp in (l.param for l in self.layers):
+            tops = set(lp.top)
+            for bn in lp.bottom:
+                if bn not in tops:
+                    consumers[bn] = consumers.get(bn, 0) + 1
+        producer = {}
+        for l in self.layers:
+            if l.param.type == "Convolution" and l.param.top:
+                producer[l.param.top[0]] = l
+        out_names = set(self.output_blob_names()) \
+            if hasattr(self, "output_blob_names") else set()
+        for l in self.layers:
+            if l.param.type != "Concat" or getattr(l, "axis", None) != 1:
+                continue
+            convs = [producer.get(bn) for bn in l.param.bottom]
+            if (any(c is None for c in convs)
+                    or any(consumers.get(bn, 0) != 1 for bn in l.param.bottom)
+                    or any(bn in out_names for bn in l.param.bottom)
+                    or len(set(id(c) for c in convs)) != len(convs)):
+                continue
+            off = 0
+            for c in convs:
+                c._concat_out = (l, off)
+                off += c.num_output
+            l._fused = True
+            l._fused_ctot = off
+            l._fused_buf = None
 
     def _prefeed_placeholder(self, layer) -> None:
         from .layers.data import CoSDataLayer, MemoryDataLayer

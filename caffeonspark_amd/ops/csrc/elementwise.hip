@@ -71,6 +71,54 @@ __global__ void relu_bwd_kernel(const u16* __restrict__ y,
   }
 }
 
+// relu backward over a row-strided channel slice (fused-concat branch:
+// y and dy are channel windows of channels_last buffers, row stride =
+// the concat's total channel count); dx written dense [rows][C]
+__global__ void relu_bwd_strided_kernel(const u16* __restrict__ y,
+                                        const u16* __restrict__ dy,
+                                        u16* __restrict__ dx, float slope,
+                                        int64_t rows, int C, int ldy,
+                                        int lddy) {
+  int nc8 = C / 8;
+  bool vec = (C % 8 == 0);
+  int64_t total = rows * (vec ? nc8 : C);
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    if (vec) {
+      int64_t r = i / nc8;
+      int c = (int)(i % nc8) * 8;
+      u16x8 vy = *reinterpret_cast<const u16x8*>(y + r * ldy + c);
+      u16x8 vd = *reinterpret_cast<const u16x8*>(dy + r * lddy + c);
+      u16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        u16 ry = vy[j], rd = vd[j];
+        float fy = bf2f(*reinterpret_cast<const bf16*>(&ry));
+        float fd = bf2f(*reinterpret_cast<const bf16*>(&rd));
+        float g = fy > 0.f ? fd : fd * slope;
+        bf16 b = f2bf(g);
+        o[j] = *reinterpret_cast<u16*>(&b);
+      }
+      *reinterpret_cast<u16x8*>(dx + r * C + c) = o;
+    } else {
+      int64_t r = i / C;
+      int c = (int)(i % C);
+      float fy = ldbf(y + r * ldy + c);
+      float fd = ldbf(dy + r * lddy + c);
+      stbf(dx + r * C + c, fy > 0.f ? fd : fd * slope);
+    }
+  }
+}
+
+void relu_bwd_strided(const void* y, const void* dy, void* dx, float slope,
+                      int64_t rows, int C, int ldy, int lddy,
+                      hipStream_t stream) {
+  int64_t total = rows * ((C % 8 == 0) ? C / 8 : C);
+  int blocks = (int)hmin<int64_t>(4096, (total + 255) / 256);
+  relu_bwd_strided_kernel<<<blocks, 256, 0, stream>>>(
+      (const u16*)y, (const u16*)dy, (u16*)dx, slope, rows, C, ldy, lddy);
+}
+
 // -------------------------------------------------------------- dropout
 // counter-based RNG (splitmix-style hash of seed+index): reproducible,
 // stateless, good enough for dropout masks.

@@ -60,6 +60,9 @@ void relu_fwd(const void* x, void* y, float slope, int64_t n,
               hipStream_t stream);
 void relu_bwd(const void* y, const void* dy, void* dx, float slope, int64_t n,
               hipStream_t stream);
+void relu_bwd_strided(const void* y, const void* dy, void* dx, float slope,
+                      int64_t rows, int C, int ldy, int lddy,
+                      hipStream_t stream);
 void dropout_fwd(const void* x, void* y, void* mask, float ratio,
                  const void* seed, int64_t n, hipStream_t stream);
 void seed_bump(void* s, hipStream_t stream);
@@ -400,6 +403,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_fwd", &py_relu_fwd);
   m.def("relu_bwd", &py_relu_bwd);
   m.def("dropout_fwd", &py_dropout_fwd);
+  m.def("relu_bwd_strided", [](Tensor y, Tensor dy, Tensor dx, double slope,
+                               int64_t rows, int64_t C, int64_t ldy,
+                               int64_t lddy) {
+    cosamd::relu_bwd_strided(y.data_ptr(), dy.data_ptr(), dx.data_ptr(),
+                             (float)slope, rows, C, ldy, lddy,
+                             cur_stream());
+  });
   m.def("seed_bump", [](Tensor s) {
     cosamd::seed_bump(s.data_ptr(), cur_stream());
   });

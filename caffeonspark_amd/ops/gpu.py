@@ -100,7 +100,7 @@ def _conv_out_dim(h, k, s, p, dil):
 
 
 def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
-                   relu=False):
+                   relu=False, out_into=None):
     _check_bf16(x, "conv input")
     sh, sw = stride
     ph, pw = pad
@@ -128,16 +128,28 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
     wr = wrb[:Kout]
     bias_f = b.float().contiguous() if b is not None else None
 
-    y = torch.empty((N, Kout, P, Q), dtype=torch.bfloat16, device=x.device,
-                    memory_format=torch.channels_last)
     NPQ = N * P * Q
-    y2 = y.permute(0, 2, 3, 1).reshape(NPQ, Kout)  # NHWC flat alias (view)
+    if out_into is not None:
+        # fused concat: write this conv's output directly into its
+        # channel window of the shared concat buffer (GEMM ldc = the
+        # concat's total channel count) — no concat copy ever happens
+        buf, c_off = out_into
+        Ctot = buf.shape[1]
+        y = buf[:, c_off:c_off + Kout]
+        y2 = buf.permute(0, 2, 3, 1).reshape(NPQ, Ctot)[:, c_off:]
+        ldc_out = Ctot
+    else:
+        y = torch.empty((N, Kout, P, Q), dtype=torch.bfloat16,
+                        device=x.device,
+                        memory_format=torch.channels_last)
+        y2 = y.permute(0, 2, 3, 1).reshape(NPQ, Kout)  # NHWC flat alias
+        ldc_out = Kout
     # Winograd F(4x4,3x3) for stride-1 pad-1 3x3 convs: 2.25-4x fewer
     # MACs than im2col GEMM (SURVEY.md §3.6 "Winograd is a rebuild
     # addition").  The col matrix is then built lazily in backward (dW
     # still uses the im2col form).
     if (R == S == 3 and sh == sw == 1 and ph == pw == 1 and dil == 1
-            and G == 1 and C % 8 == 0 and Kout % 8 == 0
+            and G == 1 and C % 8 == 0 and Kout % 8 == 0 and out_into is None
             and int(os.environ.get("COS_WINOGRAD", "0"))):
         # measured on MI355X: ~1-5% slower than the tuned direct im2col
         # GEMM on AlexNet/GoogLeNet (these shapes are staging-bound, so
@@ -160,7 +172,7 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
               dil == 1 and G == 1 and C % 8 == 0)
     if is_1x1:
         x2 = xl.permute(0, 2, 3, 1).reshape(NPQ, C)
-        _gemm(x2, wr, y2, bias_f, NPQ, Kout, C, C, C, Kout,
+        _gemm(x2, wr, y2, bias_f, NPQ, Kout, C, C, C, ldc_out,
               False, False, 0, 1, relu=relu, na=wrb.shape[0])
         col = None
     else:
@@ -172,7 +184,7 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
             # C[npq, kout_g] — write into the column slice of NHWC y
             _gemm(col[g], wr[g * Kg:(g + 1) * Kg], y2[:, g * Kg:],
                   bias_f[g * Kg:(g + 1) * Kg] if bias_f is not None else None,
-                  NPQ, Kg, Kpad, Kpad, Kpad, Kout, False, False, 0, 1,
+                  NPQ, Kg, Kpad, Kpad, Kpad, ldc_out, False, False, 0, 1,
                   relu=relu, na=wrb.shape[0] - g * Kg)
     if ctx is not None:
         ctx["col"] = col
@@ -376,7 +388,33 @@ def relu_forward(x, negative_slope=0.0):
     return y
 
 
+def _cl_slice_ld(t):
+    """If t is a channel-slice view of a channels_last buffer (fused
+    concat branch), return its row stride; None otherwise."""
+    if t.dim() != 4 or t.stride(1) != 1:
+        return None
+    N, C, H, W = t.shape
+    ld = t.stride(3)
+    if ld >= C and t.stride(2) == W * ld and t.stride(0) == H * W * ld:
+        return ld
+    return None
+
+
 def relu_backward(y, dy, negative_slope=0.0):
+    ldy = _cl_slice_ld(y)
+    if ldy is not None and y.dtype == torch.bfloat16:
+        dy = dy if dy.dtype == torch.bfloat16 else dy.to(torch.bfloat16)
+        lddy = _cl_slice_ld(dy)
+        if lddy is None:
+            dy = _cl(dy.reshape(y.shape))
+            lddy = y.shape[1]
+        N, C, H, W = y.shape
+        dx = torch.empty((N, C, H, W), dtype=torch.bfloat16,
+                         device=y.device,
+                         memory_format=torch.channels_last)
+        _ext.relu_bwd_strided(y, dy, dx.permute(0, 2, 3, 1), negative_slope,
+                              N * H * W, C, ldy, lddy)
+        return dx
     dy = dy if dy.dtype == torch.bfloat16 else dy.to(torch.bfloat16)
     dx = torch.empty_like(y)
     _ext.relu_bwd(y, dy.reshape(y.shape), dx, negative_slope)
