@@ -23,7 +23,8 @@ class ConcatLayer(Layer):
         their channel windows into (allocated by the first producer each
         step, consumed by forward())."""
         buf = getattr(self, "_fused_buf", None)
-        if buf is None or buf.shape[0] != N or buf.shape[2] != P:
+        if buf is None or buf.shape[0] != N or buf.shape[2] != P \
+                or buf.shape[3] != Q:
             buf = torch.empty((N, self._fused_ctot, P, Q), dtype=dtype,
                               device=device,
                               memory_format=torch.channels_last)
