@@ -169,3 +169,74 @@ def test_bucket_assembly_covers_all_params():
     for b, off in zip(s.params, s.param_offsets):
         assert off in covered and (off + b.count - 1) in covered
     assert len(sync.layer_bucket) == len(s.layer_slices)
+
+
+def _worker_frozen(rank, ws, store_path, q):
+    import torch.distributed as dist
+
+    from caffeonspark_amd.core.solver import Solver
+    from caffeonspark_amd.parallel import DistributedSync
+    from caffeonspark_amd.proto import caffe_pb, text_format
+
+    net_text = """
+      layer { name: "d" type: "MemoryData" top: "x" top: "t"
+              memory_data_param { batch_size: 32 channels: 3 height: 8
+                                  width: 8 } }
+      layer { name: "c" type: "Convolution" bottom: "x" top: "y"
+              param { lr_mult: 0.0 } param { lr_mult: 0.0 }
+              convolution_param { num_output: 8 kernel_size: 3 pad: 1
+                weight_filler { type: "gaussian" std: 0.1 } } }
+      layer { name: "ip" type: "InnerProduct" bottom: "y" top: "z"
+              inner_product_param { num_output: 4
+                weight_filler { type: "gaussian" std: 0.1 } } }
+      layer { name: "l" type: "SoftmaxWithLoss" bottom: "z" bottom: "t"
+              top: "loss" }
+    """
+    dist.init_process_group("gloo", rank=rank, world_size=ws,
+                            init_method=f"file://{store_path}")
+    try:
+        sp = caffe_pb.SolverParameter(
+            net_param=text_format.parse(net_text, caffe_pb.NetParameter),
+            base_lr=0.1, momentum=0.9, lr_policy="fixed", max_iter=10,
+            random_seed=5 + rank)   # deliberately different init per rank
+        s = Solver(sp)
+        sync = DistributedSync(s, bucket_mb=0.001)  # many small buckets
+        sync.broadcast_params()
+        frozen0 = s.params[0].data.clone()
+        g = torch.Generator().manual_seed(200 + rank)
+        for _ in range(4):
+            x = torch.randn(32, 3, 8, 8, generator=g)
+            y = torch.randint(0, 4, (32,), generator=g).float()
+            s.net.data_layers()[0].reset(x, y)
+            s._step_one()
+        q.put((rank, s.flat_w.clone(), frozen0, s.params[0].data.clone()))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ddp_frozen_params_stay_synchronized():
+    """lr_mult=0 params inside all-reduce buckets: ranks stay bit-identical
+    and the frozen blob never moves from its broadcast value."""
+    _retry(_run_frozen)
+
+
+def _run_frozen():
+    import tempfile
+    import uuid
+    ws = 2
+    store = tempfile.mktemp(prefix=f"cosamd_ddpf_{uuid.uuid4().hex}_")
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_worker_frozen, args=(r, ws, store, q))
+             for r in range(ws)]
+    for p in procs:
+        p.start()
+    out = {}
+    for _ in range(ws):
+        rank, flat_w, frozen0, frozen1 = q.get()
+        out[rank] = (flat_w, frozen0, frozen1)
+    for p in procs:
+        p.join(timeout=60)
+    torch.testing.assert_close(out[0][0], out[1][0], rtol=0, atol=0)
+    for r in range(ws):
+        torch.testing.assert_close(out[r][1], out[r][2], rtol=0, atol=0)
