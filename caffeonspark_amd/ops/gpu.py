@@ -262,8 +262,22 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                       Kg, Kpad, NPQ, Kout, Kpad if not is_1x1 else C,
                       Kpad, True, True, 2, sk)
             else:
-                if is_1x1 or not int(os.environ.get("COS_DW_IM2COLT", "0")):
-                    colT = _transpose(x2 if is_1x1 else col[g])
+                if is_1x1:
+                    # inception: several 1x1 branch convs share one
+                    # input — transpose it once per step, cached on the
+                    # base tensor (version-keyed like _cl)
+                    base = ctx["xl"]
+                    cached = getattr(base, "_cos_T", None)
+                    if cached is not None and cached[0] == base._version:
+                        colT = cached[1]
+                    else:
+                        colT = _transpose(x2)
+                        try:
+                            base._cos_T = (base._version, colT)
+                        except AttributeError:
+                            pass
+                elif not int(os.environ.get("COS_DW_IM2COLT", "0")):
+                    colT = _transpose(col[g])
                 else:
                     # build colT[k][npq] straight from the input (skips a
                     # pass over the col matrix but re-reads the input R*S
