@@ -160,3 +160,31 @@ def test_simulator_runs(capsys):
     simulator.main(["-iters", "8", "-batch", "4", "-size", "32"])
     out = capsys.readouterr().out
     assert "images/sec" in out or "img/s" in out or out.strip()
+
+
+def test_all_modules_import():
+    """Every package module imports cleanly (catches stale references
+    after refactors)."""
+    import importlib
+    import pkgutil
+
+    import caffeonspark_amd
+    skipped = []
+    for m in pkgutil.walk_packages(caffeonspark_amd.__path__,
+                                   prefix="caffeonspark_amd."):
+        try:
+            importlib.import_module(m.name)
+        except ImportError as e:   # optional deps only
+            skipped.append((m.name, str(e)))
+    assert not skipped, f"import failures: {skipped}"
+
+
+def test_display_utils():
+    """DisplayUtils analog renders an image/label table."""
+    import pandas as pd
+
+    from caffeonspark_amd.api.display import df_to_html
+    df = pd.DataFrame([{"data": b"xx", "label": 3},
+                       {"data": None, "label": 1}])
+    html = df_to_html(df, text_col="label")
+    assert "<table>" in html and "<img" in html and ">3<" in html
