@@ -81,3 +81,40 @@ def test_transformer_batch_shape():
     out = xf.transform(imgs)
     assert isinstance(out, torch.Tensor)
     assert out.shape == (4, 3, 8, 8)
+
+
+def test_lmdb_source_rank_sharding(tmp_path):
+    """sample_iter(rank, world): ranks see disjoint round-robin shards
+    whose union is the full dataset (reference: LmdbRDD partitioning)."""
+    from caffeonspark_amd.data.lmdb_source import LMDBSource
+    from caffeonspark_amd.proto import caffe_pb as pb
+
+    path = str(tmp_path / "db")
+    items = []
+    for i in range(17):
+        d = pb.Datum(channels=1, height=2, width=2, label=i % 5,
+                     data=bytes([i] * 4))
+        items.append((f"{i:04d}".encode(), d.SerializeToString()))
+    LmdbWriter(path).write(items)
+
+    class _Conf:
+        lmdb_partitions = 0
+
+    class _LP:
+        class memory_data_param:
+            source = path
+            batch_size = 4
+            channels, height, width = 1, 2, 2
+
+        def __getattr__(self, k):
+            raise AttributeError(k)
+
+    src = LMDBSource.__new__(LMDBSource)
+    from caffeonspark_amd.data.lmdb_io import LmdbReader
+    src.reader = LmdbReader(path)
+    shards = [[s.id for s in src.sample_iter(rank=r, world=3, epochs=1)]
+              for r in range(3)]
+    everything = sorted(sum(shards, []))
+    assert everything == sorted(k.decode() for k, _ in items)
+    flat = sum(shards, [])
+    assert len(set(flat)) == len(flat)   # disjoint
