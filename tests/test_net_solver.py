@@ -223,3 +223,30 @@ def test_lenet_converges_synthetic():
     loss = float(test_net.blob_by_name("loss").data)
     assert acc > 0.8, f"accuracy {acc}"
     assert loss < 0.5, f"loss {loss}"
+
+
+def test_graph_step_cpu_fallback():
+    """graph_step on CPU (or any non-capturable config) must silently run
+    the eager step."""
+    net_text = """
+      layer { name: "d" type: "MemoryData" top: "x" top: "t"
+              memory_data_param { batch_size: 8 channels: 4 height: 1
+                                  width: 1 } }
+      layer { name: "ip" type: "InnerProduct" bottom: "x" top: "y"
+              inner_product_param { num_output: 3
+                weight_filler { type: "gaussian" std: 0.1 } } }
+      layer { name: "l" type: "SoftmaxWithLoss" bottom: "y" bottom: "t"
+              top: "loss" }
+    """
+    sp = caffe_pb.SolverParameter(
+        net_param=text_format.parse(net_text, caffe_pb.NetParameter),
+        base_lr=0.1, lr_policy="fixed", max_iter=10, random_seed=2)
+    s = Solver(sp)
+    g = torch.Generator().manual_seed(0)
+    x = torch.randn(8, 4, 1, 1, generator=g)
+    t_ = torch.randint(0, 3, (8,), generator=g).float()
+    s.net.data_layers()[0].reset(x, t_)
+    l1 = s.graph_step()
+    l2 = s.graph_step()
+    assert s.iter == 2
+    assert l1 > 0 and l2 > 0
