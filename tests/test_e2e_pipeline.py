@@ -160,6 +160,19 @@ def test_features_extraction(workdir):
     assert os.path.exists(out)
     CaffeProcessor.reset_instance()
 
+    # parquet output format (reference -outputFormat parquet path)
+    pytest.importorskip("pyarrow")
+    out_pq = str(workdir / "features.parquet")
+    conf2 = Config(["-conf", _solver_file(workdir, "solver4b.prototxt"),
+                    "-features", "ip2", "-label", "label",
+                    "-output", out_pq, "-outputFormat", "parquet"])
+    cos2 = CaffeOnSpark(conf2)
+    cos2.features(max_samples=20)
+    import pandas as pd
+    back = pd.read_parquet(out_pq)
+    assert "ip2" in back.columns and len(back) >= 20
+    CaffeProcessor.reset_instance()
+
 
 def test_seqfile_source_roundtrip(workdir, tmp_path):
     """SourceTest analog: LMDB -> SequenceFile -> read back."""
