@@ -142,3 +142,22 @@ def test_merge_semantics():
     b.CopyFrom(a)
     b.layer[0].name = "changed"
     assert a.layer[0].name == "l1"  # deep copy
+
+
+def test_text_format_string_escapes():
+    """Escaped strings round-trip through the text codec (caffe prototxts
+    in the wild use quotes and backslashes in paths)."""
+    from caffeonspark_amd.proto import caffe_pb, text_format
+    p = caffe_pb.NetParameter(name='weird "name"\\with\tescapes\n')
+    txt = text_format.dumps(p)
+    back = text_format.parse(txt, caffe_pb.NetParameter)
+    assert back.name == p.name
+
+
+def test_bytes_field_roundtrip():
+    """Datum.data (bytes) with all byte values survives text + binary."""
+    from caffeonspark_amd.proto import caffe_pb
+    raw = bytes(range(256))
+    d = caffe_pb.Datum(channels=1, height=16, width=16, label=3, data=raw)
+    back = caffe_pb.Datum.FromString(d.SerializeToString())
+    assert bytes(back.data) == raw and back.label == 3
