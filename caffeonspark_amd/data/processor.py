@@ -58,10 +58,13 @@ class CaffeProcessor:
         self.dtype = conf.dtype
         self.solver = Solver(conf.solver_param, device=self.device,
                              dtype=self.dtype, proto_dir=conf.proto_dir)
-        if conf.weights:
-            self.solver.load_weights(conf.weights)
+        # order matters: reference setLearnedNet (CaffeNet.cpp:333-365)
+        # rewrites the state's learned_net pointer so an explicit
+        # -weights file WINS over the snapshot's own model
         if conf.snapshot_state:
             self.solver.restore(conf.snapshot_state)
+        if conf.weights:
+            self.solver.load_weights(conf.weights)
         self.queues = [QueuePair() for _ in sources]
         self.threads: List[threading.Thread] = []
         self.solver_thread: Optional[threading.Thread] = None

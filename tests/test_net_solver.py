@@ -250,3 +250,29 @@ def test_graph_step_cpu_fallback():
     l2 = s.graph_step()
     assert s.iter == 2
     assert l1 > 0 and l2 > 0
+
+
+def test_resume_with_explicit_weights(tmp_path):
+    """-snapshot + -weights: the explicit weights win over the state's
+    learned_net (reference setLearnedNet semantics)."""
+    os.chdir(tmp_path)
+    try:
+        root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        s = solver_from_prototxt(os.path.join(root, LENET_SOLVER))
+        g = torch.Generator().manual_seed(0)
+        x, y = synth_batch(64, g)
+        s.net.data_layers()[0].reset(x, y)
+        s.step(3)
+        s.snapshot()
+        state3 = s.snapshot_filename("state")
+        s.step(3)
+        model6 = s.snapshot()
+        w6 = s.flat_w.clone()
+
+        s2 = solver_from_prototxt(os.path.join(root, LENET_SOLVER))
+        s2.restore(state3)          # state from iter 3
+        s2.load_weights(model6)     # explicit newer weights win
+        assert s2.iter == 3
+        torch.testing.assert_close(s2.flat_w, w6)
+    finally:
+        os.chdir(os.path.dirname(os.path.dirname(__file__)))
