@@ -126,8 +126,19 @@ class Blob:
             src = torch.from_numpy(
                 np.asarray(proto.data, dtype=np.float32).copy()
             ).reshape(shape)
-        self.data = src.to(device=device, dtype=dtype)
+        src = src.to(device=device, dtype=dtype)
+        if list(self.data.shape) == shape:
+            # copy IN PLACE: param blobs are views of the solver's flat
+            # arena — reassignment would silently detach them from the
+            # optimizer/all-reduce/bf16-shadow paths
+            self.data.copy_(src)
+        else:
+            self.data = src
         if proto.has_field("diff"):
-            self.diff = torch.tensor(proto.diff, dtype=torch.float32).reshape(
+            d = torch.tensor(proto.diff, dtype=torch.float32).reshape(
                 shape).to(device=device, dtype=dtype)
+            if self.diff is not None and list(self.diff.shape) == shape:
+                self.diff.copy_(d)
+            else:
+                self.diff = d
         return self
