@@ -276,3 +276,34 @@ def test_resume_with_explicit_weights(tmp_path):
         torch.testing.assert_close(s2.flat_w, w6)
     finally:
         os.chdir(os.path.dirname(os.path.dirname(__file__)))
+
+
+def test_finetune_then_train_updates_weights(tmp_path):
+    """Regression: load_weights must keep param blobs as arena views so
+    subsequent training actually moves the weights the net reads."""
+    os.chdir(tmp_path)
+    try:
+        root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        s = solver_from_prototxt(os.path.join(root, LENET_SOLVER))
+        g = torch.Generator().manual_seed(1)
+        x, y = synth_batch(64, g)
+        s.net.data_layers()[0].reset(x, y)
+        s.step(2)
+        model = s.snapshot()
+
+        s2 = solver_from_prototxt(os.path.join(root, LENET_SOLVER))
+        s2.load_weights(model)
+        # blobs must still alias the arena
+        assert s2.params[0].data.data_ptr() >= s2.flat_w.data_ptr()
+        assert s2.params[0].data.data_ptr() < \
+            s2.flat_w.data_ptr() + s2.flat_w.numel() * 4
+        w0 = s2.flat_w.clone()
+        s2.net.data_layers()[0].reset(x, y)
+        loss_a = s2._step_one()
+        assert not torch.equal(w0, s2.flat_w), "training froze after load"
+        # and the net's own view moved too
+        assert not torch.equal(
+            w0.narrow(0, 0, s2.params[0].count),
+            s2.params[0].data.reshape(-1))
+    finally:
+        os.chdir(os.path.dirname(os.path.dirname(__file__)))
