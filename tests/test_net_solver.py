@@ -307,3 +307,47 @@ def test_finetune_then_train_updates_weights(tmp_path):
             s2.params[0].data.reshape(-1))
     finally:
         os.chdir(os.path.dirname(os.path.dirname(__file__)))
+
+
+def test_iter_size_accumulation():
+    """iter_size=2 over two half-batches equals one update with the
+    averaged gradient (caffe gradient-scaling rule, CaffeNet.cpp:620)."""
+    net_text = """
+      layer { name: "d" type: "MemoryData" top: "x" top: "t"
+              memory_data_param { batch_size: 16 channels: 6 height: 1
+                                  width: 1 } }
+      layer { name: "ip" type: "InnerProduct" bottom: "x" top: "y"
+              inner_product_param { num_output: 3
+                weight_filler { type: "gaussian" std: 0.1 } } }
+      layer { name: "l" type: "SoftmaxWithLoss" bottom: "y" bottom: "t"
+              top: "loss" }
+    """
+    g = torch.Generator().manual_seed(3)
+    x = torch.randn(32, 6, 1, 1, generator=g)
+    t_ = torch.randint(0, 3, (32,), generator=g).float()
+
+    def make(iter_size):
+        sp = caffe_pb.SolverParameter(
+            net_param=text_format.parse(net_text, caffe_pb.NetParameter),
+            base_lr=0.5, momentum=0.0, weight_decay=0.0,
+            lr_policy="fixed", max_iter=4, random_seed=11,
+            iter_size=iter_size)
+        return Solver(sp)
+
+    s = make(2)
+    s.net.data_layers()[0].reset(x, t_)
+    w0 = s.flat_w.clone()
+    s._step_one()                         # two micro-batches of 16
+    dw_acc = s.flat_w - w0
+
+    # manual: average of the two half-batch gradients
+    s2 = make(1)
+    grads = []
+    for half in (slice(0, 16), slice(16, 32)):
+        s2.net.zero_param_diffs()
+        s2.flat_w.copy_(w0)
+        s2.net.data_layers()[0].reset(x[half], t_[half])
+        s2.net.forward_backward()
+        grads.append(s2.flat_g.clone())
+    expected = -0.5 * 0.5 * (grads[0] + grads[1])   # lr * avg grad
+    torch.testing.assert_close(dw_acc, expected, rtol=1e-4, atol=1e-6)
