@@ -351,3 +351,49 @@ def test_iter_size_accumulation():
         grads.append(s2.flat_g.clone())
     expected = -0.5 * 0.5 * (grads[0] + grads[1])   # lr * avg grad
     torch.testing.assert_close(dw_acc, expected, rtol=1e-4, atol=1e-6)
+
+
+def test_gradient_clipping_and_l1():
+    """clip_gradients rescales by global L2 norm; L1 regularization adds
+    sign(w)*decay (reference SGDSolver::ClipGradients / Regularize)."""
+    net_text = """
+      layer { name: "d" type: "MemoryData" top: "x" top: "t"
+              memory_data_param { batch_size: 8 channels: 5 height: 1
+                                  width: 1 } }
+      layer { name: "ip" type: "InnerProduct" bottom: "x" top: "y"
+              inner_product_param { num_output: 2
+                weight_filler { type: "gaussian" std: 0.5 } } }
+      layer { name: "l" type: "SoftmaxWithLoss" bottom: "y" bottom: "t"
+              top: "loss" }
+    """
+    g = torch.Generator().manual_seed(4)
+    x = torch.randn(8, 5, 1, 1, generator=g) * 5
+    t_ = torch.randint(0, 2, (8,), generator=g).float()
+
+    def run(**kw):
+        sp = caffe_pb.SolverParameter(
+            net_param=text_format.parse(net_text, caffe_pb.NetParameter),
+            base_lr=1.0, momentum=0.0, lr_policy="fixed", max_iter=4,
+            random_seed=9, **kw)
+        s = Solver(sp)
+        s.net.data_layers()[0].reset(x, t_)
+        w0 = s.flat_w.clone()
+        s.net.zero_param_diffs()
+        s.net.forward_backward()
+        grad = s.flat_g.clone()
+        s.apply_update()
+        return w0, grad, s
+
+    # clipping: with a tiny threshold the applied step has that norm
+    clip = 0.01
+    w0, grad, s = run(weight_decay=0.0, clip_gradients=clip)
+    step = w0 - s.flat_w
+    norm = float(grad.norm())
+    if norm > clip:
+        assert abs(float(step.norm()) - clip) / clip < 1e-3
+
+    # L1: update includes sign(w) * decay
+    wd = 0.05
+    w0, grad, s = run(weight_decay=wd, regularization_type="L1")
+    expected = w0 - (grad + wd * torch.sign(w0))
+    torch.testing.assert_close(s.flat_w, expected, rtol=1e-4, atol=1e-6)
