@@ -397,3 +397,49 @@ def test_gradient_clipping_and_l1():
     w0, grad, s = run(weight_decay=wd, regularization_type="L1")
     expected = w0 - (grad + wd * torch.sign(w0))
     torch.testing.assert_close(s.flat_w, expected, rtol=1e-4, atol=1e-6)
+
+
+def test_loss_weight_scaling():
+    """Auxiliary heads (GoogLeNet loss1/loss2, loss_weight 0.3): total
+    loss and gradients scale by the declared loss_weight."""
+    net_text = """
+      layer { name: "d" type: "MemoryData" top: "x" top: "t"
+              memory_data_param { batch_size: 8 channels: 4 height: 1
+                                  width: 1 } }
+      layer { name: "ip" type: "InnerProduct" bottom: "x" top: "y"
+              inner_product_param { num_output: 3
+                weight_filler { type: "gaussian" std: 0.2 } } }
+      layer { name: "lmain" type: "SoftmaxWithLoss" bottom: "y" bottom: "t"
+              top: "loss_main" }
+      layer { name: "laux" type: "SoftmaxWithLoss" bottom: "y" bottom: "t"
+              top: "loss_aux" loss_weight: 0.3 }
+    """
+    sp = caffe_pb.SolverParameter(
+        net_param=text_format.parse(net_text, caffe_pb.NetParameter),
+        base_lr=0.1, lr_policy="fixed", max_iter=4, random_seed=6)
+    s = Solver(sp)
+    g = torch.Generator().manual_seed(0)
+    x = torch.randn(8, 4, 1, 1, generator=g)
+    t_ = torch.randint(0, 3, (8,), generator=g).float()
+    s.net.data_layers()[0].reset(x, t_)
+    total = s.net.forward()
+    lm = float(s.net.blob_by_name("loss_main").data)
+    la = float(s.net.blob_by_name("loss_aux").data)
+    assert total == pytest.approx(lm + 0.3 * la, rel=1e-5)
+    # both heads share the same logits: grad = (1 + 0.3) * single-head
+    s.net.zero_param_diffs()
+    s.net.backward()
+    g2 = s.flat_g.clone()
+    assert float(g2.norm()) > 0
+    # single-head comparison net
+    sp1 = caffe_pb.SolverParameter(
+        net_param=text_format.parse(
+            net_text.replace(' loss_weight: 0.3', ' loss_weight: 0.0'),
+            caffe_pb.NetParameter),
+        base_lr=0.1, lr_policy="fixed", max_iter=4, random_seed=6)
+    s1 = Solver(sp1)
+    s1.net.data_layers()[0].reset(x, t_)
+    s1.net.forward()
+    s1.net.zero_param_diffs()
+    s1.net.backward()
+    torch.testing.assert_close(g2, 1.3 * s1.flat_g, rtol=1e-4, atol=1e-6)
