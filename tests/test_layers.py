@@ -239,3 +239,20 @@ def test_accuracy():
     layer.setup([bx, bt], top)
     layer.forward([bx, bt], top)
     assert top[0].data.item() == pytest.approx(2.0 / 3.0)
+
+
+def test_accuracy_top_k():
+    """Accuracy with top_k > 1 (reference AccuracyLayer top-k argmax)."""
+    layer = make_layer("""name: "a" type: "Accuracy" bottom: "x" bottom: "t"
+        top: "acc" accuracy_param { top_k: 2 }""")
+    x = torch.tensor([[0.1, 0.9, 0.5],     # top2 = {1, 2}
+                      [0.8, 0.1, 0.7],     # top2 = {0, 2}
+                      [0.3, 0.2, 0.1]])    # top2 = {0, 1}
+    t_ = torch.tensor([2.0, 1.0, 0.0])     # hits: yes, no, yes
+    from caffeonspark_amd.core.blob import Blob
+    bx, bt = Blob(x.shape), Blob(t_.shape)
+    bx.data, bt.data = x, t_
+    top = [Blob([0])]
+    layer.setup([bx, bt], top)
+    layer.forward([bx, bt], top)
+    assert float(top[0].data) == pytest.approx(2.0 / 3.0)
