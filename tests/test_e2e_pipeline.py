@@ -380,23 +380,13 @@ def test_vector_mean_and_config_parse():
     col = pd.Series([[1.0, 2.0], [3.0, 4.0], [5.0, 6.0]])
     assert vector_mean(col) == [3.0, 4.0]
 
-    conf = Config(["-conf", "x.prototxt", "-train", "-devices", "4",
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    solver = os.path.join(root, "caffeonspark_amd", "models",
+                          "lenet_memory_solver.prototxt")
+    conf = Config(["-conf", solver, "-train", "-devices", "4",
                    "-clusterSize", "2", "-connection", "ethernet",
                    "-outputFormat", "parquet", "-lmdb_partitions", "7",
-                   "-captionLength", "15", "-vocabSize", "999"],
-                  parse_only=True) if "parse_only" in \
-        Config.__init__.__code__.co_varnames else None
-    if conf is None:
-        import inspect
-        # fall back: parse without proto resolution by pointing at a
-        # real solver file
-        root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-        solver = os.path.join(root, "caffeonspark_amd", "models",
-                              "lenet_memory_solver.prototxt")
-        conf = Config(["-conf", solver, "-train", "-devices", "4",
-                       "-clusterSize", "2", "-connection", "ethernet",
-                       "-outputFormat", "parquet", "-lmdb_partitions", "7",
-                       "-captionLength", "15", "-vocabSize", "999"])
+                   "-captionLength", "15", "-vocabSize", "999"])
     assert conf.devices == 4
     assert conf.clusterSize == 2
     assert conf.outputFormat == "parquet"
