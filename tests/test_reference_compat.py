@@ -14,13 +14,19 @@ import pytest
 from caffeonspark_amd.proto import caffe_pb, text_format
 
 REF = "/root/reference/data"
+REF_DIRS = [REF,
+            "/root/reference/caffe-grid/src/test/resources",
+            "/root/reference/caffe-distri/src/test/resources"]
 
 pytestmark = pytest.mark.skipif(not os.path.isdir(REF),
                                 reason="reference mount not present")
 
 
 def _files(pattern):
-    return sorted(glob.glob(os.path.join(REF, pattern)))
+    out = []
+    for d in REF_DIRS:
+        out += glob.glob(os.path.join(d, pattern))
+    return sorted(out)
 
 
 @pytest.mark.parametrize("path", _files("*solver*.prototxt"))
