@@ -74,3 +74,28 @@ def test_reference_net_parses_and_builds(path):
         pytest.skip("no feedable data layer in TRAIN phase (deploy net)")
     net = Net(np_, state)
     assert len(net.layers) == len(filtered.layer)
+
+
+def test_reference_caffenet_trains_cpu():
+    """The reference's own JNI-test net (CaffeNetTest.testTrain analog,
+    CaffeNetTest.java:271-320): build from its caffenet_solver.prototxt
+    and run real CPU train steps; loss must be finite and move."""
+    import math
+
+    import torch
+
+    from caffeonspark_amd.core.solver import Solver
+    sf = os.path.join("/root/reference/caffe-distri/src/test/resources",
+                      "caffenet_solver.prototxt")
+    sp = text_format.parse_file(sf, caffe_pb.SolverParameter)
+    sp.display = 0
+    s = Solver(sp, proto_dir=os.path.dirname(sf))
+    dl = s.net.data_layers()[0]
+    g = torch.Generator().manual_seed(0)
+    x = torch.randn(4, 3, 227, 227, generator=g)
+    y = torch.randint(0, 2, (4,), generator=g).float()  # 2-class test net
+    dl.reset(x, y)
+    l0 = s._step_one()
+    l1 = s._step_one()
+    assert math.isfinite(l0) and math.isfinite(l1)
+    assert s.iter == 2
