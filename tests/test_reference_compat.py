@@ -99,3 +99,33 @@ def test_reference_caffenet_trains_cpu():
     l1 = s._step_one()
     assert math.isfinite(l0) and math.isfinite(l1)
     assert s.iter == 2
+
+
+def test_reference_coco_fixture_pipeline(tmp_path):
+    """Run the COCO converter + vocab + embedding pipeline on the
+    reference's OWN coco fixtures (ToolTest.scala:36-135 analog)."""
+    pytest.importorskip("pandas")
+    res = "/root/reference/caffe-grid/src/test/resources"
+    if not os.path.exists(os.path.join(res, "coco.json")):
+        pytest.skip("fixture missing")
+    from caffeonspark_amd.tools.coco import coco_to_dataframe, \
+        embed_captions
+    from caffeonspark_amd.tools.vocab import Vocab
+
+    df_out = str(tmp_path / "captions.parquet")
+    n = coco_to_dataframe(os.path.join(res, "coco.json"), res, df_out)
+    assert n >= 2
+    import pandas as pd
+    df = pd.read_parquet(df_out)
+    assert {"id", "data", "caption"} <= set(df.columns)
+
+    vocab_path = str(tmp_path / "vocab.json")
+    v = Vocab.build([c for c in df["caption"]], size=100)
+    v.save(vocab_path)
+    assert len(v.words) > 5
+
+    emb_out = str(tmp_path / "embedded.parquet")
+    embed_captions(df_out, vocab_path, emb_out, caption_length=20)
+    emb = pd.read_parquet(emb_out)
+    assert "input_sentence" in emb.columns or "caption_ids" in emb.columns \
+        or len(emb.columns) >= 2
