@@ -129,3 +129,54 @@ def test_reference_coco_fixture_pipeline(tmp_path):
     emb = pd.read_parquet(emb_out)
     assert "input_sentence" in emb.columns or "caption_ids" in emb.columns \
         or len(emb.columns) >= 2
+
+
+def test_reference_images_seqfile_train(tmp_path):
+    """SourceTest.scala analog on the reference's own 4 jpgs: build a
+    SequenceFile with Binary2Sequence, feed it through SeqImageSource
+    into the reference's caffenet test net, and step the solver."""
+    imgs = "/root/reference/data/images"
+    if not os.path.exists(os.path.join(imgs, "labels.txt")):
+        pytest.skip("fixture missing")
+    try:
+        import PIL  # noqa: F401
+    except ImportError:
+        pytest.skip("PIL needed for decode")
+    from caffeonspark_amd.api import CaffeOnSpark, Config
+    from caffeonspark_amd.data.processor import CaffeProcessor
+    from caffeonspark_amd.tools.converters import binary2sequence
+
+    seq = str(tmp_path / "imgs.seq")
+    n = binary2sequence(imgs, os.path.join(imgs, "labels.txt"), seq)
+    assert n == 4
+
+    # caffenet test net reads a MemoryData source_class SeqImage source;
+    # clone the reference solver/net with our source wiring
+    res = "/root/reference/caffe-distri/src/test/resources"
+    npar = text_format.parse(
+        open(os.path.join(res, "caffenet_train_net.prototxt")).read(),
+        caffe_pb.NetParameter)
+    for lp in npar.layer:
+        if lp.type == "MemoryData":
+            lp.source_class = "com.yahoo.ml.caffe.SeqImageDataSource"
+            lp.memory_data_param.source = seq
+            lp.memory_data_param.batch_size = 2
+    sp = text_format.parse_file(os.path.join(res, "caffenet_solver.prototxt"),
+                                caffe_pb.SolverParameter)
+    sp.net_param = npar
+    sp.max_iter = 3
+    sp.display = 0
+    sp.snapshot_after_train = True
+    sp.snapshot_prefix = str(tmp_path / "refimg")
+    solver_file = tmp_path / "solver.prototxt"
+    solver_file.write_text(text_format.dumps(sp))
+
+    CaffeProcessor.reset_instance()
+    os.chdir(tmp_path)
+    conf = Config(["-conf", str(solver_file), "-train", "-label", "label"])
+    cos = CaffeOnSpark(conf)
+    cos.train()
+    # training ran to max_iter and snapshotted from real decoded jpgs
+    snaps = [f for f in os.listdir(tmp_path) if ".caffemodel" in f]
+    assert snaps, "training did not complete/snapshot"
+    CaffeProcessor.reset_instance()
