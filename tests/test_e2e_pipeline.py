@@ -394,3 +394,25 @@ def test_vector_mean_and_config_parse():
     assert conf.captionLength == 15
     assert conf.vocabSize == 999
     assert conf.isTraining
+
+
+def test_cli_main_train_and_test(workdir, capsys):
+    """The reference CLI entry (CaffeOnSpark.main): -train then -test
+    through the module main()."""
+    from caffeonspark_amd.api import caffe_on_spark
+    CaffeProcessor.reset_instance()
+    os.chdir(workdir)
+    caffe_on_spark.main(["-conf",
+                         _solver_file(workdir, "solver_cli.prototxt",
+                                      max_iter=60),
+                         "-train", "-label", "label"])
+    snaps = sorted(f for f in os.listdir(workdir)
+                   if f.endswith(".caffemodel"))
+    assert snaps
+    CaffeProcessor.reset_instance()
+    caffe_on_spark.main(["-conf",
+                         _solver_file(workdir, "solver_cli2.prototxt"),
+                         "-test", "-weights", str(workdir / snaps[-1])])
+    out = capsys.readouterr().out
+    assert "accuracy" in out
+    CaffeProcessor.reset_instance()
