@@ -188,3 +188,18 @@ def test_display_utils():
                        {"data": None, "label": 1}])
     html = df_to_html(df, text_col="label")
     assert "<table>" in html and "<img" in html and ">3<" in html
+
+
+def test_time_net_per_layer(capsys):
+    """caffe-time analog: per-layer fwd/bwd timing table."""
+    import os
+
+    from caffeonspark_amd.tools import time_net
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    time_net.main(["-conf",
+                   os.path.join(root, "caffeonspark_amd", "models",
+                                "lenet_memory_solver.prototxt"),
+                   "-iters", "2", "-batch", "8"])
+    out = capsys.readouterr().out
+    assert "TOTAL" in out
+    assert "conv1" in out and "ip2" in out
