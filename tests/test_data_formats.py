@@ -118,3 +118,40 @@ def test_lmdb_source_rank_sharding(tmp_path):
     assert everything == sorted(k.decode() for k, _ in items)
     flat = sum(shards, [])
     assert len(set(flat)) == len(flat)   # disjoint
+
+
+def test_cos_top_type_fills():
+    """CoSDataFrameSource._fill_top covers all typed tops (reference
+    DataFrameSource.Top, DataFrameSource.scala:315-353): scalars, arrays
+    with transpose (time-major), and RAW_IMAGE."""
+    from caffeonspark_amd.data.dataframe_source import CoSDataFrameSource
+    from caffeonspark_amd.proto import caffe_pb
+
+    src = CoSDataFrameSource.__new__(CoSDataFrameSource)
+    src.transformers = [None] * 4
+    T = caffe_pb.CoSTopType
+    rows = [{"lab": 3, "seq": [1, 2, 3], "img": bytes(range(12)),
+             "score": 0.5},
+            {"lab": 1, "seq": [4], "img": bytes(range(12, 24)),
+             "score": 1.5}]
+
+    cfg = caffe_pb.CoSTopParameter(name="lab", type=T.INT, channels=1)
+    t = src._fill_top(0, cfg, rows)
+    assert t.shape == (2, 1) and float(t[0, 0]) == 3.0
+
+    cfg = caffe_pb.CoSTopParameter(name="seq", type=T.INT_ARRAY, channels=5,
+                              transpose=True)
+    t = src._fill_top(1, cfg, rows)
+    assert t.shape == (5, 2)          # time-major [T, N]
+    assert float(t[0, 0]) == 1.0 and float(t[0, 1]) == 4.0
+    assert float(t[1, 1]) == 0.0      # zero-padded tail
+
+    cfg = caffe_pb.CoSTopParameter(name="img", type=T.RAW_IMAGE, channels=3,
+                              height=2, width=2)
+    t = src._fill_top(2, cfg, rows)
+    assert t.shape == (2, 3, 2, 2)
+    assert float(t[0, 0, 0, 1]) == 1.0   # CHW pixel order preserved
+
+    cfg = caffe_pb.CoSTopParameter(name="score", type=T.FLOAT, channels=1)
+    t = src._fill_top(3, cfg, rows)
+    assert float(t[1, 0]) == 1.5
