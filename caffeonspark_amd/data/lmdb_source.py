@@ -24,23 +24,24 @@ class LMDBSource(ImageDataSource):
 
     def sample_iter(self, rank: int = 0, world: int = 1,
                     epochs: int = -1) -> Iterator[ImageSample]:
-        epoch = 0
-        while epochs < 0 or epoch < epochs:
-            for i, (key, raw) in enumerate(self.reader.items()):
-                if i % world != rank:
-                    continue
-                d = caffe_pb.Datum.FromString(raw)
-                if d.float_data:
-                    import numpy as np
-                    data = np.asarray(list(d.float_data),
-                                      dtype=np.float32).tobytes()
-                    yield FloatSample(key.decode(), d.label, d.channels,
-                                      d.height, d.width, data)
-                else:
-                    yield ImageSample(key.decode(), d.label, d.channels,
-                                      d.height, d.width, bool(d.encoded),
-                                      bytes(d.data))
-            epoch += 1
+        return self.persisted_epochs(lambda: self._epoch(rank, world),
+                                     epochs)
+
+    def _epoch(self, rank: int, world: int) -> Iterator[ImageSample]:
+        for i, (key, raw) in enumerate(self.reader.items()):
+            if i % world != rank:
+                continue
+            d = caffe_pb.Datum.FromString(raw)
+            if d.float_data:
+                import numpy as np
+                data = np.asarray(list(d.float_data),
+                                  dtype=np.float32).tobytes()
+                yield FloatSample(key.decode(), d.label, d.channels,
+                                  d.height, d.width, data)
+            else:
+                yield ImageSample(key.decode(), d.label, d.channels,
+                                  d.height, d.width, bool(d.encoded),
+                                  bytes(d.data))
 
 
 class FloatSample(ImageSample):

@@ -25,15 +25,16 @@ class SeqImageDataSource(ImageDataSource):
 
     def sample_iter(self, rank: int = 0, world: int = 1,
                     epochs: int = -1) -> Iterator[ImageSample]:
-        epoch = 0
-        while epochs < 0 or epoch < epochs:
-            i = 0
-            for f in self.files:
-                for key, raw in SequenceFileReader(f).items():
-                    if i % world == rank:
-                        d = caffe_pb.Datum.FromString(raw)
-                        yield ImageSample(key.decode(), d.label, d.channels,
-                                          d.height, d.width, bool(d.encoded),
-                                          bytes(d.data))
-                    i += 1
-            epoch += 1
+        return self.persisted_epochs(lambda: self._epoch(rank, world),
+                                     epochs)
+
+    def _epoch(self, rank: int, world: int) -> Iterator[ImageSample]:
+        i = 0
+        for f in self.files:
+            for key, raw in SequenceFileReader(f).items():
+                if i % world == rank:
+                    d = caffe_pb.Datum.FromString(raw)
+                    yield ImageSample(key.decode(), d.label, d.channels,
+                                      d.height, d.width, bool(d.encoded),
+                                      bytes(d.data))
+                i += 1

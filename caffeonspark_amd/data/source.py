@@ -55,6 +55,25 @@ class DataSource:
         """Yield raw samples for this rank (round-robin partitioning)."""
         raise NotImplementedError
 
+    def persisted_epochs(self, epoch_fn, epochs: int) -> Iterator[Any]:
+        """RDD.persist analog (`-persistent`, reference Config
+        isRddPersistent): cache this rank's raw samples in memory during
+        the first epoch and replay from memory afterwards, skipping
+        storage re-reads."""
+        use_cache = bool(getattr(self.conf, "isRddPersistent", False))
+        cache = [] if use_cache else None
+        epoch = 0
+        while epochs < 0 or epoch < epochs:
+            if cache is not None and epoch > 0:
+                for s in cache:
+                    yield s
+            else:
+                for s in epoch_fn():
+                    if cache is not None:
+                        cache.append(s)
+                    yield s
+            epoch += 1
+
     # -- queue feeding (reference: offer/nextBatch) -------------------------
     def offer(self, sample, timeout: float = 0.2) -> bool:
         """Bounded put; returns False on timeout so the feeder can re-check

@@ -109,7 +109,9 @@ def test_lmdb_source_rank_sharding(tmp_path):
         def __getattr__(self, k):
             raise AttributeError(k)
 
+    import types
     src = LMDBSource.__new__(LMDBSource)
+    src.conf = types.SimpleNamespace(isRddPersistent=False)
     from caffeonspark_amd.data.lmdb_io import LmdbReader
     src.reader = LmdbReader(path)
     shards = [[s.id for s in src.sample_iter(rank=r, world=3, epochs=1)]
@@ -155,3 +157,29 @@ def test_cos_top_type_fills():
     cfg = caffe_pb.CoSTopParameter(name="score", type=T.FLOAT, channels=1)
     t = src._fill_top(3, cfg, rows)
     assert float(t[1, 0]) == 1.5
+
+
+def test_persistent_sample_cache(tmp_path):
+    """-persistent (RDD.persist analog): after the first epoch, samples
+    replay from memory — storage can vanish and epoch 2 still yields."""
+    import shutil
+    import types
+
+    from caffeonspark_amd.data.lmdb_source import LMDBSource
+    from caffeonspark_amd.proto import caffe_pb as pb
+
+    path = str(tmp_path / "db")
+    items = [(f"{i:03d}".encode(),
+              pb.Datum(channels=1, height=1, width=1, label=i,
+                       data=bytes([i])).SerializeToString())
+             for i in range(5)]
+    LmdbWriter(path).write(items)
+    src = LMDBSource.__new__(LMDBSource)
+    src.conf = types.SimpleNamespace(isRddPersistent=True)
+    from caffeonspark_amd.data.lmdb_io import LmdbReader
+    src.reader = LmdbReader(path)
+    it = src.sample_iter(rank=0, world=1, epochs=2)
+    first = [next(it).id for _ in range(5)]
+    src.reader = None          # storage gone: must replay from cache
+    second = [s.id for s in it]
+    assert first == second == [f"{i:03d}" for i in range(5)]
