@@ -50,22 +50,23 @@ class ImageDataFrameSource(ImageDataSource):
 
     def sample_iter(self, rank: int = 0, world: int = 1,
                     epochs: int = -1) -> Iterator[ImageSample]:
-        epoch = 0
-        while epochs < 0 or epoch < epochs:
-            for i, row in enumerate(_read_parquet_rows(self.source_path,
-                                                       self.columns)):
-                if i % world != rank:
-                    continue
-                yield ImageSample(
-                    str(row.get("id", i)), float(row.get("label", 0.0)),
-                    int(row.get("channels", self.channels)),
-                    int(row.get("height", self.height)),
-                    int(row.get("width", self.width)),
-                    bool(row.get("encoded",
-                                 self.layer_param.memory_data_param
-                                 .image_encoded)),
-                    bytes(row["data"]))
-            epoch += 1
+        return self.persisted_epochs(lambda: self._epoch(rank, world),
+                                     epochs)
+
+    def _epoch(self, rank: int, world: int) -> Iterator[ImageSample]:
+        for i, row in enumerate(_read_parquet_rows(self.source_path,
+                                                   self.columns)):
+            if i % world != rank:
+                continue
+            yield ImageSample(
+                str(row.get("id", i)), float(row.get("label", 0.0)),
+                int(row.get("channels", self.channels)),
+                int(row.get("height", self.height)),
+                int(row.get("width", self.width)),
+                bool(row.get("encoded",
+                             self.layer_param.memory_data_param
+                             .image_encoded)),
+                bytes(row["data"]))
 
 
 class CoSDataFrameSource(DataSource):
@@ -90,13 +91,13 @@ class CoSDataFrameSource(DataSource):
 
     def sample_iter(self, rank: int = 0, world: int = 1,
                     epochs: int = -1) -> Iterator[dict]:
-        epoch = 0
-        cols = [t.name for t in self.tops]
-        while epochs < 0 or epoch < epochs:
-            for i, row in enumerate(_read_parquet_rows(self.source_path)):
-                if i % world == rank:
-                    yield row
-            epoch += 1
+        return self.persisted_epochs(lambda: self._epoch(rank, world),
+                                     epochs)
+
+    def _epoch(self, rank: int, world: int) -> Iterator[dict]:
+        for i, row in enumerate(_read_parquet_rows(self.source_path)):
+            if i % world == rank:
+                yield row
 
     def _fill_top(self, i: int, cfg, rows: List[dict]) -> torch.Tensor:
         T = caffe_pb.CoSTopType
