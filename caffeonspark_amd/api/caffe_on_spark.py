@@ -36,6 +36,13 @@ class CaffeOnSpark:
     def _setup(self, sources: List[DataSource]) -> CaffeProcessor:
         """Reference setupTraining (CaffeOnSpark.scala:105-158): construct
         processors, exchange addresses (rendezvous), start threads."""
+        if self.conf.clusterSize and self.conf.clusterSize != self.world:
+            # reference asserts actual executor count == clusterSize
+            # (CaffeOnSpark.scala:127-133) — fail fast on a mis-launched
+            # torchrun the same way
+            raise RuntimeError(
+                f"-clusterSize {self.conf.clusterSize} but WORLD_SIZE is "
+                f"{self.world}: launch one process per device")
         for s in sources:
             s.init()
         proc = CaffeProcessor.instance(sources, self.rank, self.conf)
