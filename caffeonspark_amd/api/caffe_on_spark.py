@@ -29,7 +29,14 @@ from .config import Config
 class CaffeOnSpark:
     def __init__(self, conf: Config):
         self.conf = conf
-        self.rank = init_distributed()
+        # reference maps "ethernet" -> SOCKET; here RCCL is the GPU
+        # transport and gloo the CPU/ethernet one (Config.scala:452-457)
+        backend = {"rccl": "nccl", "nccl": "nccl", "gloo": "gloo",
+                   "ethernet": "gloo"}.get(
+                       (conf.connection or "").lower())
+        if backend == "nccl" and not torch.cuda.is_available():
+            backend = "gloo"           # CPU runs: RCCL needs devices
+        self.rank = init_distributed(backend)
         self.world = int(os.environ.get("WORLD_SIZE", "1"))
 
     # ------------------------------------------------------------- training
