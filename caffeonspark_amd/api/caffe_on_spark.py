@@ -54,7 +54,6 @@ class CaffeOnSpark:
             s.init()
         proc = CaffeProcessor.instance(sources, self.rank, self.conf)
         if self.world > 1:
-            import torch.distributed as dist
             sync = DistributedSync(proc.solver)
             sync.broadcast_params()
             proc.dist_sync = sync
@@ -163,7 +162,7 @@ class CaffeOnSpark:
         source = source or get_source(self.conf, False)
         net_param = self.conf.net_param
         # validation outputs = tops of TEST-phase output layers
-        from ..core.net import Net, filter_net
+        from ..core.net import filter_net
         state = caffe_pb.NetState(phase=caffe_pb.Phase.TEST)
         test_param = filter_net(net_param, state)
         consumed = set()

@@ -13,7 +13,6 @@ from typing import List, Optional
 
 import torch
 
-from ...proto import caffe_pb
 from .. import fillers
 from .base import Layer, register_layer
 

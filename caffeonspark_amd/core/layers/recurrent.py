@@ -19,7 +19,6 @@ from __future__ import annotations
 import torch
 
 from ... import ops
-from ...proto import caffe_pb
 from .base import Layer, register_layer
 
 

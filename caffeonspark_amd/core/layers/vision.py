@@ -8,8 +8,6 @@ Semantics match upstream Caffe (reference layer census: SURVEY.md §2.5 row
 
 from __future__ import annotations
 
-from typing import List
-
 import torch
 
 from ... import ops

@@ -12,7 +12,7 @@ from __future__ import annotations
 
 import math
 import os
-from typing import Callable, List, Optional
+from typing import List, Optional
 
 import torch
 

@@ -17,10 +17,7 @@ import queue
 import threading
 from typing import List, Optional
 
-import torch
-
 from ..core.solver import Solver
-from ..proto import caffe_pb
 from .source import STOP_MARK, DataSource
 
 _instance_lock = threading.Lock()

@@ -11,7 +11,7 @@ wrapper).
 from __future__ import annotations
 
 import io
-from typing import List, Optional, Sequence
+from typing import Optional, Sequence
 
 import numpy as np
 import torch

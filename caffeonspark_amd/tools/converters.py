@@ -12,7 +12,7 @@ from __future__ import annotations
 
 import argparse
 import os
-from typing import Iterator, List, Tuple
+from typing import List, Tuple
 
 from ..proto import caffe_pb
 from .seq_value import datum_from_image_file

@@ -28,7 +28,6 @@ def _worker(rank: int, world: int, port: int, argv: List[str]):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     from ..api import main as cos_main
-    from ..data.processor import CaffeProcessor
 
     def handler(signum, frame):
         global _proc

@@ -18,7 +18,7 @@ is by-construction from the spec.
 from __future__ import annotations
 
 import struct
-from typing import Dict, List, Optional, Tuple, Union
+from typing import Dict, List, Tuple, Union
 
 import numpy as np
 

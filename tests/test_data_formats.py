@@ -162,7 +162,6 @@ def test_cos_top_type_fills():
 def test_persistent_sample_cache(tmp_path):
     """-persistent (RDD.persist analog): after the first epoch, samples
     replay from memory — storage can vanish and epoch 2 still yields."""
-    import shutil
     import types
 
     from caffeonspark_amd.data.lmdb_source import LMDBSource

@@ -5,7 +5,6 @@ multi-rank harness as a gap to fill)."""
 
 import os
 
-import pytest
 import torch
 import torch.multiprocessing as mp
 
@@ -31,7 +30,8 @@ def _worker(rank, ws, store_path, q):
             y = torch.randint(0, 10, (64,), generator=g).float()
             s.net.data_layers()[0].reset(x, y)
             s._step_one()
-        q.put((rank, s.flat_w.clone(), s.flat_g.clone()))
+        # plain numpy: tensor fd-sharing dies with the child process
+        q.put((rank, s.flat_w.numpy().copy(), s.flat_g.numpy().copy()))
     finally:
         dist.destroy_process_group()
 
@@ -72,8 +72,10 @@ def _run_two_ranks_identical():
         assert p.exitcode == 0
     w0, g0 = results[0]
     w1, g1 = results[1]
-    torch.testing.assert_close(w0, w1, rtol=0, atol=0)
-    torch.testing.assert_close(g0, g1, rtol=1e-6, atol=1e-7)
+    torch.testing.assert_close(torch.from_numpy(w0), torch.from_numpy(w1),
+                               rtol=0, atol=0)
+    torch.testing.assert_close(torch.from_numpy(g0), torch.from_numpy(g1),
+                               rtol=1e-6, atol=1e-7)
 
 
 def _seeded_solver():
@@ -107,7 +109,7 @@ def _worker_equiv(rank, ws, store_path, q, data):
         s.net.data_layers()[0].reset(x[rank * half:(rank + 1) * half],
                                      y[rank * half:(rank + 1) * half])
         s._step_one()
-        q.put((rank, s.flat_w.clone()))
+        q.put((rank, s.flat_w.numpy().copy()))
     finally:
         dist.destroy_process_group()
 
@@ -140,7 +142,9 @@ def _run_matches_single():
         p.join(timeout=60)
         assert p.exitcode == 0
 
-    torch.testing.assert_close(results[0], results[1], rtol=0, atol=0)
+    torch.testing.assert_close(torch.from_numpy(results[0]),
+                               torch.from_numpy(results[1]),
+                               rtol=0, atol=0)
 
     # single-process run over the whole batch from the same seeded init:
     # DP-averaged update must match the full-batch update (both losses are
@@ -148,7 +152,7 @@ def _run_matches_single():
     single = _seeded_solver()
     single.net.data_layers()[0].reset(x, y)
     single._step_one()
-    torch.testing.assert_close(results[0], single.flat_w,
+    torch.testing.assert_close(torch.from_numpy(results[0]), single.flat_w,
                                rtol=1e-4, atol=1e-6)
 
 
@@ -209,7 +213,9 @@ def _worker_frozen(rank, ws, store_path, q):
             y = torch.randint(0, 4, (32,), generator=g).float()
             s.net.data_layers()[0].reset(x, y)
             s._step_one()
-        q.put((rank, s.flat_w.clone(), frozen0, s.params[0].data.clone()))
+        q.put((rank, s.flat_w.numpy().copy(),
+               frozen0.numpy().copy(),
+               s.params[0].data.detach().numpy().copy()))
     finally:
         dist.destroy_process_group()
 
@@ -237,6 +243,10 @@ def _run_frozen():
         out[rank] = (flat_w, frozen0, frozen1)
     for p in procs:
         p.join(timeout=60)
-    torch.testing.assert_close(out[0][0], out[1][0], rtol=0, atol=0)
+    torch.testing.assert_close(torch.from_numpy(out[0][0]),
+                               torch.from_numpy(out[1][0]),
+                               rtol=0, atol=0)
     for r in range(ws):
-        torch.testing.assert_close(out[r][1], out[r][2], rtol=0, atol=0)
+        torch.testing.assert_close(torch.from_numpy(out[r][1]),
+                                   torch.from_numpy(out[r][2]),
+                                   rtol=0, atol=0)

@@ -367,7 +367,6 @@ def test_batchnorm_parity():
 def test_winograd_conv_parity(monkeypatch):
     """F(2x2,3x3) Winograd path (COS_WINOGRAD=1) vs the direct im2col
     path: fwd/dx/dw/db must agree within bf16 training noise."""
-    import os as _os
     from caffeonspark_amd.ops import gpu as g
 
     torch.manual_seed(5)

@@ -6,7 +6,6 @@ import torch
 
 from caffeonspark_amd.core.blob import Blob
 from caffeonspark_amd.core.layers.base import create_layer
-from caffeonspark_amd.core.net import Net
 from caffeonspark_amd.proto import caffe_pb, text_format
 
 torch.manual_seed(0)

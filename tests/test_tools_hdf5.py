@@ -4,7 +4,6 @@ import json
 import os
 
 import numpy as np
-import pytest
 import torch
 
 from caffeonspark_amd.proto import caffe_pb, text_format
