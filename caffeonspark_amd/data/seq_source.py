@@ -29,12 +29,22 @@ class SeqImageDataSource(ImageDataSource):
                                      epochs)
 
     def _epoch(self, rank: int, world: int) -> Iterator[ImageSample]:
+        from .javaser import key_id_label
         i = 0
         for f in self.files:
             for key, raw in SequenceFileReader(f).items():
                 if i % world == rank:
-                    d = caffe_pb.Datum.FromString(raw)
-                    yield ImageSample(key.decode(), d.label, d.channels,
-                                      d.height, d.width, bool(d.encoded),
-                                      bytes(d.data))
+                    jk = key_id_label(key)
+                    if jk is not None:
+                        # reference-written file (Binary2Sequence.scala):
+                        # key = java-serialized (id, label) tuple, value
+                        # = raw encoded image bytes
+                        sid, label = jk
+                        yield ImageSample(sid, label, 3, 0, 0, True,
+                                          bytes(raw))
+                    else:
+                        d = caffe_pb.Datum.FromString(raw)
+                        yield ImageSample(key.decode(), d.label, d.channels,
+                                          d.height, d.width,
+                                          bool(d.encoded), bytes(d.data))
                 i += 1

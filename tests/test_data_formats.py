@@ -182,3 +182,54 @@ def test_persistent_sample_cache(tmp_path):
     src.reader = None          # storage gone: must replay from cache
     second = [s.id for s in it]
     assert first == second == [f"{i:03d}" for i in range(5)]
+
+
+def _java_tuple2_bytes(a, b):
+    """Hand-assembled ObjectOutputStream stream for a scala Tuple2 of two
+    strings, per the Java Object Serialization Spec grammar (magic,
+    TC_OBJECT, TC_CLASSDESC with raw-UTF names, then TC_STRING values)."""
+    import struct
+
+    def utf(s):
+        raw = s.encode("utf-8")
+        return struct.pack(">H", len(raw)) + raw
+
+    out = b"\xac\xed\x00\x05"            # STREAM_MAGIC, VERSION
+    out += b"\x73"                         # TC_OBJECT
+    out += b"\x72" + utf("scala.Tuple2")   # TC_CLASSDESC + class name
+    out += b"\x00" * 8                     # serialVersionUID
+    out += b"\x02"                         # SC_SERIALIZABLE
+    out += struct.pack(">H", 2)            # field count
+    out += b"L" + utf("_1") + b"\x74" + utf("Ljava/lang/Object;")
+    out += b"L" + utf("_2") + b"\x74" + utf("Ljava/lang/Object;")
+    out += b"\x78\x70"                    # TC_ENDBLOCKDATA, null super
+    out += b"\x74" + utf(a)                # _1 value
+    out += b"\x74" + utf(b)                # _2 value
+    return out
+
+
+def test_reference_seqfile_key_decoding(tmp_path):
+    """SeqImageSource reads reference-written SequenceFiles whose keys
+    are java-serialized (filename, label) tuples and whose values are
+    raw encoded image bytes (Binary2Sequence.scala:54-72)."""
+    import types
+
+    from caffeonspark_amd.data.javaser import key_id_label
+    from caffeonspark_amd.data.seq_source import SeqImageDataSource
+
+    key = _java_tuple2_bytes("cat1.jpg", "0")
+    assert key_id_label(key) == ("cat1.jpg", 0.0)
+    key2 = _java_tuple2_bytes("dog2.jpg", "1")
+
+    p = str(tmp_path / "ref.seq")
+    with SequenceFileWriter(p) as w:
+        w.append(key, b"JPGBYTES0")
+        w.append(key2, b"JPGBYTES1")
+
+    src = SeqImageDataSource.__new__(SeqImageDataSource)
+    src.conf = types.SimpleNamespace(isRddPersistent=False)
+    src.files = [p]
+    samples = list(src.sample_iter(epochs=1))
+    assert [(s.id, s.label, s.encoded) for s in samples] == \
+        [("cat1.jpg", 0.0, True), ("dog2.jpg", 1.0, True)]
+    assert samples[0].data == b"JPGBYTES0"
