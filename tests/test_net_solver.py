@@ -104,6 +104,8 @@ def test_lr_policies():
         (dict(lr_policy="step", gamma=0.1, stepsize=10), 25, 0.01),
         (dict(lr_policy="inv", gamma=0.5, power=1.0), 2, 0.5),
         (dict(lr_policy="poly", power=1.0), 50, 0.5),
+        (dict(lr_policy="exp", gamma=0.5), 3, 0.125),
+        (dict(lr_policy="sigmoid", gamma=-1.0, stepsize=10), 10, 0.5),
     ]
     net_text = ('layer { name: "x" type: "DummyData" top: "x" '
                 'dummy_data_param { shape { dim: 1 dim: 1 } } }')
@@ -114,6 +116,19 @@ def test_lr_policies():
         s = Solver(sp)
         s.iter = it
         assert s.get_lr() == pytest.approx(expect, rel=1e-6), kw
+
+    # multistep: gamma applied at each stepvalue crossing (stateful)
+    sp = caffe_pb.SolverParameter(
+        net_param=text_format.parse(net_text, caffe_pb.NetParameter),
+        lr_policy="multistep", gamma=0.1, stepvalue=[5, 8], **base)
+    s = Solver(sp)
+    seen = []
+    for it in range(10):
+        s.iter = it
+        seen.append(round(s.get_lr(), 6))
+    assert seen[:5] == [1.0] * 5          # before first step
+    assert seen[5:8] == [0.1, 0.1, 0.1]   # after stepvalue 5
+    assert seen[8:] == [0.01, 0.01]       # after stepvalue 8
 
 
 def test_sgd_update_matches_manual():
