@@ -458,3 +458,40 @@ def test_loss_weight_scaling():
     s1.net.zero_param_diffs()
     s1.net.backward()
     torch.testing.assert_close(g2, 1.3 * s1.flat_g, rtol=1e-4, atol=1e-6)
+
+
+def test_seed_determinism_and_loss_smoothing():
+    """random_seed gives identical init; smoothed loss averages the last
+    average_loss iterations (reference UpdateSmoothedLoss)."""
+    net_text = """
+      layer { name: "d" type: "MemoryData" top: "x" top: "t"
+              memory_data_param { batch_size: 4 channels: 3 height: 1
+                                  width: 1 } }
+      layer { name: "ip" type: "InnerProduct" bottom: "x" top: "y"
+              inner_product_param { num_output: 2
+                weight_filler { type: "gaussian" std: 0.3 } } }
+      layer { name: "l" type: "SoftmaxWithLoss" bottom: "y" bottom: "t"
+              top: "loss" }
+    """
+
+    def make():
+        sp = caffe_pb.SolverParameter(
+            net_param=text_format.parse(net_text, caffe_pb.NetParameter),
+            base_lr=0.0, lr_policy="fixed", max_iter=9, random_seed=21,
+            average_loss=3)
+        return Solver(sp)
+
+    a, b = make(), make()
+    torch.testing.assert_close(a.flat_w, b.flat_w, rtol=0, atol=0)
+
+    # lr 0: weights frozen, per-step losses vary only with data; smoothing
+    # window = mean of the last 3 step losses
+    g = torch.Generator().manual_seed(5)
+    losses = []
+    for _ in range(5):
+        x = torch.randn(4, 3, 1, 1, generator=g)
+        t_ = torch.randint(0, 2, (4,), generator=g).float()
+        a.net.data_layers()[0].reset(x, t_)
+        losses.append(a._step_one())
+    expect = sum(losses[-3:]) / 3
+    assert a.smoothed_loss == pytest.approx(expect, rel=1e-6)
