@@ -1,0 +1,71 @@
+"""Property-based round-trip tests of the clean-room proto codec — the
+analog of the reference's randomized Python-bridge conversion tests
+(TestResources.py).  hypothesis drives random field values through text
+and wire round-trips."""
+
+import math
+
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from caffeonspark_amd.proto import caffe_pb, text_format
+
+f32 = st.floats(min_value=-3.0000000054977558e+38, max_value=3.0000000054977558e+38, width=32)
+i32 = st.integers(min_value=-(2 ** 31), max_value=2 ** 31 - 1)
+u32 = st.integers(min_value=0, max_value=2 ** 32 - 1)
+name = st.text(
+    st.characters(whitelist_categories=("L", "N"), max_codepoint=0x2FF),
+    max_size=24)
+
+
+@settings(max_examples=200, deadline=None)
+@given(base_lr=f32, max_iter=u32, gamma=f32, momentum=f32,
+       snapshot_prefix=name, stepvalue=st.lists(u32, max_size=6),
+       test_iter=st.lists(u32, max_size=4), iter_size=i32)
+def test_solver_param_roundtrips(base_lr, max_iter, gamma, momentum,
+                                 snapshot_prefix, stepvalue, test_iter,
+                                 iter_size):
+    p = caffe_pb.SolverParameter(
+        base_lr=base_lr, max_iter=max_iter, gamma=gamma,
+        momentum=momentum, snapshot_prefix=snapshot_prefix,
+        stepvalue=stepvalue, test_iter=test_iter, iter_size=iter_size)
+    # binary wire
+    b = caffe_pb.SolverParameter.FromString(p.SerializeToString())
+    # text format
+    t = text_format.parse(text_format.dumps(p), caffe_pb.SolverParameter)
+    for q in (b, t):
+        assert math.isclose(q.base_lr, base_lr, rel_tol=1e-6, abs_tol=1e-37)
+        assert q.max_iter == max_iter
+        assert q.snapshot_prefix == snapshot_prefix
+        assert list(q.stepvalue) == stepvalue
+        assert list(q.test_iter) == test_iter
+        assert q.iter_size == iter_size
+
+
+@settings(max_examples=100, deadline=None)
+@given(label=i32, channels=u32, data=st.binary(max_size=512),
+       floats=st.lists(f32, max_size=32))
+def test_datum_roundtrips(label, channels, data, floats):
+    d = caffe_pb.Datum(label=label, channels=channels, data=data,
+                       float_data=floats)
+    b = caffe_pb.Datum.FromString(d.SerializeToString())
+    assert b.label == label and b.channels == channels
+    assert bytes(b.data) == data
+    assert len(b.float_data) == len(floats)
+    for a, e in zip(b.float_data, floats):
+        assert math.isclose(a, e, rel_tol=1e-6, abs_tol=1e-37)
+
+
+@settings(max_examples=100, deadline=None)
+@given(nm=name, lr=st.lists(f32, max_size=3), bottoms=st.lists(name,
+                                                               max_size=4))
+def test_layer_param_roundtrips(nm, lr, bottoms):
+    lp = caffe_pb.LayerParameter(name=nm, type="ReLU", bottom=bottoms,
+                                 loss_weight=lr)
+    np_ = caffe_pb.NetParameter(name=nm, layer=[lp])
+    b = caffe_pb.NetParameter.FromString(np_.SerializeToString())
+    t = text_format.parse(text_format.dumps(np_), caffe_pb.NetParameter)
+    for q in (b, t):
+        assert q.name == nm
+        assert q.layer[0].name == nm
+        assert list(q.layer[0].bottom) == bottoms
