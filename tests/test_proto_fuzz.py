@@ -69,3 +69,41 @@ def test_layer_param_roundtrips(nm, lr, bottoms):
         assert q.name == nm
         assert q.layer[0].name == nm
         assert list(q.layer[0].bottom) == bottoms
+
+
+# --------------------------------------------------------- format fuzz
+
+kv_key = st.binary(min_size=1, max_size=48)
+kv_val = st.binary(max_size=3000)
+
+
+@settings(max_examples=40, deadline=None)
+@given(items=st.dictionaries(kv_key, kv_val, max_size=120))
+def test_lmdb_roundtrip_fuzz(items, tmp_path_factory):
+    """Random key/value sets (sizes spanning inline and overflow pages)
+    survive the clean-room LMDB writer -> reader round-trip in key
+    order."""
+    from caffeonspark_amd.data.lmdb_io import LmdbReader, LmdbWriter
+    d = tmp_path_factory.mktemp("lmdbfuzz")
+    path = str(d / "db")
+    pairs = sorted(items.items())
+    LmdbWriter(path).write(list(items.items()))
+    got = list(LmdbReader(path).items())
+    assert got == pairs
+
+
+@settings(max_examples=40, deadline=None)
+@given(items=st.lists(st.tuples(st.binary(min_size=1, max_size=64),
+                                st.binary(max_size=2000)),
+                      max_size=60))
+def test_seqfile_roundtrip_fuzz(items, tmp_path_factory):
+    """Random records round-trip through the Hadoop SequenceFile codec
+    (sync markers land wherever they land)."""
+    from caffeonspark_amd.data.seqfile import SequenceFileReader, \
+        SequenceFileWriter
+    d = tmp_path_factory.mktemp("seqfuzz")
+    p = str(d / "f.seq")
+    with SequenceFileWriter(p) as w:
+        for k, v in items:
+            w.append(k, v)
+    assert list(SequenceFileReader(p).items()) == items
