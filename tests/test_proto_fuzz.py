@@ -107,3 +107,36 @@ def test_seqfile_roundtrip_fuzz(items, tmp_path_factory):
         for k, v in items:
             w.append(k, v)
     assert list(SequenceFileReader(p).items()) == items
+
+
+@settings(max_examples=30, deadline=None)
+@given(arrays=st.dictionaries(
+    st.text(st.characters(whitelist_categories=("L", "N")), min_size=1,
+            max_size=12),
+    st.lists(f32,
+             min_size=1, max_size=64),
+    min_size=1, max_size=8),
+    attr=st.text(max_size=40))
+def test_hdf5_roundtrip_fuzz(arrays, attr, tmp_path_factory):
+    """Random group trees of float arrays + a string attribute survive
+    the minimal HDF5 v0 writer -> reader round-trip."""
+    import numpy as np
+
+    from caffeonspark_amd.utils.hdf5 import H5Group, load, save
+    d = tmp_path_factory.mktemp("h5fuzz")
+    p = str(d / "t.h5")
+    root = H5Group()
+    root.attrs["note"] = attr
+    sub = H5Group()
+    for name, vals in arrays.items():
+        sub[name] = np.asarray(vals, dtype=np.float32)
+    root["data"] = sub
+    save(p, root)
+    back = load(p)
+    assert back.attrs.get("note", "") == attr
+    got = back["data"]
+    assert set(got.keys()) == set(arrays.keys())
+    for name, vals in arrays.items():
+        np.testing.assert_allclose(
+            np.asarray(got[name], dtype=np.float32),
+            np.asarray(vals, dtype=np.float32), rtol=1e-6)
