@@ -17,8 +17,10 @@ from torch.utils import cpp_extension  # noqa: E402
 ROOT = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(ROOT, "caffeonspark_amd", "ops", "csrc")
 
-sources = sorted(glob.glob(os.path.join(CSRC, "*.cpp")) +
-                 glob.glob(os.path.join(CSRC, "*.hip")))
+sources = sorted(
+    glob.glob(os.path.join(CSRC, "*.cpp")) +
+    [f for f in glob.glob(os.path.join(CSRC, "*.hip"))
+     if not f.endswith("_hip.hip")])   # skip hipify-generated copies
 
 setup(
     name="cosamd-hip",
