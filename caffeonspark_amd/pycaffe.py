@@ -80,6 +80,25 @@ class Net:
         return {name: self.blobs[name].data
                 for name in self._net.output_blob_names()}
 
+    def backward(self) -> Dict[str, np.ndarray]:
+        """pycaffe-style backward: populate blob diffs from the loss tops
+        and return input-blob gradients."""
+        self._net.backward()
+        out = {}
+        for name, b in self._net.blob_map.items():
+            if b.diff is not None:
+                out[name] = b.diff.detach().float().cpu().numpy()
+        return out
+
+    def save(self, path: str) -> None:
+        """Write learned weights as a .caffemodel (binary NetParameter)."""
+        from .proto import write_binary_proto
+        write_binary_proto(path, self._net.to_proto())
+
     @property
     def layer_names(self) -> List[str]:
         return [l.name for l in self._net.layers]
+
+    @property
+    def outputs(self) -> List[str]:
+        return list(self._net.output_blob_names())

@@ -189,6 +189,32 @@ def test_display_utils():
     assert "<table>" in html and "<img" in html and ">3<" in html
 
 
+def test_pycaffe_backward_and_save(tmp_path):
+    """pycaffe shim: backward returns input gradients; save writes a
+    loadable .caffemodel."""
+    import numpy as np
+
+    import caffeonspark_amd.pycaffe as caffe
+    from caffeonspark_amd.proto import caffe_pb
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    proto = os.path.join(root, "caffeonspark_amd", "models",
+                         "lenet_memory_train_test.prototxt")
+    net = caffe.Net(proto, phase=caffe_pb.Phase.TRAIN)
+    dl = net._net.data_layers()[0]
+    import torch
+    dl.reset(torch.randn(4, 1, 28, 28), torch.randint(0, 10, (4,)).float())
+    net._net.forward()
+    grads = net.backward()
+    assert any(np.abs(g).sum() > 0 for g in grads.values())
+    out = str(tmp_path / "m.caffemodel")
+    net.save(out)
+    net2 = caffe.Net(proto, weights=out, phase=caffe_pb.Phase.TRAIN)
+    np.testing.assert_allclose(net.params["ip2"][0].data,
+                               net2.params["ip2"][0].data)
+    assert "loss" in net.outputs or net.outputs
+
+
 def test_time_net_per_layer(capsys):
     """caffe-time analog: per-layer fwd/bwd timing table."""
     import os
