@@ -495,3 +495,18 @@ def test_seed_determinism_and_loss_smoothing():
         losses.append(a._step_one())
     expect = sum(losses[-3:]) / 3
     assert a.smoothed_loss == pytest.approx(expect, rel=1e-6)
+
+
+def test_model_zoo_generator_in_sync():
+    """The committed model-zoo prototxts match what generate.py emits
+    (regeneration is reproducible and nothing was hand-edited).  The
+    generator writes in place; git must see no resulting change."""
+    import subprocess
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    models = os.path.join(root, "caffeonspark_amd", "models")
+    import caffeonspark_amd.models.generate as gen
+    gen.main()
+    out = subprocess.run(["git", "diff", "--name-only", "--", models],
+                         capture_output=True, text=True, cwd=root)
+    assert out.stdout.strip() == "", out.stdout
