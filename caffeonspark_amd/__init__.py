@@ -10,8 +10,3 @@ mirroring the reference's driver/executor API.
 __version__ = "0.1.0"
 
 from . import proto  # noqa: F401
-
-
-def _lazy(name):
-    import importlib
-    return importlib.import_module(f".{name}", __name__)
