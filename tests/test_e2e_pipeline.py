@@ -416,3 +416,23 @@ def test_cli_main_train_and_test(workdir, capsys):
     out = capsys.readouterr().out
     assert "accuracy" in out
     CaffeProcessor.reset_instance()
+
+
+def test_multiclass_lr_example(workdir):
+    """MultiClassLogisticRegression.py analog end-to-end."""
+    pytest.importorskip("sklearn")
+    import sys
+    sys.path.insert(0, os.path.join(ROOT, "examples"))
+    import multiclass_logistic_regression as mlr
+
+    CaffeProcessor.reset_instance()
+    os.chdir(workdir)
+    snaps = sorted(f for f in os.listdir(workdir)
+                   if f.endswith(".caffemodel"))
+    if not snaps:
+        pytest.skip("depends on an earlier training snapshot")
+    acc = mlr.main(["-conf", str(workdir / "solver.prototxt"),
+                    "-weights", str(workdir / snaps[-1]),
+                    "-features", "ip1", "-label", "label"])
+    assert acc > 0.5
+    CaffeProcessor.reset_instance()
