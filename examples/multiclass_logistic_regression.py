@@ -33,7 +33,7 @@ def main(argv=None):
 
     from sklearn.linear_model import LogisticRegression
     n_train = int(0.8 * len(X))
-    clf = LogisticRegression(max_iter=200, multi_class="multinomial")
+    clf = LogisticRegression(max_iter=200)
     clf.fit(X[:n_train], y[:n_train])
     acc = clf.score(X[n_train:], y[n_train:])
     print(f"logistic-regression holdout accuracy: {acc:.4f}")
