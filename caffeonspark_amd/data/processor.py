@@ -183,11 +183,13 @@ class CaffeProcessor:
     # ---------------------------------------------------------------- misc
     def snapshot(self) -> str:
         model_file = self.solver.snapshot()
-        out_dir = getattr(self.conf, "model_path", None)
-        if out_dir:
+        out = getattr(self.conf, "model_path", None)
+        if out:
+            if out.startswith("file:"):
+                out = out[5:]          # reference FSUtils URI handling
             import shutil
-            os.makedirs(os.path.dirname(out_dir) or ".", exist_ok=True)
-            shutil.copyfile(model_file, out_dir)
+            os.makedirs(os.path.dirname(out) or ".", exist_ok=True)
+            shutil.copyfile(model_file, out)
         return model_file
 
     def sync(self) -> None:

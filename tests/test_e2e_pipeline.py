@@ -436,3 +436,18 @@ def test_multiclass_lr_example(workdir):
                     "-features", "ip1", "-label", "label"])
     assert acc > 0.5
     CaffeProcessor.reset_instance()
+
+
+def test_model_flag_copies_snapshot(workdir):
+    """-model file:PATH: the final model is copied there (reference
+    FSUtils.GenModelOrState)."""
+    CaffeProcessor.reset_instance()
+    os.chdir(workdir)
+    dest = str(workdir / "final_model.caffemodel")
+    conf = Config(["-conf", _solver_file(workdir, "solver_m.prototxt",
+                                         max_iter=40),
+                   "-train", "-label", "label",
+                   "-model", "file:" + dest])
+    CaffeOnSpark(conf).train()
+    assert os.path.exists(dest)
+    CaffeProcessor.reset_instance()
