@@ -451,3 +451,16 @@ def test_model_flag_copies_snapshot(workdir):
     CaffeOnSpark(conf).train()
     assert os.path.exists(dest)
     CaffeProcessor.reset_instance()
+
+
+def test_features_multi_blob(workdir):
+    """-features a,b: one row carries every requested blob."""
+    CaffeProcessor.reset_instance()
+    os.chdir(workdir)
+    conf = Config(["-conf", _solver_file(workdir, "solver_mb.prototxt"),
+                   "-features", "ip1,ip2", "-label", "label"])
+    cos = CaffeOnSpark(conf)
+    df = cos.features(max_samples=50)
+    assert {"SampleID", "label", "ip1", "ip2"} <= set(df.columns)
+    assert len(df["ip1"][0]) == 500 and len(df["ip2"][0]) == 10
+    CaffeProcessor.reset_instance()
