@@ -462,5 +462,5 @@ def test_features_multi_blob(workdir):
     cos = CaffeOnSpark(conf)
     df = cos.features(max_samples=50)
     assert {"SampleID", "label", "ip1", "ip2"} <= set(df.columns)
-    assert len(df["ip1"][0]) == 500 and len(df["ip2"][0]) == 10
+    assert len(df["ip1"][0]) == 100 and len(df["ip2"][0]) == 10
     CaffeProcessor.reset_instance()
