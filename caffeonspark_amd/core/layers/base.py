@@ -60,13 +60,26 @@ class Layer:
     def add_param(self, shape, filler: caffe_pb.FillerParameter = None,
                   name: str = "") -> Blob:
         from .. import fillers
+        spec = self.param_spec(len(self.blobs))
+        share_name = spec.name
+        if share_name and share_name in self.net.shared_params:
+            # caffe named-param sharing (param { name: "w" } on several
+            # layers aliases ONE blob: siamese nets, unrolled recurrences)
+            b = self.net.shared_params[share_name]
+            if list(b.shape) != [int(d) for d in shape]:
+                raise ValueError(
+                    f"shared param {share_name!r}: shape {list(b.shape)} "
+                    f"!= {list(shape)} requested by layer {self.name!r}")
+            self.blobs.append(b)
+            return b
         b = Blob(shape, name=name or f"{self.name}_p{len(self.blobs)}",
                  dtype=torch.float32, device=self.device, alloc_diff=True)
         if filler is not None:
             fillers.fill(b.data, filler, generator=self.net.generator)
-        spec = self.param_spec(len(self.blobs))
         b._lr_mult = spec.lr_mult
         b._decay_mult = spec.decay_mult
+        if share_name:
+            self.net.shared_params[share_name] = b
         self.blobs.append(b)
         return b
 

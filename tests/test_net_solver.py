@@ -510,3 +510,51 @@ def test_model_zoo_generator_in_sync():
     out = subprocess.run(["git", "diff", "--name-only", "--", models],
                          capture_output=True, text=True, cwd=root)
     assert out.stdout.strip() == "", out.stdout
+
+
+def test_named_param_sharing():
+    """caffe param { name } sharing: two layers alias one blob, its
+    gradient accumulates both layers' contributions, and the solver
+    counts it once (siamese-style)."""
+    net_text = """
+      layer { name: "d" type: "MemoryData" top: "x" top: "t"
+              memory_data_param { batch_size: 6 channels: 4 height: 1
+                                  width: 1 } }
+      layer { name: "ipa" type: "InnerProduct" bottom: "x" top: "a"
+              param { name: "w_shared" } param { name: "b_shared" }
+              inner_product_param { num_output: 4
+                weight_filler { type: "gaussian" std: 0.2 } } }
+      layer { name: "ipb" type: "InnerProduct" bottom: "a" top: "b"
+              param { name: "w_shared" } param { name: "b_shared" }
+              inner_product_param { num_output: 4 } }
+      layer { name: "l" type: "SoftmaxWithLoss" bottom: "b" bottom: "t"
+              top: "loss" }
+    """
+    sp = caffe_pb.SolverParameter(
+        net_param=text_format.parse(net_text, caffe_pb.NetParameter),
+        base_lr=0.1, lr_policy="fixed", max_iter=4, random_seed=13)
+    s = Solver(sp)
+    ipa = next(l for l in s.net.layers if l.name == "ipa")
+    ipb = next(l for l in s.net.layers if l.name == "ipb")
+    assert ipa.blobs[0] is ipb.blobs[0]
+    assert ipa.blobs[1] is ipb.blobs[1]
+    # deduped in the solver arena
+    assert len(s.params) == 2
+
+    g = torch.Generator().manual_seed(2)
+    x = torch.randn(6, 4, 1, 1, generator=g)
+    t_ = torch.randint(0, 4, (6,), generator=g).float()
+    s.net.data_layers()[0].reset(x, t_)
+    s.net.zero_param_diffs()
+    s.net.forward_backward()
+    shared_grad = ipa.blobs[0].diff.clone()
+
+    # autograd oracle for the weight-tied two-layer chain
+    w = ipa.blobs[0].data.detach().clone().requires_grad_(True)
+    bias = ipa.blobs[1].data.detach().clone().requires_grad_(True)
+    xf = x.reshape(6, 4)
+    h = torch.nn.functional.linear(xf, w, bias)
+    out = torch.nn.functional.linear(h, w, bias)
+    loss = torch.nn.functional.cross_entropy(out, t_.long())
+    loss.backward()
+    torch.testing.assert_close(shared_grad, w.grad, rtol=1e-4, atol=1e-6)
