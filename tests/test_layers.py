@@ -169,6 +169,9 @@ def test_concat_eltwise_slice_grads():
     layer = make_layer("""name: "e" type: "Eltwise" bottom: "a" bottom: "b"
         top: "y" eltwise_param { operation: MAX }""")
     run_grad_check(layer, [torch.randn(3, 4), torch.randn(3, 4)])
+    layer = make_layer("""name: "e" type: "Eltwise" bottom: "a" bottom: "b"
+        top: "y" eltwise_param { operation: PROD }""")
+    run_grad_check(layer, [torch.randn(3, 4) + 2, torch.randn(3, 4) + 2])
 
 
 def test_power_flatten_reshape_split_grads():
