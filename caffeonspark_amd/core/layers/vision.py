@@ -409,17 +409,21 @@ class BatchNormLayer(Layer):
             self.blobs[1].data.mul_(self.maf).add_(
                 (var * bias_corr).to(self.blobs[1].data.dtype))
             self.blobs[2].data.mul_(self.maf).add_(1.0)
-        # caffe BN has no affine term, so the output IS xhat: backward
-        # reads it from the top blob (in-place safe)
-        self._cache = inv_std
+        # caffe BN has no affine term, so the output IS xhat — but a
+        # downstream in-place layer (the standard BN->Scale->ReLU idiom)
+        # replaces the shared blob's data before our backward runs, so
+        # hold a reference to the xhat tensor itself (Caffe's x_norm_;
+        # layers here assign fresh tensors, never mutate, so the held
+        # reference stays valid)
+        self._cache = (inv_std, y)
         top[0].data = y
         return 0.0
 
     def backward(self, top, propagate_down, bottom):
         if not propagate_down[0]:
             return
-        inv_std = self._cache
-        dx = ops.bn_backward(top[0].data, top[0].diff, inv_std,
+        inv_std, xhat = self._cache
+        dx = ops.bn_backward(xhat, top[0].diff, inv_std,
                              train=not self.use_global)
         self.acc_blob_diff(bottom[0], dx.to(bottom[0].data.dtype),
                            top[0] is bottom[0])
@@ -445,6 +449,12 @@ class ScaleLayer(Layer):
 
     def forward(self, bottom, top):
         x = bottom[0].data
+        # stash the input tensor: when this layer runs in-place
+        # (top is bottom — the BN->Scale idiom), assigning top[0].data
+        # below repoints the shared blob at y, and backward's dscale
+        # needs the original x (Caffe ScaleLayer stashes its in-place
+        # bottom the same way)
+        self._x = x
         scale = bottom[1].data if len(bottom) > 1 else self.weight(0)
         y = x * scale.reshape(self._shape(x))
         if self.bias_term:
@@ -453,7 +463,7 @@ class ScaleLayer(Layer):
         return 0.0
 
     def backward(self, top, propagate_down, bottom):
-        x = bottom[0].data
+        x = self._x
         dy = top[0].diff
         dims = [d for d in range(x.dim()) if d != self.axis]
         if len(bottom) == 1:

@@ -258,3 +258,45 @@ def test_accuracy_top_k():
     layer.setup([bx, bt], top)
     layer.forward([bx, bt], top)
     assert float(top[0].data) == pytest.approx(2.0 / 3.0)
+
+
+def test_inplace_bn_scale_relu_chain_grads():
+    """The standard Caffe in-place BN->Scale->ReLU idiom: all three layers
+    read and write ONE blob. Scale's forward overwrites the blob holding
+    BN's xhat and ReLU overwrites Scale's output, so each layer must cache
+    what its backward needs (Caffe: BatchNorm x_norm_, ScaleLayer's stashed
+    in-place bottom). Verifies dx and dscale/dbias against autograd of the
+    composed chain."""
+    bn = make_layer('name: "bn" type: "BatchNorm" bottom: "x" top: "x"')
+    sc = make_layer("""name: "sc" type: "Scale" bottom: "x" top: "x"
+        scale_param { bias_term: true
+          filler { type: "gaussian" std: 1.0 }
+          bias_filler { type: "gaussian" std: 0.5 } }""")
+    relu = make_layer('name: "r" type: "ReLU" bottom: "x" top: "x"')
+
+    x0 = torch.randn(4, 3, 5, 5)
+    x = x0.detach().clone().requires_grad_(True)
+    blob = Blob(x.shape)
+    blob.data = x
+    chain = [bn, sc, relu]
+    with torch.enable_grad():
+        for layer in chain:
+            layer.setup([blob], [blob])
+        leaves = [x, sc.blobs[0].data.requires_grad_(True),
+                  sc.blobs[1].data.requires_grad_(True)]
+        for layer in chain:
+            layer.forward([blob], [blob])
+        y = blob.data
+        dy = torch.randn_like(y)
+        auto_dx, auto_dscale, auto_dbias = torch.autograd.grad(y, leaves, dy)
+    for p in sc.blobs:
+        p.data = p.data.detach()
+        p.zero_diff()
+    blob.diff = dy
+    for layer in reversed(chain):
+        layer.backward([blob], [True], [blob])
+    torch.testing.assert_close(blob.diff, auto_dx, rtol=2e-3, atol=1e-4)
+    torch.testing.assert_close(sc.blobs[0].diff, auto_dscale,
+                               rtol=2e-3, atol=1e-4)
+    torch.testing.assert_close(sc.blobs[1].diff, auto_dbias,
+                               rtol=2e-3, atol=1e-4)
