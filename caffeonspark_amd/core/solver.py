@@ -123,12 +123,24 @@ class Solver:
             else:
                 self.segments.append((o, b.count, *key))
         # per-layer arena slices in reverse-layer order (backward completion
-        # order) for overlap-friendly bucket all-reduce
+        # order) for overlap-friendly bucket all-reduce.  Each unique param
+        # is attributed ONLY to its first (earliest forward-order) owner:
+        # with caffe named-param sharing (param { name }) a blob is consumed
+        # by several layers and its gradient accumulates across all their
+        # backwards — the earliest layer's backward runs LAST in reverse
+        # order, so a bucket keyed on it fires exactly once and only after
+        # every sharer has accumulated.  First-appearance attribution also
+        # keeps reverse-order arena ranges descending and non-overlapping
+        # (the arena itself is laid out in first-appearance order above).
         by_id = {id(b): o for b, o in zip(self.params, self.param_offsets)}
+        claimed: set = set()
         self.layer_slices = []  # (layer_name, [(off, count), ...])
         for layer in self.net.layers:
-            sl = [(by_id[id(b)], b.count) for b in layer.blobs
-                  if id(b) in by_id]
+            sl = []
+            for b in layer.blobs:
+                if id(b) in by_id and id(b) not in claimed:
+                    claimed.add(id(b))
+                    sl.append((by_id[id(b)], b.count))
             if sl:
                 self.layer_slices.append((layer.name, sl))
 

@@ -87,6 +87,21 @@ class DistributedSync(Callback):
             self.buckets.append((lo, hi))
             for l in pend_layers:
                 self.layer_bucket[l] = idx
+        # invariant: buckets partition the arena — every element
+        # all-reduced exactly once (shared params would otherwise be
+        # scaled by an extra world_size factor and raced by overlapping
+        # async all_reduce calls)
+        spans = sorted(self.buckets)
+        for (a0, a1), (b0, b1) in zip(spans, spans[1:]):
+            if b0 < a1:
+                raise AssertionError(
+                    f"overlapping gradient buckets ({a0},{a1}) / ({b0},{b1})")
+        covered = sum(hi - lo for lo, hi in spans)
+        want = sum(n for _, sl in solver.layer_slices for _, n in sl)
+        if covered != want:
+            raise AssertionError(
+                f"gradient buckets cover {covered} elements, "
+                f"arena slices total {want}")
         # layers remaining per bucket before it can fire
         self._layers_per_bucket = [0] * len(self.buckets)
         for l, i in self.layer_bucket.items():

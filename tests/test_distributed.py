@@ -250,3 +250,192 @@ def _run_frozen():
         torch.testing.assert_close(torch.from_numpy(out[r][1]),
                                    torch.from_numpy(out[r][2]),
                                    rtol=0, atol=0)
+
+
+SHARED_NET = """
+  layer { name: "d" type: "MemoryData" top: "x" top: "t"
+          memory_data_param { batch_size: 16 channels: 1 height: 8
+                              width: 8 } }
+  layer { name: "ip1" type: "InnerProduct" bottom: "x" top: "h1"
+          param { name: "shared_w" } param { name: "shared_b" }
+          inner_product_param { num_output: 64
+            weight_filler { type: "gaussian" std: 0.1 } } }
+  layer { name: "r1" type: "ReLU" bottom: "h1" top: "h1" }
+  layer { name: "ip2" type: "InnerProduct" bottom: "h1" top: "h2"
+          inner_product_param { num_output: 64
+            weight_filler { type: "gaussian" std: 0.1 } } }
+  layer { name: "r2" type: "ReLU" bottom: "h2" top: "h2" }
+  layer { name: "ip3" type: "InnerProduct" bottom: "h2" top: "h3"
+          param { name: "shared_w" } param { name: "shared_b" }
+          inner_product_param { num_output: 64
+            weight_filler { type: "gaussian" std: 0.1 } } }
+  layer { name: "ip4" type: "InnerProduct" bottom: "h3" top: "z"
+          inner_product_param { num_output: 4
+            weight_filler { type: "gaussian" std: 0.1 } } }
+  layer { name: "l" type: "SoftmaxWithLoss" bottom: "z" bottom: "t"
+          top: "loss" }
+"""
+
+
+def _shared_solver(seed=7):
+    from caffeonspark_amd.core.solver import Solver
+    from caffeonspark_amd.proto import caffe_pb, text_format
+
+    sp = caffe_pb.SolverParameter(
+        net_param=text_format.parse(SHARED_NET, caffe_pb.NetParameter),
+        base_lr=0.05, momentum=0.9, lr_policy="fixed", max_iter=10,
+        random_seed=seed, display=0)
+    return Solver(sp)
+
+
+def test_shared_param_buckets_exactly_once():
+    """caffe named-param sharing (param { name }): two layers alias one
+    arena range. Buckets must partition the arena — the old per-layer
+    min/max span assembly emitted overlapping buckets that all-reduced
+    the shared range twice (x world_size gradient inflation) and raced
+    two async all_reduce calls on the same buffer."""
+    from caffeonspark_amd.parallel.ddp import DistributedSync
+
+    s = _shared_solver()
+    sync = DistributedSync(s, bucket_mb=0.01)  # force several buckets
+    assert len(sync.buckets) >= 2
+    hits = torch.zeros(int(s.flat_g.numel()), dtype=torch.int32)
+    for lo, hi in sync.buckets:
+        hits[lo:hi] += 1
+    assert int(hits.min()) == 1 and int(hits.max()) == 1
+
+    # the shared blob's bucket fires only at its EARLIEST layer's backward
+    # (the last of its consumers to run in reverse order)
+    shared_off = s.param_offsets[0]  # ip1_w is first param in the arena
+    shared_bucket = next(i for i, (lo, hi) in enumerate(sync.buckets)
+                         if lo <= shared_off < hi)
+    assert sync.layer_bucket.get("ip1") == shared_bucket
+    assert "ip3" not in sync.layer_bucket or \
+        sync.layer_bucket["ip3"] != shared_bucket or \
+        sync.layer_bucket.get("ip1") == sync.layer_bucket.get("ip3")
+
+
+def _worker_shared(rank, ws, store_path, q, data):
+    import torch.distributed as dist
+
+    from caffeonspark_amd.parallel import DistributedSync
+
+    dist.init_process_group("gloo", rank=rank, world_size=ws,
+                            init_method=f"file://{store_path}")
+    try:
+        s = _shared_solver()
+        sync = DistributedSync(s, bucket_mb=0.02)
+        sync.broadcast_params()
+        x, y = data
+        half = x.shape[1] // ws
+        for it in range(3):
+            s.net.data_layers()[0].reset(
+                x[it][rank * half:(rank + 1) * half],
+                y[it][rank * half:(rank + 1) * half])
+            s._step_one()
+        q.put((rank, s.flat_w.numpy().copy()))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ddp_shared_params_match_single():
+    """2-rank DP with param{name} weight sharing == single process on the
+    full batch. The old overlapping-bucket assembly double-reduced the
+    shared range and diverged."""
+    _retry(_run_shared_match)
+
+
+def _run_shared_match():
+    import tempfile, uuid
+    torch.manual_seed(11)
+    x = torch.stack([torch.randn(32, 1, 8, 8) for _ in range(3)])
+    y = torch.stack([torch.randint(0, 4, (32,)).float() for _ in range(3)])
+    ws = 2
+    store = tempfile.mktemp(prefix=f"cosamd_ddps_{uuid.uuid4().hex}_")
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_shared,
+                         args=(r, ws, store, q, (x, y))) for r in range(ws)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(ws):
+        rank, w = q.get(timeout=300)
+        results[rank] = w
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    torch.testing.assert_close(torch.from_numpy(results[0]),
+                               torch.from_numpy(results[1]),
+                               rtol=0, atol=0)
+    single = _shared_solver()
+    for it in range(3):
+        single.net.data_layers()[0].reset(x[it], y[it])
+        single._step_one()
+    torch.testing.assert_close(torch.from_numpy(results[0]), single.flat_w,
+                               rtol=1e-4, atol=1e-6)
+
+
+def _worker_lrcn(rank, ws, store_path, q):
+    import torch.distributed as dist
+
+    from caffeonspark_amd.core import solver_from_prototxt
+    from caffeonspark_amd.core.layers.data import CoSDataLayer
+    from caffeonspark_amd.parallel import DistributedSync
+
+    dist.init_process_group("gloo", rank=rank, world_size=ws,
+                            init_method=f"file://{store_path}")
+    try:
+        s = solver_from_prototxt(
+            os.path.join(os.path.dirname(__file__), "..",
+                         "caffeonspark_amd", "models",
+                         "lrcn_solver.prototxt"))
+        s.param.display = 0
+        sync = DistributedSync(s, bucket_mb=4.0)
+        sync.broadcast_params()
+        dl = s.net.data_layers()[0]
+        assert isinstance(dl, CoSDataLayer)
+        T = int(dl.tops_cfg[2].channels)
+        V, n = 8801, 2
+        g = torch.Generator().manual_seed(50 + rank)
+        dl.batch_size = n
+        for _ in range(2):
+            x = torch.randn(n, 3, 227, 227, generator=g)
+            cont = torch.ones(T, n)
+            cont[0] = 0
+            inp = torch.randint(0, V, (T, n), generator=g).float()
+            tgt = torch.randint(0, V, (T, n), generator=g).float()
+            dl.reset([x, torch.zeros(n, 1), cont, inp, tgt])
+            s._step_one()
+        q.put((rank, s.flat_w.numpy().copy()))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ddp_lrcn_two_ranks_identical():
+    """The actual LRCN net (CNN+LSTM, the BASELINE 4-GPU config) under
+    2-rank gloo DP: ranks stay bit-identical through synchronized steps
+    on different data."""
+    _retry(_run_lrcn)
+
+
+def _run_lrcn():
+    import tempfile, uuid
+    ws = 2
+    store = tempfile.mktemp(prefix=f"cosamd_lrcn_{uuid.uuid4().hex}_")
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_lrcn, args=(r, ws, store, q))
+             for r in range(ws)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(ws):
+        rank, w = q.get(timeout=600)
+        results[rank] = w
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    torch.testing.assert_close(torch.from_numpy(results[0]),
+                               torch.from_numpy(results[1]),
+                               rtol=0, atol=0)
