@@ -280,8 +280,8 @@ class DropoutLayer(Layer):
 
     def forward(self, bottom, top):
         if self.phase == caffe_pb.Phase.TRAIN:
-            top[0].data, self._mask = ops.dropout_forward(bottom[0].data,
-                                                          self.ratio)
+            top[0].data, self._mask = ops.dropout_forward(
+                bottom[0].data, self.ratio, generator=self.net.generator)
         else:
             top[0].data = bottom[0].data
         return 0.0

@@ -438,13 +438,17 @@ def relu_backward(y, dy, negative_slope=0.0):
 _dropout_seeds = {}
 
 
-def _dropout_seed(device):
+def _dropout_seed(device, generator=None):
     """Per-device RNG seed in DEVICE memory, advanced on-GPU by seed_bump:
     no host sync per call, and a hipGraph-captured step re-draws the mask
-    on every replay (a host kernel-arg seed would freeze it)."""
+    on every replay (a host kernel-arg seed would freeze it).  The initial
+    value derives from the net's seeded CPU generator when one is given,
+    so solver random_seed makes GPU dropout reproducible too."""
     t = _dropout_seeds.get(device.index)
     if t is None:
-        t = torch.randint(1, 2 ** 62, (1,), dtype=torch.int64, device=device)
+        init = torch.randint(1, 2 ** 62, (1,), dtype=torch.int64,
+                             generator=generator)
+        t = init.to(device)
         _dropout_seeds[device.index] = t
     return t
 
@@ -452,7 +456,7 @@ def _dropout_seed(device):
 def dropout_forward(x, ratio, generator=None):
     y = torch.empty_like(x)
     mask = torch.empty_like(x)
-    seed = _dropout_seed(x.device)
+    seed = _dropout_seed(x.device, generator)
     _ext.seed_bump(seed)
     _ext.dropout_fwd(x, y, mask, ratio, seed)
     return y, mask

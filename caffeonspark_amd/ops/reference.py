@@ -208,7 +208,11 @@ def softmax_loss_backward(prob, label, ignore_label: Optional[int], scale, axis=
 
 def dropout_forward(x, ratio, generator=None):
     keep = 1.0 - ratio
-    mask = (torch.rand_like(x, dtype=torch.float32) < keep).to(x.dtype) / keep
+    # draw through the net's seeded generator so solver random_seed makes
+    # masks reproducible (upstream Caffe's seeded RNG covers dropout)
+    r = torch.rand(x.shape, dtype=torch.float32, device=x.device,
+                   generator=generator)
+    mask = (r < keep).to(x.dtype) / keep
     return x * mask, mask
 
 

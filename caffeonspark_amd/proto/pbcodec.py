@@ -160,7 +160,11 @@ class Message(metaclass=MessageMeta):
         if isinstance(v, list):
             return len(v) > 0
         if isinstance(v, Message):
-            return True
+            # mirror the serializer: a nested message auto-vivified by a
+            # mere attribute read (and still content-free) is NOT present
+            # — otherwise reading solver.net_param anywhere would flip
+            # Solver._resolve_net_param into building an empty net
+            return bool(v._present())
         return True
 
     def clear_field(self, name: str) -> None:
