@@ -139,7 +139,10 @@ class CaffeOnSpark:
                     for n in blob_names}
             label = batch[1].float().cpu() if len(batch) > 1 else None
             for i in range(bs):
-                row = {"SampleID": f"{len(rows)}"}
+                # rank-qualified SampleID: globally unique under torchrun
+                sid = f"{self.rank}_{len(rows)}" if self.world > 1 \
+                    else f"{len(rows)}"
+                row = {"SampleID": sid}
                 if label is not None and self.conf.label:
                     row[self.conf.label] = float(label[i])
                 for name, t in outs.items():
@@ -151,9 +154,26 @@ class CaffeOnSpark:
             if max_samples is not None and len(rows) >= max_samples:
                 break
         proc.stop()
+        rows = self._gather_rows(rows)
         df = pd.DataFrame(rows)
         self._write_output(df)
         return df
+
+    def _gather_rows(self, rows: list) -> list:
+        """Collect feature rows from every rank (reference features2
+        collects executor Rows into one driver DataFrame —
+        CaffeOnSpark.scala:445-506).  Every rank returns the full set so
+        test()'s aggregate is identical everywhere; _write_output then
+        writes once from rank 0 without dropping (world-1)/world of the
+        output."""
+        if self.world <= 1:
+            return rows
+        import torch.distributed as dist
+        if not dist.is_initialized():
+            return rows
+        gathered: List[Optional[list]] = [None] * self.world
+        dist.all_gather_object(gathered, rows)
+        return [r for part in gathered for r in (part or [])]
 
     def test(self, source: Optional[DataSource] = None,
              max_samples: Optional[int] = None) -> dict:

@@ -167,18 +167,42 @@ class CaffeProcessor:
         tn = s.test_nets[0]
         iters = s.param.test_iter[0] if s.param.test_iter else 1
         dl = tn.data_layers()[0]
+        names = list(tn.output_blob_names())
         sums: dict = {}
+        done = 0
         for _ in range(iters):
             batch = self._get_full(1)
             if batch is STOP_MARK:
-                return
+                break
             self._reset_layer(dl, batch)
             tn.forward()
-            for name in tn.output_blob_names():
+            for name in names:
                 v = tn.blob_by_name(name).data.float().mean().item()
                 sums[name] = sums.get(name, 0.0) + v
-        self.validation_results.append(
-            {k: v / iters for k, v in sums.items()})
+            done += 1
+        # cross-rank aggregation (reference: validation scores accumulate
+        # across the cluster and are reported once — CaffeOnSpark.scala:
+        # 284-341 + CaffeNet.cpp:34-62).  Every rank validates its own
+        # 1/world shard of the stream; all-reducing [count, sums...] makes
+        # the reported metric the mean over the FULL stream and identical
+        # on every rank.  Ranks step in lockstep, so all of them reach
+        # this collective at the same iteration (a STOP_MARK break above
+        # still falls through to here — no rank may early-return).
+        import torch
+        import torch.distributed as dist
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            dev = self.device if self.device.type == "cuda" and \
+                dist.get_backend() == "nccl" else "cpu"
+            vec = torch.tensor(
+                [float(done)] + [sums.get(n, 0.0) for n in names],
+                dtype=torch.float64, device=dev)
+            dist.all_reduce(vec)
+            vec = vec.cpu()
+            done = max(1.0, float(vec[0]))
+            sums = {n: float(vec[1 + i]) for i, n in enumerate(names)}
+        if done:
+            self.validation_results.append(
+                {k: v / done for k, v in sums.items()})
 
     # ---------------------------------------------------------------- misc
     def snapshot(self) -> str:
