@@ -26,6 +26,10 @@ def main():
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=0)
     ap.add_argument("--model", type=str, default="alexnet")
+    ap.add_argument("--dtype", type=str, default="",
+                    choices=["", "bf16", "fp32"],
+                    help="compute dtype (default bf16 on GPU; fp32 runs the "
+                         "declared rocBLAS/MIOpen fp32 path)")
     ap.add_argument("--graph", type=int, default=-1,
                     help="hipGraph-captured steps when single-process GPU "
                          "(-1 = auto: on for launch-bound sub-ms models)")
@@ -41,7 +45,7 @@ def main():
     if use_gpu:
         torch.cuda.set_device(local_rank)
         dev = torch.device("cuda", local_rank)
-        dtype = torch.bfloat16
+        dtype = torch.float32 if args.dtype == "fp32" else torch.bfloat16
     else:
         dev = torch.device("cpu")
         dtype = torch.float32
@@ -129,7 +133,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if use_gpu else "fp32",
+            "dtype": "fp32" if dtype == torch.float32 else "bf16",
             "data": "synthetic",
             "config": {"model": {"alexnet": "alexnet(bvlc_reference/CaffeNet)"
                                  }.get(args.model, args.model),

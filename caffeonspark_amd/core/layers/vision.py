@@ -66,7 +66,7 @@ class ConvolutionLayer(Layer):
         self._ctx = {}
         kw = {}
         co = getattr(self, "_concat_out", None)
-        if co is not None and x.is_cuda:
+        if co is not None and x.is_cuda and x.dtype == torch.bfloat16:
             concat_l, c_off = co
             eff_h = (self.kh - 1) * self.dil + 1
             eff_w = (self.kw - 1) * self.dil + 1

@@ -209,9 +209,14 @@ def softmax_loss_backward(prob, label, ignore_label: Optional[int], scale, axis=
 def dropout_forward(x, ratio, generator=None):
     keep = 1.0 - ratio
     # draw through the net's seeded generator so solver random_seed makes
-    # masks reproducible (upstream Caffe's seeded RNG covers dropout)
-    r = torch.rand(x.shape, dtype=torch.float32, device=x.device,
-                   generator=generator)
+    # masks reproducible (upstream Caffe's seeded RNG covers dropout); a
+    # CPU generator serving a CUDA tensor (fp32 GPU path) draws host-side
+    if generator is not None and generator.device != x.device:
+        r = torch.rand(x.shape, dtype=torch.float32,
+                       generator=generator).to(x.device)
+    else:
+        r = torch.rand(x.shape, dtype=torch.float32, device=x.device,
+                       generator=generator)
     mask = (r < keep).to(x.dtype) / keep
     return x * mask, mask
 
