@@ -199,16 +199,23 @@ class CaffeOnSpark:
         path = self.conf.outputPath
         if not path or self.rank != 0:
             return
-        if path.startswith("file:"):
-            path = path[5:]
+        from ..utils.fsio import copy_to_uri, is_remote, split_scheme
         fmt = self.conf.outputFormat.lower()
-        os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
+        if is_remote(path):
+            import tempfile
+            local = tempfile.mktemp(suffix=f".{fmt}")
+        else:
+            local = split_scheme(path)[1] or path
+            os.makedirs(os.path.dirname(local) or ".", exist_ok=True)
         if fmt == "json":
-            df.to_json(path, orient="records", lines=True)
+            df.to_json(local, orient="records", lines=True)
         elif fmt == "parquet":
-            df.to_parquet(path)
+            df.to_parquet(local)
         else:
             raise ValueError(f"unknown outputFormat {fmt!r}")
+        if is_remote(path):
+            copy_to_uri(local, path)
+            os.unlink(local)
 
 
 def main(argv: Optional[List[str]] = None) -> None:

@@ -17,6 +17,7 @@ from typing import Dict, List, Optional
 
 import torch
 
+from .. import ops
 from ..proto import caffe_pb, text_format
 from .blob import Blob
 from .layers import base as layer_base
@@ -79,6 +80,13 @@ class Net:
         self.layer_prop_down: List[List[bool]] = []
         self.shared_params: Dict[str, Blob] = {}
         self._loss_tops: List[tuple] = []  # (layer_idx, top_idx, weight)
+        # ops-dispatch mode this net declares while its forward/backward
+        # runs: fp32-on-GPU routes activation ops to the reference torch
+        # impls (the declared fp32 path), bf16 forces the native kernels
+        if self.device.type == "cuda":
+            self._ops_mode = "fp32" if dtype == torch.float32 else "bf16"
+        else:
+            self._ops_mode = None
         self._build()
 
     # ------------------------------------------------------------------ build
@@ -244,9 +252,13 @@ class Net:
         refresh = getattr(self, "_bf16_refresh", None)
         if refresh is not None:
             refresh()      # re-cast the fp32 master arena to bf16 once
-        for layer, bottoms, tops in zip(self.layers, self.layer_bottoms,
-                                        self.layer_tops):
-            layer.forward(bottoms, tops)
+        ops.set_active_gpu_mode(self._ops_mode)
+        try:
+            for layer, bottoms, tops in zip(self.layers, self.layer_bottoms,
+                                            self.layer_tops):
+                layer.forward(bottoms, tops)
+        finally:
+            ops.set_active_gpu_mode(None)
         if read_loss:
             for (li, ti, w) in self._loss_tops:
                 loss += w * float(self.layer_tops[li][ti].data.float().sum())
@@ -262,20 +274,24 @@ class Net:
             top = self.layer_tops[li][ti]
             top.diff = torch.full_like(top.data, w, dtype=torch.float32)
             top._loss_weight = w  # host-side copy: layers avoid a sync
-        for i in range(len(self.layers) - 1, -1, -1):
-            if not self.layer_need_backward[i]:
-                continue
-            tops = self.layer_tops[i]
-            if all(t.diff is None for t in tops) and not any(
-                    b._lr_mult != 0 for b in self.layers[i].blobs):
-                continue
-            for t in tops:
-                if t.diff is None:
-                    t.ensure_diff()
-            self.layers[i].backward(tops, self.layer_prop_down[i],
-                                    self.layer_bottoms[i])
-            if on_layer_done is not None and self.layers[i].blobs:
-                on_layer_done(self.layers[i])
+        ops.set_active_gpu_mode(self._ops_mode)
+        try:
+            for i in range(len(self.layers) - 1, -1, -1):
+                if not self.layer_need_backward[i]:
+                    continue
+                tops = self.layer_tops[i]
+                if all(t.diff is None for t in tops) and not any(
+                        b._lr_mult != 0 for b in self.layers[i].blobs):
+                    continue
+                for t in tops:
+                    if t.diff is None:
+                        t.ensure_diff()
+                self.layers[i].backward(tops, self.layer_prop_down[i],
+                                        self.layer_bottoms[i])
+                if on_layer_done is not None and self.layers[i].blobs:
+                    on_layer_done(self.layers[i])
+        finally:
+            ops.set_active_gpu_mode(None)
 
     def forward_backward(self) -> float:
         loss = self.forward()

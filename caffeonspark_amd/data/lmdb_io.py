@@ -41,12 +41,17 @@ def _resolve(path: str) -> str:
 
 class LmdbReader:
     def __init__(self, path: str):
+        import mmap
         self.path = _resolve(path)
         self._f = open(self.path, "rb")
-        self.data = self._f.read()  # datasets fit memory for our use
+        # mmap, not read(): partitioned key-range readers touch only the
+        # pages of their range (reference LmdbRDD gives each executor a
+        # disjoint key range — LmdbRDD.scala:41-95)
+        self.data = mmap.mmap(self._f.fileno(), 0, access=mmap.ACCESS_READ)
         self._load_meta()
 
     def close(self):
+        self.data.close()
         self._f.close()
 
     def __enter__(self):
@@ -111,6 +116,92 @@ class LmdbReader:
         if self.root == P_INVALID or self.entries == 0:
             return
         yield from self._iter_page(self.root)
+
+    # ---- key-range partitioned access (reference LmdbRDD.scala:41-95:
+    # partitions are key ranges; each executor reads a disjoint range) ----
+
+    def _branch_children(self, pgno: int) -> List[Tuple[bytes, int]]:
+        """(separator_key, child_pgno) pairs of a branch page.  Child i
+        holds keys in [key_i, key_{i+1}); key_0 is b'' (= lower bound of
+        the parent's range)."""
+        pg = self._page(pgno)
+        flags, lower = struct.unpack_from("<HH", pg, 10)
+        assert flags & P_BRANCH
+        nkeys = (lower - PAGEHDRSZ) >> 1
+        out = []
+        for i in range(nkeys):
+            noff = struct.unpack_from("<H", pg, PAGEHDRSZ + 2 * i)[0]
+            lo, hi, nflags, ksize = struct.unpack_from("<HHHH", pg, noff)
+            child = lo | (hi << 16) | (nflags << 32)
+            key = bytes(pg[noff + 8:noff + 8 + ksize])
+            out.append((key, child))
+        return out
+
+    def _page_flags(self, pgno: int) -> int:
+        return struct.unpack_from("<H", self._page(pgno), 10)[0]
+
+    def _iter_range(self, pgno: int, start, end
+                    ) -> Iterator[Tuple[bytes, bytes]]:
+        """In-order scan of keys in [start, end) touching only pages that
+        can intersect the range (start/end None = unbounded)."""
+        flags = self._page_flags(pgno)
+        if flags & P_BRANCH:
+            ch = self._branch_children(pgno)
+            for i, (key, child) in enumerate(ch):
+                nxt = ch[i + 1][0] if i + 1 < len(ch) else None
+                if end is not None and i > 0 and key >= end:
+                    break
+                if start is not None and nxt is not None and nxt <= start:
+                    continue
+                yield from self._iter_range(child, start, end)
+        elif flags & P_LEAF:
+            for key, val in self._iter_page(pgno):
+                if start is not None and key < start:
+                    continue
+                if end is not None and key >= end:
+                    break
+                yield key, val
+
+    def items_range(self, start=None, end=None
+                    ) -> Iterator[Tuple[bytes, bytes]]:
+        if self.root == P_INVALID or self.entries == 0:
+            return
+        yield from self._iter_range(self.root, start, end)
+
+    def partition_ranges(self, n: int) -> List[Tuple[bytes, bytes]]:
+        """n (start, end) key ranges covering the whole DB disjointly
+        (start None = -inf, end None = +inf).  Split keys come from branch
+        separator keys, so discovery reads only branch pages — no full
+        scan (the reference discovers ranges with a stride iterator over
+        all keys, LmdbRDD.scala:41-95; the B+tree gives them for free)."""
+        n = max(1, int(n))
+        if self.root == P_INVALID or self.entries == 0:
+            return [(None, None)]
+        level: List[Tuple[bytes, int]] = [(b"", self.root)]
+        while len(level) < n:
+            nxt: List[Tuple[bytes, int]] = []
+            grew = False
+            for key, pgno in level:
+                if self._page_flags(pgno) & P_BRANCH:
+                    ch = self._branch_children(pgno)
+                    # first child inherits the parent's lower bound
+                    nxt.append((key, ch[0][1]))
+                    nxt.extend(ch[1:])
+                    grew = True
+                else:
+                    nxt.append((key, pgno))
+            level = nxt
+            if not grew:
+                break
+        m = len(level)
+        n = min(n, m)
+        bounds = [level[i * m // n][0] for i in range(n)]
+        ranges: List[Tuple[bytes, bytes]] = []
+        for i in range(n):
+            start = bounds[i] if i > 0 else None
+            end = bounds[i + 1] if i + 1 < n else None
+            ranges.append((start, end))
+        return ranges
 
     def __len__(self):
         return self.entries

@@ -209,11 +209,10 @@ class CaffeProcessor:
         model_file = self.solver.snapshot()
         out = getattr(self.conf, "model_path", None)
         if out:
-            if out.startswith("file:"):
-                out = out[5:]          # reference FSUtils URI handling
-            import shutil
-            os.makedirs(os.path.dirname(out) or ".", exist_ok=True)
-            shutil.copyfile(model_file, out)
+            # reference FSUtils.GenModelOrState: snapshot locally, then
+            # move/upload to the -model URI (hdfs:// etc. via fsspec)
+            from ..utils.fsio import copy_to_uri
+            copy_to_uri(model_file, out)
         return model_file
 
     def sync(self) -> None:

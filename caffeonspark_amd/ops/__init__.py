@@ -82,6 +82,25 @@ _FP32_REF_ON_GPU = {
     "bn_backward", "accuracy", "bias_add",
 }
 
+# Ops whose fp32 routing can ALSO be decided by dtype sniffing when no net
+# has declared a mode (direct op calls in tests/pycaffe).  Backward ops
+# that consume an opaque forward context are excluded: the bf16 native
+# forward packs fp32 tensors into its context, so dtype alone cannot tell
+# the paths apart — those rely on the net-declared mode below.
+_FP32_SNIFF_OK = _FP32_REF_ON_GPU - {"softmax_loss_backward",
+                                     "lstm_unit_backward",
+                                     "maxpool_backward", "bn_backward"}
+
+# Net.forward/backward declare their compute mode here ("fp32" routes the
+# _FP32_REF_ON_GPU set to the reference impls; "bf16" forces native; None
+# = no net active, fall back to dtype sniffing)
+_ACTIVE_GPU_MODE = None
+
+
+def set_active_gpu_mode(mode):
+    global _ACTIVE_GPU_MODE
+    _ACTIVE_GPU_MODE = mode
+
 
 def _args_on_gpu(args) -> bool:
     for a in args:
@@ -115,7 +134,11 @@ def _dispatch(name: str):
         on_gpu = _args_on_gpu(args)
         if on_gpu:
             native()  # force extension load (raises loudly if missing)
-            if name in _FP32_REF_ON_GPU and not _any_bf16(args):
+            if name in _FP32_REF_ON_GPU and (
+                    _ACTIVE_GPU_MODE == "fp32"
+                    or (_ACTIVE_GPU_MODE is None
+                        and name in _FP32_SNIFF_OK
+                        and not _any_bf16(args))):
                 # declared fp32 GPU path (see _FP32_REF_ON_GPU docstring)
                 return ref_fn(*args, **kwargs)
             fn = dispatcher.gpu_impls.get(name)
