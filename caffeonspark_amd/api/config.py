@@ -50,6 +50,7 @@ class Config:
         a("-dtype", dest="dtype_name", default="",
           help="compute dtype: bf16|fp32 (default: bf16 on GPU)")
         ns, _ = p.parse_known_args(args or [])
+        self.raw_args = list(args or [])   # re-parse on Spark executors
         self.__dict__.update(vars(ns))
         self.__dict__.update(kw)
 

@@ -42,6 +42,13 @@ class CaffeProcessor:
             return _instance
 
     @classmethod
+    def current(cls) -> Optional["CaffeProcessor"]:
+        """The executor-resident singleton, or None (reference
+        CaffeProcessor.instance() companion-object lookup)."""
+        with _instance_lock:
+            return _instance
+
+    @classmethod
     def reset_instance(cls):
         global _instance
         with _instance_lock:

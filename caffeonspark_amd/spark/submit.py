@@ -1,0 +1,89 @@
+"""spark-submit entry point.
+
+Mirrors the reference launch
+(`spark-submit --master ... --class com.yahoo.ml.caffe.CaffeOnSpark
+caffe-grid-*-jar-with-dependencies.jar -train -conf ...`):
+
+    python -m caffeonspark_amd.spark.submit --master "local[8]" \
+        --class com.yahoo.ml.caffe.CaffeOnSpark \
+        -conf lenet_memory_solver.prototxt -train -clusterSize 8 -devices 1
+
+or with a python app file (PySpark style):
+
+    python -m caffeonspark_amd.spark.submit --master "local[*]" app.py args...
+
+With real pyspark installed, delegates to the genuine spark-submit when
+found on PATH; otherwise runs the app in-process against the bundled
+local engine (COS_SPARK_MASTER carries the master to the driver).
+"""
+
+from __future__ import annotations
+
+import os
+import runpy
+import shutil
+import subprocess
+import sys
+from typing import List
+
+
+def main(argv: List[str] = None) -> int:
+    argv = list(sys.argv[1:] if argv is None else argv)
+    master = "local[*]"
+    app_class = ""
+    confs: List[str] = []
+    rest: List[str] = []
+    i = 0
+    while i < len(argv):
+        a = argv[i]
+        if a == "--master":
+            master = argv[i + 1]
+            i += 2
+        elif a == "--class":
+            app_class = argv[i + 1]
+            i += 2
+        elif a == "--conf":
+            confs.append(argv[i + 1])
+            i += 2
+        elif a in ("--files", "--jars", "--py-files", "--num-executors",
+                   "--executor-memory", "--driver-memory",
+                   "--executor-cores", "--driver-library-path",
+                   "--driver-class-path", "--queue", "--name"):
+            i += 2                      # accepted-and-ignored launch knobs
+        elif a.endswith(".jar"):
+            i += 1                      # the reference's assembly jar slot
+        else:
+            rest.append(a)
+            i += 1
+
+    real = shutil.which("spark-submit")
+    if real and os.environ.get("COS_USE_REAL_SPARK", "1") != "0":
+        try:
+            import pyspark  # noqa: F401
+            cmd = [real, "--master", master]
+            for c in confs:
+                cmd += ["--conf", c]
+            app = rest[0] if rest and rest[0].endswith(".py") else None
+            if app:
+                cmd += rest
+            else:
+                # scala-class launches map onto our python driver module
+                cmd += ["-m", "caffeonspark_amd.spark.driver"] + rest
+            return subprocess.call(cmd)
+        except ImportError:
+            pass
+
+    os.environ["COS_SPARK_MASTER"] = master
+    if rest and rest[0].endswith(".py"):
+        app, app_args = rest[0], rest[1:]
+        sys.argv = [app] + app_args
+        runpy.run_path(app, run_name="__main__")
+        return 0
+    # --class com.yahoo.ml.caffe.CaffeOnSpark (or no app file): run the
+    # built-in driver with the reference CLI flags
+    from .driver import main as driver_main
+    return driver_main(rest)
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())
