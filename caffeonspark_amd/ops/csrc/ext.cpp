@@ -16,6 +16,7 @@ void gemm_bf16(const void* A, const void* B, void* C, const float* bias,
                bool trans_a, bool trans_b, int store_mode, int splitk,
                bool relu, float alpha, int m_alloc, int n_alloc,
                hipStream_t stream);
+void tr16_probe(float* out, int mode, hipStream_t stream);
 void wino_conv(const void* x, const float* w, const float* bias, void* y,
                void* U, void* V, void* Mbuf, int N, int H, int W, int Cin,
                int K, int wK, int wC, int u_rows_alloc, bool flip,
@@ -256,6 +257,10 @@ void py_bias_act_cast(Tensor in, c10::optional<Tensor> bias, Tensor out,
                         in.size(0), in.size(1), relu, cur_stream());
 }
 
+void py_tr16_probe(Tensor out, int64_t mode) {
+  cosamd::tr16_probe(out.data_ptr<float>(), (int)mode, cur_stream());
+}
+
 void py_transpose(Tensor in, Tensor out, int64_t R, int64_t C) {
   CHECK_BF16(in); CHECK_BF16(out);
   cosamd::transpose_bf16(in.data_ptr(), out.data_ptr(), R, C, cur_stream());
@@ -418,6 +423,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_update_multi", &py_sgd_update_multi);
   m.def("colsum", &py_colsum);
   m.def("transpose", &py_transpose);
+  m.def("tr16_probe", &py_tr16_probe);
   m.def("bias_act_cast", &py_bias_act_cast);
   m.def("lstm_seq_fwd", &py_lstm_seq_fwd);
   m.def("lstm_seq_bwd", &py_lstm_seq_bwd);

@@ -74,8 +74,58 @@ __device__ __forceinline__ void stage_direct_guarded(
   }
 }
 
-// transposed ([K][rows]) staging: vector load 8 contiguous along rows from
-// one K-row, scatter into LDS columns.
+// ---- transposed ([K][rows]) operand staging into the CANONICAL layout
+// with u32 k-pair packing: each slot loads the same 8 output-dim columns
+// of TWO adjacent K-rows (two coalesced 16-byte global reads), packs
+// (k, k+1) element pairs into u32s, and writes 8 ds_write_b32 — half
+// the write instructions at 4x wider grain than the previous 16 scalar
+// b16 scatters, and ~4-way instead of 16-way bank conflicts (row*64B
+// strides alias banks; the k-pair dimension now spreads them).
+// Fragment reads stay the standard swizzled b128 path.
+__device__ __forceinline__ void stage_trans_pair_guarded(
+    bf16* lds_, const bf16* g_, int row0, int rows, int ld,
+    int k0, int kend, int tid, int nthreads = NTHREADS,
+    int tile_rows = BM) {
+  auto* lds = reinterpret_cast<unsigned short*>(lds_);
+  auto* g = reinterpret_cast<const unsigned short*>(g_);
+  int rslots = tile_rows >> 3;          // 8-column octets per k-pair
+  int nslots = 16 * rslots;             // 16 k-pairs per BK=32 tile
+  for (int slot = tid; slot < nslots; slot += nthreads) {
+    int kp = slot & 15;                 // k-pair index (adjacent lanes
+    int m0 = (slot >> 4) * 8;           //   get distinct kp: bank spread)
+    unsigned short va[8], vb[8];
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      unsigned short* v = h ? vb : va;
+      int gk = k0 + kp * 2 + h;
+      int gr = row0 + m0;
+      if (gk < kend && gr + 8 <= rows && ((ld | gr) % 8 == 0)) {
+        *reinterpret_cast<short8v*>(v) =
+            *reinterpret_cast<const short8v*>(g + (int64_t)gk * ld + gr);
+      } else if (gk < kend) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          v[j] = (gr + j < rows) ? g[(int64_t)gk * ld + gr + j] : 0;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v[j] = 0;
+      }
+    }
+    int kk = kp * 2;
+    int chunk0 = kk >> 3, kin = kk & 7;  // position inside a 16B chunk
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int row = m0 + j;
+      unsigned int packed = (unsigned int)va[j] |
+                            ((unsigned int)vb[j] << 16);
+      *reinterpret_cast<unsigned int*>(
+          lds + row * BK + (swz_chunk(row, chunk0) << 3) + kin) = packed;
+    }
+  }
+}
+
+// legacy transposed staging into the canonical layout (scalar scatter;
+// kept for A/B comparison via COS_GEMM_TRANS_PAIR=0).
 __device__ __forceinline__ void stage_trans_guarded(
     bf16* lds_, const bf16* g_, int row0, int rows, int ld,
     int k0, int kend, int tid, int nthreads = NTHREADS,
@@ -144,8 +194,10 @@ __global__ __launch_bounds__(WNT, 1) void gemm_tt_wide_kernel(
   f32x4 acc[4][4] = {};
 
   for (int k0 = k_begin; k0 < k_end; k0 += BK) {
-    stage_trans_guarded(As, A, tile_m, M, lda, k0, k_end, tid, WNT, BM);
-    stage_trans_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid, WNT, WBN);
+    stage_trans_pair_guarded(As, A, tile_m, M, lda, k0, k_end, tid, WNT,
+                             BM);
+    stage_trans_pair_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid, WNT,
+                             WBN);
     __syncthreads();
     bf16x8 afrag[4], bfrag[4];
 #pragma unroll
@@ -295,14 +347,14 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
       if (a_fast) {
         stage_direct_fast(As, A, tile_m, lda, k0, tid);
       } else if (TRANS_A) {
-        stage_trans_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
+        stage_trans_pair_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
       } else {
         stage_direct_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
       }
       if (b_fast) {
         stage_direct_fast(Bs, B, tile_n, ldb, k0, tid);
       } else if (TRANS_B) {
-        stage_trans_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
+        stage_trans_pair_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
       } else {
         stage_direct_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
       }
@@ -356,6 +408,32 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
       }
     }
   }
+}
+
+// ---- tr16 semantics probe: fills LDS with elem index, every lane does
+// one ds_read_b64_tr_b16 at its own address and dumps v[0..3] — verifies
+// the (addr + j*16) element pattern on hardware.
+typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
+
+__global__ void tr16_probe_kernel(float* out, int mode) {
+  __shared__ bf16 lds[2048];
+  for (int i = threadIdx.x; i < 2048; i += blockDim.x)
+    lds[i] = f2bf((float)i);
+  __syncthreads();
+  int lane = threadIdx.x & 63;
+  int base;
+  if (mode == 0) base = lane;                       // low bits sweep
+  else if (mode == 1) base = lane * 16;             // tile sweep
+  else if (mode == 2) base = (lane & 3) + (lane >> 2) * 16;  // quad cols
+  else base = (lane & 15) + (lane >> 4) * 128;      // frag pattern
+  auto* p = (__attribute__((address_space(3))) bf16x4*)(lds + base);
+  bf16x4 v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[lane * 4 + j] = bf2f(v[j]);
+}
+
+void tr16_probe(float* out, int mode, hipStream_t stream) {
+  tr16_probe_kernel<<<1, 64, 0, stream>>>(out, mode);
 }
 
 // ------------------------------------------------------------------- host

@@ -68,6 +68,18 @@ def _splitk_for(mb: int, nb: int, K: int) -> int:
     return max(1, min(K // 256, target // max(1, tiles)))
 
 
+def _splitk_tt(M: int, N: int, K: int) -> int:
+    """Split-K for the fused trans/trans GEMMs: the wide kernel re-tiles
+    N to 256, and the sweep (scratch/sweep_tt_sk.py, MI355X) shows the
+    optimum at ~448 8-wave blocks with >=512 K per block."""
+    mb = (M + 127) // 128
+    nb = (N + 255) // 256 if N > 128 else (N + 127) // 128
+    if K < 1024:
+        return 1
+    target = int(os.environ.get("COS_SPLITK_TARGET", "448"))
+    return max(1, min(target // max(1, mb * nb), K // 512))
+
+
 def _pad128(n: int) -> int:
     return (n + 127) // 128 * 128
 
@@ -251,16 +263,19 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         # torch copy kernels) turns the GEMM into the fast NT direct/direct
         # form with global_load_lds staging instead of LDS scatter staging.
         dwp = torch.zeros((Kout, Kpad), dtype=torch.float32, device=dy.device)
-        use_tt = bool(int(os.environ.get("COS_DW_TT", "0")))
+        # fused trans/trans dw (u32 k-pair staged; round 2): reads dy2/col
+        # exactly once instead of transpose kernels + an extra HBM pass
+        use_tt = bool(int(os.environ.get("COS_DW_TT", "1")))
         dyT = None if use_tt else _transpose(dy2)
         for g in range(G):
             mb, nb = (Kg + 127) // 128, (Kpad + 127) // 128
             sk = _splitk_for(mb, nb, NPQ)
             if use_tt:
                 src = x2 if is_1x1 else col[g]
+                sk_tt = _splitk_tt(Kg, Kpad, NPQ)
                 _gemm(dy2[:, g * Kg:], src, dwp[g * Kg:], None,
                       Kg, Kpad, NPQ, Kout, Kpad if not is_1x1 else C,
-                      Kpad, True, True, 2, sk)
+                      Kpad, True, True, 2, sk_tt)
             else:
                 if is_1x1:
                     # inception: several 1x1 branch convs share one
@@ -380,13 +395,11 @@ def fc_backward(x, w, dy, need_dx=True, bias=True):
         else:
             _gemm(dy, wT, dx, None, M, K, Nout, Nout, Nout, K, False,
                   False, 0, 1, na=_pad128(K))
-    # dw = dy^T @ x: transpose both (small vs the GEMM) -> NT fast form
-    dyT = _transpose(dy)
-    xT = _transpose(x)
+    # dw = dy^T @ x: fused trans/trans (u32 k-pair staging) — no
+    # operand transpose kernels, one read of each operand
     dwp = torch.zeros((Nout, K), dtype=torch.float32, device=x.device)
-    mb, nb = (Nout + 127) // 128, (K + 127) // 128
-    _gemm(dyT, xT, dwp, None, Nout, K, M, M, M, K, False, False, 2,
-          _splitk_for(mb, nb, M), ma=_pad128(Nout), na=_pad128(K))
+    _gemm(dy, x, dwp, None, Nout, K, M, Nout, K, K, True, True, 2,
+          _splitk_tt(Nout, K, M))
     dw = dwp
     if bias:
         db = torch.zeros(Nout, dtype=torch.float32, device=x.device)
