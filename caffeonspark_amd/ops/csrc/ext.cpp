@@ -17,6 +17,13 @@ void gemm_bf16(const void* A, const void* B, void* C, const float* bias,
                bool relu, float alpha, int m_alloc, int n_alloc,
                hipStream_t stream);
 void tr16_probe(float* out, int mode, hipStream_t stream);
+int lstm_persist_fwd(const void* xg, const void* w_hc, const void* cont,
+                     void* h, float* c, float* act, void* h_in, float* hg,
+                     int T, int N, int H, void* bar, hipStream_t stream);
+int lstm_persist_bwd(const void* dy, const void* w_hcT, const void* cont,
+                     const void* h, const float* c, const float* act,
+                     void* dxg, float* dh_acc, float* dc_a, float* dc_b,
+                     int T, int N, int H, void* bar, hipStream_t stream);
 void wino_conv(const void* x, const float* w, const float* bias, void* y,
                void* U, void* V, void* Mbuf, int N, int H, int W, int Cin,
                int K, int wK, int wC, int u_rows_alloc, bool flip,
@@ -257,6 +264,29 @@ void py_bias_act_cast(Tensor in, c10::optional<Tensor> bias, Tensor out,
                         in.size(0), in.size(1), relu, cur_stream());
 }
 
+int64_t py_lstm_persist_fwd(Tensor xg, Tensor w_hc, Tensor cont,
+                            Tensor h, Tensor c, Tensor act, Tensor h_in,
+                            Tensor hg, int64_t T, int64_t N, int64_t H,
+                            Tensor bar) {
+  return cosamd::lstm_persist_fwd(
+      xg.data_ptr(), w_hc.data_ptr(), cont.data_ptr(), h.data_ptr(),
+      c.data_ptr<float>(), act.data_ptr<float>(), h_in.data_ptr(),
+      hg.data_ptr<float>(), (int)T, (int)N, (int)H, bar.data_ptr(),
+      cur_stream());
+}
+
+int64_t py_lstm_persist_bwd(Tensor dy, Tensor w_hcT, Tensor cont,
+                            Tensor h, Tensor c, Tensor act, Tensor dxg,
+                            Tensor dh_acc, Tensor dc_a, Tensor dc_b,
+                            int64_t T, int64_t N, int64_t H, Tensor bar) {
+  return cosamd::lstm_persist_bwd(
+      dy.data_ptr(), w_hcT.data_ptr(), cont.data_ptr(), h.data_ptr(),
+      c.data_ptr<float>(), act.data_ptr<float>(), dxg.data_ptr(),
+      dh_acc.data_ptr<float>(), dc_a.data_ptr<float>(),
+      dc_b.data_ptr<float>(), (int)T, (int)N, (int)H, bar.data_ptr(),
+      cur_stream());
+}
+
 void py_tr16_probe(Tensor out, int64_t mode) {
   cosamd::tr16_probe(out.data_ptr<float>(), (int)mode, cur_stream());
 }
@@ -424,6 +454,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum", &py_colsum);
   m.def("transpose", &py_transpose);
   m.def("tr16_probe", &py_tr16_probe);
+  m.def("lstm_persist_fwd", &py_lstm_persist_fwd);
+  m.def("lstm_persist_bwd", &py_lstm_persist_bwd);
   m.def("bias_act_cast", &py_bias_act_cast);
   m.def("lstm_seq_fwd", &py_lstm_seq_fwd);
   m.def("lstm_seq_bwd", &py_lstm_seq_bwd);

@@ -687,9 +687,64 @@ def lstm_seq_forward(xg, w_hc, cont):
                              device=dev_)
     h_in = h_in_store[:T * N * H].view(T, N, H)
     hg = torch.empty((N, H4), dtype=torch.float32, device=dev_)
+    if _lstm_persist_ok(N):
+        # persistent whole-sequence kernel: recurrent weights stationary
+        # in registers, grid-resident with in-kernel barriers (falls back
+        # to the per-step loop when the watchdog reports non-residency)
+        bar = torch.zeros(260, dtype=torch.int32, device=dev_)
+        rc = _ext.lstm_persist_fwd(xg, whc[:H4].contiguous() if not
+                                   whc[:H4].is_contiguous() else whc[:H4],
+                                   cont_b, h, c, act, h_in, hg, T, N, H,
+                                   bar)
+        if rc == 0:
+            _persist_track(bar)
+            return h, (cont_b, h, c, act, h_in, T, N, H)
     _ext.lstm_seq_fwd(xg, whc[:H4], cont_b, h, c, act, h_in, hg,
                       T, N, H, whc.shape[0])
     return h, (cont_b, h, c, act, h_in, T, N, H)
+
+
+_persist_pending = []
+
+
+def _persist_check_pending(force=False):
+    """Deferred watchdog check: err flags are copied D2H asynchronously;
+    inspect completed copies (or all, when force=True) and raise if the
+    persistent kernel ever aborted — outputs are also NaN-poisoned, so a
+    missed check cannot silently corrupt training."""
+    global _persist_pending
+    keep = []
+    for ev, flag in _persist_pending:
+        if force or ev.query():
+            if force:
+                ev.synchronize()
+            if int(flag[0]):
+                raise RuntimeError(
+                    "persistent LSTM grid barrier watchdog fired (blocks "
+                    "not co-resident?); set COS_LSTM_PERSIST=0")
+        else:
+            keep.append((ev, flag))
+    _persist_pending = keep
+
+
+def _persist_track(bar):
+    flag = torch.empty(1, dtype=torch.int32, pin_memory=True)
+    flag.copy_(bar[258:259], non_blocking=True)
+    ev = torch.cuda.Event()
+    ev.record()
+    _persist_pending.append((ev, flag))
+
+
+def _lstm_persist_ok(N):
+    _persist_check_pending(force=len(_persist_pending) > 16)
+    if N > 64 or int(os.environ.get("COS_LSTM_PERSIST", "1")) == 0:
+        return False
+    if torch.cuda.is_current_stream_capturing():
+        return False        # host sync inside is capture-illegal
+    import torch.distributed as dist
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        return False        # RCCL kernels may break grid residency
+    return True
 
 
 def lstm_seq_backward(dy, w_hc, cache):
@@ -706,18 +761,26 @@ def lstm_seq_backward(dy, w_hc, cache):
     dh_f = torch.empty((N, H), dtype=torch.float32, device=dev_)
     dc_a = torch.empty((N, H), dtype=torch.float32, device=dev_)
     dc_b = torch.empty((N, H), dtype=torch.float32, device=dev_)
-    _ext.lstm_seq_bwd(dy, whcT, cont_b, h, c, act, dxg, dh_rec, dh_f,
-                      dc_a, dc_b, T, N, H, _pad128(H4))
-    # dw_hc = dgates_all^T @ h_in_all — one big NT GEMM
+    done = False
+    if _lstm_persist_ok(N):
+        bar = torch.zeros(260, dtype=torch.int32, device=dev_)
+        rc = _ext.lstm_persist_bwd(dy, whcT[:H].contiguous() if not
+                                   whcT[:H].is_contiguous() else whcT[:H],
+                                   cont_b, h, c, act, dxg, dh_f, dc_a,
+                                   dc_b, T, N, H, bar)
+        done = (rc == 0)
+        if done:
+            _persist_track(bar)
+    if not done:
+        _ext.lstm_seq_bwd(dy, whcT, cont_b, h, c, act, dxg, dh_rec, dh_f,
+                          dc_a, dc_b, T, N, H, _pad128(H4))
+    # dw_hc = dgates_all^T @ h_in_all — fused trans/trans GEMM (u32
+    # k-pair staging), no operand transposes
     dxg_flat = dxg.reshape(T * N, H4)
     h_in_flat = h_in.reshape(T * N, H)
-    dgT = _transpose(dxg_flat)
-    hiT = _transpose(h_in_flat)
     dwhc = torch.zeros((H4, H), dtype=torch.float32, device=dev_)
-    mb, nb = (H4 + 127) // 128, (H + 127) // 128
-    _gemm(dgT, hiT, dwhc, None, H4, H, T * N, T * N, T * N, H,
-          False, False, 2, _splitk_for(mb, nb, T * N),
-          ma=_pad128(H4), na=_pad128(H))
+    _gemm(dxg_flat, h_in_flat, dwhc, None, H4, H, T * N, H4, H, H,
+          True, True, 2, _splitk_tt(H4, H, T * N))
     return dxg, dwhc
 
 
