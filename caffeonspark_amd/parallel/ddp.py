@@ -113,14 +113,33 @@ class DistributedSync(Callback):
         self._works = []
         self._fired = [False] * len(self.buckets)
 
+    @property
+    def _bf16_wire(self) -> bool:
+        """bf16 gradient exchange over RCCL/xGMI: halves the wire bytes
+        of the ~244 MB AlexNet gradient all-reduce (ring time is per-link
+        bound at ~153 GB/s).  Default on for the nccl(=RCCL) backend —
+        the convergence gates (LeNet accuracy>0.8 on GPU e2e) validate
+        it; gloo/CPU stays fp32 so the 2-rank bit-identity tests hold.
+        Override with COS_DDP_BF16=0/1."""
+        env = os.environ.get("COS_DDP_BF16")
+        if env is not None:
+            return env == "1"
+        return dist.is_initialized() and dist.get_backend() == "nccl"
+
     def _fire(self, idx: int) -> None:
         if self._fired[idx] or self.world_size <= 1:
             return
         self._fired[idx] = True
         lo, hi = self.buckets[idx]
         buf = self.solver.flat_g.narrow(0, lo, hi - lo)
-        self._works.append(dist.all_reduce(buf, op=dist.ReduceOp.SUM,
-                                           async_op=True))
+        if self._bf16_wire:
+            b16 = buf.to(torch.bfloat16)
+            work = dist.all_reduce(b16, op=dist.ReduceOp.SUM, async_op=True)
+            self._works.append((work, buf, b16))
+        else:
+            self._works.append(
+                (dist.all_reduce(buf, op=dist.ReduceOp.SUM, async_op=True),
+                 None, None))
 
     # ---- Solver callback hooks -------------------------------------------
     def on_start(self) -> None:
@@ -137,8 +156,10 @@ class DistributedSync(Callback):
     def on_gradients_ready(self) -> None:
         for i in range(len(self.buckets)):
             self._fire(i)
-        for w in self._works:
+        for w, buf, b16 in self._works:
             w.wait()
+            if buf is not None:
+                buf.copy_(b16)      # cast the reduced bf16 sum back
         self._reset()
 
     def broadcast_params(self) -> None:
