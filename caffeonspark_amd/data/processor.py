@@ -23,6 +23,34 @@ from .source import STOP_MARK, DataSource
 _instance_lock = threading.Lock()
 _instance: Optional["CaffeProcessor"] = None
 
+# ---- InputAdapter registry (reference InputAdapter.hpp:70-76
+# REGISTER_INPUT_ADAPTER: layer-type string -> feed function, extensible
+# by user layer types without touching the processor) ------------------
+INPUT_ADAPTERS: dict = {}
+
+
+def register_input_adapter(layer_type: str):
+    def deco(fn):
+        INPUT_ADAPTERS[layer_type] = fn
+        return fn
+    return deco
+
+
+def get_input_adapter(layer_type: str):
+    return INPUT_ADAPTERS.get(layer_type)
+
+
+@register_input_adapter("MemoryData")
+def _feed_memory_data(dl, batch):
+    # reference MemoryInputAdapter.cpp:24-33 (MemoryDataLayer::Reset)
+    dl.reset(batch[0], batch[1] if len(batch) > 1 else None)
+
+
+@register_input_adapter("CoSData")
+def _feed_cos_data(dl, batch):
+    # reference CoSInputAdapter.cpp:23-34 (CoSDataLayer::Reset)
+    dl.reset(batch)
+
 
 class QueuePair:
     """Bounded Full queue of prepared batches (cap 2: double buffering)."""
@@ -161,11 +189,9 @@ class CaffeProcessor:
 
     @staticmethod
     def _reset_layer(dl, batch) -> None:
-        from ..core.layers.data import CoSDataLayer, MemoryDataLayer
-        if isinstance(dl, MemoryDataLayer):
-            dl.reset(batch[0], batch[1] if len(batch) > 1 else None)
-        elif isinstance(dl, CoSDataLayer):
-            dl.reset(batch)
+        adapter = get_input_adapter(type(dl).TYPE)
+        if adapter is not None:
+            adapter(dl, batch)
 
     def _run_validation(self) -> None:
         """Interleaved validation (CaffeProcessor.scala:429-440): consume

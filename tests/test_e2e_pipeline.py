@@ -464,3 +464,20 @@ def test_features_multi_blob(workdir):
     assert {"SampleID", "label", "ip1", "ip2"} <= set(df.columns)
     assert len(df["ip1"][0]) == 100 and len(df["ip2"][0]) == 10
     CaffeProcessor.reset_instance()
+
+
+def test_notebooks_valid():
+    """Reference ships example notebooks; ours must be valid nbformat-4
+    JSON whose code cells at least compile."""
+    import json
+
+    nbdir = os.path.join(ROOT, "examples", "notebooks")
+    nbs = [f for f in os.listdir(nbdir) if f.endswith(".ipynb")]
+    assert len(nbs) >= 2
+    for name in nbs:
+        with open(os.path.join(nbdir, name)) as f:
+            nb = json.load(f)
+        assert nb["nbformat"] == 4
+        for cell in nb["cells"]:
+            if cell["cell_type"] == "code":
+                compile("".join(cell["source"]), name, "exec")
