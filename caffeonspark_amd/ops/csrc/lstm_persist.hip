@@ -1,19 +1,21 @@
-// Persistent whole-sequence LSTM for gfx950: ONE kernel runs all T
-// timesteps with the recurrent weights STATIONARY in registers.
+// Persistent whole-sequence LSTM for gfx950, v2: hidden-slice-owned.
 //
-// The per-timestep path (lstm_seq.hip) launches 3 kernels + a memset per
-// step and re-reads the 4H x H hidden weights from HBM every step: at
-// LRCN's shape ([N=64,4H=4000] @= [64,H=1000] per t) each recurrent GEMM
-// is launch/ramp-bound (~27 us for 9 MB of traffic).  Here each block
-// owns a [64-col x 256-k] weight slice, loaded ONCE into VGPR MFMA
-// fragments; per timestep the blocks compute their partial gate
-// pre-activations (fp32 atomics into hg), grid-sync, apply the fused
-// LSTM unit math (which also zeroes hg and gates h into h_in for t+1),
-// and grid-sync again.  The grid is sized <= the CU count so every block
-// is resident (plain launch, device otherwise idle — the Python glue
-// falls back to the loop path under multi-GPU overlap); the barrier is
-// sense-reversing with a watchdog that aborts via an error flag instead
-// of hanging the device.
+// ONE kernel runs all T timesteps.  Each block owns a 16-wide slice of
+// the hidden dimension and computes, for its slice, ALL FOUR gate
+// columns over the FULL K (the four 16-row weight strips i/f/o/g stay
+// stationary in VGPR MFMA fragments, ~128 VGPRs/lane), so
+//   * gate sums never cross blocks: no fp32 atomics, no gate buffer --
+//     cross-wave K-partials reduce through LDS inside the block;
+//   * the LSTM unit math for the slice is block-local, and the cell
+//     state c (forward) / dc and dh (backward) live in LDS across the
+//     whole sequence;
+//   * exactly ONE grid barrier per timestep (between writing h_in_{t+1}
+//     / dxg_t and the other blocks' reads of them) -- v1's
+//     4H-column-partitioned design needed two plus 16 MB/t of atomics.
+// The barrier is store-slot arrival polled by block 0's threads, with a
+// watchdog that sets an error flag and NaN-poisons the output instead
+// of hanging if blocks are not co-resident (the Python glue checks the
+// flag asynchronously and gates this path to single-process runs).
 //
 // Reference scope: recurrent LSTM path of lrcn_cos.prototxt (SURVEY.md
 // §5 long-context: time-major, cont gating).
@@ -36,15 +38,11 @@ __device__ __forceinline__ void stbf(u16* p, float v) {
   *p = *reinterpret_cast<u16*>(&b);
 }
 
-// Grid barrier with store-slot arrival: every block plain-stores its
-// running barrier index into its own slot (no same-address RMW chain —
-// a flat atomic arrival serialized at ~25 us for 252 blocks, a 16-group
-// tree still ~8 us); block 0's 256 threads poll all slots in parallel
-// and release the generation word.  Watchdog aborts via err instead of
-// hanging the device.
-// bar layout (unsigned words): [0..255] arrival slots, [256] generation,
-// [258] error flag.
-
+// Grid barrier with store-slot arrival (each block plain-stores its
+// running barrier index into its own slot; block 0's threads poll all
+// slots in parallel and release the generation word).  Flat same-address
+// atomic arrival measured ~25 us at 252 blocks; this lands ~2-4 us.
+// bar layout (unsigned words): [0..255] slots, [256] gen, [258] err.
 __device__ __forceinline__ bool grid_sync(unsigned* bar, unsigned nblk,
                                           int* err, unsigned my_count) {
   unsigned* arr = bar;
@@ -74,7 +72,6 @@ __device__ __forceinline__ bool grid_sync(unsigned* bar, unsigned nblk,
       __builtin_amdgcn_s_sleep(2);
     }
     if (slot == 0) {
-      // acquire all blocks' pre-barrier writes, then release the gen
       __hip_atomic_load(arr, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
       __hip_atomic_store(gen, my_count, __ATOMIC_RELEASE,
                          __HIP_MEMORY_SCOPE_AGENT);
@@ -106,258 +103,219 @@ __device__ __forceinline__ bool grid_sync(unsigned* bar, unsigned nblk,
                            __HIP_MEMORY_SCOPE_AGENT) == 0;
 }
 
-// stationary weight fragments: wave covers cols [nbase, nbase+64) and
-// k [kbase, kbase+64) of W[rows=ncols_total][K=kdim]; guarded, zero-fill.
-struct WFrags {
-  bf16x8v b[4][2];
-};
-
-__device__ __forceinline__ void load_wfrags(
-    WFrags& wf, const u16* W, int64_t ldw, int nbase, int ncols,
-    int kbase, int kdim, int lane) {
-  int lrow = lane & 15;
-  int lk8 = (lane >> 4) * 8;
-#pragma unroll
-  for (int nf = 0; nf < 4; ++nf) {
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      int n = nbase + nf * 16 + lrow;
-      int k = kbase + ks * 32 + lk8;
-      if (n < ncols && k + 8 <= kdim) {
-        wf.b[nf][ks] = *reinterpret_cast<const bf16x8v*>(
-            W + (int64_t)n * ldw + k);
-      } else {
-        bf16 z = f2bf(0.f);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) wf.b[nf][ks][j] = z;
-      }
-    }
-  }
-}
-
-// A fragments from the [rows<=64][kdim] activation slab (row stride lda);
-// rows beyond `rows` read junk inside the over-allocated slab (outputs
-// for those rows are never consumed); k guarded to kdim.
-__device__ __forceinline__ void load_afrags(
-    bf16x8v (&a)[4][2], const u16* A, int64_t lda, int kbase, int kdim,
-    int lane) {
-  int lrow = lane & 15;
-  int lk8 = (lane >> 4) * 8;
-#pragma unroll
-  for (int mf = 0; mf < 4; ++mf) {
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      int m = mf * 16 + lrow;
-      int k = kbase + ks * 32 + lk8;
-      if (k + 8 <= kdim) {
-        a[mf][ks] = *reinterpret_cast<const bf16x8v*>(
-            A + (int64_t)m * lda + k);
-      } else {
-        bf16 z = f2bf(0.f);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) a[mf][ks][j] = z;
-      }
-    }
-  }
-}
-
-__device__ __forceinline__ void wave_reduce(f32x4v (&acc)[4][4], int wave,
-                                            int lane) {
-  __shared__ float red_buf[2][64 * 64];
-  // stage 1: waves 1,3 publish; waves 0,2 accumulate
-  if (wave & 1) {
-    float* dst = red_buf[wave >> 1] + lane * 64;
-#pragma unroll
-    for (int mf = 0; mf < 4; ++mf)
-#pragma unroll
-      for (int nf = 0; nf < 4; ++nf)
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          dst[(mf * 4 + nf) * 4 + r] = acc[mf][nf][r];
-  }
-  __syncthreads();
-  if (!(wave & 1)) {
-    const float* src = red_buf[wave >> 1] + lane * 64;
-#pragma unroll
-    for (int mf = 0; mf < 4; ++mf)
-#pragma unroll
-      for (int nf = 0; nf < 4; ++nf)
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          acc[mf][nf][r] += src[(mf * 4 + nf) * 4 + r];
-  }
-  __syncthreads();
-  // stage 2: wave 2 publishes; wave 0 accumulates
-  if (wave == 2) {
-    float* dst = red_buf[0] + lane * 64;
-#pragma unroll
-    for (int mf = 0; mf < 4; ++mf)
-#pragma unroll
-      for (int nf = 0; nf < 4; ++nf)
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          dst[(mf * 4 + nf) * 4 + r] = acc[mf][nf][r];
-  }
-  __syncthreads();
-  if (wave == 0) {
-    const float* src = red_buf[0] + lane * 64;
-#pragma unroll
-    for (int mf = 0; mf < 4; ++mf)
-#pragma unroll
-      for (int nf = 0; nf < 4; ++nf)
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          acc[mf][nf][r] += src[(mf * 4 + nf) * 4 + r];
-  }
-}
-
 }  // namespace
 
 // ------------------------------------------------------------- forward
-// grid = NCH x KCH blocks (NCH = ceil(4H/64), KCH = ceil(H/256));
-// per wave: stationary W slice + per-t A loads -> acc -> fp32 atomics.
+// grid = ceil(H/16) blocks, 256 threads (4 waves k-partitioned).
+// Per wave: 4 weight strips (i/f/o/g rows of its k-window) stationary;
+// per t: full-K gate MFMA -> cross-wave LDS reduce -> block-local unit.
 __global__ __launch_bounds__(256, 1) void lstm_persist_fwd_kernel(
     const u16* __restrict__ xg, const u16* __restrict__ w_hc,
     const u16* __restrict__ cont, u16* __restrict__ h,
     float* __restrict__ c, float* __restrict__ act,
-    u16* __restrict__ h_in, float* __restrict__ hg,
-    int T, int N, int H, unsigned* bar, int* err,
-    unsigned long long* prof) {
+    u16* __restrict__ h_in, int T, int N, int H,
+    unsigned* bar, int* err) {
+  __shared__ float red0[64 * 16 * 4], red1[64 * 16 * 4];  // wave partials
+  __shared__ float gates[64 * 64];                // [n][strip*16 + jj]
+  __shared__ float c_lds[64 * 16];                // cell state, persists
   int H4 = 4 * H;
-  int KCH = (H + 255) >> 8;
   int nblk = gridDim.x;
-  int kch = blockIdx.x % KCH;
-  int nch = blockIdx.x / KCH;
+  int j0 = blockIdx.x * 16;
   int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  int nbase = nch * 64;
-  int kbase = kch * 256 + wave * 64;
+  int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+  int Hp4 = ((H + 127) & ~127) >> 2;      // per-wave K window (mult 32)
+  int ksteps = Hp4 >> 5;
+  int kw0 = wave * Hp4;
 
-  WFrags wf;
-  load_wfrags(wf, w_hc, H, nbase, H4, kbase, H, lane);
+  // stationary weights: strip s covers gate-rows s*H + [j0, j0+16)
+  bf16x8v wf[4][8];
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    for (int ks = 0; ks < 8; ++ks) {
+      int row = s * H + j0 + lrow;
+      int k = kw0 + ks * 32 + lk8;
+      if (ks < ksteps && (j0 + lrow) < H && k + 8 <= H) {
+        wf[s][ks] = *reinterpret_cast<const bf16x8v*>(
+            w_hc + (int64_t)row * H + k);
+      } else {
+        bf16 z = f2bf(0.f);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) wf[s][ks][j] = z;
+      }
+    }
+  }
+  for (int i = threadIdx.x; i < 64 * 16; i += blockDim.x)
+    c_lds[i] = 0.f;
+  __syncthreads();
 
   int64_t nh = (int64_t)N * H;
-  int lrow = lane & 15;
-
-  unsigned long long t0, t1, t2, t3;
   unsigned sync_idx = 0;
   for (int t = 0; t < T; ++t) {
-    t0 = clock64();
-    // ---- gate GEMM phase: hg[0..63][nbase..nbase+64) += h_in_t @ W^T
+    // ---- gate GEMM over the full K for this block's 64 gate columns
     const u16* A = h_in + (int64_t)t * nh;
-    bf16x8v a[4][2];
-    load_afrags(a, A, H, kbase, H, lane);
-    f32x4v acc[4][4] = {};
+    f32x4v acc[4][4] = {};    // [mf][strip]
+    for (int ks = 0; ks < ksteps; ++ks) {
+      int k = kw0 + ks * 32 + lk8;
+      bf16x8v a[4];
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks)
+      for (int mf = 0; mf < 4; ++mf) {
+        int m = mf * 16 + lrow;
+        if (k + 8 <= H) {
+          a[mf] = *reinterpret_cast<const bf16x8v*>(
+              A + (int64_t)m * H + k);
+        } else {
+          bf16 z = f2bf(0.f);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) a[mf][j] = z;
+        }
+      }
 #pragma unroll
       for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
-        for (int nf = 0; nf < 4; ++nf)
-          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a[mf][ks], wf.b[nf][ks], acc[mf][nf], 0, 0, 0);
-    // cross-wave LDS reduction (the 4 waves hold k-partials of the SAME
-    // [64x64] tile): two pairwise stages leave wave 0 with the sum, so
-    // only 1/4 of the fp32 atomics (and their pre-barrier drain) remain
-    wave_reduce(acc, wave, lane);
-    if (wave == 0) {
-      int crow0 = (lane >> 4) << 2;
-      int ccol0 = nbase + lrow;
-#pragma unroll
-      for (int mf = 0; mf < 4; ++mf) {
-#pragma unroll
-        for (int nf = 0; nf < 4; ++nf) {
-          int col = ccol0 + nf * 16;
-          if (col >= H4) continue;
-#pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            int row = crow0 + mf * 16 + r;
-            if (row >= N) continue;
-            atomicAdd(hg + (int64_t)row * H4 + col, acc[mf][nf][r]);
-          }
-        }
-      }
+        for (int s = 0; s < 4; ++s)
+          acc[mf][s] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mf], wf[s][ks < 8 ? ks : 0], acc[mf][s], 0, 0, 0);
     }
-    t1 = clock64();
-    if (!grid_sync(bar, nblk, err, ++sync_idx)) return;
-    t2 = clock64();
+    // ---- cross-wave reduce into gates[] (pairwise through LDS)
+    {
+      float* mybuf = (wave >> 1) ? red1 : red0;
+      if (wave & 1) {
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+          for (int s = 0; s < 4; ++s)
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              mybuf[(lane * 16 + mf * 4 + s) * 4 + r] = acc[mf][s][r];
+      }
+      __syncthreads();
+      if (!(wave & 1)) {
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+          for (int s = 0; s < 4; ++s)
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              acc[mf][s][r] += mybuf[(lane * 16 + mf * 4 + s) * 4 + r];
+      }
+      __syncthreads();
+      if (wave == 2) {
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+          for (int s = 0; s < 4; ++s)
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              red0[(lane * 16 + mf * 4 + s) * 4 + r] = acc[mf][s][r];
+      }
+      __syncthreads();
+      if (wave == 0) {
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+          for (int s = 0; s < 4; ++s) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              int row = (lane >> 4) * 4 + mf * 16 + r;
+              float v = acc[mf][s][r] +
+                        red0[(lane * 16 + mf * 4 + s) * 4 + r];
+              gates[row * 64 + s * 16 + lrow] = v;
+            }
+          }
+      }
+      __syncthreads();
+    }
 
-    // ---- fused unit phase over N*H elements (all blocks cooperate)
+    // ---- block-local fused unit over (n < N) x (16 slice cols)
     const u16* cont_t = cont + (int64_t)t * N;
     const u16* xg_t = xg + (int64_t)t * N * H4;
-    const float* c_prev = (t > 0) ? c + (int64_t)(t - 1) * nh : nullptr;
     float* c_t = c + (int64_t)t * nh;
     u16* h_t = h + (int64_t)t * nh;
     float* act_t = act + (int64_t)t * N * H4;
     u16* h_in_next = (t + 1 < T) ? h_in + (int64_t)(t + 1) * nh : nullptr;
     const u16* cont_next = (t + 1 < T) ? cont + (int64_t)(t + 1) * N
                                        : nullptr;
-    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < nh; i += (int64_t)nblk * blockDim.x) {
-      int64_t row = i / H;
-      int hh = (int)(i % H);
-      int64_t g0 = row * H4 + hh;
-      float* hgp = hg + g0;
-      float gi = 1.f / (1.f + __expf(-(ldbf(xg_t + g0) + hgp[0])));
-      float gf = 1.f / (1.f + __expf(-(ldbf(xg_t + g0 + H) + hgp[H])));
+    for (int i = threadIdx.x; i < N * 16; i += blockDim.x) {
+      int n = i >> 4;
+      int jj = i & 15;
+      int jg = j0 + jj;
+      if (jg >= H) continue;
+      int64_t g0 = (int64_t)n * H4 + jg;
+      float gi = 1.f / (1.f + __expf(-(ldbf(xg_t + g0) +
+                                       gates[n * 64 + jj])));
+      float gf = 1.f / (1.f + __expf(-(ldbf(xg_t + g0 + H) +
+                                       gates[n * 64 + 16 + jj])));
       float go = 1.f / (1.f + __expf(-(ldbf(xg_t + g0 + 2 * H) +
-                                       hgp[2 * H])));
-      float gg = tanhf(ldbf(xg_t + g0 + 3 * H) + hgp[3 * H]);
-      hgp[0] = 0.f; hgp[H] = 0.f; hgp[2 * H] = 0.f; hgp[3 * H] = 0.f;
-      float ct = ldbf(cont_t + row);
-      float cp = (c_prev != nullptr) ? c_prev[i] : 0.f;
+                                       gates[n * 64 + 32 + jj])));
+      float gg = tanhf(ldbf(xg_t + g0 + 3 * H) +
+                       gates[n * 64 + 48 + jj]);
+      float ct = ldbf(cont_t + n);
+      float cp = c_lds[i];
       float cv = gf * cp * ct + gi * gg;
-      c_t[i] = cv;
+      c_lds[i] = cv;
+      int64_t hi = (int64_t)n * H + jg;
+      c_t[hi] = cv;
       float hv = go * tanhf(cv);
-      stbf(h_t + i, hv);
+      stbf(h_t + hi, hv);
       float* ap = act_t + g0;
       ap[0] = gi; ap[H] = gf; ap[2 * H] = go; ap[3 * H] = gg;
       if (h_in_next != nullptr)
-        stbf(h_in_next + i, hv * ldbf(cont_next + row));
+        stbf(h_in_next + hi, hv * ldbf(cont_next + n));
     }
-    t3 = clock64();
-    if (!grid_sync(bar, nblk, err, ++sync_idx)) return;
-    if (prof != nullptr && threadIdx.x == 0 && blockIdx.x == 0) {
-      unsigned long long t4 = clock64();
-      atomicAdd(prof + 0, t1 - t0);   // gemm
-      atomicAdd(prof + 1, t2 - t1);   // barrier 1
-      atomicAdd(prof + 2, t3 - t2);   // unit
-      atomicAdd(prof + 3, t4 - t3);   // barrier 2
+    if (!grid_sync(bar, nblk, err, ++sync_idx)) {
+      if (threadIdx.x == 0) h[(int64_t)t * nh] = 0x7FC0;  // bf16 NaN
+      return;
     }
   }
 }
 
 // ------------------------------------------------------------ backward
-// weights = w_hcT [H][4H]; NCH = ceil(H/64), KCH = ceil(4H/256).
-// Reverse-time loop: unit phase computes dgates_t (consuming the dh
-// accumulator produced by the previous iteration's GEMM phase and
-// zeroing it), then the GEMM phase accumulates dh for t-1.
+// grid = ceil(H/16); block owns dh slice [j0, j0+16): unit math (dgates)
+// is slice-local (dh/dc persist in LDS), then ONE barrier publishes
+// dxg_t before every block's dh GEMM over the full 4H.
 __global__ __launch_bounds__(256, 1) void lstm_persist_bwd_kernel(
     const u16* __restrict__ dy, const u16* __restrict__ w_hcT,
     const u16* __restrict__ cont, const u16* __restrict__ h,
     const float* __restrict__ c, const float* __restrict__ act,
-    u16* __restrict__ dxg, float* __restrict__ dh_acc,
-    float* __restrict__ dc_a, float* __restrict__ dc_b,
-    int T, int N, int H, unsigned* bar, int* err) {
+    u16* __restrict__ dxg, int T, int N, int H,
+    unsigned* bar, int* err) {
+  __shared__ float red0[64 * 16], red1[64 * 16];
+  __shared__ float dh_lds[64 * 16];
+  __shared__ float dc_lds[64 * 16];
   int H4 = 4 * H;
-  int KCH = (H4 + 255) >> 8;
   int nblk = gridDim.x;
-  int kch = blockIdx.x % KCH;
-  int nch = blockIdx.x / KCH;
+  int j0 = blockIdx.x * 16;
   int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  int nbase = nch * 64;
-  int kbase = kch * 256 + wave * 64;
+  int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+  int Hp4 = ((H4 + 127) & ~127) >> 2;     // per-wave K window over 4H
+  int ksteps = Hp4 >> 5;
+  int kw0 = wave * Hp4;
 
-  WFrags wf;
-  load_wfrags(wf, w_hcT, H4, nbase, H, kbase, H4, lane);
+  // stationary weights: w_hcT[H][4H], rows = this block's dh columns
+  // (16 rows), k over 4H -- up to 32 k-steps/wave (128 VGPRs)
+  bf16x8v wf[32];
+  for (int ks = 0; ks < 32; ++ks) {
+    int row = j0 + lrow;
+    int k = kw0 + ks * 32 + lk8;
+    if (ks < ksteps && row < H && k + 8 <= H4) {
+      wf[ks] = *reinterpret_cast<const bf16x8v*>(
+          w_hcT + (int64_t)row * H4 + k);
+    } else {
+      bf16 z = f2bf(0.f);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) wf[ks][j] = z;
+    }
+  }
+  for (int i = threadIdx.x; i < 64 * 16; i += blockDim.x) {
+    dh_lds[i] = 0.f;
+    dc_lds[i] = 0.f;
+  }
+  __syncthreads();
 
   int64_t nh = (int64_t)N * H;
-  int lrow = lane & 15;
-
   unsigned sync_idx = 0;
   for (int t = T - 1; t >= 0; --t) {
-    float* dc_next = ((T - 1 - t) & 1) ? dc_b : dc_a;
-    float* dc_prev = ((T - 1 - t) & 1) ? dc_a : dc_b;
-    // ---- unit phase: dgates_t from dy_t (+ dh_acc from t+1)
+    // ---- slice-local unit: dgates_t from dy_t + dh_lds/dc_lds
     const u16* cont_t = cont + (int64_t)t * N;
     const u16* cont_t1 = (t + 1 < T) ? cont + (int64_t)(t + 1) * N
                                      : nullptr;
@@ -366,30 +324,27 @@ __global__ __launch_bounds__(256, 1) void lstm_persist_bwd_kernel(
     const float* act_t = act + (int64_t)t * N * H4;
     const u16* dy_t = dy + (int64_t)t * nh;
     u16* dxg_t = dxg + (int64_t)t * N * H4;
-    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < nh; i += (int64_t)nblk * blockDim.x) {
-      int64_t row = i / H;
-      int hh = (int)(i % H);
-      const float* a = act_t + row * H4 + hh;
+    for (int i = threadIdx.x; i < N * 16; i += blockDim.x) {
+      int n = i >> 4;
+      int jj = i & 15;
+      int jg = j0 + jj;
+      if (jg >= H) continue;
+      int64_t g0 = (int64_t)n * H4 + jg;
+      int64_t hi = (int64_t)n * H + jg;
+      const float* a = act_t + g0;
       float gi = a[0], gf = a[H], go = a[2 * H], gg = a[3 * H];
-      float tc = tanhf(c_t[i]);
-      float ct = ldbf(cont_t + row);
-      // dh from t+1's recurrent GEMM, gated by cont_{t+1}; zero after use
-      float dhrec = 0.f;
-      if (t + 1 < T) {
-        dhrec = dh_acc[i] * ldbf(cont_t1 + row);
-        dh_acc[i] = 0.f;
-      }
-      float dhv = ldbf(dy_t + i) + dhrec;
-      float dcn = (t + 1 < T) ? dc_next[i] : 0.f;
-      float dc = dcn + dhv * go * (1.f - tc * tc);
+      float tc = tanhf(c_t[hi]);
+      float ct = ldbf(cont_t + n);
+      float dhrec = (t + 1 < T) ? dh_lds[i] * ldbf(cont_t1 + n) : 0.f;
+      float dhv = ldbf(dy_t + hi) + dhrec;
+      float dc = dc_lds[i] + dhv * go * (1.f - tc * tc);
       float dov = dhv * tc;
       float div = dc * gg;
       float dgv = dc * gi;
-      float cp = (c_prev != nullptr) ? c_prev[i] : 0.f;
+      float cp = (c_prev != nullptr) ? c_prev[hi] : 0.f;
       float dfv = dc * cp * ct;
-      dc_prev[i] = dc * gf * ct;
-      u16* dg = dxg_t + row * H4 + hh;
+      dc_lds[i] = dc * gf * ct;    // dc for t-1 (slice-local)
+      u16* dg = dxg_t + g0;
       stbf(dg, div * gi * (1.f - gi));
       stbf(dg + H, dfv * gf * (1.f - gf));
       stbf(dg + 2 * H, dov * go * (1.f - go));
@@ -400,96 +355,106 @@ __global__ __launch_bounds__(256, 1) void lstm_persist_bwd_kernel(
       return;
     }
 
-    // ---- GEMM phase: dh_acc[0..N)[nbase..] += dxg_t @ w_hcT^T
+    // ---- dh GEMM for t-1: dh_slice = dxg_t @ w_hcT^T (full 4H K)
     if (t > 0) {
-      bf16x8v a[4][2];
-      load_afrags(a, dxg_t, H4, kbase, H4, lane);
-      f32x4v acc[4][4] = {};
+      f32x4v acc[4] = {};     // [mf] x 16 cols
+      for (int ks = 0; ks < ksteps; ++ks) {
+        int k = kw0 + ks * 32 + lk8;
+        bf16x8v a[4];
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
+        for (int mf = 0; mf < 4; ++mf) {
+          int m = mf * 16 + lrow;
+          if (k + 8 <= H4) {
+            a[mf] = *reinterpret_cast<const bf16x8v*>(
+                dxg_t + (int64_t)m * H4 + k);
+          } else {
+            bf16 z = f2bf(0.f);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) a[mf][j] = z;
+          }
+        }
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+          acc[mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mf], wf[ks < 32 ? ks : 0], acc[mf], 0, 0, 0);
+      }
+      // cross-wave reduce -> dh_lds
+      float* mybuf = (wave >> 1) ? red1 : red0;
+      if (wave & 1) {
 #pragma unroll
         for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
-          for (int nf = 0; nf < 4; ++nf)
-            acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a[mf][ks], wf.b[nf][ks], acc[mf][nf], 0, 0, 0);
-      wave_reduce(acc, wave, lane);
+          for (int r = 0; r < 4; ++r)
+            mybuf[(lane * 4 + mf) * 4 + r] = acc[mf][r];
+      }
+      __syncthreads();
+      if (!(wave & 1)) {
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            acc[mf][r] += mybuf[(lane * 4 + mf) * 4 + r];
+      }
+      __syncthreads();
+      if (wave == 2) {
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            red0[(lane * 4 + mf) * 4 + r] = acc[mf][r];
+      }
+      __syncthreads();
       if (wave == 0) {
-        int crow0 = (lane >> 4) << 2;
-        int ccol0 = nbase + lrow;
 #pragma unroll
         for (int mf = 0; mf < 4; ++mf) {
 #pragma unroll
-          for (int nf = 0; nf < 4; ++nf) {
-            int col = ccol0 + nf * 16;
-            if (col >= H) continue;
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              int row = crow0 + mf * 16 + r;
-              if (row >= N) continue;
-              atomicAdd(dh_acc + (int64_t)row * H + col, acc[mf][nf][r]);
-            }
+          for (int r = 0; r < 4; ++r) {
+            int row = (lane >> 4) * 4 + mf * 16 + r;
+            dh_lds[row * 16 + lrow] =
+                acc[mf][r] + red0[(lane * 4 + mf) * 4 + r];
           }
         }
       }
+      __syncthreads();
     }
-    if (!grid_sync(bar, nblk, err, ++sync_idx)) return;
   }
 }
 
 // --------------------------------------------------------------- hosts
+// hg / dh / dc scratch buffers from the v1 API are unused (state lives
+// in LDS); kept in the signatures so the bindings stay stable.
+// returns 0 ok, 1 = not applicable (shape); watchdog aborts are read
+// asynchronously from bar[258] by the Python glue.
 
-// returns 0 ok, 1 = not applicable (shape), 2 = watchdog abort (caller
-// must fall back and recompute)
 int lstm_persist_fwd(const void* xg, const void* w_hc, const void* cont,
                      void* h, float* c, float* act, void* h_in, float* hg,
-                     int T, int N, int H, void* bar /*3 ints zeroed*/,
-                     hipStream_t stream) {
-  if (N > 64 || H % 8 != 0) return 1;
-  int NCH = (4 * H + 63) / 64;
-  int KCH = (H + 255) / 256;
-  int nblk = NCH * KCH;
-  if (nblk > 256) return 1;
+                     int T, int N, int H, void* bar, hipStream_t stream) {
+  (void)hg;
+  if (N > 64 || H % 8 != 0 || H > 1024) return 1;  // wf[4][8] k-window
+  int nblk = (H + 15) / 16;
+  if (nblk < 2 || nblk > 256) return 1;
   int64_t nh = (int64_t)N * H;
   COS_CHECK_HIP(hipMemsetAsync(h_in, 0, nh * 2, stream));
-  COS_CHECK_HIP(hipMemsetAsync(hg, 0, (int64_t)N * 4 * H * 4, stream));
   unsigned* cnt = reinterpret_cast<unsigned*>(bar);
-  const char* pf = getenv("COS_LSTM_PROF");
-  static unsigned long long* prof_buf = nullptr;
-  if (pf && !prof_buf)
-    COS_CHECK_HIP(hipMalloc(&prof_buf, 4 * sizeof(unsigned long long)));
-  if (pf) COS_CHECK_HIP(hipMemsetAsync(prof_buf, 0, 32, stream));
   lstm_persist_fwd_kernel<<<nblk, 256, 0, stream>>>(
       (const u16*)xg, (const u16*)w_hc, (const u16*)cont, (u16*)h, c, act,
-      (u16*)h_in, hg, T, N, H, cnt, (int*)(cnt + 258),
-      pf ? prof_buf : nullptr);
-  if (pf) {
-    unsigned long long hostp[4];
-    COS_CHECK_HIP(hipMemcpyAsync(hostp, prof_buf, 32,
-                                 hipMemcpyDeviceToHost, stream));
-    COS_CHECK_HIP(hipStreamSynchronize(stream));
-    printf("[lstm_persist_fwd] cycles gemm=%llu bar1=%llu unit=%llu "
-           "bar2=%llu\n", hostp[0], hostp[1], hostp[2], hostp[3]);
-  }
-  return 0;   // watchdog flag read asynchronously by the caller
+      (u16*)h_in, T, N, H, cnt, (int*)(cnt + 258));
+  return 0;
 }
 
 int lstm_persist_bwd(const void* dy, const void* w_hcT, const void* cont,
                      const void* h, const float* c, const float* act,
                      void* dxg, float* dh_acc, float* dc_a, float* dc_b,
                      int T, int N, int H, void* bar, hipStream_t stream) {
-  if (N > 64 || H % 8 != 0) return 1;
-  int NCH = (H + 63) / 64;
-  int KCH = (4 * H + 255) / 256;
-  int nblk = NCH * KCH;
-  if (nblk > 256) return 1;
-  COS_CHECK_HIP(hipMemsetAsync(dh_acc, 0, (int64_t)N * H * 4, stream));
+  (void)dh_acc; (void)dc_a; (void)dc_b;
+  if (N > 64 || H % 8 != 0 || H > 1024) return 1;  // wf[32] k-window
+  int nblk = (H + 15) / 16;
+  if (nblk < 2 || nblk > 256) return 1;
   unsigned* cnt = reinterpret_cast<unsigned*>(bar);
   lstm_persist_bwd_kernel<<<nblk, 256, 0, stream>>>(
       (const u16*)dy, (const u16*)w_hcT, (const u16*)cont, (const u16*)h,
-      c, act, (u16*)dxg, dh_acc, dc_a, dc_b, T, N, H, cnt,
-      (int*)(cnt + 258));
-  return 0;   // watchdog flag read asynchronously by the caller
+      c, act, (u16*)dxg, T, N, H, cnt, (int*)(cnt + 258));
+  return 0;
 }
 
 }  // namespace cosamd

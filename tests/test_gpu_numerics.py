@@ -599,3 +599,38 @@ def test_fp32_gpu_dropout_lrn_ops():
     gen2 = torch.Generator().manual_seed(5)
     out2, m2 = ops.dropout_forward(x, 0.5, generator=gen2)
     torch.testing.assert_close(m1.cpu(), m2.cpu(), rtol=0, atol=0)
+
+
+def test_lstm_persistent_vs_loop_path(monkeypatch):
+    """The persistent whole-sequence LSTM kernel (grid-resident,
+    slice-owned) against the per-step loop path: identical gate math,
+    so h/c/act and the backward dxg/dwhc must agree tightly."""
+    from caffeonspark_amd.ops import gpu as g
+
+    T, N, H = 7, 48, 64        # H=64 -> 4 blocks: persistent-eligible
+    H4 = 4 * H
+    torch.manual_seed(3)
+    whc = (torch.randn(H4, H) * 0.2).to(torch.bfloat16).to(dev())
+    xg = (torch.randn(T, N, H4) * 0.5).to(torch.bfloat16).to(dev())
+    cont = torch.ones(T, N)
+    cont[0] = 0
+    cont[3, :5] = 0            # mid-sequence resets exercise the gating
+    cont = cont.to(torch.bfloat16).to(dev())
+    dy = torch.randn(T, N, H).to(torch.bfloat16).to(dev())
+
+    def run(persist):
+        monkeypatch.setenv("COS_LSTM_PERSIST", "1" if persist else "0")
+        h, cache = g.lstm_seq_forward(xg, whc, cont)
+        dxg, dwhc = g.lstm_seq_backward(dy, whc, cache)
+        torch.cuda.synchronize()
+        return (h.float().cpu(), cache[2].cpu(), dxg.float().cpu(),
+                dwhc.cpu())
+
+    h_p, c_p, dxg_p, dwhc_p = run(True)
+    h_l, c_l, dxg_l, dwhc_l = run(False)
+    assert not torch.isnan(h_p).any()
+    torch.testing.assert_close(h_p, h_l, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(c_p, c_l, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(dxg_p, dxg_l, rtol=3e-2, atol=3e-2)
+    rel = (dwhc_p - dwhc_l).norm() / dwhc_l.norm().clamp_min(1e-6)
+    assert rel < 2e-2, f"dwhc relL2={rel}"
