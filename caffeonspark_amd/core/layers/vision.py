@@ -141,12 +141,25 @@ class InnerProductLayer(Layer):
         if self.transpose:
             w = w.t()
         need_dw = self.blobs[0]._lr_mult != 0
+        wb = self.blobs[0]
+        dw_out = None
+        if need_dw and not self.transpose and \
+                getattr(wb, "_grad_virgin", False):
+            # first gradient write this step: let the dw GEMM store
+            # straight into the fp32 arena slice (no dwp->arena copy)
+            d = wb.ensure_diff()
+            if d.dtype == torch.float32 and d.is_contiguous():
+                dw_out = d.view(wb.shape)
         dx, dw, db = ops.fc_backward(x, w, dy,
                                      need_dx=propagate_down[0],
                                      bias=self.bias_term and
-                                     self.blobs[1]._lr_mult != 0)
+                                     self.blobs[1]._lr_mult != 0,
+                                     dw_out=dw_out)
         if dw is not None and need_dw:
-            self.acc_param_diff(0, dw.t() if self.transpose else dw)
+            if dw is dw_out:
+                wb._grad_virgin = False
+            else:
+                self.acc_param_diff(0, dw.t() if self.transpose else dw)
         if db is not None:
             self.acc_param_diff(1, db)
         if propagate_down[0]:

@@ -48,13 +48,11 @@ __device__ __forceinline__ int swz_chunk(int row, int chunk) {
 // 16-byte slot is fully in-bounds, scalar otherwise; zero-fills the rest.
 __device__ __forceinline__ void stage_direct_guarded(
     bf16* lds_, const bf16* g_, int row0, int rows, int ld,
-    int k0, int kend, int tid) {
+    int k0, int kend, int tid, int nthreads = NTHREADS) {
   auto* lds = reinterpret_cast<unsigned short*>(lds_);
   auto* g = reinterpret_cast<const unsigned short*>(g_);
-  // 4096 elements, 512 slots of 8; 256 threads -> 2 slots each
-#pragma unroll
-  for (int it = 0; it < 2; ++it) {
-    int slot = tid + it * NTHREADS;
+  // 4096 elements, 512 slots of 8
+  for (int slot = tid; slot < 512; slot += nthreads) {
     int row = slot >> 2;            // 4 slots of 8 per 32-wide row
     int kk = (slot & 3) * 8;
     int grow = row0 + row;
@@ -236,12 +234,14 @@ __global__ __launch_bounds__(WNT, 1) void gemm_tt_wide_kernel(
 }
 
 // fast path: interior tile, aligned — direct-to-LDS DMA, 16 B per lane.
+template <int WAVES = 4>
 __device__ __forceinline__ void stage_direct_fast(
     bf16* lds, const bf16* g, int row0, int ld, int k0, int tid) {
   int wave = tid >> 6, lane = tid & 63;
+  constexpr int CPW = 8 / WAVES;         // chunks per wave
 #pragma unroll
-  for (int it = 0; it < 2; ++it) {
-    int chunk = wave * 2 + it;           // 8 chunks of 512 elements
+  for (int it = 0; it < CPW; ++it) {
+    int chunk = wave * CPW + it;         // 8 chunks of 512 elements
     int idx = chunk * 512 + lane * 8;    // linear element index in tile
     int row = idx >> 5;                  // /32
     int kk = swz_chunk(row, (idx & 31) >> 3) << 3;  // pre-swizzled source
@@ -255,8 +255,11 @@ __device__ __forceinline__ void stage_direct_fast(
 
 // ------------------------------------------------------------------ kernel
 
-template <bool TRANS_A, bool TRANS_B, int STORE_MODE>
-__global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
+// WAVES=4: 2x2 wave grid of 64x64 (2 blocks/CU = 8 waves/CU);
+// WAVES=8: 4x2 grid of 32x64 (2 blocks/CU = 16 waves/CU — doubles the
+// wave-level parallelism hiding the staging/fragment latency).
+template <bool TRANS_A, bool TRANS_B, int STORE_MODE, int WAVES = 4>
+__global__ __launch_bounds__(WAVES * 64, 2) void gemm_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     int M, int N, int K, int lda, int ldb, int ldc,
@@ -279,12 +282,15 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
   int k_begin = blockIdx.y * ksplit;
   int k_end = min(K, k_begin + ksplit);
 
+  constexpr int NT = WAVES * 64;
+  constexpr int MROWS = BM / (WAVES / 2);   // rows per wave (64 or 32)
+  constexpr int MF = MROWS / 16;            // m-fragments per wave
   int tid = threadIdx.x;
   int wave = tid >> 6, lane = tid & 63;
-  int wm = wave >> 1, wn = wave & 1;     // 2x2 wave grid, 64x64 each
+  int wm = wave >> 1, wn = wave & 1;     // (WAVES/2) x 2 wave grid
   int lrow = lane & 15, lk8 = (lane >> 4) * 8;
 
-  f32x4 acc[4][4] = {};
+  f32x4 acc[MF][4] = {};
 
   // fast path only needs the STAGED rows to be readable: callers may
   // over-allocate operand buffers (m_alloc/n_alloc >= M/N) so edge tiles
@@ -302,33 +308,42 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
     // than __syncthreads' full drain — cdna_hip_programming.md §5.5
     // T3/T4: "never drain vmcnt to 0 in the main loop".
     int cur = 0;
-    stage_direct_fast(Asb[0], A, tile_m, lda, k_begin, tid);
-    stage_direct_fast(Bsb[0], B, tile_n, ldb, k_begin, tid);
+    stage_direct_fast<WAVES>(Asb[0], A, tile_m, lda, k_begin, tid);
+    stage_direct_fast<WAVES>(Bsb[0], B, tile_n, ldb, k_begin, tid);
     for (int k0 = k_begin; k0 < k_end; k0 += BK) {
       if (k0 + BK < k_end) {
-        stage_direct_fast(Asb[cur ^ 1], A, tile_m, lda, k0 + BK, tid);
-        stage_direct_fast(Bsb[cur ^ 1], B, tile_n, ldb, k0 + BK, tid);
+        stage_direct_fast<WAVES>(Asb[cur ^ 1], A, tile_m, lda, k0 + BK,
+                                 tid);
+        stage_direct_fast<WAVES>(Bsb[cur ^ 1], B, tile_n, ldb, k0 + BK,
+                                 tid);
       }
-      // wait for the CURRENT tile's DMA loads; the prefetched tile's 4
-      // (2 per operand per wave) remain outstanding
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      // wait for the CURRENT tile's DMA loads; the prefetched tile's
+      // (2 per operand per wave at WAVES=4, 1 each at 8) remain
+      // outstanding
+      if (WAVES == 4)
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
       __builtin_amdgcn_sched_barrier(0);
       __builtin_amdgcn_s_barrier();
       const bf16* Asp = Asb[cur];
       const bf16* Bsp = Bsb[cur];
-      bf16x8 afrag[4], bfrag[4];
+      bf16x8 afrag[MF], bfrag[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        int ra = wm * 64 + f * 16 + lrow;
         int rb = wn * 64 + f * 16 + lrow;
-        afrag[f] = *reinterpret_cast<const bf16x8*>(
-            Asp + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
         bfrag[f] = *reinterpret_cast<const bf16x8*>(
             Bsp + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
       }
+#pragma unroll
+      for (int f = 0; f < MF; ++f) {
+        int ra = wm * MROWS + f * 16 + lrow;
+        afrag[f] = *reinterpret_cast<const bf16x8*>(
+            Asp + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
+      }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int fm = 0; fm < 4; ++fm)
+      for (int fm = 0; fm < MF; ++fm)
 #pragma unroll
         for (int fn = 0; fn < 4; ++fn)
           acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -345,33 +360,38 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
     bf16* Bs = Bsb[0];
     for (int k0 = k_begin; k0 < k_end; k0 += BK) {
       if (a_fast) {
-        stage_direct_fast(As, A, tile_m, lda, k0, tid);
+        stage_direct_fast<WAVES>(As, A, tile_m, lda, k0, tid);
       } else if (TRANS_A) {
-        stage_trans_pair_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
+        stage_trans_pair_guarded(As, A, tile_m, M, lda, k0, k_end, tid,
+                                 NT);
       } else {
-        stage_direct_guarded(As, A, tile_m, M, lda, k0, k_end, tid);
+        stage_direct_guarded(As, A, tile_m, M, lda, k0, k_end, tid, NT);
       }
       if (b_fast) {
-        stage_direct_fast(Bs, B, tile_n, ldb, k0, tid);
+        stage_direct_fast<WAVES>(Bs, B, tile_n, ldb, k0, tid);
       } else if (TRANS_B) {
-        stage_trans_pair_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
+        stage_trans_pair_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid,
+                                 NT);
       } else {
-        stage_direct_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid);
+        stage_direct_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid, NT);
       }
       __syncthreads();
 
-      bf16x8 afrag[4], bfrag[4];
+      bf16x8 afrag[MF], bfrag[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        int ra = wm * 64 + f * 16 + lrow;
         int rb = wn * 64 + f * 16 + lrow;
-        afrag[f] = *reinterpret_cast<const bf16x8*>(
-            As + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
         bfrag[f] = *reinterpret_cast<const bf16x8*>(
             Bs + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
       }
 #pragma unroll
-      for (int fm = 0; fm < 4; ++fm)
+      for (int f = 0; f < MF; ++f) {
+        int ra = wm * MROWS + f * 16 + lrow;
+        afrag[f] = *reinterpret_cast<const bf16x8*>(
+            As + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
+      }
+#pragma unroll
+      for (int fm = 0; fm < MF; ++fm)
 #pragma unroll
         for (int fn = 0; fn < 4; ++fn)
           acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -382,10 +402,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_kernel(
   }
 
   // ------------------------------------------------------------- epilogue
-  int crow0 = tile_m + wm * 64 + ((lane >> 4) << 2);
+  int crow0 = tile_m + wm * MROWS + ((lane >> 4) << 2);
   int ccol0 = tile_n + wn * 64 + (lane & 15);
 #pragma unroll
-  for (int fm = 0; fm < 4; ++fm) {
+  for (int fm = 0; fm < MF; ++fm) {
 #pragma unroll
     for (int fn = 0; fn < 4; ++fn) {
       int col = ccol0 + fn * 16;
@@ -458,9 +478,23 @@ void gemm_bf16_batched(const void* A_, const void* B_, void* C,
   dim3 grid(mblocks * nblocks, zblocks, batch);
   dim3 block(NTHREADS);
 
+  // 8-wave blocks win on skinny shapes (one of M/N <= 256: fc6 +12%,
+  // conv2/conv5 fwd +6-7%, conv1 +6% measured) and lose on fat ones
+  // (conv3 -15%); COS_GEMM_W8=0/1 forces either variant for A/B.
+  static const int w8_env = [] {
+    const char* e = getenv("COS_GEMM_W8");
+    return e ? (e[0] == '1' ? 1 : 0) : -1;
+  }();
+  bool w8 = (w8_env == -1) ? (M <= 256 || N <= 256) : (w8_env == 1);
+
 #define COS_GEMM_CASE(TA, TB, SM)                                          \
   gemm_kernel<TA, TB, SM><<<grid, block, 0, stream>>>(                      \
       A, B, C, bias, M, N, K, lda, ldb, ldc, ksplit, relu ? 1 : 0, alpha,    \
+      m_alloc, n_alloc, sA, sB, sC_bytes)
+
+#define COS_GEMM_CASE8(SM)                                                 \
+  gemm_kernel<false, false, SM, 8><<<grid, dim3(512), 0, stream>>>(        \
+      A, B, C, bias, M, N, K, lda, ldb, ldc, ksplit, relu ? 1 : 0, alpha,   \
       m_alloc, n_alloc, sA, sB, sC_bytes)
 
 #define COS_GEMM_SM(TA, TB)                                                 \
@@ -478,12 +512,21 @@ void gemm_bf16_batched(const void* A_, const void* B_, void* C,
         alpha);
     return;
   }
-  if (!trans_a && !trans_b)      COS_GEMM_SM(false, false);
+  if (!trans_a && !trans_b) {
+    if (w8) {
+      if (store_mode == 0)      COS_GEMM_CASE8(0);
+      else if (store_mode == 1) COS_GEMM_CASE8(1);
+      else                      COS_GEMM_CASE8(2);
+      return;
+    }
+    COS_GEMM_SM(false, false);
+  }
   else if (!trans_a && trans_b)  COS_GEMM_SM(false, true);
   else if (trans_a && !trans_b)  COS_GEMM_SM(true, false);
   else                           COS_GEMM_SM(true, true);
 #undef COS_GEMM_SM
 #undef COS_GEMM_CASE
+#undef COS_GEMM_CASE8
 }
 
 void gemm_bf16(const void* A_, const void* B_, void* C, const float* bias,

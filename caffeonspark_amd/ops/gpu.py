@@ -373,7 +373,7 @@ def fc_forward(x, w, b, relu=False):
     return y
 
 
-def fc_backward(x, w, dy, need_dx=True, bias=True):
+def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None):
     _check_bf16(dy, "fc dy")
     x = x.contiguous()
     dy = dy.contiguous()
@@ -396,8 +396,16 @@ def fc_backward(x, w, dy, need_dx=True, bias=True):
             _gemm(dy, wT, dx, None, M, K, Nout, Nout, Nout, K, False,
                   False, 0, 1, na=_pad128(K))
     # dw = dy^T @ x: fused trans/trans (u32 k-pair staging) — no
-    # operand transpose kernels, one read of each operand
-    dwp = torch.zeros((Nout, K), dtype=torch.float32, device=x.device)
+    # operand transpose kernels, one read of each operand.  dw_out (the
+    # solver's fp32 arena slice, when the layer owns the only gradient
+    # write this step) receives the GEMM output directly, skipping the
+    # dwp-to-arena copy (fc6 alone is 151 MB fp32)
+    if dw_out is not None and dw_out.dtype == torch.float32 \
+            and dw_out.is_contiguous() and tuple(dw_out.shape) == (Nout, K):
+        dwp = dw_out
+        dwp.zero_()
+    else:
+        dwp = torch.zeros((Nout, K), dtype=torch.float32, device=x.device)
     _gemm(dy, x, dwp, None, Nout, K, M, Nout, K, K, True, True, 2,
           _splitk_tt(Nout, K, M))
     dw = dwp

@@ -47,9 +47,12 @@ def fc_forward(x, w, b, relu=False):
     return F.relu(y) if relu else y
 
 
-def fc_backward(x, w, dy, need_dx=True, bias=True):
+def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None):
     dx = dy @ w if need_dx else None
     dw = dy.t() @ x
+    if dw_out is not None:
+        dw_out.copy_(dw)
+        dw = dw_out
     db = dy.sum(dim=0) if bias else None
     return dx, dw, db
 
