@@ -154,9 +154,11 @@ __global__ __launch_bounds__(256, 1) void lstm_persist_fwd_kernel(
     // ---- gate GEMM over the full K for this block's 64 gate columns
     const u16* A = h_in + (int64_t)t * nh;
     f32x4v acc[4][4] = {};    // [mf][strip]
-    for (int ks = 0; ks < ksteps; ++ks) {
+    // software-pipelined a-loads: fetch ks+1 while ks's MFMAs run (the
+    // 4-MFMA bursts are too short to hide the L2 load chain otherwise)
+    bf16x8v a_cur[4], a_nxt[4];
+    auto load_a = [&](bf16x8v (&a)[4], int ks) {
       int k = kw0 + ks * 32 + lk8;
-      bf16x8v a[4];
 #pragma unroll
       for (int mf = 0; mf < 4; ++mf) {
         int m = mf * 16 + lrow;
@@ -169,12 +171,18 @@ __global__ __launch_bounds__(256, 1) void lstm_persist_fwd_kernel(
           for (int j = 0; j < 8; ++j) a[mf][j] = z;
         }
       }
+    };
+    load_a(a_cur, 0);
+    for (int ks = 0; ks < ksteps; ++ks) {
+      if (ks + 1 < ksteps) load_a(a_nxt, ks + 1);
 #pragma unroll
       for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
         for (int s = 0; s < 4; ++s)
           acc[mf][s] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a[mf], wf[s][ks < 8 ? ks : 0], acc[mf][s], 0, 0, 0);
+              a_cur[mf], wf[s][ks < 8 ? ks : 0], acc[mf][s], 0, 0, 0);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) a_cur[mf] = a_nxt[mf];
     }
     // ---- cross-wave reduce into gates[] (pairwise through LDS)
     {
@@ -358,9 +366,9 @@ __global__ __launch_bounds__(256, 1) void lstm_persist_bwd_kernel(
     // ---- dh GEMM for t-1: dh_slice = dxg_t @ w_hcT^T (full 4H K)
     if (t > 0) {
       f32x4v acc[4] = {};     // [mf] x 16 cols
-      for (int ks = 0; ks < ksteps; ++ks) {
+      bf16x8v a_cur[4], a_nxt[4];
+      auto load_a = [&](bf16x8v (&a)[4], int ks) {
         int k = kw0 + ks * 32 + lk8;
-        bf16x8v a[4];
 #pragma unroll
         for (int mf = 0; mf < 4; ++mf) {
           int m = mf * 16 + lrow;
@@ -373,10 +381,16 @@ __global__ __launch_bounds__(256, 1) void lstm_persist_bwd_kernel(
             for (int j = 0; j < 8; ++j) a[mf][j] = z;
           }
         }
+      };
+      load_a(a_cur, 0);
+      for (int ks = 0; ks < ksteps; ++ks) {
+        if (ks + 1 < ksteps) load_a(a_nxt, ks + 1);
 #pragma unroll
         for (int mf = 0; mf < 4; ++mf)
           acc[mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a[mf], wf[ks < 32 ? ks : 0], acc[mf], 0, 0, 0);
+              a_cur[mf], wf[ks < 32 ? ks : 0], acc[mf], 0, 0, 0);
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf) a_cur[mf] = a_nxt[mf];
       }
       // cross-wave reduce -> dh_lds
       float* mybuf = (wave >> 1) ? red1 : red0;
