@@ -284,3 +284,56 @@ class SparkContext:
         for ex in jobs:
             results.append(ex.recv())
         return results
+
+
+class Partitioner:
+    """Spark Partitioner base (getPartition(key) -> partition index)."""
+
+    def __init__(self, numPartitions: int):
+        self.numPartitions = numPartitions
+
+    def getPartition(self, key) -> int:
+        return hash(key) % self.numPartitions
+
+
+class FixedSizePartitioner(Partitioner):
+    """Reference FixedSizePartitioner.scala: integer keys map to
+    key // part_size — used by trainWithValidation to slice the train
+    stream into fixed-size per-executor partitions."""
+
+    def __init__(self, numPartitions: int, part_size: int):
+        super().__init__(numPartitions)
+        self.part_size = max(1, part_size)
+
+    def getPartition(self, key) -> int:
+        return int(key) // self.part_size % self.numPartitions
+
+
+def partition_by(sc: "SparkContext", pairs, partitioner: Partitioner
+                 ) -> RDD:
+    """RDD.partitionBy analog for (key, value) pairs."""
+    parts: List[List[Any]] = [[] for _ in range(partitioner.numPartitions)]
+    for k, v in pairs:
+        parts[partitioner.getPartition(k)].append((k, v))
+    return RDD(sc, parts, lazy=False)
+
+
+def union_with_locations(sc: "SparkContext", rdds: List[RDD],
+                         locations: Optional[List[List[str]]] = None
+                         ) -> RDD:
+    """Reference UnionRDDWLocsSpecified.scala: union whose partitions
+    carry preferred executor locations so each executor receives a copy
+    of the same validation partition.  The local engine's deterministic
+    partition->executor placement (i % n) realises the location hint by
+    ORDER: partition i of the union lands on executor i % n, so passing
+    one copy of the validation data per executor reproduces the
+    reference's co-location."""
+    parts: List[Any] = []
+    lazy = any(r._lazy for r in rdds)
+    for r in rdds:
+        if lazy and not r._lazy:
+            for p in r._parts:
+                parts.append(lambda p=p: list(p))
+        else:
+            parts.extend(r._parts)
+    return RDD(sc, parts, lazy=lazy)

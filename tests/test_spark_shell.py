@@ -157,3 +157,41 @@ def test_sparkshell_train_with_validation(tmp_path):
         assert "accuracy" in rows[-1]
     finally:
         sc.stop()
+
+
+def test_fixed_size_partitioner_and_union():
+    """Reference FixedSizePartitioner + UnionRDDWLocsSpecified semantics
+    on the local engine: fixed-size key slicing, and a union that
+    co-locates one copy of the same partition per executor."""
+    from caffeonspark_amd.spark.local import (FixedSizePartitioner,
+                                              SparkContext, partition_by,
+                                              union_with_locations)
+
+    sc = SparkContext(master="local[2]")
+    try:
+        part = FixedSizePartitioner(4, part_size=25)
+        pairs = [(i, f"v{i}") for i in range(100)]
+        rdd = partition_by(sc, pairs, part)
+        assert rdd.getNumPartitions() == 4
+        groups = rdd.mapPartitions(
+            lambda it: [sorted(k for k, _ in it)]).collect()
+        # each partition holds one contiguous 25-key slice
+        assert sorted(g[0] for g in groups if g) == [0, 25, 50, 75]
+        for g in groups:
+            assert g == list(range(g[0], g[0] + 25))
+
+        # union replicating a validation partition to both executors
+        val = sc.parallelize([("val", 1), ("val", 2)], 1)
+        u = union_with_locations(sc, [val, val])
+        assert u.getNumPartitions() == 2
+
+        def where(it):
+            import os
+            yield (os.environ.get("COS_SPARK_EXECUTOR_ID"),
+                   sorted(v for _, v in it))
+        got = u.mapPartitions(where).collect()
+        # both executors received the SAME validation rows
+        assert sorted(ex for ex, _ in got) == ["0", "1"]
+        assert all(rows == [1, 2] for _, rows in got)
+    finally:
+        sc.stop()
