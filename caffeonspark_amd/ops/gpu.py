@@ -381,19 +381,34 @@ def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None):
     M, K = x.shape
     Nout = wb.shape[0]
     dx = dw = db = None
+    # odd output widths (LRCN fc8: vocab 8801) misalign every dy row and
+    # force the GEMMs onto the guarded staging path; one zero-padded
+    # copy restores the 16-byte DMA path for both dx (K dim) and dw
+    # (trans lda), and the zero columns contribute nothing
+    Np = Nout
+    if Nout % 8:
+        Np = _pad8(Nout)
+        dy_p = torch.zeros((M, Np), dtype=torch.bfloat16, device=dy.device)
+        dy_p[:, :Nout] = dy
+        dy = dy_p
+        wb_p = torch.zeros((Np, K), dtype=torch.bfloat16, device=dy.device)
+        wb_p[:Nout] = wb
+        wb = wb_p
     if need_dx:
-        # dx = dy @ w: transpose w once -> NT direct/direct fast staging
+        # dx = dy @ w: transpose w once -> NT direct/direct fast staging.
+        # With padded dy the K dim runs to Np (the pad columns multiply
+        # wT's over-allocated junk rows by zero).
         wT = _transpose(wb)
         dx = torch.empty((M, K), dtype=torch.bfloat16, device=x.device)
         mb, nb = (M + 127) // 128, (K + 127) // 128
-        sk = _splitk_for(mb, nb, Nout)
+        sk = _splitk_for(mb, nb, Np)
         if sk > 1:
             wsp = torch.zeros((M, K), dtype=torch.float32, device=x.device)
-            _gemm(dy, wT, wsp, None, M, K, Nout, Nout, Nout, K, False,
+            _gemm(dy, wT, wsp, None, M, K, Np, Np, Np, K, False,
                   False, 2, sk, na=_pad128(K))
             _ext.bias_act_cast(wsp, None, dx, False)
         else:
-            _gemm(dy, wT, dx, None, M, K, Nout, Nout, Nout, K, False,
+            _gemm(dy, wT, dx, None, M, K, Np, Np, Np, K, False,
                   False, 0, 1, na=_pad128(K))
     # dw = dy^T @ x: fused trans/trans (u32 k-pair staging) — no
     # operand transpose kernels, one read of each operand.  dw_out (the
@@ -406,12 +421,12 @@ def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None):
         dwp.zero_()
     else:
         dwp = torch.zeros((Nout, K), dtype=torch.float32, device=x.device)
-    _gemm(dy, x, dwp, None, Nout, K, M, Nout, K, K, True, True, 2,
+    _gemm(dy, x, dwp, None, Nout, K, M, Np, K, K, True, True, 2,
           _splitk_tt(Nout, K, M))
     dw = dwp
     if bias:
         db = torch.zeros(Nout, dtype=torch.float32, device=x.device)
-        _ext.colsum(dy, db, M, Nout, Nout)
+        _ext.colsum(dy, db, M, Nout, Np)
     return dx, dw, db
 
 
