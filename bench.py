@@ -96,12 +96,11 @@ def main():
         if use_gpu:
             torch.cuda.synchronize()
 
-    # measured on MI355X: whole-step hipGraph replay wins for sub-ms
-    # launch-bound steps (cifar 106k -> 142k img/s) and is neutral to
-    # slightly negative for the big models (alexnet 32.0k eager vs 31.5k
-    # graphed) — auto mode picks per model
-    use_graph = args.graph == 1 or (
-        args.graph == -1 and args.model == "cifar10_quick")
+    # measured on MI355X: whole-step hipGraph replay wins everywhere at
+    # ws=1 after the round-2 kernel work shifted the steps launch-bound
+    # (cifar 106k -> 142k round 1; round 2: googlenet +2.2%, alexnet
+    # +1.1%, lrcn +0.4%) — auto mode turns it on for single-process runs
+    use_graph = args.graph != 0
     step = solver.graph_step if (use_graph and ws == 1 and use_gpu) \
         else solver._step_one
     for _ in range(args.warmup):

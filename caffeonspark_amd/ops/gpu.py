@@ -751,6 +751,8 @@ def _persist_check_pending(force=False):
 
 
 def _persist_track(bar):
+    if torch.cuda.is_current_stream_capturing():
+        return              # poison-only safety under graph capture
     flag = torch.empty(1, dtype=torch.int32, pin_memory=True)
     flag.copy_(bar[258:259], non_blocking=True)
     ev = torch.cuda.Event()
@@ -759,11 +761,14 @@ def _persist_track(bar):
 
 
 def _lstm_persist_ok(N):
-    _persist_check_pending(force=len(_persist_pending) > 16)
     if N > 64 or int(os.environ.get("COS_LSTM_PERSIST", "1")) == 0:
         return False
     if torch.cuda.is_current_stream_capturing():
-        return False        # host sync inside is capture-illegal
+        # capture-legal (async memset + one launch); the deferred err
+        # check can't run per-replay, but a watchdog abort NaN-poisons
+        # the outputs so it cannot pass silently
+        return True
+    _persist_check_pending(force=len(_persist_pending) > 16)
     import torch.distributed as dist
     if dist.is_initialized() and dist.get_world_size() > 1:
         return False        # RCCL kernels may break grid residency
