@@ -110,7 +110,12 @@ class Solver:
             for b, o in zip(self.params, self.param_offsets):
                 b.data._cos_bf16 = self.flat_wb.narrow(
                     0, o, b.count).view(b.shape)
-            refresh = lambda: self.flat_wb.copy_(self.flat_w)  # noqa: E731
+            def refresh():
+                self.flat_wb.copy_(self.flat_w)
+                # fused per-step conv weight repack (GEMM layouts follow
+                # the shadow arena in one kernel instead of 2/conv)
+                from ..ops import gpu as _gops
+                _gops.refresh_packed_weights()
             self.net._bf16_refresh = refresh
         # contiguous segments sharing (lr_mult, decay_mult) for fused updates
         self.segments = []  # (off, n, lr_mult, decay_mult)

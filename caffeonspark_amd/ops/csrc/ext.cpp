@@ -17,6 +17,8 @@ void gemm_bf16(const void* A, const void* B, void* C, const float* bias,
                bool relu, float alpha, int m_alloc, int n_alloc,
                hipStream_t stream);
 void tr16_probe(float* out, int mode, hipStream_t stream);
+void repack_weights(const int64_t* table, int ndesc, int64_t max_total,
+                    hipStream_t stream);
 int lstm_persist_fwd(const void* xg, const void* w_hc, const void* cont,
                      void* h, float* c, float* act, void* h_in, float* hg,
                      int T, int N, int H, void* bar, hipStream_t stream);
@@ -264,6 +266,11 @@ void py_bias_act_cast(Tensor in, c10::optional<Tensor> bias, Tensor out,
                         in.size(0), in.size(1), relu, cur_stream());
 }
 
+void py_repack_weights(Tensor table, int64_t ndesc, int64_t max_total) {
+  cosamd::repack_weights(table.data_ptr<int64_t>(), (int)ndesc, max_total,
+                         cur_stream());
+}
+
 int64_t py_lstm_persist_fwd(Tensor xg, Tensor w_hc, Tensor cont,
                             Tensor h, Tensor c, Tensor act, Tensor h_in,
                             Tensor hg, int64_t T, int64_t N, int64_t H,
@@ -454,6 +461,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum", &py_colsum);
   m.def("transpose", &py_transpose);
   m.def("tr16_probe", &py_tr16_probe);
+  m.def("repack_weights", &py_repack_weights);
   m.def("lstm_persist_fwd", &py_lstm_persist_fwd);
   m.def("lstm_persist_bwd", &py_lstm_persist_bwd);
   m.def("bias_act_cast", &py_bias_act_cast);

@@ -1,0 +1,50 @@
+// Fused conv-weight repack: ONE launch refreshes every conv layer's
+// GEMM-ready weight layouts per step, replacing per-layer permute-copy
+// + transpose kernels (GoogLeNet: 57 convs -> ~120 tiny launches/step,
+// each dominated by wave-ramp overhead).
+//
+// Per descriptor: src  [Kout][Cg][R][S] bf16 (the solver's bf16 shadow
+//                 view of the caffe-layout master weight)
+//                 dst  [>=Kout][Kpad]   (forward GEMM B operand,
+//                 [k][(r*S+s)*Cg + c]; pad columns stay zero)
+//                 dstT [>=Kpad][Kout]   (optional, 0 = skip: the dx
+//                 GEMM's transposed operand)
+// Table rows: [src, dst, dstT, Kout, Cg, R, S, Kpad] as int64.
+
+#include "common.h"
+
+namespace cosamd {
+
+typedef unsigned short u16;
+
+__global__ void repack_weights_kernel(const int64_t* __restrict__ table,
+                                      int ndesc) {
+  const int64_t* e = table + (int64_t)blockIdx.y * 8;
+  const u16* src = reinterpret_cast<const u16*>(e[0]);
+  u16* dst = reinterpret_cast<u16*>(e[1]);
+  u16* dstT = reinterpret_cast<u16*>(e[2]);
+  int Kout = (int)e[3], Cg = (int)e[4], R = (int)e[5], S = (int)e[6];
+  int Kpad = (int)e[7];
+  int RS = R * S;
+  int Kcol = RS * Cg;
+  int64_t total = (int64_t)Kout * Kcol;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int k = (int)(i / Kcol);
+    int col = (int)(i % Kcol);
+    int rs = col / Cg, c = col % Cg;
+    u16 v = src[((int64_t)k * Cg + c) * RS + rs];
+    dst[(int64_t)k * Kpad + col] = v;
+    if (dstT != nullptr) dstT[(int64_t)col * Kout + k] = v;
+  }
+}
+
+void repack_weights(const int64_t* table, int ndesc, int64_t max_total,
+                    hipStream_t stream) {
+  if (ndesc <= 0) return;
+  int bx = (int)hmin<int64_t>(512, (max_total + 255) / 256);
+  dim3 grid(bx, ndesc);
+  repack_weights_kernel<<<grid, 256, 0, stream>>>(table, ndesc);
+}
+
+}  // namespace cosamd

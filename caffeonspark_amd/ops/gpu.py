@@ -95,6 +95,40 @@ def _transpose(t: torch.Tensor) -> torch.Tensor:
     return out
 
 
+# ---- fused weight repack registry: one kernel per step refreshes every
+# conv's GEMM-ready layouts (wrb and, for G==1, its transpose) from the
+# bf16 shadow arena, replacing ~2 tiny launches per conv per step.
+# Items hold tensor refs so the device pointers in the table stay live.
+_repack = {"items": [], "table": None, "max": 0, "epoch": 0, "dirty": False}
+
+
+def _repack_register(src, wrb, wrT, Kout, Cg, R, S, Kpad):
+    _repack["items"].append((src, wrb, wrT, Kout, Cg, R, S, Kpad))
+    _repack["dirty"] = True
+
+
+def refresh_packed_weights():
+    """Called from the solver's per-step bf16 shadow refresh."""
+    items = _repack["items"]
+    if not items:
+        _repack["epoch"] += 1
+        return
+    if _repack["dirty"]:
+        rows = []
+        mx = 0
+        for (src, wrb, wrT, Kout, Cg, R, S, Kpad) in items:
+            rows.append([src.data_ptr(), wrb.data_ptr(),
+                         0 if wrT is None else wrT.data_ptr(),
+                         Kout, Cg, R, S, Kpad])
+            mx = max(mx, Kout * R * S * Cg)
+        _repack["table"] = torch.tensor(rows, dtype=torch.int64).to(
+            items[0][1].device)
+        _repack["max"] = mx
+        _repack["dirty"] = False
+    _ext.repack_weights(_repack["table"], len(items), _repack["max"])
+    _repack["epoch"] += 1
+
+
 def _gemm(A, B, C, bias, M, N, K, lda, ldb, ldc, ta, tb, store, splitk,
           relu=False, alpha=1.0, ma=0, na=0):
     # staging-allocation bounds: number of rows safely readable past M/N
@@ -136,7 +170,28 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
         wrb = torch.zeros((_pad128(Kout), Kpad), dtype=torch.bfloat16,
                           device=x.device)
         w._cos_wrb = wrb
-    wrb[:Kout, :Kcol] = _as_bf16(w).permute(0, 2, 3, 1).reshape(Kout, Kcol)
+        shadow = getattr(w, "_cos_bf16", None)
+        if shadow is not None and shadow.is_contiguous():
+            # solver-managed weight: the fused per-step repack kernel
+            # (refresh_packed_weights, run with the shadow refresh)
+            # maintains wrb — and for G==1 also the dx GEMM's transposed
+            # layout — from here on
+            wrT = None
+            if G == 1:
+                wrT = torch.zeros((_pad128(Kpad), Kout),
+                                  dtype=torch.bfloat16, device=x.device)
+                w._cos_wrT = wrT
+            _repack_register(shadow, wrb, wrT, Kout, Cg, R, S, Kpad)
+            w._cos_repacked = True
+    if getattr(w, "_cos_repack_epoch", -1) != _repack["epoch"] or \
+            not getattr(w, "_cos_repacked", False):
+        wrb[:Kout, :Kcol] = _as_bf16(w).permute(0, 2, 3, 1) \
+            .reshape(Kout, Kcol)
+        wrT = getattr(w, "_cos_wrT", None)
+        if wrT is not None:
+            wrT[:Kcol] = wrb[:Kout, :Kcol].t()
+        if getattr(w, "_cos_repacked", False):
+            w._cos_repack_epoch = _repack["epoch"]
     wr = wrb[:Kout]
     bias_f = b.float().contiguous() if b is not None else None
 
@@ -205,6 +260,7 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
         ctx["shape"] = (N, C, H, W, P, Q, R, S, sh, sw, ph, pw, dil, G, Cg,
                         Kg, Kpad, Kcol)
         ctx["wr"] = wr
+        ctx["w_ref"] = w
     return y
 
 
@@ -329,9 +385,13 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         dcol = torch.empty((NPQ, Kpad), dtype=torch.bfloat16,
                            device=dy.device)
         for g in range(G):
-            # dcol[npq, kpad] = dy_g[npq, kg] @ w_g[kg, kpad]: transpose the
-            # (small) packed weights so B is [kpad][kg] = NT direct form
-            wrT = _transpose(wr[g * Kg:(g + 1) * Kg].contiguous())
+            # dcol[npq, kpad] = dy_g[npq, kg] @ w_g[kg, kpad]: transposed
+            # packed weights — maintained by the fused repack kernel for
+            # G==1, transposed here otherwise
+            wrT = getattr(ctx.get("w_ref"), "_cos_wrT", None) \
+                if G == 1 else None
+            if wrT is None:
+                wrT = _transpose(wr[g * Kg:(g + 1) * Kg].contiguous())
             _gemm(dy2[:, g * Kg:], wrT, dcol, None,
                   NPQ, Kpad, Kg, Kout, Kg, Kpad, False, False, 0, 1,
                   na=_pad128(Kpad))
