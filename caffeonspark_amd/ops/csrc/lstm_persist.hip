@@ -69,7 +69,7 @@ __device__ __forceinline__ bool grid_sync(unsigned* bar, unsigned nblk,
                              __HIP_MEMORY_SCOPE_SYSTEM);
         break;
       }
-      __builtin_amdgcn_s_sleep(2);
+      if (spins > 2048) __builtin_amdgcn_s_sleep(2);
     }
     if (slot == 0) {
       __hip_atomic_load(arr, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
@@ -85,7 +85,7 @@ __device__ __forceinline__ bool grid_sync(unsigned* bar, unsigned nblk,
       long spins = 0;
       while (__hip_atomic_load(gen, __ATOMIC_RELAXED,
                                __HIP_MEMORY_SCOPE_AGENT) < my_count) {
-        __builtin_amdgcn_s_sleep(4);
+        if (spins > 2048) __builtin_amdgcn_s_sleep(4);
         if (++spins > (1L << 24)) {
           __hip_atomic_store(err, 1, __ATOMIC_RELEASE,
                              __HIP_MEMORY_SCOPE_SYSTEM);
