@@ -39,11 +39,12 @@ def main():
     from caffeonspark_amd.parallel import DistributedSync, init_distributed
 
     ws = int(os.environ.get("WORLD_SIZE", "1"))
-    rank = init_distributed()
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     use_gpu = torch.cuda.is_available()
     if use_gpu:
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(local_rank)   # before NCCL/RCCL init
+    rank = init_distributed()
+    if use_gpu:
         dev = torch.device("cuda", local_rank)
         dtype = torch.float32 if args.dtype == "fp32" else torch.bfloat16
     else:
