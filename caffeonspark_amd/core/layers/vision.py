@@ -86,13 +86,22 @@ class ConvolutionLayer(Layer):
         w = self.weight(0)
         need_dw = self.blobs[0]._lr_mult != 0
         need_db = self.bias_term and self.blobs[1]._lr_mult != 0
+        wb = self.blobs[0]
+        dw_out = None
+        if need_dw and getattr(wb, "_grad_virgin", False):
+            d = wb.ensure_diff()
+            if d.dtype == torch.float32 and d.is_contiguous():
+                dw_out = d.view(wb.shape)
         dx, dw, db = ops.conv2d_backward(
             x, w, dy, (self.sh, self.sw), (self.ph, self.pw),
             (self.dil, self.dil), self.groups,
             need_dx=propagate_down[0], need_dw=need_dw, bias=need_db,
-            ctx=getattr(self, "_ctx", None))
+            ctx=getattr(self, "_ctx", None), dw_out=dw_out)
         if dw is not None:
-            self.acc_param_diff(0, dw)
+            if dw is dw_out:
+                wb._grad_virgin = False
+            else:
+                self.acc_param_diff(0, dw)
         if db is not None:
             self.acc_param_diff(1, db)
         if propagate_down[0]:

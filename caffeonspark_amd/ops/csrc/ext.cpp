@@ -19,6 +19,9 @@ void gemm_bf16(const void* A, const void* B, void* C, const float* bias,
 void tr16_probe(float* out, int mode, hipStream_t stream);
 void repack_weights(const int64_t* table, int ndesc, int64_t max_total,
                     hipStream_t stream);
+void dw_unpack_acc(const float* dwp, float* arena, int Kout, int Cg,
+                   int R, int S, int Kpad, bool accumulate,
+                   hipStream_t stream);
 int lstm_persist_fwd(const void* xg, const void* w_hc, const void* cont,
                      void* h, float* c, float* act, void* h_in, float* hg,
                      int T, int N, int H, void* bar, hipStream_t stream);
@@ -266,6 +269,14 @@ void py_bias_act_cast(Tensor in, c10::optional<Tensor> bias, Tensor out,
                         in.size(0), in.size(1), relu, cur_stream());
 }
 
+void py_dw_unpack_acc(Tensor dwp, Tensor arena, int64_t Kout, int64_t Cg,
+                      int64_t R, int64_t S, int64_t Kpad,
+                      bool accumulate) {
+  cosamd::dw_unpack_acc(dwp.data_ptr<float>(), arena.data_ptr<float>(),
+                        (int)Kout, (int)Cg, (int)R, (int)S, (int)Kpad,
+                        accumulate, cur_stream());
+}
+
 void py_repack_weights(Tensor table, int64_t ndesc, int64_t max_total) {
   cosamd::repack_weights(table.data_ptr<int64_t>(), (int)ndesc, max_total,
                          cur_stream());
@@ -462,6 +473,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("transpose", &py_transpose);
   m.def("tr16_probe", &py_tr16_probe);
   m.def("repack_weights", &py_repack_weights);
+  m.def("dw_unpack_acc", &py_dw_unpack_acc);
   m.def("lstm_persist_fwd", &py_lstm_persist_fwd);
   m.def("lstm_persist_bwd", &py_lstm_persist_bwd);
   m.def("bias_act_cast", &py_bias_act_cast);

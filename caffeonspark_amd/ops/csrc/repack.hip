@@ -47,4 +47,38 @@ void repack_weights(const int64_t* table, int ndesc, int64_t max_total,
   repack_weights_kernel<<<grid, 256, 0, stream>>>(table, ndesc);
 }
 
+// conv weight-gradient unpack + arena accumulate: the dw GEMM produces
+// dwp[Kout][Kpad] in the GEMM's [k][(r*S+s)*Cg + c] layout; this writes
+// it into the caffe-layout fp32 arena slice [Kout][Cg][R][S] in one
+// pass (overwrite or +=), replacing a permute-contiguous copy plus the
+// arena accumulate.
+__global__ void dw_unpack_acc_kernel(const float* __restrict__ dwp,
+                                     float* __restrict__ arena,
+                                     int Kout, int Cg, int R, int S,
+                                     int Kpad, int accumulate) {
+  int RS = R * S;
+  int Kcol = RS * Cg;
+  int64_t total = (int64_t)Kout * Kcol;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int k = (int)(i / Kcol);
+    int crs = (int)(i % Kcol);   // arena-order within one filter
+    int c = crs / RS, rs = crs % RS;
+    float v = dwp[(int64_t)k * Kpad + rs * Cg + c];
+    if (accumulate)
+      arena[i] += v;
+    else
+      arena[i] = v;
+  }
+}
+
+void dw_unpack_acc(const float* dwp, float* arena, int Kout, int Cg,
+                   int R, int S, int Kpad, bool accumulate,
+                   hipStream_t stream) {
+  int64_t total = (int64_t)Kout * R * S * Cg;
+  int blocks = (int)hmin<int64_t>(2048, (total + 255) / 256);
+  dw_unpack_acc_kernel<<<blocks, 256, 0, stream>>>(
+      dwp, arena, Kout, Cg, R, S, Kpad, accumulate ? 1 : 0);
+}
+
 }  // namespace cosamd

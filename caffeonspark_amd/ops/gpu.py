@@ -287,7 +287,8 @@ def _wino_run(xl, w, b, y, N, P, Q, C, K, relu, flip=False):
 
 
 def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
-                    need_dx=True, need_dw=True, bias=True, ctx=None):
+                    need_dx=True, need_dw=True, bias=True, ctx=None,
+                    dw_out=None):
     _check_bf16(dy, "conv dy")
     if ctx is None or "col" not in ctx:
         ctx = ctx if ctx is not None else {}
@@ -361,8 +362,16 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                 _gemm(dyT[g * Kg:(g + 1) * Kg], colT, dwp[g * Kg:],
                       None, Kg, Kpad, NPQ, NPQ, NPQ, Kpad, False, False,
                       2, sk, ma=_pad128(Kout) - g * Kg, na=_pad128(Kpad))
-        dw = dwp[:, :Kcol].reshape(Kout, R, S, Cg).permute(0, 3, 1, 2) \
-            .contiguous()
+        if dw_out is not None and dw_out.dtype == torch.float32 \
+                and dw_out.is_contiguous() \
+                and tuple(dw_out.shape) == (Kout, Cg, R, S):
+            # unpack straight into the fp32 arena slice (one kernel
+            # instead of permute-contiguous + arena copy)
+            _ext.dw_unpack_acc(dwp, dw_out, Kout, Cg, R, S, Kpad, False)
+            dw = dw_out
+        else:
+            dw = dwp[:, :Kcol].reshape(Kout, R, S, Cg) \
+                .permute(0, 3, 1, 2).contiguous()
     if bias:
         db = torch.zeros(Kout, dtype=torch.float32, device=dy.device)
         _ext.colsum(dy2, db, NPQ, Kout, Kout)

@@ -24,7 +24,8 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
 
 
 def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
-                    need_dx=True, need_dw=True, bias=True, ctx=None):
+                    need_dx=True, need_dw=True, bias=True, ctx=None,
+                    dw_out=None):
     dx = dw = db = None
     if need_dx:
         dx = torch.nn.grad.conv2d_input(list(x.shape), w, dy, stride=stride,
@@ -34,6 +35,9 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         dw = torch.nn.grad.conv2d_weight(x, list(w.shape), dy, stride=stride,
                                          padding=pad, dilation=dilation,
                                          groups=groups)
+        if dw_out is not None:
+            dw_out.copy_(dw)
+            dw = dw_out
     if bias:
         db = dy.sum(dim=(0, 2, 3))
     return dx, dw, db
