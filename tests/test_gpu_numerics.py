@@ -634,3 +634,59 @@ def test_lstm_persistent_vs_loop_path(monkeypatch):
     torch.testing.assert_close(dxg_p, dxg_l, rtol=3e-2, atol=3e-2)
     rel = (dwhc_p - dwhc_l).norm() / dwhc_l.norm().clamp_min(1e-6)
     assert rel < 2e-2, f"dwhc relL2={rel}"
+
+
+def test_gpu_iter_size_accumulation():
+    """iter_size=2 gradient accumulation on the GPU path: the fused
+    dw-to-arena write (micro-batch 1, virgin overwrite) must compose
+    with the accumulate path (micro-batch 2) — equals iter_size=1 over
+    the concatenated batch within bf16 tolerance."""
+    from caffeonspark_amd.core.solver import Solver
+    from caffeonspark_amd.proto import caffe_pb, text_format
+
+    net_text = """
+      layer { name: "d" type: "MemoryData" top: "x" top: "t"
+              memory_data_param { batch_size: 16 channels: 8 height: 9
+                                  width: 9 } }
+      layer { name: "c" type: "Convolution" bottom: "x" top: "y"
+              convolution_param { num_output: 16 kernel_size: 3 pad: 1
+                weight_filler { type: "gaussian" std: 0.1 } } }
+      layer { name: "r" type: "ReLU" bottom: "y" top: "y" }
+      layer { name: "ip" type: "InnerProduct" bottom: "y" top: "z"
+              inner_product_param { num_output: 5
+                weight_filler { type: "gaussian" std: 0.1 } } }
+      layer { name: "l" type: "SoftmaxWithLoss" bottom: "z" bottom: "t"
+              top: "loss" }
+    """
+    torch.manual_seed(11)
+    xs = [torch.randn(16, 8, 9, 9).to(torch.bfloat16) for _ in range(2)]
+    ys = [torch.randint(0, 5, (16,)).float() for _ in range(2)]
+
+    def run(iter_size):
+        sp = caffe_pb.SolverParameter(
+            net_param=text_format.parse(net_text, caffe_pb.NetParameter),
+            base_lr=0.1, momentum=0.9, lr_policy="fixed", max_iter=4,
+            random_seed=3, display=0, iter_size=iter_size)
+        s = Solver(sp, device=dev(), dtype=torch.bfloat16)
+        dl = s.net.data_layers()[0]
+        if iter_size == 2:
+            feed = iter([(xs[0], ys[0]), (xs[1], ys[1])] * 4)
+            orig_forward = s.net.forward
+
+            def forward(*a, **k):
+                x, y = next(feed)
+                dl.reset(x.to(dev()), y.to(dev()))
+                return orig_forward(*a, **k)
+            s.net.forward = forward
+            s._step_one()
+        else:
+            dl.batch_size = 32
+            dl.reset(torch.cat(xs).to(dev()), torch.cat(ys).to(dev()))
+            s._step_one()
+        return s.flat_w.cpu()
+
+    w2 = run(2)
+    w1 = run(1)
+    # iter_size=2 averages two half-batch losses == full-batch mean
+    rel = (w2 - w1).norm() / w1.norm().clamp_min(1e-6)
+    assert rel < 5e-3, f"iter_size accumulation diverged: relL2={rel}"
