@@ -73,8 +73,15 @@ class Solver:
                 tn._bf16_refresh = self.net._bf16_refresh
             self.test_nets.append(tn)
 
-        self.history2 = [torch.zeros_like(b.data) for b in self.params] \
-            if self.type in ("Adam", "AdaDelta") else []
+        if self.type in ("Adam", "AdaDelta"):
+            # second-moment arena (views per blob): lets the fused
+            # whole-arena Adam kernel run like the SGD one
+            self.flat_m2 = torch.zeros_like(self.flat_m)
+            self.history2 = [self.flat_m2.narrow(0, o, b.count).view(b.shape)
+                             for o, b in zip(self.param_offsets, self.params)]
+        else:
+            self.flat_m2 = None
+            self.history2 = []
         self._losses: List[float] = []
         self.smoothed_loss = 0.0
 
@@ -330,6 +337,19 @@ class Solver:
                                self.flat_m.narrow(0, off, n),
                                rate * lrm, p.momentum, wd * dm)
             return
+        if reg == "L2" and self.device.type == "cuda":
+            # fused whole-arena Nesterov / Adam (round-2: non-SGD updates
+            # previously ran through torch glue per blob)
+            if self.type == "Nesterov":
+                fn = ops.gpu_op("nesterov_update_multi_arena")
+                if fn is not None and fn(self, rate, p.momentum, wd):
+                    return
+            elif self.type == "Adam":
+                fn = ops.gpu_op("adam_update_multi_arena")
+                if fn is not None and fn(self, rate, p.momentum,
+                                         p.momentum2, p.delta, wd,
+                                         self.iter + 1):
+                    return
         for i, b in enumerate(self.params):
             if b._lr_mult == 0 or b.diff is None:
                 continue

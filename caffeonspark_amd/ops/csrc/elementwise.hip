@@ -350,6 +350,92 @@ void sgd_update_multi(float* p, const float* g, float* v,
       p, g, v, seg_off, seg_lr, seg_wd, nseg, mu, total4);
 }
 
+// Nesterov momentum, same whole-arena structure (reference solver
+// family, SURVEY.md §2.5: caffe Nesterov p -= (1+mu)*v_new - mu*v_old)
+__global__ void nesterov_update_multi_kernel(
+    float* __restrict__ p, const float* __restrict__ g,
+    float* __restrict__ v, const int64_t* __restrict__ seg_off,
+    const float* __restrict__ seg_lr, const float* __restrict__ seg_wd,
+    int nseg, float mu, int64_t total4) {
+  __shared__ int64_t soff[COS_SGD_MAX_SEG + 1];
+  __shared__ float slr[COS_SGD_MAX_SEG], swd[COS_SGD_MAX_SEG];
+  for (int i = threadIdx.x; i <= nseg; i += blockDim.x) {
+    soff[i] = seg_off[i];
+    if (i < nseg) { slr[i] = seg_lr[i]; swd[i] = seg_wd[i]; }
+  }
+  __syncthreads();
+  int s = 0;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total4; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t e0 = i * 4;
+    while (s + 1 < nseg && e0 >= soff[s + 1]) ++s;
+    float lr = slr[s], wd = swd[s];
+    if (lr == 0.f) continue;
+    int64_t end = min(soff[s + 1], e0 + 4);
+    for (int64_t k = e0; k < end; ++k) {
+      float gg = g[k] + wd * p[k];
+      float vprev = v[k];
+      float vnew = mu * vprev + lr * gg;
+      v[k] = vnew;
+      p[k] -= (1.f + mu) * vnew - mu * vprev;
+    }
+  }
+}
+
+void nesterov_update_multi(float* p, const float* g, float* v,
+                           const int64_t* seg_off, const float* seg_lr,
+                           const float* seg_wd, int nseg, float mu,
+                           int64_t total, hipStream_t stream) {
+  int64_t total4 = (total + 3) / 4;
+  int blocks = (int)hmin<int64_t>(2048, (total4 + 255) / 256);
+  nesterov_update_multi_kernel<<<blocks, 256, 0, stream>>>(
+      p, g, v, seg_off, seg_lr, seg_wd, nseg, mu, total4);
+}
+
+// Adam, whole-arena: m/v moment arenas; the bias-correction factor is
+// folded into seg_lr host-side (it is global per step)
+__global__ void adam_update_multi_kernel(
+    float* __restrict__ p, const float* __restrict__ g,
+    float* __restrict__ m, float* __restrict__ v,
+    const int64_t* __restrict__ seg_off,
+    const float* __restrict__ seg_lr, const float* __restrict__ seg_wd,
+    int nseg, float b1, float b2, float eps, int64_t total4) {
+  __shared__ int64_t soff[COS_SGD_MAX_SEG + 1];
+  __shared__ float slr[COS_SGD_MAX_SEG], swd[COS_SGD_MAX_SEG];
+  for (int i = threadIdx.x; i <= nseg; i += blockDim.x) {
+    soff[i] = seg_off[i];
+    if (i < nseg) { slr[i] = seg_lr[i]; swd[i] = seg_wd[i]; }
+  }
+  __syncthreads();
+  int s = 0;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total4; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t e0 = i * 4;
+    while (s + 1 < nseg && e0 >= soff[s + 1]) ++s;
+    float lr = slr[s], wd = swd[s];
+    if (lr == 0.f) continue;
+    int64_t end = min(soff[s + 1], e0 + 4);
+    for (int64_t k = e0; k < end; ++k) {
+      float gg = g[k] + wd * p[k];
+      float mm = b1 * m[k] + (1.f - b1) * gg;
+      float vv = b2 * v[k] + (1.f - b2) * gg * gg;
+      m[k] = mm;
+      v[k] = vv;
+      p[k] -= lr * mm / (sqrtf(vv) + eps);
+    }
+  }
+}
+
+void adam_update_multi(float* p, const float* g, float* m, float* v,
+                       const int64_t* seg_off, const float* seg_lr,
+                       const float* seg_wd, int nseg, float b1, float b2,
+                       float eps, int64_t total, hipStream_t stream) {
+  int64_t total4 = (total + 3) / 4;
+  int blocks = (int)hmin<int64_t>(2048, (total4 + 255) / 256);
+  adam_update_multi_kernel<<<blocks, 256, 0, stream>>>(
+      p, g, m, v, seg_off, seg_lr, seg_wd, nseg, b1, b2, eps, total4);
+}
+
 // ---------------------------------------------------------- column reduce
 // out[c] += sum_r in[r*ld + c]  (bias gradients; in bf16, out fp32)
 // Block = 256 threads as [4 row-groups x 64 cols]: coalesced 64-wide column

@@ -83,6 +83,14 @@ void mul_bf16(const void* a, const void* b, void* y, int64_t n,
               hipStream_t stream);
 void sgd_update(float* p, const float* g, float* v, float lr, float mu,
                 float wd, int64_t n, hipStream_t stream);
+void nesterov_update_multi(float* p, const float* g, float* v,
+                           const int64_t* seg_off, const float* seg_lr,
+                           const float* seg_wd, int nseg, float mu,
+                           int64_t total, hipStream_t stream);
+void adam_update_multi(float* p, const float* g, float* m, float* v,
+                       const int64_t* seg_off, const float* seg_lr,
+                       const float* seg_wd, int nseg, float b1, float b2,
+                       float eps, int64_t total, hipStream_t stream);
 void sgd_update_multi(float* p, const float* g, float* v,
                       const int64_t* seg_off, const float* seg_lr,
                       const float* seg_wd, int nseg, float mu,
@@ -250,6 +258,31 @@ void py_sgd_update_multi(Tensor p, Tensor g, Tensor v, Tensor seg_off,
                            seg_wd.data_ptr<float>(),
                            (int)seg_lr.numel(), mu, p.numel(),
                            cur_stream());
+}
+
+void py_nesterov_update_multi(Tensor p, Tensor g, Tensor v, Tensor seg_off,
+                              Tensor seg_lr, Tensor seg_wd, double mu) {
+  CHECK_F32(p); CHECK_F32(g); CHECK_F32(v);
+  cosamd::nesterov_update_multi(p.data_ptr<float>(), g.data_ptr<float>(),
+                                v.data_ptr<float>(),
+                                seg_off.data_ptr<int64_t>(),
+                                seg_lr.data_ptr<float>(),
+                                seg_wd.data_ptr<float>(),
+                                (int)seg_lr.numel(), mu, p.numel(),
+                                cur_stream());
+}
+
+void py_adam_update_multi(Tensor p, Tensor g, Tensor m, Tensor v,
+                          Tensor seg_off, Tensor seg_lr, Tensor seg_wd,
+                          double b1, double b2, double eps) {
+  CHECK_F32(p); CHECK_F32(g); CHECK_F32(m); CHECK_F32(v);
+  cosamd::adam_update_multi(p.data_ptr<float>(), g.data_ptr<float>(),
+                            m.data_ptr<float>(), v.data_ptr<float>(),
+                            seg_off.data_ptr<int64_t>(),
+                            seg_lr.data_ptr<float>(),
+                            seg_wd.data_ptr<float>(),
+                            (int)seg_lr.numel(), b1, b2, eps, p.numel(),
+                            cur_stream());
 }
 
 void py_sgd_update(Tensor p, Tensor g, Tensor v, double lr, double mu,
@@ -469,6 +502,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mul", &py_mul);
   m.def("sgd_update", &py_sgd_update);
   m.def("sgd_update_multi", &py_sgd_update_multi);
+  m.def("nesterov_update_multi", &py_nesterov_update_multi);
+  m.def("adam_update_multi", &py_adam_update_multi);
   m.def("colsum", &py_colsum);
   m.def("transpose", &py_transpose);
   m.def("tr16_probe", &py_tr16_probe);
