@@ -27,8 +27,9 @@ from .transformer import DataTransformer, decode_image
 
 def _read_parquet_rows(path: str, columns=None) -> Iterator[dict]:
     import pyarrow.parquet as pq
-    if path.startswith("file:"):
-        path = path[5:]
+
+    from ..utils.fsio import ensure_local
+    path = ensure_local(path)
     files = [path]
     if os.path.isdir(path):
         files = sorted(glob.glob(os.path.join(path, "*.parquet"))) or \

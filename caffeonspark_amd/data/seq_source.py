@@ -13,9 +13,8 @@ from .seqfile import SequenceFileReader
 
 class SeqImageDataSource(ImageDataSource):
     def init(self) -> None:
-        path = self.source_path
-        if path.startswith("file:"):
-            path = path[5:]
+        from ..utils.fsio import ensure_local
+        path = ensure_local(self.source_path)
         if os.path.isdir(path):
             self.files = sorted(
                 f for f in glob.glob(os.path.join(path, "*"))
