@@ -182,16 +182,17 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
                                   dtype=torch.bfloat16, device=x.device)
                 w._cos_wrT = wrT
             _repack_register(shadow, wrb, wrT, Kout, Cg, R, S, Kpad)
-            w._cos_repacked = True
-    if getattr(w, "_cos_repack_epoch", -1) != _repack["epoch"] or \
-            not getattr(w, "_cos_repacked", False):
+            w._cos_register_epoch = _repack["epoch"]
+    # once a refresh has run AFTER registration, the fused kernel owns
+    # wrb/wrT; until then (first step, or unregistered weights) repack
+    # inline
+    reg_epoch = getattr(w, "_cos_register_epoch", None)
+    if reg_epoch is None or _repack["epoch"] <= reg_epoch:
         wrb[:Kout, :Kcol] = _as_bf16(w).permute(0, 2, 3, 1) \
             .reshape(Kout, Kcol)
         wrT = getattr(w, "_cos_wrT", None)
         if wrT is not None:
             wrT[:Kcol] = wrb[:Kout, :Kcol].t()
-        if getattr(w, "_cos_repacked", False):
-            w._cos_repack_epoch = _repack["epoch"]
     wr = wrb[:Kout]
     bias_f = b.float().contiguous() if b is not None else None
 
