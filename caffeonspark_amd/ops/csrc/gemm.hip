@@ -172,7 +172,7 @@ constexpr int WNT = 512;  // 8 waves: 2 (M) x 4 (N)
 __global__ __launch_bounds__(WNT, 1) void gemm_tt_wide_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     float* __restrict__ C, int M, int N, int K, int lda, int ldb, int ldc,
-    int ksplit, float alpha) {
+    int ksplit, float alpha, float* __restrict__ db) {
   __shared__ bf16 As[BM * BK];
   __shared__ bf16 Bs[WBN * BK];
 
@@ -191,12 +191,25 @@ __global__ __launch_bounds__(WNT, 1) void gemm_tt_wide_kernel(
 
   f32x4 acc[4][4] = {};
 
+  // fused bias gradient: db[m] += sum_k A[k][m] — the A tiles are in
+  // LDS anyway; the bn==0 blocks of each k-split cover every k exactly
+  // once (colsum was 4-5%% of the AlexNet/GoogLeNet steps as separate
+  // ramp-bound launches over the same dy operand)
+  bool do_db = (db != nullptr) && (bn == 0);
   for (int k0 = k_begin; k0 < k_end; k0 += BK) {
     stage_trans_pair_guarded(As, A, tile_m, M, lda, k0, k_end, tid, WNT,
                              BM);
     stage_trans_pair_guarded(Bs, B, tile_n, N, ldb, k0, k_end, tid, WNT,
                              WBN);
     __syncthreads();
+    if (do_db && tid < BM && tile_m + tid < M) {
+      float acc_b = 0.f;
+      const bf16* row = As + tid * BK;
+#pragma unroll
+      for (int kk = 0; kk < BK; ++kk)
+        acc_b += bf2f(row[(swz_chunk(tid, kk >> 3) << 3) | (kk & 7)]);
+      atomicAdd(db + tile_m + tid, acc_b);
+    }
     bf16x8 afrag[4], bfrag[4];
 #pragma unroll
     for (int f = 0; f < 4; ++f) {
@@ -509,7 +522,7 @@ void gemm_bf16_batched(const void* A_, const void* B_, void* C,
     dim3 gw(mblocks * nb_w, zblocks);
     gemm_tt_wide_kernel<<<gw, WNT, 0, stream>>>(
         A, B, reinterpret_cast<float*>(C), M, N, K, lda, ldb, ldc, ksplit,
-        alpha);
+        alpha, const_cast<float*>(bias));   // bias slot = fused db out
     return;
   }
   if (!trans_a && !trans_b) {

@@ -315,6 +315,16 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         x2 = ctx["xl"].permute(0, 2, 3, 1).reshape(NPQ, C)
 
     dx = dw = db = None
+    # fused bias gradient inside the TT-wide dw GEMM: measured SLOWER
+    # (googlenet 9.19k -> 8.73k, alexnet 38.9k -> 38.1k same-box A/B) —
+    # the bn==0 blocks doing the extra LDS column pass become the
+    # kernel's stragglers, costing more than the separate ramp-bound
+    # colsum launches saved.  Kept behind COS_FUSE_DB=1 for evidence.
+    fuse_db = bias and need_dw and Kpad > 128 and \
+        bool(int(os.environ.get("COS_DW_TT", "1"))) and \
+        bool(int(os.environ.get("COS_FUSE_DB", "0")))
+    if fuse_db:
+        db = torch.zeros(Kout, dtype=torch.float32, device=dy.device)
     if need_dw:
         # dw[kout][kpad] = sum_npq dy[npq][kout] * col[npq][kpad].
         # Both operands are K(=npq)-major; transposing them once (cheap
@@ -331,7 +341,8 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
             if use_tt:
                 src = x2 if is_1x1 else col[g]
                 sk_tt = _splitk_tt(Kg, Kpad, NPQ)
-                _gemm(dy2[:, g * Kg:], src, dwp[g * Kg:], None,
+                db_slice = db.narrow(0, g * Kg, Kg) if fuse_db else None
+                _gemm(dy2[:, g * Kg:], src, dwp[g * Kg:], db_slice,
                       Kg, Kpad, NPQ, Kout, Kpad if not is_1x1 else C,
                       Kpad, True, True, 2, sk_tt)
             else:
@@ -373,7 +384,7 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         else:
             dw = dwp[:, :Kcol].reshape(Kout, R, S, Cg) \
                 .permute(0, 3, 1, 2).contiguous()
-    if bias:
+    if bias and not fuse_db:
         db = torch.zeros(Kout, dtype=torch.float32, device=dy.device)
         _ext.colsum(dy2, db, NPQ, Kout, Kout)
     if need_dx:
@@ -491,10 +502,14 @@ def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None):
         dwp.zero_()
     else:
         dwp = torch.zeros((Nout, K), dtype=torch.float32, device=x.device)
-    _gemm(dy, x, dwp, None, Nout, K, M, Np, K, K, True, True, 2,
-          _splitk_tt(Nout, K, M))
+    fuse_db = bias and K > 128 and \
+        bool(int(os.environ.get("COS_FUSE_DB", "0")))
+    if fuse_db:
+        db = torch.zeros(Nout, dtype=torch.float32, device=x.device)
+    _gemm(dy, x, dwp, db if fuse_db else None, Nout, K, M, Np, K, K,
+          True, True, 2, _splitk_tt(Nout, K, M))
     dw = dwp
-    if bias:
+    if bias and not fuse_db:
         db = torch.zeros(Nout, dtype=torch.float32, device=x.device)
         _ext.colsum(dy, db, M, Nout, Np)
     return dx, dw, db
