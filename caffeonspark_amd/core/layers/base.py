@@ -102,7 +102,18 @@ class Layer:
 
     # -- helpers -------------------------------------------------------------
     def cast(self, t: torch.Tensor) -> torch.Tensor:
-        return t.to(self.dtype) if t.dtype != self.dtype else t
+        if t.dtype == self.dtype:
+            return t
+        if self.dtype == torch.bfloat16:
+            # solver-managed params carry a bf16 shadow-arena view
+            # (refreshed once per step): returning IT keeps the weight
+            # tensor identity stable across steps, so per-weight caches
+            # (packed GEMM layouts, fused repack registration) hit —
+            # a fresh .to() copy per forward defeated them all
+            sh = getattr(t, "_cos_bf16", None)
+            if sh is not None:
+                return sh
+        return t.to(self.dtype)
 
     def weight(self, i: int = 0) -> torch.Tensor:
         """Param i's data in compute dtype."""

@@ -115,8 +115,9 @@ class Solver:
                                             dtype=torch.bfloat16)
             self.flat_wb.copy_(self.flat_w)
             for b, o in zip(self.params, self.param_offsets):
-                b.data._cos_bf16 = self.flat_wb.narrow(
-                    0, o, b.count).view(b.shape)
+                view = self.flat_wb.narrow(0, o, b.count).view(b.shape)
+                view._cos_stable = True   # identity persists across steps
+                b.data._cos_bf16 = view
             def refresh():
                 self.flat_wb.copy_(self.flat_w)
                 # fused per-step conv weight repack (GEMM layouts follow

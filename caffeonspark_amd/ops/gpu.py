@@ -170,7 +170,9 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
         wrb = torch.zeros((_pad128(Kout), Kpad), dtype=torch.bfloat16,
                           device=x.device)
         w._cos_wrb = wrb
-        shadow = getattr(w, "_cos_bf16", None)
+        shadow = w if (w.dtype == torch.bfloat16 and
+                       getattr(w, "_cos_stable", False)) \
+            else getattr(w, "_cos_bf16", None)
         if shadow is not None and shadow.is_contiguous():
             # solver-managed weight: the fused per-step repack kernel
             # (refresh_packed_weights, run with the shadow refresh)
