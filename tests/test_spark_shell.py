@@ -195,3 +195,22 @@ def test_fixed_size_partitioner_and_union():
         assert all(rows == [1, 2] for _, rows in got)
     finally:
         sc.stop()
+
+
+def test_engine_task_failure_propagates():
+    """A failing task surfaces as a driver-side error with the executor
+    traceback (reference: CHECK-crash fail-stop semantics)."""
+    from caffeonspark_amd.spark.local import SparkContext
+
+    sc = SparkContext(master="local[2]")
+    try:
+        def boom(it):
+            raise ValueError("exec-side failure")
+            yield  # pragma: no cover
+
+        with pytest.raises(RuntimeError, match="exec-side failure"):
+            sc.parallelize(range(4), 2).mapPartitions(boom).collect()
+        # the pool survives a failed job
+        assert sc.parallelize([1, 2], 2).count() == 2
+    finally:
+        sc.stop()
