@@ -279,10 +279,19 @@ class SparkContext:
                 data = part
             ex.submit(fn_bytes, data, i)
             jobs.append(ex)
-        # receive in submission order per executor (FIFO pipes)
+        # receive in submission order per executor (FIFO pipes); drain
+        # every pending result even on failure so the pool stays usable
         results: List[List[Any]] = []
+        first_err: Optional[BaseException] = None
         for ex in jobs:
-            results.append(ex.recv())
+            try:
+                results.append(ex.recv())
+            except RuntimeError as e:
+                if first_err is None:
+                    first_err = e
+                results.append([])
+        if first_err is not None:
+            raise first_err
         return results
 
 
