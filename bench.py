@@ -101,7 +101,11 @@ def main():
     # ws=1 after the round-2 kernel work shifted the steps launch-bound
     # (cifar 106k -> 142k round 1; round 2: googlenet +2.2%, alexnet
     # +1.1%, lrcn +0.4%) — auto mode turns it on for single-process runs
-    use_graph = args.graph != 0
+    # fp32 runs the rocBLAS/MIOpen reference route: graph capture forces
+    # capture-safe MIOpen algorithms (measured 10.2k -> 2.9k img/s), so
+    # auto mode keeps fp32 eager
+    use_graph = args.graph == 1 or (args.graph == -1
+                                    and dtype != torch.float32)
     step = solver.graph_step if (use_graph and ws == 1 and use_gpu) \
         else solver._step_one
     for _ in range(args.warmup):

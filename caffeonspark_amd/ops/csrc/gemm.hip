@@ -122,45 +122,6 @@ __device__ __forceinline__ void stage_trans_pair_guarded(
   }
 }
 
-// legacy transposed staging into the canonical layout (scalar scatter;
-// kept for A/B comparison via COS_GEMM_TRANS_PAIR=0).
-__device__ __forceinline__ void stage_trans_guarded(
-    bf16* lds_, const bf16* g_, int row0, int rows, int ld,
-    int k0, int kend, int tid, int nthreads = NTHREADS,
-    int tile_rows = BM) {
-  auto* lds = reinterpret_cast<unsigned short*>(lds_);
-  auto* g = reinterpret_cast<const unsigned short*>(g_);
-  // slots: (k, r0): 32 k-rows x tile_rows/8 slots of 8 rows
-  int rslots = tile_rows >> 3;
-  int nslots = 32 * rslots;
-  for (int slot = tid; slot < nslots; slot += nthreads) {
-    int kk = slot / rslots;
-    int r0 = (slot % rslots) * 8;
-    int gk = k0 + kk;
-    unsigned short vals[8];
-    if (gk < kend && row0 + r0 + 8 <= rows && (ld | (row0 + r0)) % 8 == 0) {
-      short8v v = *reinterpret_cast<const short8v*>(
-          g + (int64_t)gk * ld + row0 + r0);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) vals[j] = v[j];
-    } else if (gk < kend) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        vals[j] = (row0 + r0 + j < rows)
-                      ? g[(int64_t)gk * ld + row0 + r0 + j] : 0;
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) vals[j] = 0;
-    }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int row = r0 + j;
-      int dst_k = (swz_chunk(row, kk >> 3) << 3) | (kk & 7);
-      lds[row * BK + dst_k] = vals[j];
-    }
-  }
-}
-
 // --------------------------------------------------- wide-N dw kernel
 // Specialized for the skinny trans/trans weight-gradient GEMMs
 // (dw[Kout][Kcol] = dy^T @ col, K = N*P*Q huge): BN=256 halves the

@@ -98,12 +98,18 @@ def _transpose(t: torch.Tensor) -> torch.Tensor:
 # ---- fused weight repack registry: one kernel per step refreshes every
 # conv's GEMM-ready layouts (wrb and, for G==1, its transpose) from the
 # bf16 shadow arena, replacing ~2 tiny launches per conv per step.
-# Items hold tensor refs so the device pointers in the table stay live.
+# Sources are held by WEAK reference so dead solvers' buffers drop out
+# of the table instead of leaking (tests construct many solvers); the
+# wrb/wrT buffers are kept alive by the weight tensors they cache on.
+import weakref as _weakref
+
 _repack = {"items": [], "table": None, "max": 0, "epoch": 0, "dirty": False}
 
 
 def _repack_register(src, wrb, wrT, Kout, Cg, R, S, Kpad):
-    _repack["items"].append((src, wrb, wrT, Kout, Cg, R, S, Kpad))
+    _repack["items"].append((_weakref.ref(src), _weakref.ref(wrb),
+                             None if wrT is None else _weakref.ref(wrT),
+                             Kout, Cg, R, S, Kpad))
     _repack["dirty"] = True
 
 
@@ -113,16 +119,28 @@ def refresh_packed_weights():
     if not items:
         _repack["epoch"] += 1
         return
+    live = [(it, it[0](), it[1]()) for it in items]
+    if any(src is None or wrb is None for _, src, wrb in live):
+        _repack["items"] = items = [it for it, src, wrb in live
+                                    if src is not None and wrb is not None]
+        live = [(it, s, w) for it, s, w in live
+                if s is not None and w is not None]
+        _repack["dirty"] = True
+    if not items:
+        _repack["epoch"] += 1
+        return
     if _repack["dirty"]:
         rows = []
         mx = 0
-        for (src, wrb, wrT, Kout, Cg, R, S, Kpad) in items:
+        for (it, src, wrb) in live:
+            wrT = None if it[2] is None else it[2]()
+            _, _, _, Kout, Cg, R, S, Kpad = it
             rows.append([src.data_ptr(), wrb.data_ptr(),
                          0 if wrT is None else wrT.data_ptr(),
                          Kout, Cg, R, S, Kpad])
             mx = max(mx, Kout * R * S * Cg)
         _repack["table"] = torch.tensor(rows, dtype=torch.int64).to(
-            items[0][1].device)
+            live[0][2].device)
         _repack["max"] = mx
         _repack["dirty"] = False
     _ext.repack_weights(_repack["table"], len(items), _repack["max"])
