@@ -214,3 +214,25 @@ def test_engine_task_failure_propagates():
         assert sc.parallelize([1, 2], 2).count() == 2
     finally:
         sc.stop()
+
+
+def test_sparkshell_test_aggregation(tmp_path):
+    """test() through the Spark driver: features over the validation
+    output blobs + VectorMean per column (CaffeOnSpark.scala:396-418)."""
+    sf = _prep_lenet(tmp_path, max_iter=10)
+    os.chdir(tmp_path)
+    from caffeonspark_amd.data.processor import CaffeProcessor
+    CaffeProcessor.reset_instance()
+    from caffeonspark_amd.api.config import Config
+    from caffeonspark_amd.spark import CaffeOnSpark, SparkContext
+
+    sc = SparkContext(master="local[2]")
+    try:
+        conf = Config(["-conf", sf, "-test", "-label", "label",
+                       "-clusterSize", "2"])
+        cos = CaffeOnSpark(sc, conf)
+        result = cos.test()
+        assert "accuracy" in result and "loss" in result
+        assert 0.0 <= result["accuracy"][0] <= 1.0
+    finally:
+        sc.stop()
