@@ -232,7 +232,12 @@ class CaffeOnSpark:
                 def feed(proc=proc, rank=rank):
                     for sample in proc.sources[0].sample_iter(rank, n,
                                                               epochs=1):
-                        if not proc.feed_queue(0, sample):
+                        # bounded queue: retry until accepted (a single
+                        # failed offer is just a momentarily full queue)
+                        while not proc.stop_flag.is_set():
+                            if proc.feed_queue(0, sample):
+                                break
+                        if proc.stop_flag.is_set():
                             break
                     proc.stop_feeding(0)
                 t = threading.Thread(target=feed, daemon=True)

@@ -236,3 +236,37 @@ def test_sparkshell_test_aggregation(tmp_path):
         assert 0.0 <= result["accuracy"][0] <= 1.0
     finally:
         sc.stop()
+
+
+def test_sparkshell_features_large_dataset(tmp_path):
+    """features over a dataset larger than the bounded source queue
+    (1024): the feeder must retry full-queue offers, not truncate."""
+    import sys as _sys
+    _sys.path.insert(0, os.path.join(ROOT, "tests"))
+    from test_e2e_pipeline import LENET_NET, SOLVER, make_synthetic_lmdb
+    d = str(tmp_path)
+    make_synthetic_lmdb(os.path.join(d, "train_lmdb"), 100, seed=1)
+    make_synthetic_lmdb(os.path.join(d, "test_lmdb"), 2600, seed=2)
+    with open(os.path.join(d, "lenet.prototxt"), "w") as f:
+        f.write(LENET_NET.format(train=os.path.join(d, "train_lmdb"),
+                                 test=os.path.join(d, "test_lmdb")))
+    sf = os.path.join(d, "solver.prototxt")
+    with open(sf, "w") as f:
+        f.write(SOLVER.format(net=os.path.join(d, "lenet.prototxt"),
+                              test_interval=0, max_iter=10,
+                              prefix=os.path.join(d, "lenet")))
+    os.chdir(d)
+    from caffeonspark_amd.data.processor import CaffeProcessor
+    CaffeProcessor.reset_instance()
+    from caffeonspark_amd.api.config import Config
+    from caffeonspark_amd.spark import CaffeOnSpark, SparkContext
+
+    sc = SparkContext(master="local[2]")
+    try:
+        conf = Config(["-conf", sf, "-features", "ip2",
+                       "-clusterSize", "2"])
+        cos = CaffeOnSpark(sc, conf)
+        df = cos.features(conf)
+        assert df.count() == 2600
+    finally:
+        sc.stop()
