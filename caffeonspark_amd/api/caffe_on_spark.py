@@ -34,10 +34,16 @@ class CaffeOnSpark:
         backend = {"rccl": "nccl", "nccl": "nccl", "gloo": "gloo",
                    "ethernet": "gloo"}.get(
                        (conf.connection or "").lower())
-        if backend == "nccl" and not torch.cuda.is_available():
-            backend = "gloo"           # CPU runs: RCCL needs devices
+        world = int(os.environ.get("WORLD_SIZE", "1"))
+        if backend == "nccl" and (
+                not torch.cuda.is_available()
+                or world > max(1, torch.cuda.device_count())):
+            # RCCL needs one device per rank: CPU runs, and oversubscribed
+            # single-node launches (more ranks than visible GPUs), fall
+            # back to gloo instead of hanging in communicator init
+            backend = "gloo"
         self.rank = init_distributed(backend)
-        self.world = int(os.environ.get("WORLD_SIZE", "1"))
+        self.world = world
 
     # ------------------------------------------------------------- training
     def _setup(self, sources: List[DataSource]) -> CaffeProcessor:

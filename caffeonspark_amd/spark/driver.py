@@ -98,7 +98,11 @@ class CaffeOnSpark:
                 proc = CaffeProcessor.current()
                 host, port = master.value
                 if n > 1 and not dist.is_initialized():
-                    backend = "nccl" if torch.cuda.is_available() else "gloo"
+                    # RCCL needs one device per rank; oversubscribed
+                    # launches (n ranks > visible GPUs) use gloo
+                    backend = ("nccl" if torch.cuda.is_available()
+                               and n <= torch.cuda.device_count()
+                               else "gloo")
                     dist.init_process_group(
                         backend, rank=rank, world_size=n,
                         init_method=f"tcp://{host}:{port}")

@@ -31,7 +31,9 @@ def _worker_features(rank, ws, port, workdir, q):
         out = os.path.join(workdir, "feat2r.json")
         conf = Config(["-conf", os.path.join(workdir, "mr_solver.prototxt"),
                        "-features", "ip2", "-label", "label",
-                       "-output", out, "-outputFormat", "json"])
+                       "-output", out, "-outputFormat", "json",
+                       "-connection", "gloo"])  # CPU semantics even on
+                                                # a GPU box (2 ranks)
         cos = CaffeOnSpark(conf)
         df = cos.features()
         q.put((rank, len(df), sorted(df["SampleID"]), os.path.exists(out)))
@@ -50,7 +52,7 @@ def _worker_validation(rank, ws, port, workdir, q):
         CaffeProcessor.reset_instance()
         os.chdir(workdir)
         conf = Config(["-conf", os.path.join(workdir, "mrv_solver.prototxt"),
-                       "-train"])
+                       "-train", "-connection", "gloo"])
         cos = CaffeOnSpark(conf)
         results = cos.train_with_validation()
         q.put((rank, results))
