@@ -275,6 +275,12 @@ class Net:
             top.diff = torch.full_like(top.data, w, dtype=torch.float32)
             top._loss_weight = w  # host-side copy: layers avoid a sync
         ops.set_active_gpu_mode(self._ops_mode)
+        if self._ops_mode == "bf16":
+            # reset the per-backward scratch sequence + one fused zero of
+            # every split-K/colsum accumulator (replaces ~2 FillFunctor
+            # launches per conv per step)
+            from ..ops import gpu as _gops
+            _gops.begin_backward()
         try:
             for i in range(len(self.layers) - 1, -1, -1):
                 if not self.layer_need_backward[i]:

@@ -130,6 +130,7 @@ __device__ __forceinline__ void stage_trans_pair_guarded(
 constexpr int WBN = 256;
 constexpr int WNT = 512;  // 8 waves: 2 (M) x 4 (N)
 
+template <int STORE>  // 1 = plain fp32 store (no split-K), 2 = atomic
 __global__ __launch_bounds__(WNT, 1) void gemm_tt_wide_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     float* __restrict__ C, int M, int N, int K, int lda, int ldb, int ldc,
@@ -202,7 +203,10 @@ __global__ __launch_bounds__(WNT, 1) void gemm_tt_wide_kernel(
       for (int r = 0; r < 4; ++r) {
         int row = crow0 + fm * 16 + r;
         if (row >= M) continue;
-        atomicAdd(C + (int64_t)row * ldc + col, acc[fm][fn][r] * alpha);
+        if (STORE == 1)
+          C[(int64_t)row * ldc + col] = acc[fm][fn][r] * alpha;
+        else
+          atomicAdd(C + (int64_t)row * ldc + col, acc[fm][fn][r] * alpha);
       }
     }
 }
@@ -478,12 +482,17 @@ void gemm_bf16_batched(const void* A_, const void* B_, void* C,
     else                      COS_GEMM_CASE(TA, TB, 2);                     \
   } while (0)
 
-  if (trans_a && trans_b && store_mode == 2 && N > 128 && batch == 1) {
+  if (trans_a && trans_b && store_mode >= 1 && N > 128 && batch == 1) {
     int nb_w = (N + WBN - 1) / WBN;
     dim3 gw(mblocks * nb_w, zblocks);
-    gemm_tt_wide_kernel<<<gw, WNT, 0, stream>>>(
-        A, B, reinterpret_cast<float*>(C), M, N, K, lda, ldb, ldc, ksplit,
-        alpha, const_cast<float*>(bias));   // bias slot = fused db out
+    if (store_mode == 1)   // single-split: exclusive tiles, plain store
+      gemm_tt_wide_kernel<1><<<gw, WNT, 0, stream>>>(
+          A, B, reinterpret_cast<float*>(C), M, N, K, lda, ldb, ldc,
+          ksplit, alpha, const_cast<float*>(bias));
+    else
+      gemm_tt_wide_kernel<2><<<gw, WNT, 0, stream>>>(
+          A, B, reinterpret_cast<float*>(C), M, N, K, lda, ldb, ldc,
+          ksplit, alpha, const_cast<float*>(bias));   // bias = fused db
     return;
   }
   if (!trans_a && !trans_b) {

@@ -147,6 +147,56 @@ def refresh_packed_weights():
     _repack["epoch"] += 1
 
 
+# ---- per-backward scratch: the split-K GEMMs and colsum accumulate with
+# fp32 atomics, so their outputs need zero-init every backward.  Fresh
+# torch.zeros per call costs one FillFunctor launch each (GoogLeNet:
+# ~140 tiny fills/step, 4-5% of the step).  Instead, buffers persist
+# across steps keyed by their call ORDER within the backward (the net
+# graph is fixed, so the sequence is deterministic), and ONE fused
+# _foreach_zero_ at backward start re-zeroes them all.  Buffers with
+# zero_each_step=False only need their padding zeroed once (padded-copy
+# staging: the live region is fully overwritten every call).
+_scratch = {"seq": 0, "bufs": [], "zero_list": [], "dirty": False,
+            "retired": []}
+
+
+def begin_backward():
+    """Reset the scratch sequence + fused-zero all registered buffers.
+    Called by Net.backward on the GPU path (captured into hipGraphs like
+    any other kernel, so replayed steps re-zero too)."""
+    _scratch["seq"] = 0
+    if _scratch["dirty"]:
+        _scratch["zero_list"] = [t for t, z in _scratch["bufs"] if z]
+        _scratch["dirty"] = False
+    if _scratch["zero_list"]:
+        torch._foreach_zero_(_scratch["zero_list"])
+
+
+def _scratch_buf(shape, dtype, device, zero_each_step=True):
+    """Next persistent scratch buffer in call order, allocated zeroed.
+    A shape/dtype/device mismatch (different net interleaved) reallocates
+    in place, keeping correctness at the cost of cache churn."""
+    i = _scratch["seq"]
+    _scratch["seq"] = i + 1
+    bufs = _scratch["bufs"]
+    if i < len(bufs):
+        t, z = bufs[i]
+        if t.shape == shape and t.dtype == dtype and t.device == device \
+                and z == zero_each_step:
+            return t
+        new = torch.zeros(shape, dtype=dtype, device=device)
+        # NEVER free a replaced buffer: a captured hipGraph may hold its
+        # address baked into kernel launches (replay runs no Python)
+        _scratch["retired"].append(t)
+        bufs[i] = (new, zero_each_step)
+        _scratch["dirty"] = True
+        return new
+    t = torch.zeros(shape, dtype=dtype, device=device)
+    bufs.append((t, zero_each_step))
+    _scratch["dirty"] = True
+    return t
+
+
 def _gemm(A, B, C, bias, M, N, K, lda, ldb, ldc, ta, tb, store, splitk,
           relu=False, alpha=1.0, ma=0, na=0):
     # staging-allocation bounds: number of rows safely readable past M/N
@@ -344,27 +394,32 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         bool(int(os.environ.get("COS_DW_TT", "1"))) and \
         bool(int(os.environ.get("COS_FUSE_DB", "0")))
     if fuse_db:
-        db = torch.zeros(Kout, dtype=torch.float32, device=dy.device)
+        db = _scratch_buf((Kout,), torch.float32, dy.device)
     if need_dw:
         # dw[kout][kpad] = sum_npq dy[npq][kout] * col[npq][kpad].
-        # Both operands are K(=npq)-major; transposing them once (cheap
-        # torch copy kernels) turns the GEMM into the fast NT direct/direct
-        # form with global_load_lds staging instead of LDS scatter staging.
-        dwp = torch.zeros((Kout, Kpad), dtype=torch.float32, device=dy.device)
         # fused trans/trans dw (u32 k-pair staged; round 2): reads dy2/col
         # exactly once instead of transpose kernels + an extra HBM pass
         use_tt = bool(int(os.environ.get("COS_DW_TT", "1")))
+        sk_tt = _splitk_tt(Kg, Kpad, NPQ) if use_tt else 0
+        if use_tt and sk_tt == 1:
+            # single split: every output element is written by exactly
+            # one block — plain store, no zero-init pass needed
+            dwp = torch.empty((Kout, Kpad), dtype=torch.float32,
+                              device=dy.device)
+            store_dw = 1
+        else:
+            dwp = _scratch_buf((Kout, Kpad), torch.float32, dy.device)
+            store_dw = 2
         dyT = None if use_tt else _transpose(dy2)
         for g in range(G):
             mb, nb = (Kg + 127) // 128, (Kpad + 127) // 128
             sk = _splitk_for(mb, nb, NPQ)
             if use_tt:
                 src = x2 if is_1x1 else col[g]
-                sk_tt = _splitk_tt(Kg, Kpad, NPQ)
                 db_slice = db.narrow(0, g * Kg, Kg) if fuse_db else None
                 _gemm(dy2[:, g * Kg:], src, dwp[g * Kg:], db_slice,
                       Kg, Kpad, NPQ, Kout, Kpad if not is_1x1 else C,
-                      Kpad, True, True, 2, sk_tt)
+                      Kpad, True, True, store_dw, sk_tt)
             else:
                 if is_1x1:
                     # inception: several 1x1 branch convs share one
@@ -405,7 +460,7 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
             dw = dwp[:, :Kcol].reshape(Kout, R, S, Cg) \
                 .permute(0, 3, 1, 2).contiguous()
     if bias and not fuse_db:
-        db = torch.zeros(Kout, dtype=torch.float32, device=dy.device)
+        db = _scratch_buf((Kout,), torch.float32, dy.device)
         _ext.colsum(dy2, db, NPQ, Kout, Kout)
     if need_dx:
         dx = torch.empty((N, C, H, W), dtype=torch.bfloat16,
@@ -489,10 +544,14 @@ def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None):
     Np = Nout
     if Nout % 8:
         Np = _pad8(Nout)
-        dy_p = torch.zeros((M, Np), dtype=torch.bfloat16, device=dy.device)
+        # persistent pad buffers: live region fully overwritten each
+        # call, pad rows/cols zeroed once at allocation and never touched
+        dy_p = _scratch_buf((M, Np), torch.bfloat16, dy.device,
+                            zero_each_step=False)
         dy_p[:, :Nout] = dy
         dy = dy_p
-        wb_p = torch.zeros((Np, K), dtype=torch.bfloat16, device=dy.device)
+        wb_p = _scratch_buf((Np, K), torch.bfloat16, dy.device,
+                            zero_each_step=False)
         wb_p[:Nout] = wb
         wb = wb_p
     if need_dx:
@@ -504,7 +563,7 @@ def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None):
         mb, nb = (M + 127) // 128, (K + 127) // 128
         sk = _splitk_for(mb, nb, Np)
         if sk > 1:
-            wsp = torch.zeros((M, K), dtype=torch.float32, device=x.device)
+            wsp = _scratch_buf((M, K), torch.float32, x.device)
             _gemm(dy, wT, wsp, None, M, K, Np, Np, Np, K, False,
                   False, 2, sk, na=_pad128(K))
             _ext.bias_act_cast(wsp, None, dx, False)
@@ -516,21 +575,32 @@ def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None):
     # solver's fp32 arena slice, when the layer owns the only gradient
     # write this step) receives the GEMM output directly, skipping the
     # dwp-to-arena copy (fc6 alone is 151 MB fp32)
-    if dw_out is not None and dw_out.dtype == torch.float32 \
-            and dw_out.is_contiguous() and tuple(dw_out.shape) == (Nout, K):
+    sk_tt = _splitk_tt(Nout, K, M)
+    arena_dw = dw_out is not None and dw_out.dtype == torch.float32 \
+        and dw_out.is_contiguous() and tuple(dw_out.shape) == (Nout, K)
+    if sk_tt == 1:
+        # plain store covers every element once — no zero-init pass
+        # (fc6's arena slice alone is 151 MB: zeroing it every step was
+        # a full extra HBM pass over the largest weight in the net)
+        dwp = dw_out if arena_dw else torch.empty(
+            (Nout, K), dtype=torch.float32, device=x.device)
+        store_dw = 1
+    elif arena_dw:
         dwp = dw_out
-        dwp.zero_()
+        dwp.zero_()     # arena slice: NOT scratch (iter_size accumulates)
+        store_dw = 2
     else:
-        dwp = torch.zeros((Nout, K), dtype=torch.float32, device=x.device)
+        dwp = _scratch_buf((Nout, K), torch.float32, x.device)
+        store_dw = 2
     fuse_db = bias and K > 128 and \
         bool(int(os.environ.get("COS_FUSE_DB", "0")))
     if fuse_db:
-        db = torch.zeros(Nout, dtype=torch.float32, device=x.device)
+        db = _scratch_buf((Nout,), torch.float32, x.device)
     _gemm(dy, x, dwp, db if fuse_db else None, Nout, K, M, Np, K, K,
-          True, True, 2, _splitk_tt(Nout, K, M))
+          True, True, store_dw, sk_tt)
     dw = dwp
     if bias and not fuse_db:
-        db = torch.zeros(Nout, dtype=torch.float32, device=x.device)
+        db = _scratch_buf((Nout,), torch.float32, x.device)
         _ext.colsum(dy, db, M, Nout, Np)
     return dx, dw, db
 
@@ -764,7 +834,9 @@ def embed_forward(idx, w, b=None):
 
 def embed_backward(idx, dy, vocab_size, bias=True):
     e = dy.shape[-1]
-    dw = torch.zeros(vocab_size, e, dtype=torch.float32, device=dy.device)
+    # atomic scatter-add target (LRCN: 8801x1000 fp32 = 35 MB zeroed
+    # per step — one fused pass via the backward scratch instead)
+    dw = _scratch_buf((vocab_size, e), torch.float32, dy.device)
     _ext.embed_bwd(idx.float().contiguous(),
                    _as_bf16(dy).contiguous(), dw, vocab_size)
     db = dy.float().reshape(-1, e).sum(0) if bias else None
@@ -911,9 +983,13 @@ def lstm_seq_backward(dy, w_hc, cache):
     # k-pair staging), no operand transposes
     dxg_flat = dxg.reshape(T * N, H4)
     h_in_flat = h_in.reshape(T * N, H)
-    dwhc = torch.zeros((H4, H), dtype=torch.float32, device=dev_)
+    sk_tt = _splitk_tt(H4, H, T * N)
+    if sk_tt == 1:
+        dwhc = torch.empty((H4, H), dtype=torch.float32, device=dev_)
+    else:
+        dwhc = _scratch_buf((H4, H), torch.float32, dev_)
     _gemm(dxg_flat, h_in_flat, dwhc, None, H4, H, T * N, H4, H, H,
-          True, True, 2, _splitk_tt(H4, H, T * N))
+          True, True, 1 if sk_tt == 1 else 2, sk_tt)
     return dxg, dwhc
 
 
@@ -968,8 +1044,8 @@ def bn_backward(xhat, dy, invstd, train):
     rows = N * H * W
     x2 = xl.permute(0, 2, 3, 1).reshape(rows, C)
     dy2 = dyl.permute(0, 2, 3, 1).reshape(rows, C)
-    s1 = torch.zeros(C, dtype=torch.float32, device=xhat.device)
-    s2 = torch.zeros(C, dtype=torch.float32, device=xhat.device)
+    s1 = _scratch_buf((C,), torch.float32, xhat.device)
+    s2 = _scratch_buf((C,), torch.float32, xhat.device)
     if train:
         _ext.bn_bwd_sums(dy2, x2, s1, s2, rows, C)
     dx = torch.empty((N, C, H, W), dtype=torch.bfloat16,
