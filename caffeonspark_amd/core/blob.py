@@ -18,7 +18,7 @@ from ..proto import caffe_pb
 
 class Blob:
     __slots__ = ("name", "data", "diff", "_lr_mult", "_decay_mult",
-                 "_loss_weight", "_grad_virgin")
+                 "_loss_weight", "_grad_virgin", "_diff_prezeroed")
 
     def __init__(self, shape: Sequence[int] = (), *, name: str = "",
                  dtype: torch.dtype = torch.float32,

@@ -131,6 +131,20 @@ class Layer:
         else:
             d.add_(grad.to(d.dtype))
 
+    def _bias_arena(self):
+        """The bias param's pre-zeroed fp32 arena diff view, when backward
+        kernels may atomic-accumulate into it directly (flag set by
+        Net.zero_param_diffs for 1-D params each step)."""
+        if len(self.blobs) < 2:
+            return None
+        b = self.blobs[1]
+        if getattr(b, "_diff_prezeroed", False):
+            d = b.diff
+            if d is not None and d.dtype == torch.float32 \
+                    and d.is_contiguous():
+                return d
+        return None
+
     @staticmethod
     def acc_blob_diff(blob: Blob, dx: torch.Tensor, inplace: bool) -> None:
         """Accumulate dx into blob.diff (replace when the layer ran in-place:

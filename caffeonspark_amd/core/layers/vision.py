@@ -92,17 +92,18 @@ class ConvolutionLayer(Layer):
             d = wb.ensure_diff()
             if d.dtype == torch.float32 and d.is_contiguous():
                 dw_out = d.view(wb.shape)
+        db_out = self._bias_arena() if need_db else None
         dx, dw, db = ops.conv2d_backward(
             x, w, dy, (self.sh, self.sw), (self.ph, self.pw),
             (self.dil, self.dil), self.groups,
             need_dx=propagate_down[0], need_dw=need_dw, bias=need_db,
-            ctx=getattr(self, "_ctx", None), dw_out=dw_out)
+            ctx=getattr(self, "_ctx", None), dw_out=dw_out, db_out=db_out)
         if dw is not None:
             if dw is dw_out:
                 wb._grad_virgin = False
             else:
                 self.acc_param_diff(0, dw)
-        if db is not None:
+        if db is not None and db is not db_out:
             self.acc_param_diff(1, db)
         if propagate_down[0]:
             self.acc_blob_diff(bottom[0], dx, False)
@@ -137,7 +138,7 @@ class InnerProductLayer(Layer):
         w = self.weight(0)
         if self.transpose:
             w = w.t()
-        b = self.cast(self.blobs[1].data) if self.bias_term else None
+        b = self.blobs[1].data if self.bias_term else None  # fp32 master
         y = ops.fc_forward(x, w, b, relu=getattr(self, "_fuse_relu", False))
         out_shape = list(bottom[0].shape[:self.axis]) + [self.num_output]
         top[0].data = y.reshape(out_shape)
@@ -159,17 +160,18 @@ class InnerProductLayer(Layer):
             d = wb.ensure_diff()
             if d.dtype == torch.float32 and d.is_contiguous():
                 dw_out = d.view(wb.shape)
+        need_db = self.bias_term and self.blobs[1]._lr_mult != 0
+        db_out = self._bias_arena() if need_db else None
         dx, dw, db = ops.fc_backward(x, w, dy,
                                      need_dx=propagate_down[0],
-                                     bias=self.bias_term and
-                                     self.blobs[1]._lr_mult != 0,
-                                     dw_out=dw_out)
+                                     bias=need_db, dw_out=dw_out,
+                                     db_out=db_out)
         if dw is not None and need_dw:
             if dw is dw_out:
                 wb._grad_virgin = False
             else:
                 self.acc_param_diff(0, dw.t() if self.transpose else dw)
-        if db is not None:
+        if db is not None and db is not db_out:
             self.acc_param_diff(1, db)
         if propagate_down[0]:
             self.acc_blob_diff(bottom[0], dx.reshape(bottom[0].data.shape), False)

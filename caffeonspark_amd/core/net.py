@@ -315,11 +315,24 @@ class Net:
         return out
 
     def zero_param_diffs(self) -> None:
-        # mark rather than zero: the first acc_param_diff this step
-        # copies instead of accumulating (param regions never written by
-        # backward stay zero from the arena's init-time fill)
+        # 2D+ params: mark rather than zero — the first acc_param_diff
+        # this step copies (or the GEMM plain-stores into the arena
+        # slice) instead of accumulating.  1-D params (biases): ONE fused
+        # zero of all their diffs, so backward kernels (colsum) can
+        # atomic-add straight into the arena with no staging copy.
+        vec = []
         for b in self.learnable_params():
-            b._grad_virgin = True
+            d = b.diff
+            if b.data.dim() == 1 and d is not None \
+                    and d.dtype == torch.float32 and d.is_contiguous():
+                vec.append(d)
+                b._grad_virgin = False
+                b._diff_prezeroed = True
+            else:
+                b._grad_virgin = True
+                b._diff_prezeroed = False
+        if vec:
+            torch._foreach_zero_(vec)
 
     # ----------------------------------------------------------------- access
     def blob_by_name(self, name: str) -> Blob:

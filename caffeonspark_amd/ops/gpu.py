@@ -359,7 +359,7 @@ def _wino_run(xl, w, b, y, N, P, Q, C, K, relu, flip=False):
 
 def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                     need_dx=True, need_dw=True, bias=True, ctx=None,
-                    dw_out=None):
+                    dw_out=None, db_out=None):
     _check_bf16(dy, "conv dy")
     if ctx is None or "col" not in ctx:
         ctx = ctx if ctx is not None else {}
@@ -394,7 +394,8 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         bool(int(os.environ.get("COS_DW_TT", "1"))) and \
         bool(int(os.environ.get("COS_FUSE_DB", "0")))
     if fuse_db:
-        db = _scratch_buf((Kout,), torch.float32, dy.device)
+        db = db_out if db_out is not None \
+            else _scratch_buf((Kout,), torch.float32, dy.device)
     if need_dw:
         # dw[kout][kpad] = sum_npq dy[npq][kout] * col[npq][kpad].
         # fused trans/trans dw (u32 k-pair staged; round 2): reads dy2/col
@@ -460,7 +461,10 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
             dw = dwp[:, :Kcol].reshape(Kout, R, S, Cg) \
                 .permute(0, 3, 1, 2).contiguous()
     if bias and not fuse_db:
-        db = _scratch_buf((Kout,), torch.float32, dy.device)
+        # db_out = the solver's pre-zeroed arena slice: colsum's atomics
+        # accumulate the bias gradient in place, no staging copy
+        db = db_out if db_out is not None \
+            else _scratch_buf((Kout,), torch.float32, dy.device)
         _ext.colsum(dy2, db, NPQ, Kout, Kout)
     if need_dx:
         dx = torch.empty((N, C, H, W), dtype=torch.bfloat16,
@@ -529,7 +533,8 @@ def fc_forward(x, w, b, relu=False):
     return y
 
 
-def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None):
+def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None,
+                db_out=None):
     _check_bf16(dy, "fc dy")
     x = x.contiguous()
     dy = dy.contiguous()
@@ -595,12 +600,14 @@ def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None):
     fuse_db = bias and K > 128 and \
         bool(int(os.environ.get("COS_FUSE_DB", "0")))
     if fuse_db:
-        db = _scratch_buf((Nout,), torch.float32, x.device)
+        db = db_out if db_out is not None \
+            else _scratch_buf((Nout,), torch.float32, x.device)
     _gemm(dy, x, dwp, db if fuse_db else None, Nout, K, M, Np, K, K,
           True, True, store_dw, sk_tt)
     dw = dwp
     if bias and not fuse_db:
-        db = _scratch_buf((Nout,), torch.float32, x.device)
+        db = db_out if db_out is not None \
+            else _scratch_buf((Nout,), torch.float32, x.device)
         _ext.colsum(dy, db, M, Nout, Np)
     return dx, dw, db
 

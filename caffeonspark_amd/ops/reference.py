@@ -18,6 +18,8 @@ import torch.nn.functional as F
 
 def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
                    relu=False):
+    if b is not None and b.dtype != x.dtype:
+        b = b.to(x.dtype)   # callers pass the fp32 master bias
     y = F.conv2d(x, w, b, stride=stride, padding=pad,
                  dilation=dilation, groups=groups)
     return F.relu(y) if relu else y
@@ -25,7 +27,7 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
 
 def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                     need_dx=True, need_dw=True, bias=True, ctx=None,
-                    dw_out=None):
+                    dw_out=None, db_out=None):  # db_out: GPU-path hint
     dx = dw = db = None
     if need_dx:
         dx = torch.nn.grad.conv2d_input(list(x.shape), w, dy, stride=stride,
@@ -47,11 +49,14 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
 
 def fc_forward(x, w, b, relu=False):
     """x: [M, K], w: [N, K] (caffe layout), b: [N] or None."""
+    if b is not None and b.dtype != x.dtype:
+        b = b.to(x.dtype)   # callers pass the fp32 master bias
     y = F.linear(x, w, b)
     return F.relu(y) if relu else y
 
 
-def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None):
+def fc_backward(x, w, dy, need_dx=True, bias=True, dw_out=None,
+                db_out=None):  # db_out: GPU-path hint, unused here
     dx = dy @ w if need_dx else None
     dw = dy.t() @ x
     if dw_out is not None:
