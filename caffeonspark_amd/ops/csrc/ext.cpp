@@ -16,6 +16,14 @@ void gemm_bf16(const void* A, const void* B, void* C, const float* bias,
                bool trans_a, bool trans_b, int store_mode, int splitk,
                bool relu, float alpha, int m_alloc, int n_alloc,
                hipStream_t stream);
+void gemm_conv_fwd(const void* X, const void* B, void* C,
+                   const float* bias, int M, int N, int K, int ldb,
+                   int ldc, bool relu, const int* geom,
+                   hipStream_t stream);
+void gemm_conv_dw(const void* A, const void* X, float* C, int M, int N,
+                  int K, int lda, int ldc, int store_mode, int splitk,
+                  float alpha, float* db, const int* geom,
+                  hipStream_t stream);
 void tr16_probe(float* out, int mode, hipStream_t stream);
 void repack_weights(const int64_t* table, int ndesc, int64_t max_total,
                     hipStream_t stream);
@@ -158,6 +166,44 @@ void py_gemm(Tensor A, Tensor B, Tensor C, c10::optional<Tensor> bias,
       A.data_ptr(), B.data_ptr(), C.data_ptr(), bptr, (int)M, (int)N, (int)K, (int)lda, (int)ldb, (int)ldc,
       trans_a, trans_b, (int)store_mode, (int)splitk, relu, (float)alpha,
       (int)m_alloc, (int)n_alloc, cur_stream());
+}
+
+void py_gemm_conv_fwd(Tensor X, Tensor B, Tensor C,
+                      c10::optional<Tensor> bias, int64_t M, int64_t N,
+                      int64_t K, int64_t ldb, int64_t ldc, bool relu,
+                      std::vector<int64_t> geom) {
+  CHECK_CUDA(X); CHECK_BF16(X); CHECK_BF16(B); CHECK_BF16(C);
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    CHECK_F32(bias.value());
+    bptr = bias.value().data_ptr<float>();
+  }
+  int g[14];
+  TORCH_CHECK(geom.size() == 14, "geom must have 14 ints");
+  for (int i = 0; i < 14; ++i) g[i] = (int)geom[i];
+  cosamd::gemm_conv_fwd(X.data_ptr(), B.data_ptr(), C.data_ptr(), bptr,
+                        (int)M, (int)N, (int)K, (int)ldb, (int)ldc, relu,
+                        g, cur_stream());
+}
+
+void py_gemm_conv_dw(Tensor A, Tensor X, Tensor C,
+                     c10::optional<Tensor> db, int64_t M, int64_t N,
+                     int64_t K, int64_t lda, int64_t ldc,
+                     int64_t store_mode, int64_t splitk, double alpha,
+                     std::vector<int64_t> geom) {
+  CHECK_CUDA(A); CHECK_BF16(A); CHECK_BF16(X); CHECK_F32(C);
+  float* dbp = nullptr;
+  if (db.has_value()) {
+    CHECK_F32(db.value());
+    dbp = db.value().data_ptr<float>();
+  }
+  int g[14];
+  TORCH_CHECK(geom.size() == 14, "geom must have 14 ints");
+  for (int i = 0; i < 14; ++i) g[i] = (int)geom[i];
+  cosamd::gemm_conv_dw(A.data_ptr(), X.data_ptr(), C.data_ptr<float>(),
+                       (int)M, (int)N, (int)K, (int)lda, (int)ldc,
+                       (int)store_mode, (int)splitk, (float)alpha, dbp,
+                       g, cur_stream());
 }
 
 void py_im2col(Tensor x, Tensor col, int64_t N, int64_t H, int64_t W,
@@ -435,6 +481,8 @@ void py_softmax_loss_bwd(Tensor prob, Tensor label, Tensor dx, double scale,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm", &py_gemm);
+  m.def("gemm_conv_fwd", &py_gemm_conv_fwd);
+  m.def("gemm_conv_dw", &py_gemm_conv_dw);
   m.def("im2col", &py_im2col);
   m.def("bn_stats", [](Tensor x, Tensor sum, Tensor sq, int64_t rows,
                        int64_t C) {

@@ -408,6 +408,320 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gemm_kernel(
   }
 }
 
+// ------------------------------------------ implicit-im2col conv GEMMs
+// The col matrix's 8-element k-runs (one (r,s) tap, 8 consecutive
+// channels) are contiguous channel spans of the NHWC input whenever
+// Cg % 8 == 0, so the GEMM stages its col operand straight from x —
+// zero-filling padding taps — and the col matrix (AlexNet conv2 alone:
+// ~900 MB written + read twice per step) never materializes.
+
+struct CGeom {
+  int H, W, C;                       // input NHWC dims (C = all channels)
+  int Q, PQ;                         // output width, P*Q
+  int sh, sw, ph, pw, dil, S;
+  int c0, Cg;                        // group channel window
+  int Kcol;                          // R*S*Cg (zero beyond)
+  // magic reciprocals: floor(n/d) == (n * m) >> 40 for n < 2^20
+  unsigned long long mPQ, mQ, mCg, mS;
+};
+
+__device__ __forceinline__ unsigned mdiv(unsigned n,
+                                         unsigned long long m) {
+  return (unsigned)(((unsigned long long)n * m) >> 40);
+}
+
+static inline unsigned long long magic40(unsigned d) {
+  return ((1ull << 40) + d - 1) / d;
+}
+
+// per-block output-pixel decode: rinfo[row] = {n*H*W, p*sh-ph, q*sw-pw}
+__device__ __forceinline__ void decode_rows(int* rinfo, const CGeom& gm,
+                                            int tile_m, int M, int rows,
+                                            int tid) {
+  for (int i = tid; i < rows; i += blockDim.x) {
+    int m = tile_m + i;
+    if (m >= M) m = M - 1;            // junk rows: valid addrs, masked out
+    unsigned n = mdiv(m, gm.mPQ);
+    unsigned pq = m - n * gm.PQ;
+    unsigned p = mdiv(pq, gm.mQ);
+    unsigned q = pq - p * gm.Q;
+    rinfo[i * 3] = (int)(n * gm.H * gm.W);
+    rinfo[i * 3 + 1] = (int)p * gm.sh - gm.ph;
+    rinfo[i * 3 + 2] = (int)q * gm.sw - gm.pw;
+  }
+}
+
+// A-operand staging for the forward conv GEMM: same LDS layout and DMA
+// placement as stage_direct_fast, but the per-lane source address is the
+// im2col gather; out-of-image (or k >= Kcol) chunks ds_write zeros
+// instead (the DMA is EXEC-masked, so their slots are ours to fill).
+template <int WAVES>
+__device__ __forceinline__ void stage_implicit_fast(
+    bf16* lds_, const bf16* X, const CGeom& gm, const int* rinfo,
+    int k0, int tid) {
+  auto* lds = reinterpret_cast<unsigned short*>(lds_);
+  int wave = tid >> 6, lane = tid & 63;
+  constexpr int CPW = 8 / WAVES;
+#pragma unroll
+  for (int it = 0; it < CPW; ++it) {
+    int chunk = wave * CPW + it;
+    int idx = chunk * 512 + lane * 8;
+    int row = idx >> 5;
+    int kk = swz_chunk(row, (idx & 31) >> 3) << 3;
+    int k = k0 + kk;
+    unsigned rs = mdiv(k, gm.mCg);
+    int c = k - rs * gm.Cg;
+    unsigned r = mdiv(rs, gm.mS);
+    int s = rs - r * gm.S;
+    int h = rinfo[row * 3 + 1] + (int)r * gm.dil;
+    int w = rinfo[row * 3 + 2] + s * gm.dil;
+    bool ok = (k < gm.Kcol) && (h >= 0) && (h < gm.H) && (w >= 0) &&
+              (w < gm.W);
+    auto* lp = (__attribute__((address_space(3))) unsigned int*)(
+        lds + chunk * 512);
+    if (ok) {
+      int64_t off = ((int64_t)rinfo[row * 3] +
+                     (int64_t)(h * gm.W + w)) * gm.C + gm.c0 + c;
+      auto* gp = (const __attribute__((address_space(1))) unsigned int*)(
+          X + off);
+      __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+    } else {
+      auto* zp = (__attribute__((address_space(3))) short8v*)(
+          lds + chunk * 512 + lane * 8);
+      *zp = short8v{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+}
+
+// B-operand staging for the dw trans/trans GEMM: canonical u32 k-pair
+// layout like stage_trans_pair_guarded, but each 8-column octet (one
+// (r,s) tap, 8 channels) is gathered from x for two adjacent output
+// pixels (the k-pair).
+__device__ __forceinline__ void stage_trans_pair_implicit(
+    bf16* lds_, const bf16* X, const CGeom& gm, int row0,
+    int k0, int kend, int tid, int nthreads, int tile_rows) {
+  auto* lds = reinterpret_cast<unsigned short*>(lds_);
+  int rslots = tile_rows >> 3;
+  int nslots = 16 * rslots;
+  for (int slot = tid; slot < nslots; slot += nthreads) {
+    int kp = slot & 15;
+    int m0 = (slot >> 4) * 8;
+    int gr = row0 + m0;                 // col-matrix column octet base
+    unsigned rs = mdiv(gr, gm.mCg);
+    int c = gr - rs * gm.Cg;
+    unsigned r = mdiv(rs, gm.mS);
+    int s = rs - r * gm.S;
+    bool col_ok = gr < gm.Kcol;         // octet fully in/out (Kcol%8==0)
+    int hr = (int)r * gm.dil - gm.ph;
+    int ws = s * gm.dil - gm.pw;
+    unsigned short va[8] = {}, vb[8] = {};
+#pragma unroll
+    for (int hh = 0; hh < 2; ++hh) {
+      unsigned short* v = hh ? vb : va;
+      int gk = k0 + kp * 2 + hh;        // output-pixel (npq) index
+      if (gk < kend && col_ok) {
+        unsigned n = mdiv(gk, gm.mPQ);
+        unsigned pq = gk - n * gm.PQ;
+        unsigned p = mdiv(pq, gm.mQ);
+        unsigned q = pq - p * gm.Q;
+        int hi = (int)p * gm.sh + hr;
+        int wi = (int)q * gm.sw + ws;
+        if (hi >= 0 && hi < gm.H && wi >= 0 && wi < gm.W) {
+          const unsigned short* src =
+              reinterpret_cast<const unsigned short*>(X) +
+              ((int64_t)(n * gm.H + hi) * gm.W + wi) * gm.C + gm.c0 + c;
+          *reinterpret_cast<short8v*>(v) =
+              *reinterpret_cast<const short8v*>(src);
+        }
+      }
+    }
+    int kk = kp * 2;
+    int chunk0 = kk >> 3, kin = kk & 7;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int row = m0 + j;
+      unsigned int packed = (unsigned int)va[j] |
+                            ((unsigned int)vb[j] << 16);
+      *reinterpret_cast<unsigned int*>(
+          lds + row * BK + (swz_chunk(row, chunk0) << 3) + kin) = packed;
+    }
+  }
+}
+
+// forward conv GEMM: C[NPQ, Kg] = im2col(x) @ Wr^T, bias+ReLU fused.
+// Always the pipelined double-buffered loop: the implicit A staging
+// zero-fills k >= Kcol, so K runs over Kpad (32-aligned by the host)
+// with no guarded edge cases, and Wr's padded rows/columns are zero.
+template <int WAVES>
+__global__ __launch_bounds__(WAVES * 64, 2) void gemm_conv_fwd_kernel(
+    const bf16* __restrict__ X, const bf16* __restrict__ B,
+    bf16* __restrict__ C, const float* __restrict__ bias,
+    int M, int N, int K, int ldb, int ldc, int relu, CGeom gm) {
+  __shared__ bf16 Asb[2][BM * BK];
+  __shared__ bf16 Bsb[2][BN * BK];
+  __shared__ int rinfo[BM * 3];
+
+  int mblocks = (M + BM - 1) / BM;
+  int nblocks = (N + BN - 1) / BN;
+  int bid = xcd_swizzle(blockIdx.x, mblocks * nblocks);
+  int bm = bid / nblocks, bn = bid % nblocks;
+  int tile_m = bm * BM, tile_n = bn * BN;
+
+  constexpr int NT = WAVES * 64;
+  constexpr int MROWS = BM / (WAVES / 2);
+  constexpr int MF = MROWS / 16;
+  int tid = threadIdx.x;
+  int wave = tid >> 6, lane = tid & 63;
+  int wm = wave >> 1, wn = wave & 1;
+  int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+
+  decode_rows(rinfo, gm, tile_m, M, BM, tid);
+  __syncthreads();
+
+  f32x4 acc[MF][4] = {};
+  int cur = 0;
+  stage_implicit_fast<WAVES>(Asb[0], X, gm, rinfo, 0, tid);
+  stage_direct_fast<WAVES>(Bsb[0], B, tile_n, ldb, 0, tid);
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    if (k0 + BK < K) {
+      stage_implicit_fast<WAVES>(Asb[cur ^ 1], X, gm, rinfo, k0 + BK,
+                                 tid);
+      stage_direct_fast<WAVES>(Bsb[cur ^ 1], B, tile_n, ldb, k0 + BK,
+                               tid);
+    }
+    if (WAVES == 4)
+      asm volatile("s_waitcnt vmcnt(4) lgkmcnt(0)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(2) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_barrier();
+    const bf16* Asp = Asb[cur];
+    const bf16* Bsp = Bsb[cur];
+    bf16x8 afrag[MF], bfrag[4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      int rb = wn * 64 + f * 16 + lrow;
+      bfrag[f] = *reinterpret_cast<const bf16x8*>(
+          Bsp + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
+    }
+#pragma unroll
+    for (int f = 0; f < MF; ++f) {
+      int ra = wm * MROWS + f * 16 + lrow;
+      afrag[f] = *reinterpret_cast<const bf16x8*>(
+          Asp + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int fm = 0; fm < MF; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
+  }
+
+  int crow0 = tile_m + wm * MROWS + ((lane >> 4) << 2);
+  int ccol0 = tile_n + wn * 64 + (lane & 15);
+#pragma unroll
+  for (int fm = 0; fm < MF; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int col = ccol0 + fn * 16;
+      if (col >= N) continue;
+      float badd = (bias != nullptr) ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow0 + fm * 16 + r;
+        if (row >= M) continue;
+        float v = acc[fm][fn][r] + badd;
+        if (relu && v < 0.f) v = 0.f;
+        C[(int64_t)row * ldc + col] = f2bf(v);
+      }
+    }
+  }
+}
+
+// dw trans/trans GEMM with the col operand gathered implicitly:
+// dw[Kg][Kcol] = dy^T @ im2col(x), optional fused db, split-K atomics
+// (STORE 2) or plain store (STORE 1, single split).
+template <int STORE>
+__global__ __launch_bounds__(WNT, 1) void gemm_conv_dw_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ X,
+    float* __restrict__ C, int M, int N, int K, int lda, int ldc,
+    int ksplit, float alpha, float* __restrict__ db, CGeom gm) {
+  __shared__ bf16 As[BM * BK];
+  __shared__ bf16 Bs[WBN * BK];
+
+  int mblocks = (M + BM - 1) / BM;
+  int nblocks = (N + WBN - 1) / WBN;
+  int bid = xcd_swizzle(blockIdx.x, mblocks * nblocks);
+  int bm = bid / nblocks, bn = bid % nblocks;
+  int tile_m = bm * BM, tile_n = bn * WBN;
+  int k_begin = blockIdx.y * ksplit;
+  int k_end = min(K, k_begin + ksplit);
+
+  int tid = threadIdx.x;
+  int wave = tid >> 6, lane = tid & 63;
+  int wm = wave >> 2, wn = wave & 3;
+  int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+
+  f32x4 acc[4][4] = {};
+  bool do_db = (db != nullptr) && (bn == 0);
+  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
+    stage_trans_pair_guarded(As, A, tile_m, M, lda, k0, k_end, tid, WNT,
+                             BM);
+    stage_trans_pair_implicit(Bs, X, gm, tile_n, k0, k_end, tid, WNT,
+                              WBN);
+    __syncthreads();
+    if (do_db && tid < BM && tile_m + tid < M) {
+      float acc_b = 0.f;
+      const bf16* row = As + tid * BK;
+#pragma unroll
+      for (int kk = 0; kk < BK; ++kk)
+        acc_b += bf2f(row[(swz_chunk(tid, kk >> 3) << 3) | (kk & 7)]);
+      atomicAdd(db + tile_m + tid, acc_b);
+    }
+    bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      int ra = wm * 64 + f * 16 + lrow;
+      int rb = wn * 64 + f * 16 + lrow;
+      afrag[f] = *reinterpret_cast<const bf16x8*>(
+          As + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
+      bfrag[f] = *reinterpret_cast<const bf16x8*>(
+          Bs + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
+    }
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
+    __syncthreads();
+  }
+
+  int crow0 = tile_m + wm * 64 + ((lane >> 4) << 2);
+  int ccol0 = tile_n + wn * 64 + (lane & 15);
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int col = ccol0 + fn * 16;
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow0 + fm * 16 + r;
+        if (row >= M) continue;
+        if (STORE == 1)
+          C[(int64_t)row * ldc + col] = acc[fm][fn][r] * alpha;
+        else
+          atomicAdd(C + (int64_t)row * ldc + col, acc[fm][fn][r] * alpha);
+      }
+    }
+}
+
 // ---- tr16 semantics probe: fills LDS with elem index, every lane does
 // one ds_read_b64_tr_b16 at its own address and dumps v[0..3] — verifies
 // the (addr + j*16) element pattern on hardware.
@@ -510,6 +824,64 @@ void gemm_bf16_batched(const void* A_, const void* B_, void* C,
 #undef COS_GEMM_SM
 #undef COS_GEMM_CASE
 #undef COS_GEMM_CASE8
+}
+
+static CGeom make_geom(const int* g) {
+  // g = [H, W, C, P, Q, sh, sw, ph, pw, dil, S, c0, Cg, Kcol]
+  CGeom gm;
+  gm.H = g[0]; gm.W = g[1]; gm.C = g[2];
+  gm.Q = g[4]; gm.PQ = g[3] * g[4];
+  gm.sh = g[5]; gm.sw = g[6]; gm.ph = g[7]; gm.pw = g[8];
+  gm.dil = g[9]; gm.S = g[10];
+  gm.c0 = g[11]; gm.Cg = g[12]; gm.Kcol = g[13];
+  gm.mPQ = magic40(gm.PQ);
+  gm.mQ = magic40(gm.Q);
+  gm.mCg = magic40(gm.Cg);
+  gm.mS = magic40(gm.S);
+  return gm;
+}
+
+void gemm_conv_fwd(const void* X, const void* B, void* C,
+                   const float* bias, int M, int N, int K, int ldb,
+                   int ldc, bool relu, const int* geom,
+                   hipStream_t stream) {
+  CGeom gm = make_geom(geom);
+  int mblocks = (M + BM - 1) / BM, nblocks = (N + BN - 1) / BN;
+  dim3 grid(mblocks * nblocks);
+  static const int w8_env = [] {
+    const char* e = getenv("COS_GEMM_W8");
+    return e ? (e[0] == '1' ? 1 : 0) : -1;
+  }();
+  bool w8 = (w8_env == -1) ? (M <= 256 || N <= 256) : (w8_env == 1);
+  if (w8)
+    gemm_conv_fwd_kernel<8><<<grid, dim3(512), 0, stream>>>(
+        (const bf16*)X, (const bf16*)B, (bf16*)C, bias, M, N, K, ldb,
+        ldc, relu ? 1 : 0, gm);
+  else
+    gemm_conv_fwd_kernel<4><<<grid, dim3(256), 0, stream>>>(
+        (const bf16*)X, (const bf16*)B, (bf16*)C, bias, M, N, K, ldb,
+        ldc, relu ? 1 : 0, gm);
+}
+
+void gemm_conv_dw(const void* A, const void* X, float* C, int M, int N,
+                  int K, int lda, int ldc, int store_mode, int splitk,
+                  float alpha, float* db, const int* geom,
+                  hipStream_t stream) {
+  CGeom gm = make_geom(geom);
+  splitk = max(1, splitk);
+  int ksplit = (K + splitk - 1) / splitk;
+  ksplit = ((ksplit + BK - 1) / BK) * BK;
+  int zblocks = (K + ksplit - 1) / ksplit;
+  int mblocks = (M + BM - 1) / BM, nblocks = (N + WBN - 1) / WBN;
+  dim3 grid(mblocks * nblocks, zblocks);
+  if (store_mode == 1 && zblocks == 1)
+    gemm_conv_dw_kernel<1><<<grid, WNT, 0, stream>>>(
+        (const bf16*)A, (const bf16*)X, C, M, N, K, lda, ldc, ksplit,
+        alpha, db, gm);
+  else
+    gemm_conv_dw_kernel<2><<<grid, WNT, 0, stream>>>(
+        (const bf16*)A, (const bf16*)X, C, M, N, K, lda, ldc, ksplit,
+        alpha, db, gm);
 }
 
 void gemm_bf16(const void* A_, const void* B_, void* C, const float* bias,

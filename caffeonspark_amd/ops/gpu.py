@@ -24,6 +24,10 @@ def _pad8(k: int) -> int:
     return (k + 7) // 8 * 8
 
 
+def _pad32(k: int) -> int:
+    return (k + 31) // 32 * 32
+
+
 def _cl(x: torch.Tensor) -> torch.Tensor:
     """channels_last contiguous form of a 4-D activation; the converted
     copy is cached on the tensor (keyed by torch's in-place version
@@ -226,16 +230,29 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
     G = groups
     Kg = Kout // G
     Kcol = R * S * Cg
-    Kpad = _pad8(Kcol)
+    # implicit-im2col GEMM (round 2): stage the col operand straight
+    # from the NHWC input inside the GEMM — the col matrix is never
+    # materialized.  Needs 8-aligned channel octets (one (r,s) tap per
+    # 16-byte load) and npq/k decode within the magic-divide bound.
+    implicit = (C % 8 == 0 and Cg % 8 == 0 and not (R == S == 1)
+                and N * P * Q < (1 << 20) and R * S * Cg < (1 << 20)
+                and bool(int(os.environ.get("COS_IMPLICIT", "1"))))
+    # 32-aligned K keeps every k-tile on the pipelined fast path (and
+    # bounds the implicit kernel's B reads); pad columns stay zero
+    Kpad = _pad32(Kcol) if implicit else _pad8(Kcol)
 
     xl = _cl(x)
     # weight repack: [K, Cg, R, S] -> bf16 [K, R, S, Cg] padded to Kpad
     # columns and 128-aligned rows (GEMM fast-staging bound); the padded
     # buffer is cached on the weight tensor (pad region stays zero),
     # only the permute-copy runs per step
+    # row over-allocation: the implicit forward kernel fast-stages full
+    # 128-row B tiles per group slice, so the last group's tile span
+    # must stay inside the buffer (extra rows are zero -> contribute 0)
+    rows_w = max(_pad128(Kout), (G - 1) * Kg + _pad128(Kg))
     wrb = getattr(w, "_cos_wrb", None)
-    if wrb is None or wrb.shape != (_pad128(Kout), Kpad):
-        wrb = torch.zeros((_pad128(Kout), Kpad), dtype=torch.bfloat16,
+    if wrb is None or wrb.shape != (rows_w, Kpad):
+        wrb = torch.zeros((rows_w, Kpad), dtype=torch.bfloat16,
                           device=x.device)
         w._cos_wrb = wrb
         shadow = w if (w.dtype == torch.bfloat16 and
@@ -313,6 +330,15 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
         _gemm(x2, wr, y2, bias_f, NPQ, Kout, C, C, C, ldc_out,
               False, False, 0, 1, relu=relu, na=wrb.shape[0])
         col = None
+    elif implicit:
+        col = None
+        for g in range(G):
+            geom = [H, W, C, P, Q, sh, sw, ph, pw, dil, S, g * Cg, Cg,
+                    Kcol]
+            _ext.gemm_conv_fwd(
+                xl, wrb[g * Kg:], y2[:, g * Kg:],
+                bias_f[g * Kg:(g + 1) * Kg] if bias_f is not None
+                else None, NPQ, Kg, Kpad, Kpad, ldc_out, relu, geom)
     else:
         col = torch.empty((G, NPQ, Kpad), dtype=torch.bfloat16,
                           device=x.device)
@@ -328,6 +354,7 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
         ctx["col"] = col
         ctx["xl"] = xl
         ctx["is_1x1"] = is_1x1
+        ctx["implicit"] = implicit and not is_1x1
         ctx["shape"] = (N, C, H, W, P, Q, R, S, sh, sw, ph, pw, dil, G, Cg,
                         Kg, Kpad, Kcol)
         ctx["wr"] = wr
@@ -370,12 +397,17 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
     col, wr = ctx["col"], ctx["wr"]
     is_1x1 = ctx.get("is_1x1", False)
     wino = ctx.get("wino", False)
-    if wino and need_dw and col is None:
-        # forward skipped im2col; dW still wants the col matrix
+    implicit = ctx.get("implicit", False)
+    # implicit dw needs the wide TT kernel (N > 128); otherwise — and
+    # for Winograd forward — materialize the col matrix lazily
+    dw_implicit = implicit and Kpad > 128 and \
+        bool(int(os.environ.get("COS_DW_TT", "1")))
+    if need_dw and col is None and not is_1x1 and not dw_implicit:
         col = torch.empty((G, N * P * Q, Kpad), dtype=torch.bfloat16,
                           device=dy.device)
-        _ext.im2col(ctx["xl"], col[0], N, H, W, C, P, Q, R, S, sh, sw,
-                    ph, pw, dil, Kpad, 0, Cg)
+        for g in range(G):
+            _ext.im2col(ctx["xl"], col[g], N, H, W, C, P, Q, R, S, sh, sw,
+                        ph, pw, dil, Kpad, g * Cg, Cg)
         ctx["col"] = col
     Kout = Kg * G
     NPQ = N * P * Q
@@ -415,7 +447,15 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         for g in range(G):
             mb, nb = (Kg + 127) // 128, (Kpad + 127) // 128
             sk = _splitk_for(mb, nb, NPQ)
-            if use_tt:
+            if use_tt and dw_implicit:
+                # B operand gathered straight from the NHWC input
+                db_slice = db.narrow(0, g * Kg, Kg) if fuse_db else None
+                geom = [H, W, C, P, Q, sh, sw, ph, pw, dil, S, g * Cg,
+                        Cg, Kcol]
+                _ext.gemm_conv_dw(dy2[:, g * Kg:], ctx["xl"],
+                                  dwp[g * Kg:], db_slice, Kg, Kpad, NPQ,
+                                  Kout, Kpad, store_dw, sk_tt, 1.0, geom)
+            elif use_tt:
                 src = x2 if is_1x1 else col[g]
                 db_slice = db.narrow(0, g * Kg, Kg) if fuse_db else None
                 _gemm(dy2[:, g * Kg:], src, dwp[g * Kg:], db_slice,
