@@ -9,7 +9,10 @@
 //                 [k][(r*S+s)*Cg + c]; pad columns stay zero)
 //                 dstT [>=Kpad][Kout]   (optional, 0 = skip: the dx
 //                 GEMM's transposed operand)
-// Table rows: [src, dst, dstT, Kout, Cg, R, S, Kpad] as int64.
+//                 dstF [>=Cg][K2p]      (optional, 0 = skip: flipped
+//                 layout [c][(RS-1-rs)*Kout + k] for the implicit-dx
+//                 GEMM; pad columns stay zero)
+// Table rows: [src, dst, dstT, Kout, Cg, R, S, Kpad, dstF, K2p].
 
 #include "common.h"
 
@@ -19,12 +22,14 @@ typedef unsigned short u16;
 
 __global__ void repack_weights_kernel(const int64_t* __restrict__ table,
                                       int ndesc) {
-  const int64_t* e = table + (int64_t)blockIdx.y * 8;
+  const int64_t* e = table + (int64_t)blockIdx.y * 10;
   const u16* src = reinterpret_cast<const u16*>(e[0]);
   u16* dst = reinterpret_cast<u16*>(e[1]);
   u16* dstT = reinterpret_cast<u16*>(e[2]);
   int Kout = (int)e[3], Cg = (int)e[4], R = (int)e[5], S = (int)e[6];
   int Kpad = (int)e[7];
+  u16* dstF = reinterpret_cast<u16*>(e[8]);
+  int K2p = (int)e[9];
   int RS = R * S;
   int Kcol = RS * Cg;
   int64_t total = (int64_t)Kout * Kcol;
@@ -36,6 +41,8 @@ __global__ void repack_weights_kernel(const int64_t* __restrict__ table,
     u16 v = src[((int64_t)k * Cg + c) * RS + rs];
     dst[(int64_t)k * Kpad + col] = v;
     if (dstT != nullptr) dstT[(int64_t)col * Kout + k] = v;
+    if (dstF != nullptr)
+      dstF[(int64_t)c * K2p + (RS - 1 - rs) * Kout + k] = v;
   }
 }
 
