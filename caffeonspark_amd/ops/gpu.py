@@ -537,9 +537,13 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                          device=dy.device,
                          memory_format=torch.channels_last)
         if is_1x1:
-            # col2im is the identity: write dx's NHWC alias directly
+            # col2im is the identity: write dx's NHWC alias directly.
+            # wrT is maintained per step by the fused repack kernel for
+            # solver-managed weights; transpose inline otherwise.
             dx2 = dx.permute(0, 2, 3, 1).reshape(NPQ, C)
-            wrT = _transpose(wr.contiguous())
+            wrT = getattr(ctx.get("w_ref"), "_cos_wrT", None)
+            if wrT is None or wrT.shape[1] != Kout:
+                wrT = _transpose(wr.contiguous())
             _gemm(dy2, wrT, dx2, None, NPQ, C, Kout, Kout, Kout, C,
                   False, False, 0, 1, na=_pad128(C))
             return dx, dw, db
