@@ -9,10 +9,11 @@
 //                 [k][(r*S+s)*Cg + c]; pad columns stay zero)
 //                 dstT [>=Kpad][Kout]   (optional, 0 = skip: the dx
 //                 GEMM's transposed operand)
-//                 dstF [>=Cg][K2p]      (optional, 0 = skip: flipped
-//                 layout [c][(RS-1-rs)*Kout + k] for the implicit-dx
-//                 GEMM; pad columns stay zero)
-// Table rows: [src, dst, dstT, Kout, Cg, R, S, Kpad, dstF, K2p].
+//                 dstF [G][>=Cg][K2p]   (optional, 0 = skip: flipped
+//                 per-group layout [g][c][(RS-1-rs)*Kg + k%Kg] for the
+//                 implicit-dx GEMM; pad columns stay zero)
+// Table rows: [src, dst, dstT, Kout, Cg, R, S, Kpad, dstF, K2p, G, Cgp]
+// (Cgp = the padded row count of one dstF group slab).
 
 #include "common.h"
 
@@ -22,7 +23,7 @@ typedef unsigned short u16;
 
 __global__ void repack_weights_kernel(const int64_t* __restrict__ table,
                                       int ndesc) {
-  const int64_t* e = table + (int64_t)blockIdx.y * 10;
+  const int64_t* e = table + (int64_t)blockIdx.y * 12;
   const u16* src = reinterpret_cast<const u16*>(e[0]);
   u16* dst = reinterpret_cast<u16*>(e[1]);
   u16* dstT = reinterpret_cast<u16*>(e[2]);
@@ -30,6 +31,8 @@ __global__ void repack_weights_kernel(const int64_t* __restrict__ table,
   int Kpad = (int)e[7];
   u16* dstF = reinterpret_cast<u16*>(e[8]);
   int K2p = (int)e[9];
+  int G = (int)e[10], Cgp = (int)e[11];
+  int Kg = G > 0 ? Kout / G : Kout;
   int RS = R * S;
   int Kcol = RS * Cg;
   int64_t total = (int64_t)Kout * Kcol;
@@ -41,8 +44,11 @@ __global__ void repack_weights_kernel(const int64_t* __restrict__ table,
     u16 v = src[((int64_t)k * Cg + c) * RS + rs];
     dst[(int64_t)k * Kpad + col] = v;
     if (dstT != nullptr) dstT[(int64_t)col * Kout + k] = v;
-    if (dstF != nullptr)
-      dstF[(int64_t)c * K2p + (RS - 1 - rs) * Kout + k] = v;
+    if (dstF != nullptr) {
+      int g = k / Kg;
+      dstF[((int64_t)g * Cgp + c) * K2p +
+           (RS - 1 - rs) * Kg + (k - g * Kg)] = v;
+    }
   }
 }
 

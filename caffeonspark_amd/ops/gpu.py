@@ -111,12 +111,12 @@ _repack = {"items": [], "table": None, "max": 0, "epoch": 0, "dirty": False}
 
 
 def _repack_register(src, wrb, wrT, Kout, Cg, R, S, Kpad, wrF=None,
-                     k2p=0):
+                     k2p=0, G=1, Cgp=0):
     _repack["items"].append((_weakref.ref(src), _weakref.ref(wrb),
                              None if wrT is None else _weakref.ref(wrT),
                              Kout, Cg, R, S, Kpad,
                              None if wrF is None else _weakref.ref(wrF),
-                             k2p))
+                             k2p, G, Cgp))
     _repack["dirty"] = True
 
 
@@ -146,7 +146,8 @@ def refresh_packed_weights():
             rows.append([src.data_ptr(), wrb.data_ptr(),
                          0 if wrT is None else wrT.data_ptr(),
                          Kout, Cg, R, S, Kpad,
-                         0 if wrF is None else wrF.data_ptr(), it[9]])
+                         0 if wrF is None else wrF.data_ptr(), it[9],
+                         it[10], it[11]])
             mx = max(mx, Kout * R * S * Cg)
         _repack["table"] = torch.tensor(rows, dtype=torch.int64).to(
             live[0][2].device)
@@ -266,17 +267,17 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
         # implicit-dx eligibility: stride-1 G==1 convs compute the data
         # gradient as ONE implicit GEMM over dy with the flipped-weight
         # layout wrF[c][(RS-1-rs)*Kout + k] (no dcol, no col2im)
-        k2 = R * S * Kout
-        dx_imp = (implicit and G == 1 and sh == sw == 1
+        k2 = R * S * Kg
+        dx_imp = (implicit and sh == sw == 1
                   and bool(int(os.environ.get("COS_DX_IMP", "1")))
-                  and Kout % 8 == 0 and dil * (R - 1) - ph >= 0
+                  and Kg % 8 == 0 and dil * (R - 1) - ph >= 0
                   and dil * (S - 1) - pw >= 0 and N * H * W < (1 << 20)
                   and k2 < (1 << 20))
         wrF = None
         k2p = 0
         if dx_imp:
             k2p = _pad32(k2)
-            wrF = torch.zeros((_pad128(C), k2p),
+            wrF = torch.zeros((G * _pad128(Cg), k2p),
                               dtype=torch.bfloat16, device=x.device)
             w._cos_wrF = wrF
         if shadow is not None and shadow.is_contiguous():
@@ -290,7 +291,7 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
                                   dtype=torch.bfloat16, device=x.device)
                 w._cos_wrT = wrT
             _repack_register(shadow, wrb, wrT, Kout, Cg, R, S, Kpad,
-                             wrF, k2p)
+                             wrF, k2p, G, _pad128(Cg))
             w._cos_register_epoch = _repack["epoch"]
     # once a refresh has run AFTER registration, the fused kernel owns
     # wrb/wrT; until then (first step, or unregistered weights) repack
@@ -304,8 +305,12 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
             wrT[:Kcol] = wrb[:Kout, :Kcol].t()
         wrF = getattr(w, "_cos_wrF", None)
         if wrF is not None:
-            wrF[:C, :R * S * Kout] = _as_bf16(w).flip(2, 3) \
-                .permute(1, 2, 3, 0).reshape(C, R * S * Kout)
+            wbf = _as_bf16(w).flip(2, 3)
+            cgp = wrF.shape[0] // G
+            for g_ in range(G):
+                wrF[g_ * cgp:g_ * cgp + Cg, :R * S * Kg] = \
+                    wbf[g_ * Kg:(g_ + 1) * Kg].permute(1, 2, 3, 0) \
+                    .reshape(Cg, R * S * Kg)
     wr = wrb[:Kout]
     bias_f = b.float().contiguous() if b is not None else None
 
@@ -553,18 +558,21 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                       relu=False, flip=True)
             return dx, dw, db
         wrF = getattr(ctx.get("w_ref"), "_cos_wrF", None)
-        k2 = R * S * Kout
-        if wrF is not None and wrF.shape == (_pad128(C), _pad32(k2)) \
+        k2 = R * S * Kg
+        cgp = _pad128(Cg)
+        if wrF is not None and wrF.shape == (G * cgp, _pad32(k2)) \
                 and N * H * W < (1 << 20):
             # implicit dx: the data gradient IS a stride-1 convolution of
-            # dy with the flipped weights — one GEMM writes dx's NHWC
-            # alias directly (no dcol buffer, no col2im pass)
+            # dy with the flipped weights — one GEMM per group writes
+            # dx's NHWC alias directly (no dcol buffer, no col2im pass)
             dx2 = dx.permute(0, 2, 3, 1).reshape(N * H * W, C)
-            geom = [P, Q, Kout, H, W, 1, 1,
-                    dil * (R - 1) - ph, dil * (S - 1) - pw, dil, S,
-                    0, Kout, k2]
-            _ext.gemm_conv_fwd(dyl, wrF, dx2, None, N * H * W, C,
-                               _pad32(k2), _pad32(k2), C, False, geom)
+            for g in range(G):
+                geom = [P, Q, Kout, H, W, 1, 1,
+                        dil * (R - 1) - ph, dil * (S - 1) - pw, dil, S,
+                        g * Kg, Kg, k2]
+                _ext.gemm_conv_fwd(dyl, wrF[g * cgp:], dx2[:, g * Cg:],
+                                   None, N * H * W, Cg, _pad32(k2),
+                                   _pad32(k2), C, False, geom)
             return dx, dw, db
         dcol = torch.empty((NPQ, Kpad), dtype=torch.bfloat16,
                            device=dy.device)
