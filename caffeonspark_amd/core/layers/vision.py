@@ -62,7 +62,7 @@ class ConvolutionLayer(Layer):
     def forward(self, bottom, top):
         x = bottom[0].data
         w = self.weight(0)
-        b = self.cast(self.blobs[1].data) if self.bias_term else None
+        b = self.blobs[1].data if self.bias_term else None  # fp32 master
         self._ctx = {}
         kw = {}
         co = getattr(self, "_concat_out", None)
@@ -86,6 +86,8 @@ class ConvolutionLayer(Layer):
         w = self.weight(0)
         need_dw = self.blobs[0]._lr_mult != 0
         need_db = self.bias_term and self.blobs[1]._lr_mult != 0
+        if self.__dict__.pop("_db_done_step", False):
+            need_db = False      # fused into the ReLU backward pass
         wb = self.blobs[0]
         dw_out = None
         if need_dw and getattr(wb, "_grad_virgin", False):
@@ -161,6 +163,8 @@ class InnerProductLayer(Layer):
             if d.dtype == torch.float32 and d.is_contiguous():
                 dw_out = d.view(wb.shape)
         need_db = self.bias_term and self.blobs[1]._lr_mult != 0
+        if self.__dict__.pop("_db_done_step", False):
+            need_db = False      # fused into the ReLU backward pass
         db_out = self._bias_arena() if need_db else None
         dx, dw, db = ops.fc_backward(x, w, dy,
                                      need_dx=propagate_down[0],
@@ -190,9 +194,27 @@ class ReLULayer(Layer):
         return 0.0
 
     def backward(self, top, propagate_down, bottom):
-        if propagate_down[0]:
+        if not propagate_down[0]:
+            return
+        prod = getattr(self, "_db_producer", None)
+        db_out = None
+        import os as _os
+        if prod is not None and self.net._ops_mode == "bf16" \
+                and bool(int(_os.environ.get("COS_RELU_DB", "0"))):
+            # fused relu'+colsum measured SLOWER than the two separate
+            # passes (AlexNet 54.1k->52.6k, GoogLeNet 10.77k->10.30k,
+            # same-box A/B): the column-strip block shape the reduction
+            # needs costs more dx-write coalescing than the saved dy
+            # read.  Kept behind COS_RELU_DB=1 for evidence.
+            db_out = prod._bias_arena()
+        if db_out is not None:
+            dx, done = ops.relu_backward(top[0].data, top[0].diff,
+                                         self.slope, db_out=db_out)
+            if done:
+                prod._db_done_step = True
+        else:
             dx = ops.relu_backward(top[0].data, top[0].diff, self.slope)
-            self.acc_blob_diff(bottom[0], dx, top[0] is bottom[0])
+        self.acc_blob_diff(bottom[0], dx, top[0] is bottom[0])
 
 
 @register_layer("Sigmoid")

@@ -181,6 +181,10 @@ class Net:
                 continue
             a._fuse_relu = True
             b._fused_upstream = True
+            if len(a.blobs) > 1 and a.blobs[1].data.dim() == 1:
+                # backward fusion: the ReLU's gradient pass also column-
+                # sums the producer's bias gradient into the arena
+                b._db_producer = a
 
     def _fuse_concat_peephole(self) -> None:
         """Inception fusion (GPU): when every input of a channel Concat

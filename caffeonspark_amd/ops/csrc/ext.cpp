@@ -84,6 +84,9 @@ void relu_bwd(const void* y, const void* dy, void* dx, float slope, int64_t n,
 void relu_bwd_strided(const void* y, const void* dy, void* dx, float slope,
                       int64_t rows, int C, int ldy, int lddy,
                       hipStream_t stream);
+void relu_colsum_bwd(const void* y, const void* dy, void* dx, float* db,
+                     float slope, int64_t rows, int cols, int ldy,
+                     int lddy, hipStream_t stream);
 void dropout_fwd(const void* x, void* y, void* mask, float ratio,
                  const void* seed, int64_t n, hipStream_t stream);
 void seed_bump(void* s, hipStream_t stream);
@@ -204,6 +207,16 @@ void py_gemm_conv_dw(Tensor A, Tensor X, Tensor C,
                        (int)M, (int)N, (int)K, (int)lda, (int)ldc,
                        (int)store_mode, (int)splitk, (float)alpha, dbp,
                        g, cur_stream());
+}
+
+void py_relu_colsum_bwd(Tensor y, Tensor dy, Tensor dx, Tensor db,
+                        double slope, int64_t rows, int64_t cols,
+                        int64_t ldy, int64_t lddy) {
+  CHECK_CUDA(y); CHECK_BF16(y); CHECK_BF16(dy); CHECK_BF16(dx);
+  CHECK_F32(db);
+  cosamd::relu_colsum_bwd(y.data_ptr(), dy.data_ptr(), dx.data_ptr(),
+                          db.data_ptr<float>(), (float)slope, rows,
+                          (int)cols, (int)ldy, (int)lddy, cur_stream());
 }
 
 void py_im2col(Tensor x, Tensor col, int64_t N, int64_t H, int64_t W,
@@ -481,6 +494,7 @@ void py_softmax_loss_bwd(Tensor prob, Tensor label, Tensor dx, double scale,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm", &py_gemm);
+  m.def("relu_colsum_bwd", &py_relu_colsum_bwd);
   m.def("gemm_conv_fwd", &py_gemm_conv_fwd);
   m.def("gemm_conv_dw", &py_gemm_conv_dw);
   m.def("im2col", &py_im2col);

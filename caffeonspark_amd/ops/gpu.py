@@ -724,7 +724,7 @@ def _cl_slice_ld(t):
     return None
 
 
-def relu_backward(y, dy, negative_slope=0.0):
+def relu_backward(y, dy, negative_slope=0.0, db_out=None):
     ldy = _cl_slice_ld(y)
     if ldy is not None and y.dtype == torch.bfloat16:
         dy = dy if dy.dtype == torch.bfloat16 else dy.to(torch.bfloat16)
@@ -736,13 +736,26 @@ def relu_backward(y, dy, negative_slope=0.0):
         dx = torch.empty((N, C, H, W), dtype=torch.bfloat16,
                          device=y.device,
                          memory_format=torch.channels_last)
+        if db_out is not None and db_out.shape[0] == C:
+            # fused: the producing conv's bias gradient accumulates into
+            # its arena slice during this same pass (its colsum is skipped)
+            _ext.relu_colsum_bwd(y, dy, dx.permute(0, 2, 3, 1), db_out,
+                                 negative_slope, N * H * W, C, ldy, lddy)
+            return dx, True
         _ext.relu_bwd_strided(y, dy, dx.permute(0, 2, 3, 1), negative_slope,
                               N * H * W, C, ldy, lddy)
-        return dx
+        return (dx, False) if db_out is not None else dx
     dy = dy if dy.dtype == torch.bfloat16 else dy.to(torch.bfloat16)
+    if db_out is not None and y.dim() == 2 and y.is_contiguous() \
+            and db_out.shape[0] == y.shape[1]:
+        dx = torch.empty_like(y)
+        _ext.relu_colsum_bwd(y, dy.reshape(y.shape), dx, db_out,
+                             negative_slope, y.shape[0], y.shape[1],
+                             y.shape[1], y.shape[1])
+        return dx, True
     dx = torch.empty_like(y)
     _ext.relu_bwd(y, dy.reshape(y.shape), dx, negative_slope)
-    return dx
+    return (dx, False) if db_out is not None else dx
 
 
 _dropout_seeds = {}

@@ -72,7 +72,10 @@ def relu_forward(x, negative_slope=0.0):
     return F.leaky_relu(x, negative_slope) if negative_slope else F.relu(x)
 
 
-def relu_backward(y, dy, negative_slope=0.0):
+def relu_backward(y, dy, negative_slope=0.0, db_out=None):
+    if db_out is not None:
+        dx = torch.where(y > 0, dy, dy * negative_slope)
+        return dx, False
     """Caffe ReLU backward uses bottom data; with in-place layers only top
     data is available — sign(y) equals sign(x) for slope<1 so using y is
     exact for slope >= 0."""
