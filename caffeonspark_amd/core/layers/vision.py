@@ -95,11 +95,18 @@ class ConvolutionLayer(Layer):
             if d.dtype == torch.float32 and d.is_contiguous():
                 dw_out = d.view(wb.shape)
         db_out = self._bias_arena() if need_db else None
+        # branch fan-in (inception): when the bottom already has a diff,
+        # let the dx GEMM epilogue accumulate into it in place
+        dxb = bottom[0].diff
+        dx_into = dxb if (propagate_down[0] and dxb is not None
+                          and x.is_cuda and dxb.dtype == torch.bfloat16
+                          and list(dxb.shape) == list(x.shape)) else None
         dx, dw, db = ops.conv2d_backward(
             x, w, dy, (self.sh, self.sw), (self.ph, self.pw),
             (self.dil, self.dil), self.groups,
             need_dx=propagate_down[0], need_dw=need_dw, bias=need_db,
-            ctx=getattr(self, "_ctx", None), dw_out=dw_out, db_out=db_out)
+            ctx=getattr(self, "_ctx", None), dw_out=dw_out, db_out=db_out,
+            dx_into=dx_into)
         if dw is not None:
             if dw is dw_out:
                 wb._grad_virgin = False
@@ -107,7 +114,7 @@ class ConvolutionLayer(Layer):
                 self.acc_param_diff(0, dw)
         if db is not None and db is not db_out:
             self.acc_param_diff(1, db)
-        if propagate_down[0]:
+        if propagate_down[0] and dx is not dx_into:
             self.acc_blob_diff(bottom[0], dx, False)
 
 

@@ -398,6 +398,11 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gemm_kernel(
         int64_t off = (int64_t)row * ldc + col;
         if (STORE_MODE == 0) {
           reinterpret_cast<bf16*>(C)[off] = f2bf(v);
+        } else if (STORE_MODE == 3) {
+          // bf16 read-modify-write accumulate (dx fan-in; exclusive
+          // tiles, single split -> race-free)
+          bf16* p = reinterpret_cast<bf16*>(C) + off;
+          *p = f2bf(v + bf2f(*p));
         } else if (STORE_MODE == 1) {
           reinterpret_cast<float*>(C)[off] = v;
         } else {
@@ -556,7 +561,8 @@ template <int WAVES>
 __global__ __launch_bounds__(WAVES * 64, 2) void gemm_conv_fwd_kernel(
     const bf16* __restrict__ X, const bf16* __restrict__ B,
     bf16* __restrict__ C, const float* __restrict__ bias,
-    int M, int N, int K, int ldb, int ldc, int relu, CGeom gm) {
+    int M, int N, int K, int ldb, int ldc, int relu, int accum,
+    CGeom gm) {
   __shared__ bf16 Asb[2][BM * BK];
   __shared__ bf16 Bsb[2][BN * BK];
   __shared__ int rinfo[BM * 3];
@@ -637,7 +643,9 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gemm_conv_fwd_kernel(
         if (row >= M) continue;
         float v = acc[fm][fn][r] + badd;
         if (relu && v < 0.f) v = 0.f;
-        C[(int64_t)row * ldc + col] = f2bf(v);
+        int64_t off = (int64_t)row * ldc + col;
+        if (accum) v += bf2f(C[off]);   // dx fan-in: exclusive tiles,
+        C[off] = f2bf(v);               // so plain RMW is race-free
       }
     }
   }
@@ -767,6 +775,7 @@ void gemm_bf16_batched(const void* A_, const void* B_, void* C,
   int zblocks = (K + ksplit - 1) / ksplit;
   if (zblocks > 1 && store_mode != 2)
     throw std::runtime_error("gemm: split-K requires atomic store mode");
+  // store_mode 3 = bf16 RMW accumulate (single split only)
   dim3 grid(mblocks * nblocks, zblocks, batch);
   dim3 block(NTHREADS);
 
@@ -793,6 +802,7 @@ void gemm_bf16_batched(const void* A_, const void* B_, void* C,
   do {                                                                      \
     if (store_mode == 0)      COS_GEMM_CASE(TA, TB, 0);                     \
     else if (store_mode == 1) COS_GEMM_CASE(TA, TB, 1);                     \
+    else if (store_mode == 3) COS_GEMM_CASE(TA, TB, 3);                     \
     else                      COS_GEMM_CASE(TA, TB, 2);                     \
   } while (0)
 
@@ -813,6 +823,7 @@ void gemm_bf16_batched(const void* A_, const void* B_, void* C,
     if (w8) {
       if (store_mode == 0)      COS_GEMM_CASE8(0);
       else if (store_mode == 1) COS_GEMM_CASE8(1);
+      else if (store_mode == 3) COS_GEMM_CASE8(3);
       else                      COS_GEMM_CASE8(2);
       return;
     }
@@ -843,7 +854,7 @@ static CGeom make_geom(const int* g) {
 
 void gemm_conv_fwd(const void* X, const void* B, void* C,
                    const float* bias, int M, int N, int K, int ldb,
-                   int ldc, bool relu, const int* geom,
+                   int ldc, bool relu, bool accum, const int* geom,
                    hipStream_t stream) {
   CGeom gm = make_geom(geom);
   int mblocks = (M + BM - 1) / BM, nblocks = (N + BN - 1) / BN;
@@ -856,11 +867,11 @@ void gemm_conv_fwd(const void* X, const void* B, void* C,
   if (w8)
     gemm_conv_fwd_kernel<8><<<grid, dim3(512), 0, stream>>>(
         (const bf16*)X, (const bf16*)B, (bf16*)C, bias, M, N, K, ldb,
-        ldc, relu ? 1 : 0, gm);
+        ldc, relu ? 1 : 0, accum ? 1 : 0, gm);
   else
     gemm_conv_fwd_kernel<4><<<grid, dim3(256), 0, stream>>>(
         (const bf16*)X, (const bf16*)B, (bf16*)C, bias, M, N, K, ldb,
-        ldc, relu ? 1 : 0, gm);
+        ldc, relu ? 1 : 0, accum ? 1 : 0, gm);
 }
 
 void gemm_conv_dw(const void* A, const void* X, float* C, int M, int N,

@@ -369,7 +369,8 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
             _ext.gemm_conv_fwd(
                 xl, wrb[g * Kg:], y2[:, g * Kg:],
                 bias_f[g * Kg:(g + 1) * Kg] if bias_f is not None
-                else None, NPQ, Kg, Kpad, Kpad, ldc_out, relu, geom)
+                else None, NPQ, Kg, Kpad, Kpad, ldc_out, relu, False,
+                geom)
     else:
         col = torch.empty((G, NPQ, Kpad), dtype=torch.bfloat16,
                           device=x.device)
@@ -417,7 +418,7 @@ def _wino_run(xl, w, b, y, N, P, Q, C, K, relu, flip=False):
 
 def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                     need_dx=True, need_dw=True, bias=True, ctx=None,
-                    dw_out=None, db_out=None):
+                    dw_out=None, db_out=None, dx_into=None):
     _check_bf16(dy, "conv dy")
     if ctx is None or "col" not in ctx:
         ctx = ctx if ctx is not None else {}
@@ -538,9 +539,24 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
             else _scratch_buf((Kout,), torch.float32, dy.device)
         _ext.colsum(dy2, db, NPQ, Kout, Kout)
     if need_dx:
-        dx = torch.empty((N, C, H, W), dtype=torch.bfloat16,
-                         device=dy.device,
-                         memory_format=torch.channels_last)
+        wrF_ = getattr(ctx.get("w_ref"), "_cos_wrF", None)
+        acc_capable = not wino and (
+            is_1x1 or (wrF_ is not None
+                       and wrF_.shape == (G * _pad128(Cg),
+                                          _pad32(R * S * Kg))
+                       and N * H * W < (1 << 20)))
+        dx_acc = acc_capable and dx_into is not None and \
+            dx_into.is_contiguous(memory_format=torch.channels_last) and \
+            dx_into.dtype == torch.bfloat16 and \
+            tuple(dx_into.shape) == (N, C, H, W)
+        if dx_acc:
+            # fan-in: accumulate straight into the existing bottom diff
+            # inside the GEMM epilogue (saves the separate add pass)
+            dx = dx_into
+        else:
+            dx = torch.empty((N, C, H, W), dtype=torch.bfloat16,
+                             device=dy.device,
+                             memory_format=torch.channels_last)
         if is_1x1:
             # col2im is the identity: write dx's NHWC alias directly.
             # wrT is maintained per step by the fused repack kernel for
@@ -550,7 +566,7 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
             if wrT is None or wrT.shape[1] != Kout:
                 wrT = _transpose(wr.contiguous())
             _gemm(dy2, wrT, dx2, None, NPQ, C, Kout, Kout, Kout, C,
-                  False, False, 0, 1, na=_pad128(C))
+                  False, False, 3 if dx_acc else 0, 1, na=_pad128(C))
             return dx, dw, db
         if wino:
             # data gradient via Winograd on dy with flipped weights
@@ -572,7 +588,7 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                         g * Kg, Kg, k2]
                 _ext.gemm_conv_fwd(dyl, wrF[g * cgp:], dx2[:, g * Cg:],
                                    None, N * H * W, Cg, _pad32(k2),
-                                   _pad32(k2), C, False, geom)
+                                   _pad32(k2), C, False, dx_acc, geom)
             return dx, dw, db
         dcol = torch.empty((NPQ, Kpad), dtype=torch.bfloat16,
                            device=dy.device)

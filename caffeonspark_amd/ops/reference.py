@@ -27,7 +27,8 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
 
 def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                     need_dx=True, need_dw=True, bias=True, ctx=None,
-                    dw_out=None, db_out=None):  # db_out: GPU-path hint
+                    dw_out=None, db_out=None, dx_into=None):
+    # db_out/dx_into: GPU-path hints, unused here
     dx = dw = db = None
     if need_dx:
         dx = torch.nn.grad.conv2d_input(list(x.shape), w, dy, stride=stride,
