@@ -40,31 +40,50 @@ __global__ void lrn_fwd_kernel(const u16* __restrict__ x,
                                float* __restrict__ scale,
                                int64_t total, int C, int half, float a_over_n,
                                float beta, float k) {
-  // one thread per 8 channels: vectorized loads, per-lane sliding window
+  // one thread per 8 channels: three 16-byte octet loads (prev/mid/next)
+  // cover the [cc-half, cc+7+half] window (half<=4), float4 scale stores
   int c8s = C / 8;
   for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i8 < total; i8 += (int64_t)gridDim.x * blockDim.x) {
     int cc = (int)(i8 % c8s) * 8;
     int64_t base = (i8 / c8s) * (int64_t)C;
-    float v[16];  // window neighborhood [cc-half, cc+7+half], half<=4
+    float v[24];
+    u16x8 oct;
+    if (cc >= 8) {
+      oct = *reinterpret_cast<const u16x8*>(x + base + cc - 8);
 #pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      int c = cc - half + j;
-      v[j] = (c >= 0 && c < C && j < 8 + 2 * half)
-                 ? ld_bf(x + base + c) : 0.f;
+      for (int j = 0; j < 8; ++j) v[j] = ld_bf((const u16*)&oct + j);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = 0.f;
+    }
+    oct = *reinterpret_cast<const u16x8*>(x + base + cc);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v[8 + j] = ld_bf((const u16*)&oct + j);
+    if (cc + 16 <= C) {
+      oct = *reinterpret_cast<const u16x8*>(x + base + cc + 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[16 + j] = ld_bf((const u16*)&oct + j);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[16 + j] = 0.f;
     }
     u16x8 out;
+    float scv[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float win = 0.f;
 #pragma unroll
       for (int t = 0; t <= 8; ++t)  // window size 2*half+1 <= 9
-        if (t <= 2 * half) win += v[j + t] * v[j + t];
+        if (t <= 2 * half) win += v[8 + j - half + t] * v[8 + j - half + t];
       float sc = k + a_over_n * win;
-      scale[base + cc + j] = sc;
-      bf16 b = f2bf(v[j + half] * pow_negbeta<BMODE>(sc, beta));
+      scv[j] = sc;
+      bf16 b = f2bf(v[8 + j] * pow_negbeta<BMODE>(sc, beta));
       out[j] = *reinterpret_cast<u16*>(&b);
     }
+    auto* sp = reinterpret_cast<float4*>(scale + base + cc);
+    sp[0] = float4{scv[0], scv[1], scv[2], scv[3]};
+    sp[1] = float4{scv[4], scv[5], scv[6], scv[7]};
     *reinterpret_cast<u16x8*>(y + base + cc) = out;
   }
 }
@@ -78,12 +97,15 @@ __global__ void lrn_ratio_kernel(const u16* __restrict__ y,
     int64_t i = i8 * 8;
     u16x8 vy = *reinterpret_cast<const u16x8*>(y + i);
     u16x8 vd = *reinterpret_cast<const u16x8*>(dy + i);
+    auto* sp = reinterpret_cast<const float4*>(scale + i);
+    float4 s0 = sp[0], s1 = sp[1];
     u16x8 out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       u16 ry = vy[j], rd = vd[j];
+      float sc = j < 4 ? (&s0.x)[j] : (&s1.x)[j - 4];
       float f = bf2f(*reinterpret_cast<const bf16*>(&rd)) *
-                bf2f(*reinterpret_cast<const bf16*>(&ry)) / scale[i + j];
+                bf2f(*reinterpret_cast<const bf16*>(&ry)) / sc;
       bf16 b = f2bf(f);
       out[j] = *reinterpret_cast<u16*>(&b);
     }
@@ -104,23 +126,41 @@ __global__ void lrn_bwd_kernel(const u16* __restrict__ x,
        i8 < total8; i8 += (int64_t)gridDim.x * blockDim.x) {
     int cc = (int)(i8 % c8s) * 8;
     int64_t base = (i8 / c8s) * (int64_t)C;
-    float r[16];
+    float r[24];
+    u16x8 oct;
+    if (cc >= 8) {
+      oct = *reinterpret_cast<const u16x8*>(ratio + base + cc - 8);
 #pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      int c = cc - half + j;
-      r[j] = (c >= 0 && c < C && j < 8 + 2 * half)
-                 ? ld_bf(ratio + base + c) : 0.f;
+      for (int j = 0; j < 8; ++j) r[j] = ld_bf((const u16*)&oct + j);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) r[j] = 0.f;
     }
+    oct = *reinterpret_cast<const u16x8*>(ratio + base + cc);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) r[8 + j] = ld_bf((const u16*)&oct + j);
+    if (cc + 16 <= C) {
+      oct = *reinterpret_cast<const u16x8*>(ratio + base + cc + 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) r[16 + j] = ld_bf((const u16*)&oct + j);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) r[16 + j] = 0.f;
+    }
+    u16x8 vdy = *reinterpret_cast<const u16x8*>(dy + base + cc);
+    u16x8 vx = *reinterpret_cast<const u16x8*>(x + base + cc);
+    auto* sp = reinterpret_cast<const float4*>(scale + base + cc);
+    float4 s0 = sp[0], s1 = sp[1];
     u16x8 out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float win = 0.f;
 #pragma unroll
       for (int t = 0; t <= 8; ++t)
-        if (t <= 2 * half) win += r[j + t];
-      int64_t i = base + cc + j;
-      float v = ld_bf(dy + i) * pow_negbeta<BMODE>(scale[i], beta)
-                - ratio_coef * ld_bf(x + i) * win;
+        if (t <= 2 * half) win += r[8 + j - half + t];
+      float sc = j < 4 ? (&s0.x)[j] : (&s1.x)[j - 4];
+      float v = ld_bf((const u16*)&vdy + j) * pow_negbeta<BMODE>(sc, beta)
+                - ratio_coef * ld_bf((const u16*)&vx + j) * win;
       bf16 b = f2bf(v);
       out[j] = *reinterpret_cast<u16*>(&b);
     }
