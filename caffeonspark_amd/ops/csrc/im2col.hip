@@ -44,6 +44,14 @@ __global__ void im2col_smallc_kernel(
   // one 16-byte gather instead of 8 scalar loads (conv2-5 of AlexNet,
   // all GoogLeNet 3x3/5x5 towers; conv1's Ct=3 takes the scalar path)
   bool runs8 = (Ct % 8 == 0);
+  // pad-0 dil-1 full-channel small-C convs (AlexNet conv1): k runs are
+  // contiguous (s,c) spans of one input row, always in-bounds — ONE
+  // unaligned 16-byte load (gfx950 supports it natively) replaces the
+  // 8 scalar gathers; chunks straddling an r boundary merge two.
+  bool runfast = !runs8 && Ct == C && c0 == 0 && dil == 1 &&
+                 ph == 0 && pw == 0;
+  int SC = S * C;
+  int64_t total_x = (int64_t)N * H * W * C;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i < total8; i += (int64_t)gridDim.x * blockDim.x) {
     int64_t npq = i / kslots;
@@ -70,17 +78,39 @@ __global__ void im2col_smallc_kernel(
         out = u16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
     } else {
+      int k0f = ks * 8;
+      const u16* src = nullptr;
+      int r0 = 0, t0 = 0, sp = 8;
+      if (runfast && k0f + 8 <= Kcol) {
+        r0 = k0f / SC;
+        t0 = k0f - r0 * SC;
+        sp = SC - t0;                 // elements before r increments
+        src = xbase + (int64_t)(h0 + r0) * W * C + w0 * C + t0;
+        if (src - x + 8 > total_x) src = nullptr;  // over-read guard
+      }
+      if (src != nullptr && sp >= 8) {
+        __builtin_memcpy(&out, src, 16);
+      } else if (src != nullptr) {
+        u16x8 lo, hi;
+        __builtin_memcpy(&lo, src, 16);
+        const u16* src2 = xbase + (int64_t)(h0 + r0 + 1) * W * C + w0 * C;
+        __builtin_memcpy(&hi, src2, 16);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int ej = lut[j * 320 + ks];
-        u16 v = 0;
-        if (ej >= 0) {
-          int h = h0 + (ej >> 20);
-          int w = w0 + ((ej >> 10) & 1023);
-          if (h >= 0 && h < H && w >= 0 && w < W)
-            v = xbase[((int64_t)h * W + w) * C + (ej & 1023)];
+        for (int j = 0; j < 8; ++j)
+          out[j] = j < sp ? lo[j] : hi[j - sp];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int ej = lut[j * 320 + ks];
+          u16 v = 0;
+          if (ej >= 0) {
+            int h = h0 + (ej >> 20);
+            int w = w0 + ((ej >> 10) & 1023);
+            if (h >= 0 && h < H && w >= 0 && w < W)
+              v = xbase[((int64_t)h * W + w) * C + (ej & 1023)];
+          }
+          out[j] = v;
         }
-        out[j] = v;
       }
     }
     *reinterpret_cast<u16x8*>(col + npq * Kpad + k0) = out;
