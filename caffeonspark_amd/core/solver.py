@@ -119,7 +119,12 @@ class Solver:
                 view._cos_stable = True   # identity persists across steps
                 b.data._cos_bf16 = view
             def refresh():
-                self.flat_wb.copy_(self.flat_w)
+                # the fused update kernels write the bf16 shadow in the
+                # same pass once they have run (sticky flag: they run
+                # every step thereafter) — then the whole-arena cast
+                # here is redundant
+                if not getattr(self, "_shadow_synced", False):
+                    self.flat_wb.copy_(self.flat_w)
                 # fused per-step conv weight repack (GEMM layouts follow
                 # the shadow arena in one kernel instead of 2/conv)
                 from ..ops import gpu as _gops
@@ -462,6 +467,15 @@ class Solver:
                 self.net.copy_trained_layers_from(net_param)
             for tn in self.test_nets:
                 tn.share_trained_layers_with(self.net)
+        self.resync_shadow()
+
+    def resync_shadow(self) -> None:
+        """Re-cast the bf16 shadow after any out-of-band write to the
+        fp32 arena (weight load, broadcast): the fused update kernels
+        keep it current per step, but only for their own writes."""
+        wb = getattr(self, "flat_wb", None)
+        if wb is not None:
+            wb.copy_(self.flat_w)
 
 
 def solver_from_prototxt(path: str, **kwargs) -> Solver:

@@ -9,6 +9,7 @@ namespace cosamd {
 
 typedef unsigned short u16;
 typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
+typedef unsigned short u16x4 __attribute__((ext_vector_type(4)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 __device__ __forceinline__ float ldbf(const u16* p) {
@@ -301,7 +302,7 @@ __global__ void sgd_update_multi_kernel(
     float* __restrict__ p, const float* __restrict__ g,
     float* __restrict__ v, const int64_t* __restrict__ seg_off,
     const float* __restrict__ seg_lr, const float* __restrict__ seg_wd,
-    int nseg, float mu, int64_t total4) {
+    int nseg, float mu, int64_t total4, u16* __restrict__ sh) {
   __shared__ int64_t soff[COS_SGD_MAX_SEG + 1];
   __shared__ float slr[COS_SGD_MAX_SEG], swd[COS_SGD_MAX_SEG];
   for (int i = threadIdx.x; i <= nseg; i += blockDim.x) {
@@ -329,12 +330,27 @@ __global__ void sgd_update_multi_kernel(
       }
       *reinterpret_cast<f32x4*>(v + e0) = vv;
       *reinterpret_cast<f32x4*>(p + e0) = pv;
+      if (sh != nullptr) {
+        // bf16 shadow write fused in: the separate whole-arena cast
+        // pass (and its read of p) disappears from the step
+        u16x4 o;
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          bf16 b = f2bf(pv[k]);
+          o[k] = *reinterpret_cast<u16*>(&b);
+        }
+        *reinterpret_cast<u16x4*>(sh + e0) = o;
+      }
     } else {
       int64_t end = soff[s + 1];
       for (int64_t k = e0; k < end; ++k) {
         float gg = g[k] + wd * p[k];
         v[k] = mu * v[k] + lr * gg;
         p[k] -= v[k];
+        if (sh != nullptr) {
+          bf16 b = f2bf(p[k]);
+          sh[k] = *reinterpret_cast<u16*>(&b);
+        }
       }
     }
   }
@@ -343,11 +359,11 @@ __global__ void sgd_update_multi_kernel(
 void sgd_update_multi(float* p, const float* g, float* v,
                       const int64_t* seg_off, const float* seg_lr,
                       const float* seg_wd, int nseg, float mu,
-                      int64_t total, hipStream_t stream) {
+                      int64_t total, void* sh, hipStream_t stream) {
   int64_t total4 = (total + 3) / 4;
   int blocks = (int)hmin<int64_t>(2048, (total4 + 255) / 256);
   sgd_update_multi_kernel<<<blocks, 256, 0, stream>>>(
-      p, g, v, seg_off, seg_lr, seg_wd, nseg, mu, total4);
+      p, g, v, seg_off, seg_lr, seg_wd, nseg, mu, total4, (u16*)sh);
 }
 
 // Nesterov momentum, same whole-arena structure (reference solver
@@ -356,7 +372,7 @@ __global__ void nesterov_update_multi_kernel(
     float* __restrict__ p, const float* __restrict__ g,
     float* __restrict__ v, const int64_t* __restrict__ seg_off,
     const float* __restrict__ seg_lr, const float* __restrict__ seg_wd,
-    int nseg, float mu, int64_t total4) {
+    int nseg, float mu, int64_t total4, u16* __restrict__ sh) {
   __shared__ int64_t soff[COS_SGD_MAX_SEG + 1];
   __shared__ float slr[COS_SGD_MAX_SEG], swd[COS_SGD_MAX_SEG];
   for (int i = threadIdx.x; i <= nseg; i += blockDim.x) {
@@ -378,6 +394,10 @@ __global__ void nesterov_update_multi_kernel(
       float vnew = mu * vprev + lr * gg;
       v[k] = vnew;
       p[k] -= (1.f + mu) * vnew - mu * vprev;
+      if (sh != nullptr) {
+        bf16 b = f2bf(p[k]);
+        sh[k] = *reinterpret_cast<u16*>(&b);
+      }
     }
   }
 }
@@ -385,11 +405,11 @@ __global__ void nesterov_update_multi_kernel(
 void nesterov_update_multi(float* p, const float* g, float* v,
                            const int64_t* seg_off, const float* seg_lr,
                            const float* seg_wd, int nseg, float mu,
-                           int64_t total, hipStream_t stream) {
+                           int64_t total, void* sh, hipStream_t stream) {
   int64_t total4 = (total + 3) / 4;
   int blocks = (int)hmin<int64_t>(2048, (total4 + 255) / 256);
   nesterov_update_multi_kernel<<<blocks, 256, 0, stream>>>(
-      p, g, v, seg_off, seg_lr, seg_wd, nseg, mu, total4);
+      p, g, v, seg_off, seg_lr, seg_wd, nseg, mu, total4, (u16*)sh);
 }
 
 // Adam, whole-arena: m/v moment arenas; the bias-correction factor is
@@ -399,7 +419,8 @@ __global__ void adam_update_multi_kernel(
     float* __restrict__ m, float* __restrict__ v,
     const int64_t* __restrict__ seg_off,
     const float* __restrict__ seg_lr, const float* __restrict__ seg_wd,
-    int nseg, float b1, float b2, float eps, int64_t total4) {
+    int nseg, float b1, float b2, float eps, int64_t total4,
+    u16* __restrict__ sh) {
   __shared__ int64_t soff[COS_SGD_MAX_SEG + 1];
   __shared__ float slr[COS_SGD_MAX_SEG], swd[COS_SGD_MAX_SEG];
   for (int i = threadIdx.x; i <= nseg; i += blockDim.x) {
@@ -422,6 +443,10 @@ __global__ void adam_update_multi_kernel(
       m[k] = mm;
       v[k] = vv;
       p[k] -= lr * mm / (sqrtf(vv) + eps);
+      if (sh != nullptr) {
+        bf16 b = f2bf(p[k]);
+        sh[k] = *reinterpret_cast<u16*>(&b);
+      }
     }
   }
 }
@@ -429,11 +454,13 @@ __global__ void adam_update_multi_kernel(
 void adam_update_multi(float* p, const float* g, float* m, float* v,
                        const int64_t* seg_off, const float* seg_lr,
                        const float* seg_wd, int nseg, float b1, float b2,
-                       float eps, int64_t total, hipStream_t stream) {
+                       float eps, int64_t total, void* sh,
+                       hipStream_t stream) {
   int64_t total4 = (total + 3) / 4;
   int blocks = (int)hmin<int64_t>(2048, (total4 + 255) / 256);
   adam_update_multi_kernel<<<blocks, 256, 0, stream>>>(
-      p, g, m, v, seg_off, seg_lr, seg_wd, nseg, b1, b2, eps, total4);
+      p, g, m, v, seg_off, seg_lr, seg_wd, nseg, b1, b2, eps, total4,
+      (u16*)sh);
 }
 
 // ---------------------------------------------------------- column reduce

@@ -97,15 +97,16 @@ void sgd_update(float* p, const float* g, float* v, float lr, float mu,
 void nesterov_update_multi(float* p, const float* g, float* v,
                            const int64_t* seg_off, const float* seg_lr,
                            const float* seg_wd, int nseg, float mu,
-                           int64_t total, hipStream_t stream);
+                           int64_t total, void* sh, hipStream_t stream);
 void adam_update_multi(float* p, const float* g, float* m, float* v,
                        const int64_t* seg_off, const float* seg_lr,
                        const float* seg_wd, int nseg, float b1, float b2,
-                       float eps, int64_t total, hipStream_t stream);
+                       float eps, int64_t total, void* sh,
+                       hipStream_t stream);
 void sgd_update_multi(float* p, const float* g, float* v,
                       const int64_t* seg_off, const float* seg_lr,
                       const float* seg_wd, int nseg, float mu,
-                      int64_t total, hipStream_t stream);
+                      int64_t total, void* sh, hipStream_t stream);
 void colsum(const void* in, float* out, int64_t rows, int cols, int ld,
             hipStream_t stream);
 void transpose_bf16(const void* in, void* out, int64_t R, int64_t C,
@@ -308,40 +309,49 @@ void py_mul(Tensor a, Tensor b, Tensor y) {
 }
 
 void py_sgd_update_multi(Tensor p, Tensor g, Tensor v, Tensor seg_off,
-                         Tensor seg_lr, Tensor seg_wd, double mu) {
+                         Tensor seg_lr, Tensor seg_wd, double mu,
+                         c10::optional<Tensor> sh) {
   CHECK_F32(p); CHECK_F32(g); CHECK_F32(v);
+  void* shp = nullptr;
+  if (sh.has_value()) { CHECK_BF16(sh.value()); shp = sh->data_ptr(); }
   cosamd::sgd_update_multi(p.data_ptr<float>(), g.data_ptr<float>(),
                            v.data_ptr<float>(),
                            seg_off.data_ptr<int64_t>(),
                            seg_lr.data_ptr<float>(),
                            seg_wd.data_ptr<float>(),
-                           (int)seg_lr.numel(), mu, p.numel(),
+                           (int)seg_lr.numel(), mu, p.numel(), shp,
                            cur_stream());
 }
 
 void py_nesterov_update_multi(Tensor p, Tensor g, Tensor v, Tensor seg_off,
-                              Tensor seg_lr, Tensor seg_wd, double mu) {
+                              Tensor seg_lr, Tensor seg_wd, double mu,
+                              c10::optional<Tensor> sh) {
   CHECK_F32(p); CHECK_F32(g); CHECK_F32(v);
+  void* shp = nullptr;
+  if (sh.has_value()) { CHECK_BF16(sh.value()); shp = sh->data_ptr(); }
   cosamd::nesterov_update_multi(p.data_ptr<float>(), g.data_ptr<float>(),
                                 v.data_ptr<float>(),
                                 seg_off.data_ptr<int64_t>(),
                                 seg_lr.data_ptr<float>(),
                                 seg_wd.data_ptr<float>(),
-                                (int)seg_lr.numel(), mu, p.numel(),
+                                (int)seg_lr.numel(), mu, p.numel(), shp,
                                 cur_stream());
 }
 
 void py_adam_update_multi(Tensor p, Tensor g, Tensor m, Tensor v,
                           Tensor seg_off, Tensor seg_lr, Tensor seg_wd,
-                          double b1, double b2, double eps) {
+                          double b1, double b2, double eps,
+                          c10::optional<Tensor> sh) {
   CHECK_F32(p); CHECK_F32(g); CHECK_F32(m); CHECK_F32(v);
+  void* shp = nullptr;
+  if (sh.has_value()) { CHECK_BF16(sh.value()); shp = sh->data_ptr(); }
   cosamd::adam_update_multi(p.data_ptr<float>(), g.data_ptr<float>(),
                             m.data_ptr<float>(), v.data_ptr<float>(),
                             seg_off.data_ptr<int64_t>(),
                             seg_lr.data_ptr<float>(),
                             seg_wd.data_ptr<float>(),
                             (int)seg_lr.numel(), b1, b2, eps, p.numel(),
-                            cur_stream());
+                            shp, cur_stream());
 }
 
 void py_sgd_update(Tensor p, Tensor g, Tensor v, double lr, double mu,

@@ -48,9 +48,11 @@ __global__ void im2col_smallc_kernel(
   // contiguous (s,c) spans of one input row, always in-bounds — ONE
   // unaligned 16-byte load (gfx950 supports it natively) replaces the
   // 8 scalar gathers; chunks straddling an r boundary merge two.
-  bool runfast = !runs8 && Ct == C && c0 == 0 && dil == 1 &&
-                 ph == 0 && pw == 0;
   int SC = S * C;
+  // SC >= 8 keeps a chunk within at most two filter rows (the merge
+  // below handles exactly one straddle; LeNet's 5x1 runs would span 3)
+  bool runfast = !runs8 && Ct == C && c0 == 0 && dil == 1 &&
+                 ph == 0 && pw == 0 && SC >= 8;
   int64_t total_x = (int64_t)N * H * W * C;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i < total8; i += (int64_t)gridDim.x * blockDim.x) {

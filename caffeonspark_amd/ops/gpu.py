@@ -1212,9 +1212,12 @@ def nesterov_update_multi_arena(solver, rate, momentum, wd):
     if len(solver.segments) > 512:
         return False
     _seg_tensors(solver)
+    sh = getattr(solver, "flat_wb", None)
     _ext.nesterov_update_multi(solver.flat_w, solver.flat_g, solver.flat_m,
                                solver._seg_off, solver._seg_lrm * rate,
-                               solver._seg_dm * wd, momentum)
+                               solver._seg_dm * wd, momentum, sh)
+    if sh is not None:
+        solver._shadow_synced = True
     return True
 
 
@@ -1226,10 +1229,13 @@ def adam_update_multi_arena(solver, rate, b1, b2, eps, wd, t):
         return False
     _seg_tensors(solver)
     corr = (1.0 - b2 ** t) ** 0.5 / (1.0 - b1 ** t)
+    sh = getattr(solver, "flat_wb", None)
     _ext.adam_update_multi(solver.flat_w, solver.flat_g, solver.flat_m,
                            solver.flat_m2, solver._seg_off,
                            solver._seg_lrm * (rate * corr),
-                           solver._seg_dm * wd, b1, b2, eps)
+                           solver._seg_dm * wd, b1, b2, eps, sh)
+    if sh is not None:
+        solver._shadow_synced = True
     return True
 
 
@@ -1245,9 +1251,12 @@ def sgd_update_multi_arena(solver, rate, momentum, wd):
                             float(rate) * lrm, momentum, wd * dm)
         return
     _seg_tensors(solver)
+    sh = getattr(solver, "flat_wb", None)
     _ext.sgd_update_multi(solver.flat_w, solver.flat_g, solver.flat_m,
                           solver._seg_off, solver._seg_lrm * rate,
-                          solver._seg_dm * wd, momentum)
+                          solver._seg_dm * wd, momentum, sh)
+    if sh is not None:
+        solver._shadow_synced = True
 
 
 # --------------------------------------------------------------- registry

@@ -169,3 +169,5 @@ class DistributedSync(Callback):
             return
         dist.broadcast(self.solver.flat_w, src=0)
         dist.broadcast(self.solver.flat_m, src=0)
+        if hasattr(self.solver, "resync_shadow"):
+            self.solver.resync_shadow()   # out-of-band flat_w write
