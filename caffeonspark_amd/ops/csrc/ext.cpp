@@ -17,9 +17,9 @@ void gemm_bf16(const void* A, const void* B, void* C, const float* bias,
                bool relu, float alpha, int m_alloc, int n_alloc,
                hipStream_t stream);
 void gemm_conv_fwd(const void* X, const void* B, void* C,
-                   const float* bias, int M, int N, int K, int ldb,
-                   int ldc, bool relu, bool accum, const int* geom,
-                   hipStream_t stream);
+                   const float* bias, const void* zpage, int M, int N,
+                   int K, int ldb, int ldc, bool relu, bool accum,
+                   const int* geom, hipStream_t stream);
 void gemm_conv_dw(const void* A, const void* X, float* C, int M, int N,
                   int K, int lda, int ldc, int store_mode, int splitk,
                   float alpha, float* db, const int* geom,
@@ -173,9 +173,10 @@ void py_gemm(Tensor A, Tensor B, Tensor C, c10::optional<Tensor> bias,
 }
 
 void py_gemm_conv_fwd(Tensor X, Tensor B, Tensor C,
-                      c10::optional<Tensor> bias, int64_t M, int64_t N,
-                      int64_t K, int64_t ldb, int64_t ldc, bool relu,
-                      bool accum, std::vector<int64_t> geom) {
+                      c10::optional<Tensor> bias, Tensor zpage,
+                      int64_t M, int64_t N, int64_t K, int64_t ldb,
+                      int64_t ldc, bool relu, bool accum,
+                      std::vector<int64_t> geom) {
   CHECK_CUDA(X); CHECK_BF16(X); CHECK_BF16(B); CHECK_BF16(C);
   const float* bptr = nullptr;
   if (bias.has_value()) {
@@ -186,8 +187,9 @@ void py_gemm_conv_fwd(Tensor X, Tensor B, Tensor C,
   TORCH_CHECK(geom.size() == 14, "geom must have 14 ints");
   for (int i = 0; i < 14; ++i) g[i] = (int)geom[i];
   cosamd::gemm_conv_fwd(X.data_ptr(), B.data_ptr(), C.data_ptr(), bptr,
-                        (int)M, (int)N, (int)K, (int)ldb, (int)ldc, relu,
-                        accum, g, cur_stream());
+                        zpage.data_ptr(), (int)M, (int)N, (int)K,
+                        (int)ldb, (int)ldc, relu, accum, g,
+                        cur_stream());
 }
 
 void py_gemm_conv_dw(Tensor A, Tensor X, Tensor C,

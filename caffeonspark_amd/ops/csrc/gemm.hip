@@ -458,12 +458,15 @@ __device__ __forceinline__ void decode_rows(int* rinfo, const CGeom& gm,
 
 // A-operand staging for the forward conv GEMM: same LDS layout and DMA
 // placement as stage_direct_fast, but the per-lane source address is the
-// im2col gather; out-of-image (or k >= Kcol) chunks ds_write zeros
-// instead (the DMA is EXEC-masked, so their slots are ours to fill).
+// im2col gather.  Out-of-image (or k >= Kcol) lanes redirect their DMA
+// to a 16-byte zero page instead of branching: the instruction always
+// issues, so the pipeline's counted s_waitcnt vmcnt(N) stays exact (a
+// divergent skip would desynchronize the count), and no LDS zero-fill
+// writes are needed.
 template <int WAVES>
 __device__ __forceinline__ void stage_implicit_fast(
-    bf16* lds_, const bf16* X, const CGeom& gm, const int* rinfo,
-    int k0, int tid) {
+    bf16* lds_, const bf16* X, const bf16* zpage, const CGeom& gm,
+    const int* rinfo, int k0, int tid) {
   auto* lds = reinterpret_cast<unsigned short*>(lds_);
   int wave = tid >> 6, lane = tid & 63;
   constexpr int CPW = 8 / WAVES;
@@ -482,19 +485,13 @@ __device__ __forceinline__ void stage_implicit_fast(
     int w = rinfo[row * 3 + 2] + s * gm.dil;
     bool ok = (k < gm.Kcol) && (h >= 0) && (h < gm.H) && (w >= 0) &&
               (w < gm.W);
+    int64_t off = ((int64_t)rinfo[row * 3] +
+                   (int64_t)(h * gm.W + w)) * gm.C + gm.c0 + c;
+    const bf16* src = ok ? X + off : zpage;
     auto* lp = (__attribute__((address_space(3))) unsigned int*)(
         lds + chunk * 512);
-    if (ok) {
-      int64_t off = ((int64_t)rinfo[row * 3] +
-                     (int64_t)(h * gm.W + w)) * gm.C + gm.c0 + c;
-      auto* gp = (const __attribute__((address_space(1))) unsigned int*)(
-          X + off);
-      __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
-    } else {
-      auto* zp = (__attribute__((address_space(3))) short8v*)(
-          lds + chunk * 512 + lane * 8);
-      *zp = short8v{0, 0, 0, 0, 0, 0, 0, 0};
-    }
+    auto* gp = (const __attribute__((address_space(1))) unsigned int*)src;
+    __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
   }
 }
 
@@ -561,6 +558,7 @@ template <int WAVES>
 __global__ __launch_bounds__(WAVES * 64, 2) void gemm_conv_fwd_kernel(
     const bf16* __restrict__ X, const bf16* __restrict__ B,
     bf16* __restrict__ C, const float* __restrict__ bias,
+    const bf16* __restrict__ zpage,
     int M, int N, int K, int ldb, int ldc, int relu, int accum,
     CGeom gm) {
   __shared__ bf16 Asb[2][BM * BK];
@@ -586,19 +584,19 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gemm_conv_fwd_kernel(
 
   f32x4 acc[MF][4] = {};
   int cur = 0;
-  stage_implicit_fast<WAVES>(Asb[0], X, gm, rinfo, 0, tid);
+  stage_implicit_fast<WAVES>(Asb[0], X, zpage, gm, rinfo, 0, tid);
   stage_direct_fast<WAVES>(Bsb[0], B, tile_n, ldb, 0, tid);
   for (int k0 = 0; k0 < K; k0 += BK) {
     if (k0 + BK < K) {
-      stage_implicit_fast<WAVES>(Asb[cur ^ 1], X, gm, rinfo, k0 + BK,
-                                 tid);
+      stage_implicit_fast<WAVES>(Asb[cur ^ 1], X, zpage, gm, rinfo,
+                                 k0 + BK, tid);
       stage_direct_fast<WAVES>(Bsb[cur ^ 1], B, tile_n, ldb, k0 + BK,
                                tid);
     }
     if (WAVES == 4)
-      asm volatile("s_waitcnt vmcnt(4) lgkmcnt(0)" ::: "memory");
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     else
-      asm volatile("s_waitcnt vmcnt(2) lgkmcnt(0)" ::: "memory");
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
     __builtin_amdgcn_sched_barrier(0);
     __builtin_amdgcn_s_barrier();
     const bf16* Asp = Asb[cur];
@@ -853,9 +851,9 @@ static CGeom make_geom(const int* g) {
 }
 
 void gemm_conv_fwd(const void* X, const void* B, void* C,
-                   const float* bias, int M, int N, int K, int ldb,
-                   int ldc, bool relu, bool accum, const int* geom,
-                   hipStream_t stream) {
+                   const float* bias, const void* zpage, int M, int N,
+                   int K, int ldb, int ldc, bool relu, bool accum,
+                   const int* geom, hipStream_t stream) {
   CGeom gm = make_geom(geom);
   int mblocks = (M + BM - 1) / BM, nblocks = (N + BN - 1) / BN;
   dim3 grid(mblocks * nblocks);
@@ -866,12 +864,14 @@ void gemm_conv_fwd(const void* X, const void* B, void* C,
   bool w8 = (w8_env == -1) ? (M <= 256 || N <= 256) : (w8_env == 1);
   if (w8)
     gemm_conv_fwd_kernel<8><<<grid, dim3(512), 0, stream>>>(
-        (const bf16*)X, (const bf16*)B, (bf16*)C, bias, M, N, K, ldb,
-        ldc, relu ? 1 : 0, accum ? 1 : 0, gm);
+        (const bf16*)X, (const bf16*)B, (bf16*)C, bias,
+        (const bf16*)zpage, M, N, K, ldb, ldc, relu ? 1 : 0,
+        accum ? 1 : 0, gm);
   else
     gemm_conv_fwd_kernel<4><<<grid, dim3(256), 0, stream>>>(
-        (const bf16*)X, (const bf16*)B, (bf16*)C, bias, M, N, K, ldb,
-        ldc, relu ? 1 : 0, accum ? 1 : 0, gm);
+        (const bf16*)X, (const bf16*)B, (bf16*)C, bias,
+        (const bf16*)zpage, M, N, K, ldb, ldc, relu ? 1 : 0,
+        accum ? 1 : 0, gm);
 }
 
 void gemm_conv_dw(const void* A, const void* X, float* C, int M, int N,

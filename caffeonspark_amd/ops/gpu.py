@@ -207,6 +207,19 @@ def _scratch_buf(shape, dtype, device, zero_each_step=True):
     return t
 
 
+_zero16 = {}
+
+
+def _zpage(device):
+    """Persistent 16-byte zero page: OOB implicit-staging lanes DMA from
+    here so the instruction always issues (exact vmcnt counting)."""
+    t = _zero16.get(device.index)
+    if t is None:
+        t = torch.zeros(8, dtype=torch.bfloat16, device=device)
+        _zero16[device.index] = t
+    return t
+
+
 def _gemm(A, B, C, bias, M, N, K, lda, ldb, ldc, ta, tb, store, splitk,
           relu=False, alpha=1.0, ma=0, na=0):
     # staging-allocation bounds: number of rows safely readable past M/N
@@ -369,8 +382,8 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
             _ext.gemm_conv_fwd(
                 xl, wrb[g * Kg:], y2[:, g * Kg:],
                 bias_f[g * Kg:(g + 1) * Kg] if bias_f is not None
-                else None, NPQ, Kg, Kpad, Kpad, ldc_out, relu, False,
-                geom)
+                else None, _zpage(x.device), NPQ, Kg, Kpad, Kpad,
+                ldc_out, relu, False, geom)
     else:
         col = torch.empty((G, NPQ, Kpad), dtype=torch.bfloat16,
                           device=x.device)
@@ -587,8 +600,9 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                         dil * (R - 1) - ph, dil * (S - 1) - pw, dil, S,
                         g * Kg, Kg, k2]
                 _ext.gemm_conv_fwd(dyl, wrF[g * cgp:], dx2[:, g * Cg:],
-                                   None, N * H * W, Cg, _pad32(k2),
-                                   _pad32(k2), C, False, dx_acc, geom)
+                                   None, _zpage(dy.device), N * H * W,
+                                   Cg, _pad32(k2), _pad32(k2), C, False,
+                                   dx_acc, geom)
             return dx, dw, db
         dcol = torch.empty((NPQ, Kpad), dtype=torch.bfloat16,
                            device=dy.device)
