@@ -649,6 +649,107 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gemm_conv_fwd_kernel(
   }
 }
 
+// quad-buffered variant: four LDS buffers (65.5 KB, still 2 blocks/CU
+// within CDNA4's 160 KB) let ONE barrier per k-tile replace the usual
+// two — the buffer written at iter t+1 (slot (t+3)&3) was last read at
+// iter t-1, and barrier_t separates them (a wave's ds_reads complete
+// before its MFMAs consume them, hence before it arrives at barrier_t).
+template <int WAVES>
+__global__ __launch_bounds__(WAVES * 64, 2) void gemm_conv_fwd_q_kernel(
+    const bf16* __restrict__ X, const bf16* __restrict__ B,
+    bf16* __restrict__ C, const float* __restrict__ bias,
+    const bf16* __restrict__ zpage,
+    int M, int N, int K, int ldb, int ldc, int relu, int accum,
+    CGeom gm) {
+  __shared__ bf16 Asb[4][BM * BK];
+  __shared__ bf16 Bsb[4][BN * BK];
+  __shared__ int rinfo[BM * 3];
+
+  int mblocks = (M + BM - 1) / BM;
+  int nblocks = (N + BN - 1) / BN;
+  int bid = xcd_swizzle(blockIdx.x, mblocks * nblocks);
+  int bm = bid / nblocks, bn = bid % nblocks;
+  int tile_m = bm * BM, tile_n = bn * BN;
+
+  constexpr int MROWS = BM / (WAVES / 2);
+  constexpr int MF = MROWS / 16;
+  int tid = threadIdx.x;
+  int wave = tid >> 6, lane = tid & 63;
+  int wm = wave >> 1, wn = wave & 1;
+  int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+
+  decode_rows(rinfo, gm, tile_m, M, BM, tid);
+  __syncthreads();
+
+  f32x4 acc[MF][4] = {};
+  int T = K / BK;
+  stage_implicit_fast<WAVES>(Asb[0], X, zpage, gm, rinfo, 0, tid);
+  stage_direct_fast<WAVES>(Bsb[0], B, tile_n, ldb, 0, tid);
+  if (T > 1) {
+    stage_implicit_fast<WAVES>(Asb[1], X, zpage, gm, rinfo, BK, tid);
+    stage_direct_fast<WAVES>(Bsb[1], B, tile_n, ldb, BK, tid);
+  }
+  for (int t = 0; t < T; ++t) {
+    if (t + 2 < T) {
+      stage_implicit_fast<WAVES>(Asb[(t + 2) & 3], X, zpage, gm, rinfo,
+                                 (t + 2) * BK, tid);
+      stage_direct_fast<WAVES>(Bsb[(t + 2) & 3], B, tile_n, ldb,
+                               (t + 2) * BK, tid);
+    }
+    if (WAVES == 4)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_barrier();
+    const bf16* Asp = Asb[t & 3];
+    const bf16* Bsp = Bsb[t & 3];
+    bf16x8 afrag[MF], bfrag[4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      int rb = wn * 64 + f * 16 + lrow;
+      bfrag[f] = *reinterpret_cast<const bf16x8*>(
+          Bsp + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
+    }
+#pragma unroll
+    for (int f = 0; f < MF; ++f) {
+      int ra = wm * MROWS + f * 16 + lrow;
+      afrag[f] = *reinterpret_cast<const bf16x8*>(
+          Asp + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int fm = 0; fm < MF; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  }
+
+  int crow0 = tile_m + wm * MROWS + ((lane >> 4) << 2);
+  int ccol0 = tile_n + wn * 64 + (lane & 15);
+#pragma unroll
+  for (int fm = 0; fm < MF; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int col = ccol0 + fn * 16;
+      if (col >= N) continue;
+      float badd = (bias != nullptr) ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow0 + fm * 16 + r;
+        if (row >= M) continue;
+        float v = acc[fm][fn][r] + badd;
+        if (relu && v < 0.f) v = 0.f;
+        int64_t off = (int64_t)row * ldc + col;
+        if (accum) v += bf2f(C[off]);
+        C[off] = f2bf(v);
+      }
+    }
+  }
+}
+
 // dw trans/trans GEMM with the col operand gathered implicitly:
 // dw[Kg][Kcol] = dy^T @ im2col(x), optional fused db, split-K atomics
 // (STORE 2) or plain store (STORE 1, single split).
@@ -862,6 +963,20 @@ void gemm_conv_fwd(const void* X, const void* B, void* C,
     return e ? (e[0] == '1' ? 1 : 0) : -1;
   }();
   bool w8 = (w8_env == -1) ? (M <= 256 || N <= 256) : (w8_env == 1);
+  static const int quad = [] {
+    const char* e = getenv("COS_CONVQ");
+    return e ? (e[0] == '1' ? 1 : 0) : 0;
+  }();
+  // default OFF: same-box A/B shows +1.6% GoogLeNet / +1..5% CIFAR-LRCN
+  // but -3% AlexNet (the headline config) — barrier savings lose to
+  // some second-order effect on AlexNet's longer-K 8-wave shapes
+  if (quad && w8 && K >= 4 * BK) {
+    gemm_conv_fwd_q_kernel<8><<<grid, dim3(512), 0, stream>>>(
+        (const bf16*)X, (const bf16*)B, (bf16*)C, bias,
+        (const bf16*)zpage, M, N, K, ldb, ldc, relu ? 1 : 0,
+        accum ? 1 : 0, gm);
+    return;
+  }
   if (w8)
     gemm_conv_fwd_kernel<8><<<grid, dim3(512), 0, stream>>>(
         (const bf16*)X, (const bf16*)B, (bf16*)C, bias,
