@@ -95,26 +95,37 @@ def test_gemm_splitk_atomic():
     dict(n=2, c=24, h=13, w=13, k=48, r=3, stride=1, pad=1, groups=1),
     dict(n=2, c=32, h=9, w=9, k=16, r=1, stride=1, pad=0, groups=1),
     dict(n=2, c=32, h=8, w=8, k=16, r=1, stride=1, pad=0, groups=1),
+    # implicit fwd/dw with the dcol+col2im dx fallback (stride 2)
+    dict(n=2, c=16, h=15, w=15, k=24, r=3, stride=2, pad=1, groups=1),
+    # implicit dx at maximal pad (pad' == 0 boundary)
+    dict(n=2, c=8, h=12, w=12, k=16, r=3, stride=1, pad=2, groups=1),
+    # Cg%8 != 0: materialized-col fallback for an 8-misaligned group
+    dict(n=2, c=16, h=11, w=11, k=16, r=3, stride=1, pad=1, groups=4),
+    # dilation through the implicit decode
+    dict(n=2, c=16, h=15, w=15, k=16, r=3, stride=1, pad=2, groups=1,
+         dil=2),
 ])
 def test_conv_forward_backward(case):
     n, c, h, w = case["n"], case["c"], case["h"], case["w"]
     k, r, st, pd, g = (case["k"], case["r"], case["stride"], case["pad"],
                        case["groups"])
+    dil = case.get("dil", 1)
     x = bf(torch.randn(n, c, h, w)).to(dev())
     wt = bf(torch.randn(k, c // g, r, r) * 0.1).to(dev()).float()
     b = torch.randn(k).to(dev()).float()
     ctx = {}
-    y = ops.conv2d_forward(x, wt, b, (st, st), (pd, pd), (1, 1), g, ctx=ctx)
+    y = ops.conv2d_forward(x, wt, b, (st, st), (pd, pd), (dil, dil), g,
+                           ctx=ctx)
     x_cpu, w_cpu, b_cpu = x.float().cpu(), wt.float().cpu(), b.float().cpu()
     want = reference.conv2d_forward(x_cpu, w_cpu, b_cpu, (st, st), (pd, pd),
-                                    (1, 1), g)
+                                    (dil, dil), g)
     agree(y, want, rtol=0.05, atol=0.1)
 
     dy = bf(torch.randn_like(y.float())).to(dev())
-    dx, dw, db = ops.conv2d_backward(x, wt, dy, (st, st), (pd, pd), (1, 1),
-                                     g, ctx=ctx)
+    dx, dw, db = ops.conv2d_backward(x, wt, dy, (st, st), (pd, pd),
+                                     (dil, dil), g, ctx=ctx)
     rdx, rdw, rdb = reference.conv2d_backward(
-        x_cpu, w_cpu, dy.float().cpu(), (st, st), (pd, pd), (1, 1), g)
+        x_cpu, w_cpu, dy.float().cpu(), (st, st), (pd, pd), (dil, dil), g)
     agree(dx, rdx, rtol=0.05, atol=0.15)
     agree(dw, rdw, rtol=0.05, atol=0.3)
     agree(db, rdb, rtol=0.05, atol=0.3)
