@@ -42,9 +42,26 @@ class ConcatLayer(Layer):
 
     def backward(self, top, propagate_down, bottom):
         if getattr(self, "_fused", False):
+            relus = getattr(self, "_branch_relus", None)
+            dy = top[0].diff
+            if relus is not None and self.net._ops_mode == "bf16" \
+                    and dy is not None and dy.dtype == torch.bfloat16:
+                # every branch ends in a fused ReLU: ONE dense masked
+                # pass over the whole concat (full cache lines), then
+                # hand out pre-masked windows; the branch ReLU layers
+                # pass straight through
+                from ... import ops
+                dxm = ops.relu_backward(top[0].data, dy)
+                off = 0
+                for b, pd, rl in zip(bottom, propagate_down, relus):
+                    k = b.data.shape[1]
+                    if pd:
+                        self.acc_blob_diff(b, dxm[:, off:off + k], False)
+                        rl._mask_done_step = True
+                    off += k
+                return
             # hand each branch a channel-slice VIEW — the in-place ReLU
             # backward consumes it stride-aware, so no slicing copy
-            dy = top[0].diff
             off = 0
             for b, pd in zip(bottom, propagate_down):
                 k = b.data.shape[1]

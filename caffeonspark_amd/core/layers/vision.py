@@ -203,6 +203,11 @@ class ReLULayer(Layer):
     def backward(self, top, propagate_down, bottom):
         if not propagate_down[0]:
             return
+        if self.__dict__.pop("_mask_done_step", False):
+            # the fused concat already applied this ReLU's mask densely
+            self.acc_blob_diff(bottom[0], top[0].diff,
+                               top[0] is bottom[0])
+            return
         prod = getattr(self, "_db_producer", None)
         db_out = None
         import os as _os

@@ -226,6 +226,17 @@ class Net:
             l._fused = True
             l._fused_ctot = off
             l._fused_buf = None
+            # dense-ReLU backward: when every branch ends in a fused
+            # in-place ReLU, ONE dense relu_bwd over the whole concat
+            # replaces per-branch strided-window passes (half their
+            # cache-line utilization); branch ReLUs then pass through
+            relus = {}
+            for rl in self.layers:
+                if rl.param.type == "ReLU" and rl.param.top and \
+                        getattr(rl, "_fused_upstream", False):
+                    relus[rl.param.top[0]] = rl
+            br = [relus.get(bn) for bn in l.param.bottom]
+            l._branch_relus = br if all(r is not None for r in br) else None
 
     def _prefeed_placeholder(self, layer) -> None:
         from .layers.data import CoSDataLayer, MemoryDataLayer

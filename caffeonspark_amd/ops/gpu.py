@@ -456,8 +456,28 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         ctx["col"] = col
     Kout = Kg * G
     NPQ = N * P * Q
-    dyl = _cl(dy)
-    dy2 = dyl.permute(0, 2, 3, 1).reshape(NPQ, Kout)
+    # fused-concat channel window (dense-ReLU backward hands the branch
+    # its pre-masked slice of the concat diff): every consumer kernel
+    # reads with an explicit row stride, so no compacting copy is made
+    ld_dy = _cl_slice_ld(dy)
+    use_tt_env = bool(int(os.environ.get("COS_DW_TT", "1")))
+    if ld_dy is not None and ld_dy != Kout and not wino and use_tt_env \
+            and dy.storage_offset() % 8 == 0 and ld_dy % 8 == 0 \
+            and dy.dtype == torch.bfloat16:
+        dyl = dy
+        dy2 = None
+    else:
+        dyl = _cl(dy)
+        ld_dy = Kout
+        dy2 = dyl.permute(0, 2, 3, 1).reshape(NPQ, Kout)
+
+    def _dyg(g):
+        """dy's group-g channel slice (2-D alias when dense, 4-D
+        channel sub-window when strided — the bindings only use the
+        base pointer; lda carries the row stride)."""
+        if dy2 is not None:
+            return dy2[:, g * Kg:] if g else dy2
+        return dyl[:, g * Kg:] if g else dyl
     if is_1x1:
         x2 = ctx["xl"].permute(0, 2, 3, 1).reshape(NPQ, C)
 
@@ -497,14 +517,15 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                 db_slice = db.narrow(0, g * Kg, Kg) if fuse_db else None
                 geom = [H, W, C, P, Q, sh, sw, ph, pw, dil, S, g * Cg,
                         Cg, Kcol]
-                _ext.gemm_conv_dw(dy2[:, g * Kg:], ctx["xl"],
+                _ext.gemm_conv_dw(_dyg(g), ctx["xl"],
                                   dwp[g * Kg:], db_slice, Kg, Kpad, NPQ,
-                                  Kout, Kpad, store_dw, sk_tt, 1.0, geom)
+                                  ld_dy, Kpad, store_dw, sk_tt, 1.0,
+                                  geom)
             elif use_tt:
                 src = x2 if is_1x1 else col[g]
                 db_slice = db.narrow(0, g * Kg, Kg) if fuse_db else None
-                _gemm(dy2[:, g * Kg:], src, dwp[g * Kg:], db_slice,
-                      Kg, Kpad, NPQ, Kout, Kpad if not is_1x1 else C,
+                _gemm(_dyg(g), src, dwp[g * Kg:], db_slice,
+                      Kg, Kpad, NPQ, ld_dy, Kpad if not is_1x1 else C,
                       Kpad, True, True, store_dw, sk_tt)
             else:
                 if is_1x1:
@@ -550,7 +571,7 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
         # accumulate the bias gradient in place, no staging copy
         db = db_out if db_out is not None \
             else _scratch_buf((Kout,), torch.float32, dy.device)
-        _ext.colsum(dy2, db, NPQ, Kout, Kout)
+        _ext.colsum(dyl if dy2 is None else dy2, db, NPQ, Kout, ld_dy)
     if need_dx:
         wrF_ = getattr(ctx.get("w_ref"), "_cos_wrF", None)
         acc_capable = not wino and (
@@ -578,8 +599,9 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
             wrT = getattr(ctx.get("w_ref"), "_cos_wrT", None)
             if wrT is None or wrT.shape[1] != Kout:
                 wrT = _transpose(wr.contiguous())
-            _gemm(dy2, wrT, dx2, None, NPQ, C, Kout, Kout, Kout, C,
-                  False, False, 3 if dx_acc else 0, 1, na=_pad128(C))
+            _gemm(_dyg(0), wrT, dx2, None, NPQ, C, Kout, ld_dy, Kout,
+                  C, False, False, 3 if dx_acc else 0, 1, ma=NPQ,
+                  na=_pad128(C))
             return dx, dw, db
         if wino:
             # data gradient via Winograd on dy with flipped weights
@@ -596,7 +618,7 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
             # dx's NHWC alias directly (no dcol buffer, no col2im pass)
             dx2 = dx.permute(0, 2, 3, 1).reshape(N * H * W, C)
             for g in range(G):
-                geom = [P, Q, Kout, H, W, 1, 1,
+                geom = [P, Q, ld_dy, H, W, 1, 1,
                         dil * (R - 1) - ph, dil * (S - 1) - pw, dil, S,
                         g * Kg, Kg, k2]
                 _ext.gemm_conv_fwd(dyl, wrF[g * cgp:], dx2[:, g * Cg:],
@@ -614,9 +636,9 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                 if G == 1 else None
             if wrT is None:
                 wrT = _transpose(wr[g * Kg:(g + 1) * Kg].contiguous())
-            _gemm(dy2[:, g * Kg:], wrT, dcol, None,
-                  NPQ, Kpad, Kg, Kout, Kg, Kpad, False, False, 0, 1,
-                  na=_pad128(Kpad))
+            _gemm(_dyg(g), wrT, dcol, None,
+                  NPQ, Kpad, Kg, ld_dy, Kg, Kpad, False, False, 0, 1,
+                  ma=NPQ, na=_pad128(Kpad))
             _ext.col2im(dcol, dx, N, H, W, C, P, Q, R, S, sh, sw, ph, pw,
                         dil, Kpad, g * Cg, Cg)
     return dx, dw, db
