@@ -964,13 +964,15 @@ void gemm_conv_fwd(const void* X, const void* B, void* C,
   }();
   bool w8 = (w8_env == -1) ? (M <= 256 || N <= 256) : (w8_env == 1);
   static const int quad = [] {
+    // COS_CONVQ: 0 = off, otherwise a max-K gate (1 = "all shapes").
+    // Default 1152: same-box A/B is +2.2% CIFAR, neutral AlexNet/
+    // GoogLeNet/LRCN; ungated it was -3% AlexNet (long-K shapes).
     const char* e = getenv("COS_CONVQ");
-    return e ? (e[0] == '1' ? 1 : 0) : 0;
+    if (!e) return 1152;
+    int v = atoi(e);
+    return v == 1 ? 1 << 30 : v;
   }();
-  // default OFF: same-box A/B shows +1.6% GoogLeNet / +1..5% CIFAR-LRCN
-  // but -3% AlexNet (the headline config) — barrier savings lose to
-  // some second-order effect on AlexNet's longer-K 8-wave shapes
-  if (quad && w8 && K >= 4 * BK) {
+  if (quad && w8 && K >= 4 * BK && K <= quad) {
     gemm_conv_fwd_q_kernel<8><<<grid, dim3(512), 0, stream>>>(
         (const bf16*)X, (const bf16*)B, (bf16*)C, bias,
         (const bf16*)zpage, M, N, K, ldb, ldc, relu ? 1 : 0,
