@@ -253,7 +253,11 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
     # from the NHWC input inside the GEMM — the col matrix is never
     # materialized.  Needs 8-aligned channel octets (one (r,s) tap per
     # 16-byte load) and npq/k decode within the magic-divide bound.
-    implicit = (C % 8 == 0 and Cg % 8 == 0 and not (R == S == 1)
+    imp1x1 = (R == S == 1 and sh == sw == 1 and ph == pw == 0
+              and G == 1 and C >= 256
+              and bool(int(os.environ.get("COS_IMP_1X1", "0"))))
+    implicit = (C % 8 == 0 and Cg % 8 == 0
+                and (not (R == S == 1) or imp1x1)
                 and N * P * Q < (1 << 20) and R * S * Cg < (1 << 20)
                 and bool(int(os.environ.get("COS_IMPLICIT", "1"))))
     # 32-aligned K keeps every k-tile on the pipelined fast path (and
@@ -368,7 +372,7 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
     # 1x1/stride-1 conv: im2col is the identity in NHWC — GEMM straight
     # off the input (GoogLeNet's many 1x1 convs skip the col buffer)
     is_1x1 = (R == S == 1 and sh == sw == 1 and ph == pw == 0 and
-              dil == 1 and G == 1 and C % 8 == 0)
+              dil == 1 and G == 1 and C % 8 == 0 and not implicit)
     if is_1x1:
         x2 = xl.permute(0, 2, 3, 1).reshape(NPQ, C)
         _gemm(x2, wr, y2, bias_f, NPQ, Kout, C, C, C, ldc_out,
