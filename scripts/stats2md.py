@@ -23,7 +23,7 @@ for m, title in hdr.items():
         line = line.strip()
         if line.startswith('{"metric"'):
             js = json.loads(line)
-    out = [f"# {title} bf16 1xMI355X — round-1 final kernel profile "
+    out = [f"# {title} bf16 1xMI355X — round-2 final kernel profile "
            f"(hipGraph stepping)",
            f"{js.get('ms_per_step', '?')} ms/step under rocprofv3 "
            f"--kernel-trace (adds overhead; unprofiled numbers in "
