@@ -24,6 +24,14 @@ void gemm_conv_dw(const void* A, const void* X, float* C, int M, int N,
                   int K, int lda, int ldc, int store_mode, int splitk,
                   float alpha, float* db, const int* geom,
                   hipStream_t stream);
+void gemm_conv_fwd_sc(const void* X, const void* B, void* C,
+                      const float* bias, int M, int N, int K, int ldb,
+                      int ldc, bool relu, const int* geom,
+                      hipStream_t stream);
+void gemm_conv_dw_sc(const void* A, const void* X, float* C, int M,
+                     int N, int K, int lda, int ldc, int store_mode,
+                     int splitk, float alpha, float* db, const int* geom,
+                     hipStream_t stream);
 void tr16_probe(float* out, int mode, hipStream_t stream);
 void repack_weights(const int64_t* table, int ndesc, int64_t max_total,
                     hipStream_t stream);
@@ -220,6 +228,45 @@ void py_relu_colsum_bwd(Tensor y, Tensor dy, Tensor dx, Tensor db,
   cosamd::relu_colsum_bwd(y.data_ptr(), dy.data_ptr(), dx.data_ptr(),
                           db.data_ptr<float>(), (float)slope, rows,
                           (int)cols, (int)ldy, (int)lddy, cur_stream());
+}
+
+void py_gemm_conv_fwd_sc(Tensor X, Tensor B, Tensor C,
+                         c10::optional<Tensor> bias, int64_t M,
+                         int64_t N, int64_t K, int64_t ldb, int64_t ldc,
+                         bool relu, std::vector<int64_t> geom) {
+  CHECK_CUDA(X); CHECK_BF16(X); CHECK_BF16(B); CHECK_BF16(C);
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    CHECK_F32(bias.value());
+    bptr = bias.value().data_ptr<float>();
+  }
+  int g[14];
+  TORCH_CHECK(geom.size() == 14, "geom must have 14 ints");
+  for (int i = 0; i < 14; ++i) g[i] = (int)geom[i];
+  cosamd::gemm_conv_fwd_sc(X.data_ptr(), B.data_ptr(), C.data_ptr(),
+                           bptr, (int)M, (int)N, (int)K, (int)ldb,
+                           (int)ldc, relu, g, cur_stream());
+}
+
+void py_gemm_conv_dw_sc(Tensor A, Tensor X, Tensor C,
+                        c10::optional<Tensor> db, int64_t M, int64_t N,
+                        int64_t K, int64_t lda, int64_t ldc,
+                        int64_t store_mode, int64_t splitk, double alpha,
+                        std::vector<int64_t> geom) {
+  CHECK_CUDA(A); CHECK_BF16(A); CHECK_BF16(X); CHECK_F32(C);
+  float* dbp = nullptr;
+  if (db.has_value()) {
+    CHECK_F32(db.value());
+    dbp = db.value().data_ptr<float>();
+  }
+  int g[14];
+  TORCH_CHECK(geom.size() == 14, "geom must have 14 ints");
+  for (int i = 0; i < 14; ++i) g[i] = (int)geom[i];
+  cosamd::gemm_conv_dw_sc(A.data_ptr(), X.data_ptr(),
+                          C.data_ptr<float>(), (int)M, (int)N, (int)K,
+                          (int)lda, (int)ldc, (int)store_mode,
+                          (int)splitk, (float)alpha, dbp, g,
+                          cur_stream());
 }
 
 void py_im2col(Tensor x, Tensor col, int64_t N, int64_t H, int64_t W,
@@ -509,6 +556,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_colsum_bwd", &py_relu_colsum_bwd);
   m.def("gemm_conv_fwd", &py_gemm_conv_fwd);
   m.def("gemm_conv_dw", &py_gemm_conv_dw);
+  m.def("gemm_conv_fwd_sc", &py_gemm_conv_fwd_sc);
+  m.def("gemm_conv_dw_sc", &py_gemm_conv_dw_sc);
   m.def("im2col", &py_im2col);
   m.def("bn_stats", [](Tensor x, Tensor sum, Tensor sq, int64_t rows,
                        int64_t C) {

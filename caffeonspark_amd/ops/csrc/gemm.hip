@@ -550,6 +550,314 @@ __device__ __forceinline__ void stage_trans_pair_implicit(
   }
 }
 
+// ---- small-C implicit staging (conv1-class layers, C % 8 != 0).
+// With dil==1, c0==0 and full channels, the col matrix's k axis within
+// one filter row r is a CONTIGUOUS span of the input row (k = r*S*C +
+// s*C + c), so an 8-element slot is one unaligned 16-byte load (gfx950
+// supports it natively), merging two loads when the slot straddles a
+// filter-row boundary (S*C >= 8 keeps it to at most one straddle).
+// CGeom field reuse for the _sc kernels: Cg/mCg hold S*C, S/mS hold C.
+
+// NN A-operand [row=npq][k] into the canonical swizzled layout
+__device__ __forceinline__ void stage_implicit_sc(
+    bf16* lds_, const bf16* X, const CGeom& gm, const int* rinfo,
+    int k0, int tid, int nthreads) {
+  auto* lds = reinterpret_cast<unsigned short*>(lds_);
+  auto* x = reinterpret_cast<const unsigned short*>(X);
+  int SC = gm.Cg, C = gm.S;
+  for (int slot = tid; slot < 512; slot += nthreads) {
+    int row = slot >> 2;
+    int kk = (slot & 3) * 8;
+    int k = k0 + kk;
+    unsigned short* dst = lds + row * BK + (swz_chunk(row, slot & 3) << 3);
+    unsigned r0 = mdiv(k, gm.mCg);          // filter row
+    int t0 = k - r0 * SC;                   // s*C + c within the row
+    int h = rinfo[row * 3 + 1] + (int)r0;
+    int w0 = rinfo[row * 3 + 2];
+    int sp = SC - t0;                       // elements before r0+1
+    // fully-interior slot: the spanned w range [w0, w0+S) is in-image
+    bool interior = k + 8 <= gm.Kcol && h >= 0 && w0 >= 0 &&
+                    w0 * C + SC <= gm.W * C &&
+                    (sp >= 8 ? h < gm.H : h + 1 < gm.H);
+    if (interior) {
+      const unsigned short* a0 =
+          x + ((int64_t)rinfo[row * 3] + (int64_t)h * gm.W) * C +
+          w0 * C + t0;
+      if (sp >= 8) {
+        __builtin_memcpy(dst, a0, 16);
+      } else {
+        unsigned short lo[8], hi[8];
+        __builtin_memcpy(lo, a0, 16);
+        const unsigned short* a1 =
+            x + ((int64_t)rinfo[row * 3] + (int64_t)(h + 1) * gm.W) * C +
+            w0 * C;
+        __builtin_memcpy(hi, a1, 16);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) dst[j] = j < sp ? lo[j] : hi[j - sp];
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int kj = k + j;
+        unsigned short v = 0;
+        if (kj < gm.Kcol) {
+          unsigned r = mdiv(kj, gm.mCg);
+          int t = kj - r * SC;
+          unsigned s = mdiv(t, gm.mS);
+          int c = t - s * C;
+          int hh = rinfo[row * 3 + 1] + (int)r;
+          int ww = w0 + (int)s;
+          if (hh >= 0 && hh < gm.H && ww >= 0 && ww < gm.W)
+            v = x[((int64_t)rinfo[row * 3] +
+                   (int64_t)hh * gm.W + ww) * C + c];
+        }
+        dst[j] = v;
+      }
+    }
+  }
+}
+
+// TT B-operand [k=npq][col] u32 k-pair canonical layout
+__device__ __forceinline__ void stage_trans_pair_implicit_sc(
+    bf16* lds_, const bf16* X, const CGeom& gm, int row0,
+    int k0, int kend, int tid, int nthreads, int tile_rows) {
+  auto* lds = reinterpret_cast<unsigned short*>(lds_);
+  auto* x = reinterpret_cast<const unsigned short*>(X);
+  int SC = gm.Cg, C = gm.S;
+  int rslots = tile_rows >> 3;
+  int nslots = 16 * rslots;
+  for (int slot = tid; slot < nslots; slot += nthreads) {
+    int kp = slot & 15;
+    int m0 = (slot >> 4) * 8;
+    int gr = row0 + m0;                  // col-matrix column octet base
+    unsigned r0 = mdiv(gr, gm.mCg);
+    int t0 = gr - r0 * SC;
+    int sp = SC - t0;
+    unsigned short va[8] = {}, vb[8] = {};
+#pragma unroll
+    for (int hh2 = 0; hh2 < 2; ++hh2) {
+      unsigned short* v = hh2 ? vb : va;
+      int gk = k0 + kp * 2 + hh2;        // output-pixel index
+      if (gk >= kend) continue;
+      unsigned n = mdiv(gk, gm.mPQ);
+      unsigned pq = gk - n * gm.PQ;
+      unsigned p = mdiv(pq, gm.mQ);
+      unsigned q = pq - p * gm.Q;
+      int h = (int)p * gm.sh - gm.ph + (int)r0;
+      int w0 = (int)q * gm.sw - gm.pw;
+      bool interior = gr + 8 <= gm.Kcol && h >= 0 && w0 >= 0 &&
+                      w0 * C + SC <= gm.W * C &&
+                      (sp >= 8 ? h < gm.H : h + 1 < gm.H);
+      int64_t nHW = (int64_t)(n * gm.H);
+      if (interior) {
+        const unsigned short* a0 =
+            x + (nHW + h) * (int64_t)gm.W * C + w0 * C + t0;
+        if (sp >= 8) {
+          __builtin_memcpy(v, a0, 16);
+        } else {
+          unsigned short lo[8], hi[8];
+          __builtin_memcpy(lo, a0, 16);
+          const unsigned short* a1 =
+              x + (nHW + h + 1) * (int64_t)gm.W * C + w0 * C;
+          __builtin_memcpy(hi, a1, 16);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) v[j] = j < sp ? lo[j] : hi[j - sp];
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int kj = gr + j;
+          if (kj >= gm.Kcol) continue;
+          unsigned r = mdiv(kj, gm.mCg);
+          int t = kj - r * SC;
+          unsigned s = mdiv(t, gm.mS);
+          int c = t - s * C;
+          int hh = (int)p * gm.sh - gm.ph + (int)r;
+          int ww = w0 + (int)s;
+          if (hh >= 0 && hh < gm.H && ww >= 0 && ww < gm.W)
+            v[j] = x[(nHW + hh) * (int64_t)gm.W * C + ww * C + c];
+        }
+      }
+    }
+    int kk = kp * 2;
+    int chunk0 = kk >> 3, kin = kk & 7;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int row = m0 + j;
+      unsigned int packed = (unsigned int)va[j] |
+                            ((unsigned int)vb[j] << 16);
+      *reinterpret_cast<unsigned int*>(
+          lds + row * BK + (swz_chunk(row, chunk0) << 3) + kin) = packed;
+    }
+  }
+}
+
+// small-C forward conv GEMM: single-buffered, __syncthreads pipeline
+// (the ds_write staging has no counted-vmcnt scheme to preserve)
+template <int WAVES>
+__global__ __launch_bounds__(WAVES * 64, 2) void gemm_conv_fwd_sc_kernel(
+    const bf16* __restrict__ X, const bf16* __restrict__ B,
+    bf16* __restrict__ C, const float* __restrict__ bias,
+    int M, int N, int K, int ldb, int ldc, int relu, CGeom gm) {
+  __shared__ bf16 Asb[2][BM * BK];
+  __shared__ bf16 Bsb[2][BN * BK];
+  __shared__ int rinfo[BM * 3];
+
+  int mblocks = (M + BM - 1) / BM;
+  int nblocks = (N + BN - 1) / BN;
+  int bid = xcd_swizzle(blockIdx.x, mblocks * nblocks);
+  int bm = bid / nblocks, bn = bid % nblocks;
+  int tile_m = bm * BM, tile_n = bn * BN;
+
+  constexpr int NT = WAVES * 64;
+  constexpr int MROWS = BM / (WAVES / 2);
+  constexpr int MF = MROWS / 16;
+  int tid = threadIdx.x;
+  int wave = tid >> 6, lane = tid & 63;
+  int wm = wave >> 1, wn = wave & 1;
+  int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+
+  decode_rows(rinfo, gm, tile_m, M, BM, tid);
+  __syncthreads();
+
+  f32x4 acc[MF][4] = {};
+  // double-buffered, ONE barrier per tile: staging t+1 (into buf^1)
+  // overlaps compute of t; the end-of-iteration barrier covers both
+  // "staging t+1 done" and "all reads of buf[t&1] done" before the
+  // next iteration's stage(t+2) overwrites it
+  int T = K / BK;
+  stage_implicit_sc(Asb[0], X, gm, rinfo, 0, tid, NT);
+  stage_direct_fast<WAVES>(Bsb[0], B, tile_n, ldb, 0, tid);
+  __syncthreads();
+  for (int t = 0; t < T; ++t) {
+    if (t + 1 < T) {
+      stage_implicit_sc(Asb[(t + 1) & 1], X, gm, rinfo, (t + 1) * BK,
+                        tid, NT);
+      stage_direct_fast<WAVES>(Bsb[(t + 1) & 1], B, tile_n, ldb,
+                               (t + 1) * BK, tid);
+    }
+    const bf16* Asp = Asb[t & 1];
+    const bf16* Bsp = Bsb[t & 1];
+    bf16x8 afrag[MF], bfrag[4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      int rb = wn * 64 + f * 16 + lrow;
+      bfrag[f] = *reinterpret_cast<const bf16x8*>(
+          Bsp + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
+    }
+#pragma unroll
+    for (int f = 0; f < MF; ++f) {
+      int ra = wm * MROWS + f * 16 + lrow;
+      afrag[f] = *reinterpret_cast<const bf16x8*>(
+          Asp + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
+    }
+#pragma unroll
+    for (int fm = 0; fm < MF; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
+    __syncthreads();
+  }
+
+  int crow0 = tile_m + wm * MROWS + ((lane >> 4) << 2);
+  int ccol0 = tile_n + wn * 64 + (lane & 15);
+#pragma unroll
+  for (int fm = 0; fm < MF; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int col = ccol0 + fn * 16;
+      if (col >= N) continue;
+      float badd = (bias != nullptr) ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow0 + fm * 16 + r;
+        if (row >= M) continue;
+        float v = acc[fm][fn][r] + badd;
+        if (relu && v < 0.f) v = 0.f;
+        C[(int64_t)row * ldc + col] = f2bf(v);
+      }
+    }
+  }
+}
+
+// small-C dw trans/trans GEMM (B operand gathered contiguously from x)
+template <int STORE>
+__global__ __launch_bounds__(WNT, 1) void gemm_conv_dw_sc_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ X,
+    float* __restrict__ C, int M, int N, int K, int lda, int ldc,
+    int ksplit, float alpha, float* __restrict__ db, CGeom gm) {
+  __shared__ bf16 As[BM * BK];
+  __shared__ bf16 Bs[WBN * BK];
+
+  int mblocks = (M + BM - 1) / BM;
+  int nblocks = (N + WBN - 1) / WBN;
+  int bid = xcd_swizzle(blockIdx.x, mblocks * nblocks);
+  int bm = bid / nblocks, bn = bid % nblocks;
+  int tile_m = bm * BM, tile_n = bn * WBN;
+  int k_begin = blockIdx.y * ksplit;
+  int k_end = min(K, k_begin + ksplit);
+
+  int tid = threadIdx.x;
+  int wave = tid >> 6, lane = tid & 63;
+  int wm = wave >> 2, wn = wave & 3;
+  int lrow = lane & 15, lk8 = (lane >> 4) * 8;
+
+  f32x4 acc[4][4] = {};
+  bool do_db = (db != nullptr) && (bn == 0);
+  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
+    stage_trans_pair_guarded(As, A, tile_m, M, lda, k0, k_end, tid, WNT,
+                             BM);
+    stage_trans_pair_implicit_sc(Bs, X, gm, tile_n, k0, k_end, tid, WNT,
+                                 WBN);
+    __syncthreads();
+    if (do_db && tid < BM && tile_m + tid < M) {
+      float acc_b = 0.f;
+      const bf16* row = As + tid * BK;
+#pragma unroll
+      for (int kk = 0; kk < BK; ++kk)
+        acc_b += bf2f(row[(swz_chunk(tid, kk >> 3) << 3) | (kk & 7)]);
+      atomicAdd(db + tile_m + tid, acc_b);
+    }
+    bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      int ra = wm * 64 + f * 16 + lrow;
+      int rb = wn * 64 + f * 16 + lrow;
+      afrag[f] = *reinterpret_cast<const bf16x8*>(
+          As + ra * BK + (swz_chunk(ra, lk8 >> 3) << 3));
+      bfrag[f] = *reinterpret_cast<const bf16x8*>(
+          Bs + rb * BK + (swz_chunk(rb, lk8 >> 3) << 3));
+    }
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
+    __syncthreads();
+  }
+
+  int crow0 = tile_m + wm * 64 + ((lane >> 4) << 2);
+  int ccol0 = tile_n + wn * 64 + (lane & 15);
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int col = ccol0 + fn * 16;
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow0 + fm * 16 + r;
+        if (row >= M) continue;
+        if (STORE == 1)
+          C[(int64_t)row * ldc + col] = acc[fm][fn][r] * alpha;
+        else
+          atomicAdd(C + (int64_t)row * ldc + col, acc[fm][fn][r] * alpha);
+      }
+    }
+}
+
 // forward conv GEMM: C[NPQ, Kg] = im2col(x) @ Wr^T, bias+ReLU fused.
 // Always the pipelined double-buffered loop: the implicit A staging
 // zero-fills k >= Kcol, so K runs over Kpad (32-aligned by the host)
@@ -989,6 +1297,49 @@ void gemm_conv_fwd(const void* X, const void* B, void* C,
         (const bf16*)X, (const bf16*)B, (bf16*)C, bias,
         (const bf16*)zpage, M, N, K, ldb, ldc, relu ? 1 : 0,
         accum ? 1 : 0, gm);
+}
+
+void gemm_conv_fwd_sc(const void* X, const void* B, void* C,
+                      const float* bias, int M, int N, int K, int ldb,
+                      int ldc, bool relu, const int* geom,
+                      hipStream_t stream) {
+  CGeom gm = make_geom(geom);
+  int mblocks = (M + BM - 1) / BM, nblocks = (N + BN - 1) / BN;
+  dim3 grid(mblocks * nblocks);
+  static const int w8_env = [] {
+    const char* e = getenv("COS_GEMM_W8");
+    return e ? (e[0] == '1' ? 1 : 0) : -1;
+  }();
+  bool w8 = (w8_env == -1) ? (M <= 256 || N <= 256) : (w8_env == 1);
+  if (w8)
+    gemm_conv_fwd_sc_kernel<8><<<grid, dim3(512), 0, stream>>>(
+        (const bf16*)X, (const bf16*)B, (bf16*)C, bias, M, N, K, ldb,
+        ldc, relu ? 1 : 0, gm);
+  else
+    gemm_conv_fwd_sc_kernel<4><<<grid, dim3(256), 0, stream>>>(
+        (const bf16*)X, (const bf16*)B, (bf16*)C, bias, M, N, K, ldb,
+        ldc, relu ? 1 : 0, gm);
+}
+
+void gemm_conv_dw_sc(const void* A, const void* X, float* C, int M,
+                     int N, int K, int lda, int ldc, int store_mode,
+                     int splitk, float alpha, float* db, const int* geom,
+                     hipStream_t stream) {
+  CGeom gm = make_geom(geom);
+  splitk = max(1, splitk);
+  int ksplit = (K + splitk - 1) / splitk;
+  ksplit = ((ksplit + BK - 1) / BK) * BK;
+  int zblocks = (K + ksplit - 1) / ksplit;
+  int mblocks = (M + BM - 1) / BM, nblocks = (N + WBN - 1) / WBN;
+  dim3 grid(mblocks * nblocks, zblocks);
+  if (store_mode == 1 && zblocks == 1)
+    gemm_conv_dw_sc_kernel<1><<<grid, WNT, 0, stream>>>(
+        (const bf16*)A, (const bf16*)X, C, M, N, K, lda, ldc, ksplit,
+        alpha, db, gm);
+  else
+    gemm_conv_dw_sc_kernel<2><<<grid, WNT, 0, stream>>>(
+        (const bf16*)A, (const bf16*)X, C, M, N, K, lda, ldc, ksplit,
+        alpha, db, gm);
 }
 
 void gemm_conv_dw(const void* A, const void* X, float* C, int M, int N,

@@ -260,9 +260,21 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
                 and (not (R == S == 1) or imp1x1)
                 and N * P * Q < (1 << 20) and R * S * Cg < (1 << 20)
                 and bool(int(os.environ.get("COS_IMPLICIT", "1"))))
+    # small-C variant (conv1-class, C%8!=0): with dil==1 / full
+    # channels the k axis is CONTIGUOUS spans of the input rows, so
+    # staging uses unaligned 16-byte loads straight from x (the col
+    # matrix for AlexNet conv1 was 570 MB written + read twice)
+    # M gate: at very large NPQ (AlexNet conv1, 774k rows) the staging
+    # VALU cost beats the saved im2col (same-box A/B: -2% AlexNet,
+    # +1.5% LRCN whose conv1 has 193k rows)
+    implicit_sc = (not implicit and C == Cg and G == 1 and dil == 1
+                   and S * C >= 8 and not (R == S == 1)
+                   and N * P * Q < (1 << 18)
+                   and 256 <= Kcol < (1 << 20)
+                   and bool(int(os.environ.get("COS_IMPLICIT_SC", "1"))))
     # 32-aligned K keeps every k-tile on the pipelined fast path (and
     # bounds the implicit kernel's B reads); pad columns stay zero
-    Kpad = _pad32(Kcol) if implicit else _pad8(Kcol)
+    Kpad = _pad32(Kcol) if (implicit or implicit_sc) else _pad8(Kcol)
 
     xl = _cl(x)
     # weight repack: [K, Cg, R, S] -> bf16 [K, R, S, Cg] padded to Kpad
@@ -388,6 +400,13 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
                 bias_f[g * Kg:(g + 1) * Kg] if bias_f is not None
                 else None, _zpage(x.device), NPQ, Kg, Kpad, Kpad,
                 ldc_out, relu, False, geom)
+    elif implicit_sc:
+        col = None
+        # CGeom field reuse for the _sc kernels: S slot carries C,
+        # Cg slot carries S*C (both are magic-divide divisors there)
+        geom = [H, W, C, P, Q, sh, sw, ph, pw, dil, C, 0, S * C, Kcol]
+        _ext.gemm_conv_fwd_sc(xl, wrb, y2, bias_f, NPQ, Kout, Kpad,
+                              Kpad, ldc_out, relu, geom)
     else:
         col = torch.empty((G, NPQ, Kpad), dtype=torch.bfloat16,
                           device=x.device)
@@ -404,6 +423,7 @@ def conv2d_forward(x, w, b, stride, pad, dilation, groups, ctx=None,
         ctx["xl"] = xl
         ctx["is_1x1"] = is_1x1
         ctx["implicit"] = implicit and not is_1x1
+        ctx["implicit_sc"] = implicit_sc
         ctx["shape"] = (N, C, H, W, P, Q, R, S, sh, sw, ph, pw, dil, G, Cg,
                         Kg, Kpad, Kcol)
         ctx["wr"] = wr
@@ -447,11 +467,15 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
     is_1x1 = ctx.get("is_1x1", False)
     wino = ctx.get("wino", False)
     implicit = ctx.get("implicit", False)
+    implicit_sc = ctx.get("implicit_sc", False)
     # implicit dw needs the wide TT kernel (N > 128); otherwise — and
     # for Winograd forward — materialize the col matrix lazily
     dw_implicit = implicit and Kpad > 128 and \
         bool(int(os.environ.get("COS_DW_TT", "1")))
-    if need_dw and col is None and not is_1x1 and not dw_implicit:
+    dw_implicit_sc = implicit_sc and Kpad > 128 and \
+        bool(int(os.environ.get("COS_DW_TT", "1")))
+    if need_dw and col is None and not is_1x1 and not dw_implicit \
+            and not dw_implicit_sc:
         col = torch.empty((G, N * P * Q, Kpad), dtype=torch.bfloat16,
                           device=dy.device)
         for g in range(G):
@@ -525,6 +549,14 @@ def conv2d_backward(x, w, dy, stride, pad, dilation, groups,
                                   dwp[g * Kg:], db_slice, Kg, Kpad, NPQ,
                                   ld_dy, Kpad, store_dw, sk_tt, 1.0,
                                   geom)
+            elif use_tt and dw_implicit_sc:
+                db_slice = db.narrow(0, g * Kg, Kg) if fuse_db else None
+                geom = [H, W, C, P, Q, sh, sw, ph, pw, dil, C, 0,
+                        S * C, Kcol]
+                _ext.gemm_conv_dw_sc(_dyg(g), ctx["xl"],
+                                     dwp[g * Kg:], db_slice, Kg, Kpad,
+                                     NPQ, ld_dy, Kpad, store_dw, sk_tt,
+                                     1.0, geom)
             elif use_tt:
                 src = x2 if is_1x1 else col[g]
                 db_slice = db.narrow(0, g * Kg, Kg) if fuse_db else None
