@@ -692,7 +692,7 @@ __device__ __forceinline__ void stage_trans_pair_implicit_sc(
   }
 }
 
-// small-C forward conv GEMM: single-buffered, __syncthreads pipeline
+// small-C forward conv GEMM: double-buffered __syncthreads pipeline
 // (the ds_write staging has no counted-vmcnt scheme to preserve)
 template <int WAVES>
 __global__ __launch_bounds__(WAVES * 64, 2) void gemm_conv_fwd_sc_kernel(
