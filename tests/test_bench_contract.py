@@ -39,3 +39,25 @@ def test_bench_default_args_fast():
     mod = importlib.util.module_from_spec(spec)
     spec.loader.exec_module(mod)
     assert hasattr(mod, "main")
+
+
+def test_bench_two_rank_gloo_contract():
+    """The driver launches bench.py under torchrun for the multi-GPU
+    scaling run; lock the rendezvous + DDP + JSON contract down with a
+    2-rank gloo (CPU) run of the smallest config."""
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29713", "--no-python", sys.executable,
+         os.path.join(ROOT, "bench.py"), "--gpus", "2", "--steps", "2",
+         "--warmup", "1", "--batch", "8", "--model", "cifar10_quick"],
+        capture_output=True, text=True, timeout=600, cwd=ROOT, env=env)
+    assert out.returncode == 0, out.stderr[-800:]
+    line = [l for l in out.stdout.splitlines()
+            if l.startswith('{"metric"')]
+    assert len(line) == 1, out.stdout[-500:]  # rank 0 prints exactly one
+    rec = json.loads(line[0])
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["parallelism"] == "dp2"
