@@ -47,6 +47,7 @@ def test_bench_two_rank_gloo_contract():
     2-rank gloo (CPU) run of the smallest config."""
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
+    env["CUDA_VISIBLE_DEVICES"] = ""   # gloo contract even on a GPU box
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
