@@ -47,7 +47,11 @@ def test_bench_two_rank_gloo_contract():
     2-rank gloo (CPU) run of the smallest config."""
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
-    env["CUDA_VISIBLE_DEVICES"] = ""   # gloo contract even on a GPU box
+    # gloo contract even on a GPU box (ROCm reads the HIP/ROCR variants
+    # and treats an empty CUDA_VISIBLE_DEVICES as unset)
+    env["CUDA_VISIBLE_DEVICES"] = ""
+    env["HIP_VISIBLE_DEVICES"] = ""
+    env["ROCR_VISIBLE_DEVICES"] = ""
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
