@@ -76,9 +76,14 @@ def main(argv: List[str] = None) -> int:
     os.environ["COS_SPARK_MASTER"] = master
     if rest and rest[0].endswith(".py"):
         app, app_args = rest[0], rest[1:]
-        sys.argv = [app] + app_args
-        runpy.run_path(app, run_name="__main__")
-        return 0
+        if app_class and not os.path.exists(app):
+            # class-driven launch: the app slot carries the reference's
+            # artifact name (jar or placeholder .py) — skip it
+            rest = app_args
+        else:
+            sys.argv = [app] + app_args
+            runpy.run_path(app, run_name="__main__")
+            return 0
     # --class com.yahoo.ml.caffe.CaffeOnSpark (or no app file): run the
     # built-in driver with the reference CLI flags
     from .driver import main as driver_main
