@@ -104,6 +104,9 @@ def test_gemm_splitk_atomic():
     # dilation through the implicit decode
     dict(n=2, c=16, h=15, w=15, k=16, r=3, stride=1, pad=2, groups=1,
          dil=2),
+    # small-C implicit (contiguous-run staging) WITH padding: edge
+    # slots take the per-element guard path
+    dict(n=2, c=4, h=21, w=21, k=16, r=9, stride=2, pad=4, groups=1),
 ])
 def test_conv_forward_backward(case):
     n, c, h, w = case["n"], case["c"], case["h"], case["w"]
@@ -752,3 +755,40 @@ def test_fused_solver_updates_gpu(stype):
     rel = (cpu.flat_w - gpu.flat_w.cpu()).norm() / \
         cpu.flat_w.norm().clamp_min(1e-6)
     assert rel < 1e-5, f"{stype}: relL2={rel}"
+
+
+@pytest.mark.gpu
+def test_bf16_shadow_coherence_through_updates_and_load(tmp_path):
+    """The fused update kernels write the bf16 shadow from registers and
+    refresh() then skips the whole-arena cast; weight loads must resync
+    explicitly.  Verify the shadow tracks flat_w through steps, snapshot
+    restore, and further steps."""
+    import os
+
+    from caffeonspark_amd.core.solver import solver_from_prototxt
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    proto = os.path.join(root, "caffeonspark_amd", "models",
+                         "lenet_memory_solver.prototxt")
+    s = solver_from_prototxt(proto, device=dev(), dtype=torch.bfloat16)
+    s.param.snapshot_prefix = str(tmp_path / "snap")
+    x = torch.randn(32, 1, 28, 28).to(dev(), torch.bfloat16)
+    y = torch.randint(0, 10, (32,)).float().to(dev())
+    s.net.data_layers()[0].reset(x, y)
+
+    def shadow_ok(sv):
+        torch.cuda.synchronize()
+        want = sv.flat_w.to(torch.bfloat16)
+        assert torch.equal(sv.flat_wb, want), "bf16 shadow diverged"
+
+    s.step(3)
+    s.net.forward()          # refresh runs (skips the cast once synced)
+    shadow_ok(s)
+
+    model = s.snapshot()
+    s.step(2)
+    s.load_weights(model)    # out-of-band arena write -> resync path
+    shadow_ok(s)
+    s.step(2)
+    s.net.forward()
+    shadow_ok(s)
