@@ -300,3 +300,23 @@ def test_inplace_bn_scale_relu_chain_grads():
                                rtol=2e-3, atol=1e-4)
     torch.testing.assert_close(sc.blobs[1].diff, auto_dbias,
                                rtol=2e-3, atol=1e-4)
+
+
+def test_slice_bias_silence_layers():
+    # Slice: channel split with slice points (reference SliceLayer)
+    layer = make_layer("""name: "sl" type: "Slice" bottom: "x"
+        top: "a" top: "b" top: "c"
+        slice_param { axis: 1 slice_point: 2 slice_point: 5 }""")
+    run_grad_check(layer, [torch.randn(2, 8, 3, 3)])
+
+    # Bias: learned per-channel bias broadcast over trailing axes
+    layer = make_layer("""name: "bi" type: "Bias" bottom: "x" top: "y"
+        bias_param { axis: 1 }""")
+    run_grad_check(layer, [torch.randn(2, 4, 5, 5)])
+
+    # Silence: consumes its bottom, produces nothing, zero gradient
+    layer = make_layer('name: "si" type: "Silence" bottom: "x"')
+    x = Blob([2, 3])
+    x.data = torch.randn(2, 3)
+    assert layer.forward([x], []) == 0.0
+    layer.backward([], [True], [x])   # must not raise
